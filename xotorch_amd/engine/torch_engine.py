@@ -1,0 +1,313 @@
+"""The PyTorch-ROCm engine: our ShardedModel + HIP ops + device-resident KV.
+
+Engine parity with the reference's torchtune engine
+(/root/reference/xotorch/inference/torch/sharded_inference_engine.py:37-424),
+redesigned MI355X-first:
+- hidden states stay bf16 on-device between calls (the numpy fp32 round trip
+  exists only at the legacy Node/NumPy boundary for multi-host transport);
+- KV caches + position state are engine-owned and never serialized
+  (the reference ships mask/tokens/input_pos as JSON per hop);
+- all GPU work is serialized through a 1-thread executor (same concurrency
+  model as the reference, :46), keeping the asyncio API responsive;
+- train/evaluate are real (CE loss, backward from a received ring gradient,
+  optimizer step) — the reference declares but never implements them.
+"""
+from __future__ import annotations
+
+import asyncio
+import os
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import Dict, Optional, Tuple
+
+import numpy as np
+import torch
+
+from xotorch_amd import ops
+from xotorch_amd.engine.interface import InferenceEngine
+from xotorch_amd.engine.kvcache import ShardKVCache
+from xotorch_amd.engine.state import ShardInferenceState
+from xotorch_amd.engine.tokenizers import DummyTokenizer, resolve_tokenizer
+from xotorch_amd.models.config import ModelConfig, config_from_hf
+from xotorch_amd.models.llama import ShardedModel
+from xotorch_amd.models.registry import builtin_config, get_repo
+from xotorch_amd.models.weights import load_shard_weights, random_init
+from xotorch_amd.shard import Shard
+
+TEMP = float(os.getenv("XOT_TEMP", "0.6"))
+TOP_K = int(os.getenv("XOT_TOP_K", "35"))
+DEFAULT_MAX_GEN = int(os.getenv("XOT_MAX_GEN", "512"))
+
+
+@dataclass
+class Session:
+  cache: ShardKVCache
+  state: ShardInferenceState = field(default_factory=ShardInferenceState)
+
+
+class TorchEngine(InferenceEngine):
+  def __init__(self, shard_downloader=None, require_hip: bool = False, device: Optional[str] = None,
+               dtype: Optional[torch.dtype] = None):
+    self.shard_downloader = shard_downloader
+    self.require_hip = require_hip
+    if device is None:
+      device = "cuda" if torch.cuda.is_available() else "cpu"
+    self.device = device
+    self.dtype = dtype if dtype is not None else (torch.bfloat16 if device == "cuda" else torch.float32)
+    self.shard: Optional[Shard] = None
+    self.model: Optional[ShardedModel] = None
+    self.cfg: Optional[ModelConfig] = None
+    self.tokenizer = None
+    self.sessions: Dict[str, Session] = {}
+    self._executor = ThreadPoolExecutor(max_workers=1)
+    self._rng = torch.Generator(device=self.device)
+    self._rng.manual_seed(int(os.getenv("XOT_SEED", "1234")))
+    self.oom_count = 0
+
+  # ---------- model lifecycle ----------
+
+  async def ensure_shard(self, shard: Shard):
+    if self.shard == shard and self.model is not None:
+      return
+    model_dir: Optional[Path] = None
+    if self.shard_downloader is not None:
+      try:
+        model_dir = await self.shard_downloader.ensure_shard(shard, type(self).__name__)
+      except Exception:
+        model_dir = None
+    await asyncio.get_running_loop().run_in_executor(self._executor, self._build_model, shard, model_dir)
+    # tokenizer
+    if model_dir is not None and (Path(model_dir) / "tokenizer_config.json").exists():
+      self.tokenizer = await resolve_tokenizer(model_dir)
+    else:
+      repo = get_repo(shard.model_id, type(self).__name__)
+      if repo in (None, "dummy") or os.getenv("XOT_OFFLINE", "0") == "1":
+        self.tokenizer = DummyTokenizer()
+      else:
+        try:
+          self.tokenizer = await resolve_tokenizer(repo)
+        except Exception:
+          self.tokenizer = DummyTokenizer()
+
+  def _build_model(self, shard: Shard, model_dir: Optional[Path]):
+    cfg_dict = None
+    if model_dir is not None and (Path(model_dir) / "config.json").exists():
+      cfg_dict = str(Path(model_dir) / "config.json")
+    else:
+      cfg_dict = builtin_config(shard.model_id)
+    if cfg_dict is None:
+      raise ValueError(f"no config available for model {shard.model_id}")
+    cfg = config_from_hf(cfg_dict, model_id=shard.model_id)
+    with torch.device("meta"):
+      model = ShardedModel(cfg, shard)
+    model = model.to_empty(device=self.device)
+    model = model.to(self.dtype)
+    # recompute rope tables (to_empty leaves buffers uninitialized)
+    from xotorch_amd.ops.torch_ref import rope_cos_sin
+    cos, sin = rope_cos_sin(cfg.head_dim, cfg.max_seq_len, cfg.rope_theta, cfg.rope_scaling, device=self.device)
+    model.rope_cos.copy_(cos)
+    model.rope_sin.copy_(sin)
+    if model_dir is not None and (
+      (Path(model_dir) / "model.safetensors.index.json").exists() or (Path(model_dir) / "model.safetensors").exists()
+    ):
+      load_shard_weights(model, Path(model_dir), device="cpu")
+      model = model.to(self.device)
+    else:
+      random_init(model)
+    model.eval()
+    self.model = model
+    self.cfg = cfg
+    self.shard = shard
+    self.sessions.clear()
+
+  def _session(self, request_id: str, batch: int, total_len: int) -> Session:
+    sess = self.sessions.get(request_id)
+    if sess is None or sess.cache.batch != batch or sess.cache.capacity < total_len:
+      cache = ShardKVCache(
+        n_layers=self.shard.get_layer_count(), batch=batch, n_kv_heads=self.cfg.n_kv_heads,
+        capacity=total_len, head_dim=self.cfg.head_dim, dtype=self.dtype, device=self.device,
+      )
+      sess = Session(cache=cache, state=ShardInferenceState(total_len=total_len, batch=batch))
+      self.sessions[request_id] = sess
+    return sess
+
+  async def clear_session(self, request_id: str):
+    self.sessions.pop(request_id, None)
+
+  # ---------- tokenizer plumbing ----------
+
+  async def encode(self, shard: Shard, prompt: str) -> np.ndarray:
+    await self.ensure_shard(shard)
+    tokens = self.tokenizer.encode(prompt)
+    return np.asarray(tokens, dtype=np.int64)
+
+  async def decode(self, shard: Shard, tokens) -> str:
+    await self.ensure_shard(shard)
+    return self.tokenizer.decode(list(np.asarray(tokens).reshape(-1)))
+
+  async def sample(self, x: np.ndarray, temp: float = TEMP, top_k: int = TOP_K) -> np.ndarray:
+    logits = torch.from_numpy(np.asarray(x))
+    if logits.dim() == 3:
+      logits = logits[:, -1, :]
+    tok = ops.softmax_sample(logits, temperature=temp, top_k=top_k)
+    return tok.cpu().numpy()
+
+  # ---------- inference ----------
+
+  async def infer_tensor(
+    self, request_id: str, shard: Shard, input_data: np.ndarray, inference_state: Optional[dict] = None
+  ) -> Tuple[np.ndarray, Optional[dict]]:
+    await self.ensure_shard(shard)
+    loop = asyncio.get_running_loop()
+    return await loop.run_in_executor(self._executor, self._infer_tensor_sync, request_id, input_data, inference_state)
+
+  def _infer_tensor_sync(self, request_id: str, input_data: np.ndarray, inference_state: Optional[dict]):
+    meta = ShardInferenceState.from_dict(inference_state)
+    x = torch.from_numpy(np.ascontiguousarray(input_data))
+    is_tokens = x.dtype in (torch.int32, torch.int64)
+    B = x.shape[0]
+    S = x.shape[1]
+    total_len = meta.total_len or min(self.cfg.max_seq_len, S + DEFAULT_MAX_GEN)
+    sess = self._session(request_id, B, total_len)
+    if meta.curr_pos == 0 and not is_tokens and inference_state is not None:
+      # mid-ring stage receiving a fresh prefill: state carries positions
+      pass
+    start_pos = meta.curr_pos if inference_state is not None else sess.state.curr_pos
+    x = x.to(self.device)
+    if not is_tokens:
+      x = x.to(self.dtype)
+    positions = torch.arange(start_pos, start_pos + S, dtype=torch.int32, device=self.device)
+    is_decode = S == 1 and start_pos > 0
+    try:
+      with torch.inference_mode():
+        out = self.model(
+          x, caches=sess.cache.caches, positions=positions, start_pos=start_pos, is_decode=is_decode,
+        )
+    except torch.cuda.OutOfMemoryError:
+      self.oom_count += 1
+      self.sessions.pop(request_id, None)
+      if self.device == "cuda":
+        torch.cuda.empty_cache()
+      raise
+    new_pos = start_pos + S
+    sess.state.curr_pos = new_pos
+    sess.state.total_len = total_len
+    out_np = out.float().cpu().numpy()
+    return out_np, {"curr_pos": new_pos, "total_len": total_len, "batch": B}
+
+  # ---------- training (capability the reference declared but never built) ----------
+
+  async def train(self, request_id: str, shard: Shard, inputs, targets, lengths, loss: str = "ce", opt=None):
+    await self.ensure_shard(shard)
+    loop = asyncio.get_running_loop()
+    return await loop.run_in_executor(
+      self._executor, self._train_sync, request_id, inputs, targets, lengths, loss, opt, True
+    )
+
+  async def evaluate(self, request_id: str, shard: Shard, inputs, targets, lengths, loss: str = "ce"):
+    await self.ensure_shard(shard)
+    loop = asyncio.get_running_loop()
+    return await loop.run_in_executor(
+      self._executor, self._train_sync, request_id, inputs, targets, lengths, loss, None, False
+    )
+
+  def _forward_nocache(self, x: torch.Tensor):
+    """Training-mode forward over this shard (no KV cache, causal attention)."""
+    model = self.model
+    if x.dtype in (torch.int32, torch.int64):
+      h = model.embed_tokens(x)
+    else:
+      h = x
+    S = h.shape[1]
+    positions = torch.arange(0, S, dtype=torch.long, device=h.device)
+    cos, sin = model.rope_cos, model.rope_sin
+    from xotorch_amd.ops import torch_ref as tr
+    for lid in model.local_layer_ids:
+      layer = model.layers[str(lid)]
+      normed = tr.rmsnorm(h, layer.input_layernorm.weight, layer.eps)
+      attn = layer.self_attn
+      B = h.shape[0]
+      cfgm = attn.cfg
+      q = attn.q_proj(normed).view(B, S, cfgm.n_heads, cfgm.head_dim)
+      k = attn.k_proj(normed).view(B, S, cfgm.n_kv_heads, cfgm.head_dim)
+      v = attn.v_proj(normed).view(B, S, cfgm.n_kv_heads, cfgm.head_dim)
+      q, k = tr.rope_apply(q, k, cos, sin, positions)
+      rep = cfgm.n_heads // cfgm.n_kv_heads
+      out = torch.nn.functional.scaled_dot_product_attention(
+        q.transpose(1, 2).float(),
+        k.transpose(1, 2).repeat_interleave(rep, dim=1).float(),
+        v.transpose(1, 2).repeat_interleave(rep, dim=1).float(),
+        is_causal=True,
+      ).to(h.dtype)
+      attn_out = attn.o_proj(out.transpose(1, 2).reshape(B, S, cfgm.n_heads * cfgm.head_dim))
+      h = h + attn_out
+      normed2 = tr.rmsnorm(h, layer.post_attention_layernorm.weight, layer.eps)
+      h = h + layer.mlp(normed2)
+    if model.shard.is_last_layer:
+      h = tr.rmsnorm(h, model.norm.weight, model.cfg.norm_eps)
+      h = torch.nn.functional.linear(h, model.head_weight().to(h.dtype))
+    return h
+
+  def _train_sync(self, request_id, inputs, targets, lengths, loss, opt, do_step):
+    model = self.model
+    was_training = model.training
+    model.train(do_step)
+    x = torch.from_numpy(np.ascontiguousarray(inputs)).to(self.device)
+    is_tokens = x.dtype in (torch.int32, torch.int64)
+    if not is_tokens:
+      x = x.to(self.dtype).requires_grad_(True)
+    with torch.enable_grad():
+      out = self._forward_nocache(x)
+      if loss == "back_gradient":
+        # backward from the gradient received off the ring (targets = dL/d(out))
+        grad = torch.from_numpy(np.ascontiguousarray(targets)).to(self.device).to(out.dtype)
+        out.backward(grad)
+        loss_val = 0.0
+      else:
+        tgt = torch.from_numpy(np.ascontiguousarray(targets)).to(self.device)
+        if model.shard.is_last_layer:
+          lens = torch.from_numpy(np.ascontiguousarray(lengths)).to(self.device)
+          S = tgt.shape[1]
+          mask = torch.arange(S, device=self.device)[None, :] < lens[:, None]
+          lg = out.float().reshape(-1, out.shape[-1])
+          lt = tgt.reshape(-1).long()
+          ce = torch.nn.functional.cross_entropy(lg, lt, reduction="none").reshape(tgt.shape)
+          loss_t = (ce * mask).sum() / mask.sum().clamp(min=1)
+          loss_t.backward()
+          loss_val = float(loss_t.detach())
+        else:
+          raise ValueError("non-last shard must train with loss='back_gradient'")
+    if do_step:
+      if opt is None:
+        opt = self._default_optimizer()
+      opt.step()
+      opt.zero_grad(set_to_none=True)
+    in_grad = None
+    if not is_tokens and x.grad is not None:
+      in_grad = x.grad.float().cpu().numpy()
+    model.train(was_training)
+    return loss_val, in_grad
+
+  _opt = None
+
+  def _default_optimizer(self):
+    if self._opt is None:
+      self._opt = torch.optim.AdamW(self.model.parameters(), lr=float(os.getenv("XOT_LR", "1e-5")))
+    return self._opt
+
+  # ---------- checkpointing ----------
+
+  async def save_checkpoint(self, shard: Shard, path: str):
+    await self.ensure_shard(shard)
+    from safetensors.torch import save_file
+    Path(path).parent.mkdir(parents=True, exist_ok=True)
+    sd = {k: v.detach().contiguous().cpu() for k, v in self.model.state_dict().items() if not k.startswith("rope_")}
+    save_file(sd, path)
+
+  async def load_checkpoint(self, shard: Shard, path: str):
+    await self.ensure_shard(shard)
+    from safetensors import safe_open
+    with safe_open(path, framework="pt", device="cpu") as f:
+      sd = {k: f.get_tensor(k) for k in f.keys()}
+    self.model.load_state_dict(sd, strict=False)

@@ -1,0 +1,128 @@
+"""Model configuration: HF config.json → internal ModelConfig.
+
+Parity with the reference's config loader
+(/root/reference/xotorch/inference/torch/llm_utils.py:79-126), including the
+XOT_MAX_SEQ_LEN clamp, rewritten for this framework's own decoder: one
+generic GQA + SwiGLU + RMSNorm architecture covering the llama / qwen2 /
+mistral families with per-family RoPE variants.
+"""
+from __future__ import annotations
+
+import json
+import math
+import os
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import Optional
+
+import torch
+
+_DTYPE_MAP = {
+  "float16": torch.float16,
+  "bfloat16": torch.bfloat16,
+  "float32": torch.float32,
+}
+
+
+@dataclass
+class RopeScaling:
+  # llama3-style frequency scaling; None fields → plain RoPE
+  factor: float = 1.0
+  low_freq_factor: float = 1.0
+  high_freq_factor: float = 4.0
+  original_max_position_embeddings: int = 8192
+  rope_type: str = "default"
+
+
+@dataclass
+class ModelConfig:
+  model_id: str = "unknown"
+  vocab_size: int = 32000
+  dim: int = 4096
+  n_layers: int = 32
+  n_heads: int = 32
+  n_kv_heads: int = 8
+  head_dim: int = 128
+  intermediate_dim: int = 14336
+  norm_eps: float = 1e-5
+  rope_theta: float = 500000.0
+  rope_scaling: Optional[RopeScaling] = None
+  max_seq_len: int = 8192
+  tie_word_embeddings: bool = False
+  attn_bias: bool = False  # qwen2 q/k/v bias
+  torch_dtype: torch.dtype = torch.bfloat16
+  bos_token_id: Optional[int] = None
+  eos_token_id: Optional[int] = None
+  # MoE (Mixtral-class); n_experts == 0 → dense MLP
+  n_experts: int = 0
+  n_experts_per_tok: int = 2
+
+  @property
+  def kv_mult(self) -> int:
+    return self.n_heads // self.n_kv_heads
+
+  def estimate_param_bytes(self) -> int:
+    """Rough bf16 parameter footprint — used by the partitioner for sizing."""
+    per_layer = (
+      self.dim * (self.n_heads + 2 * self.n_kv_heads) * self.head_dim  # qkv
+      + self.n_heads * self.head_dim * self.dim  # o
+      + 3 * self.dim * self.intermediate_dim * max(1, self.n_experts)  # mlp
+      + 2 * self.dim  # norms
+    )
+    embed = self.vocab_size * self.dim * (1 if self.tie_word_embeddings else 2)
+    return 2 * (self.n_layers * per_layer + embed + self.dim)
+
+
+def config_from_hf(config_path: Path | str | dict, model_id: str = "unknown") -> ModelConfig:
+  """Parse a HF-style config.json (path or dict) into a ModelConfig."""
+  if isinstance(config_path, (str, Path)):
+    with open(config_path) as f:
+      raw = json.load(f)
+  else:
+    raw = dict(config_path)
+
+  n_heads = raw.get("num_attention_heads", 32)
+  dim = raw.get("hidden_size", 4096)
+  head_dim = raw.get("head_dim") or dim // n_heads
+  rope_scaling = None
+  rs = raw.get("rope_scaling")
+  if rs and rs.get("rope_type", rs.get("type", "default")) == "llama3":
+    rope_scaling = RopeScaling(
+      factor=rs.get("factor", 8.0),
+      low_freq_factor=rs.get("low_freq_factor", 1.0),
+      high_freq_factor=rs.get("high_freq_factor", 4.0),
+      original_max_position_embeddings=rs.get("original_max_position_embeddings", 8192),
+      rope_type="llama3",
+    )
+
+  max_seq_len = raw.get("max_position_embeddings", 8192)
+  env_max = os.getenv("XOT_MAX_SEQ_LEN")
+  if env_max:
+    max_seq_len = min(max_seq_len, int(env_max))
+
+  eos = raw.get("eos_token_id")
+  if isinstance(eos, list):
+    eos = eos[0] if eos else None
+
+  mtype = (raw.get("model_type") or "llama").lower()
+  return ModelConfig(
+    model_id=model_id,
+    vocab_size=raw.get("vocab_size", 32000),
+    dim=dim,
+    n_layers=raw.get("num_hidden_layers", 32),
+    n_heads=n_heads,
+    n_kv_heads=raw.get("num_key_value_heads", n_heads),
+    head_dim=head_dim,
+    intermediate_dim=raw.get("intermediate_size", 4 * dim),
+    norm_eps=raw.get("rms_norm_eps", 1e-5),
+    rope_theta=raw.get("rope_theta", 10000.0),
+    rope_scaling=rope_scaling,
+    max_seq_len=max_seq_len,
+    tie_word_embeddings=raw.get("tie_word_embeddings", False),
+    attn_bias=mtype in ("qwen2",) or raw.get("attention_bias", False),
+    torch_dtype=_DTYPE_MAP.get(raw.get("torch_dtype", "bfloat16"), torch.bfloat16),
+    bos_token_id=raw.get("bos_token_id"),
+    eos_token_id=eos,
+    n_experts=raw.get("num_local_experts", 0) or 0,
+    n_experts_per_tok=raw.get("num_experts_per_tok", 2),
+  )

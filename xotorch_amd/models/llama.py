@@ -1,0 +1,221 @@
+"""Sharded GQA transformer decoder for the llama / qwen2 / mistral families.
+
+This is the framework's own decoder (capability parity with the reference's
+torchtune assembly, /root/reference/xotorch/inference/torch/models/general_mha.py:23-254
+and llm_utils.py:335-489), built for MI355X execution:
+
+- every hot op goes through `xotorch_amd.ops` (HIP kernels on GPU, torch on CPU);
+- KV caches are engine-owned, device-resident, laid out [B, KVH, T, hd] for
+  the decode-attention kernel's row reads;
+- a shard holds only layers [start_layer .. end_layer]; the first shard owns
+  the token embedding, the last owns final norm + lm_head (tied embeddings
+  supported);
+- parameter names mirror HF checkpoints (q_proj/k_proj/... ) and RoPE uses
+  the HF rotate-half convention, so safetensors load with no permute.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from xotorch_amd import ops
+from xotorch_amd.models.config import ModelConfig
+from xotorch_amd.ops.torch_ref import rope_cos_sin
+from xotorch_amd.shard import Shard
+
+
+class RMSNorm(nn.Module):
+  def __init__(self, dim: int, eps: float):
+    super().__init__()
+    self.weight = nn.Parameter(torch.ones(dim))
+    self.eps = eps
+
+  def forward(self, x):
+    return ops.rmsnorm(x, self.weight, self.eps)
+
+
+class Attention(nn.Module):
+  def __init__(self, cfg: ModelConfig):
+    super().__init__()
+    self.cfg = cfg
+    H, KVH, hd, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim, cfg.dim
+    self.q_proj = nn.Linear(D, H * hd, bias=cfg.attn_bias)
+    self.k_proj = nn.Linear(D, KVH * hd, bias=cfg.attn_bias)
+    self.v_proj = nn.Linear(D, KVH * hd, bias=cfg.attn_bias)
+    self.o_proj = nn.Linear(H * hd, D, bias=False)
+
+  def forward(self, x, cos, sin, positions, k_cache, v_cache, start_pos: int, is_decode: bool, seq_lens=None):
+    B, S, _ = x.shape
+    cfg = self.cfg
+    q = self.q_proj(x).view(B, S, cfg.n_heads, cfg.head_dim)
+    k = self.k_proj(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+    v = self.v_proj(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+    q = ops.rope_kv_append(q, k, v, cos, sin, positions, k_cache, v_cache, start_pos)
+    if is_decode:
+      sl = seq_lens if seq_lens is not None else start_pos + 1
+      out = ops.attn_decode(q, k_cache, v_cache, sl)
+    else:
+      out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S)
+    return self.o_proj(out.reshape(B, S, cfg.n_heads * cfg.head_dim))
+
+
+class MLP(nn.Module):
+  def __init__(self, cfg: ModelConfig):
+    super().__init__()
+    self.gate_proj = nn.Linear(cfg.dim, cfg.intermediate_dim, bias=False)
+    self.up_proj = nn.Linear(cfg.dim, cfg.intermediate_dim, bias=False)
+    self.down_proj = nn.Linear(cfg.intermediate_dim, cfg.dim, bias=False)
+
+  def forward(self, x):
+    return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class MoEMLP(nn.Module):
+  """Mixtral-class sparse MoE block (top-k routed SwiGLU experts)."""
+
+  def __init__(self, cfg: ModelConfig):
+    super().__init__()
+    self.n_experts = cfg.n_experts
+    self.top_k = cfg.n_experts_per_tok
+    self.gate = nn.Linear(cfg.dim, cfg.n_experts, bias=False)
+    self.experts = nn.ModuleList([MLP(cfg) for _ in range(cfg.n_experts)])
+
+  def forward(self, x):
+    B, S, D = x.shape
+    flat = x.view(-1, D)
+    router = self.gate(flat).float()
+    weights, selected = torch.topk(torch.softmax(router, dim=-1), self.top_k, dim=-1)
+    weights = weights / weights.sum(dim=-1, keepdim=True)
+    out = torch.zeros_like(flat, dtype=torch.float32)
+    for e in range(self.n_experts):
+      token_idx, k_idx = torch.where(selected == e)
+      if token_idx.numel() == 0:
+        continue
+      expert_out = self.experts[e](flat[token_idx]).float()
+      out.index_add_(0, token_idx, expert_out * weights[token_idx, k_idx, None])
+    return out.view(B, S, D).to(x.dtype)
+
+
+class DecoderLayer(nn.Module):
+  def __init__(self, cfg: ModelConfig):
+    super().__init__()
+    self.input_layernorm = RMSNorm(cfg.dim, cfg.norm_eps)
+    self.self_attn = Attention(cfg)
+    self.post_attention_layernorm = RMSNorm(cfg.dim, cfg.norm_eps)
+    self.mlp = MoEMLP(cfg) if cfg.n_experts > 0 else MLP(cfg)
+    self.eps = cfg.norm_eps
+
+  def forward(self, x, cos, sin, positions, k_cache, v_cache, start_pos, is_decode, seq_lens=None):
+    attn_out = self.self_attn(
+      self.input_layernorm(x), cos, sin, positions, k_cache, v_cache, start_pos, is_decode, seq_lens
+    )
+    # fused residual-add + norm: h = x + attn_out; normed = rmsnorm(h)
+    normed, h = ops.rmsnorm_residual(attn_out, x, self.post_attention_layernorm.weight, self.eps)
+    return h + self.mlp(normed)
+
+
+class ShardedModel(nn.Module):
+  """The layer range [shard.start_layer .. shard.end_layer] of one model."""
+
+  def __init__(self, cfg: ModelConfig, shard: Shard):
+    super().__init__()
+    self.cfg = cfg
+    self.shard = shard
+    if shard.is_first_layer:
+      self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.dim)
+    self.layers = nn.ModuleDict({str(i): DecoderLayer(cfg) for i in range(shard.start_layer, shard.end_layer + 1)})
+    if shard.is_last_layer:
+      self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+      if not cfg.tie_word_embeddings:
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+      elif not shard.is_first_layer:
+        # tied embeddings but the first shard (owner of embed_tokens) is
+        # elsewhere: keep a local copy of the embedding matrix as the head.
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+    cos, sin = rope_cos_sin(cfg.head_dim, cfg.max_seq_len, cfg.rope_theta, cfg.rope_scaling)
+    self.register_buffer("rope_cos", cos, persistent=False)
+    self.register_buffer("rope_sin", sin, persistent=False)
+
+  @property
+  def local_layer_ids(self) -> List[int]:
+    return list(range(self.shard.start_layer, self.shard.end_layer + 1))
+
+  def head_weight(self):
+    if self.cfg.tie_word_embeddings and hasattr(self, "embed_tokens"):
+      return self.embed_tokens.weight
+    return self.lm_head.weight
+
+  def forward(
+    self,
+    x: torch.Tensor,
+    caches: List[Tuple[torch.Tensor, torch.Tensor]],
+    positions: torch.Tensor,
+    start_pos: int,
+    is_decode: bool = False,
+    seq_lens: Optional[torch.Tensor] = None,
+    last_only: bool = True,
+  ) -> torch.Tensor:
+    """Run this shard.
+
+    x: [B,S] int tokens (first shard) or [B,S,D] hidden states.
+    caches: one (k_cache, v_cache) pair per LOCAL layer.
+    positions: absolute position ids [S] or [B,S] (device tensor).
+    Returns hidden [B,S,D] for non-last shards; logits for the last
+    ([B,V] when is_decode/last_only, else [B,S,V]).
+    """
+    if x.dtype in (torch.int32, torch.int64):
+      assert self.shard.is_first_layer, "token input requires the first shard"
+      h = self.embed_tokens(x)
+    else:
+      h = x
+    cos, sin = self.rope_cos, self.rope_sin
+    for idx, lid in enumerate(self.local_layer_ids):
+      k_cache, v_cache = caches[idx]
+      h = self.layers[str(lid)](h, cos, sin, positions, k_cache, v_cache, start_pos, is_decode, seq_lens)
+    if not self.shard.is_last_layer:
+      return h
+    if last_only and h.shape[1] > 1:
+      h = h[:, -1:, :]
+    h = self.norm(h)
+    logits = torch.nn.functional.linear(h, self.head_weight().to(h.dtype))
+    if is_decode or last_only:
+      return logits[:, -1, :]
+    return logits
+
+
+def hf_key_map(shard: Shard, cfg: ModelConfig):
+  """Map HF checkpoint keys → this ShardedModel's state-dict keys (shard-aware).
+
+  Returns a dict {hf_key: our_key}; HF keys outside the shard map to None.
+  Naming mirrors HF so this is near-identity (no q/k permute —
+  we use HF's rotate-half RoPE directly).
+  """
+  mapping = {}
+  if shard.is_first_layer:
+    mapping["model.embed_tokens.weight"] = "embed_tokens.weight"
+  if shard.is_last_layer:
+    mapping["model.norm.weight"] = "norm.weight"
+    if not cfg.tie_word_embeddings:
+      mapping["lm_head.weight"] = "lm_head.weight"
+    elif not shard.is_first_layer:
+      mapping["model.embed_tokens.weight"] = "lm_head.weight"
+  for lid in range(shard.start_layer, shard.end_layer + 1):
+    hf = f"model.layers.{lid}."
+    ours = f"layers.{lid}."
+    for sub in (
+      "self_attn.q_proj.weight", "self_attn.k_proj.weight", "self_attn.v_proj.weight",
+      "self_attn.o_proj.weight", "self_attn.q_proj.bias", "self_attn.k_proj.bias",
+      "self_attn.v_proj.bias", "mlp.gate_proj.weight", "mlp.up_proj.weight",
+      "mlp.down_proj.weight", "input_layernorm.weight", "post_attention_layernorm.weight",
+    ):
+      mapping[hf + sub] = ours + sub
+    if cfg.n_experts > 0:
+      mapping[hf + "block_sparse_moe.gate.weight"] = ours + "mlp.gate.weight"
+      for e in range(cfg.n_experts):
+        mapping[hf + f"block_sparse_moe.experts.{e}.w1.weight"] = ours + f"mlp.experts.{e}.gate_proj.weight"
+        mapping[hf + f"block_sparse_moe.experts.{e}.w3.weight"] = ours + f"mlp.experts.{e}.up_proj.weight"
+        mapping[hf + f"block_sparse_moe.experts.{e}.w2.weight"] = ours + f"mlp.experts.{e}.down_proj.weight"
+  return mapping

@@ -1,0 +1,192 @@
+"""Model registry: model cards, pretty names, shard builders.
+
+Capability parity with the reference's registry
+(/root/reference/xotorch/models.py:4-278): id → layer count + HF repo +
+per-engine support, plus `build_base_shard`/`build_full_shard`.
+
+Additionally each well-known architecture carries a BUILTIN HF-style config
+so models can be constructed with random-init weights offline (the bench and
+tests run with no network — BASELINE.json "synthetic data / random-init").
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+from xotorch_amd.shard import Shard
+
+# --- builtin HF-style configs for offline/random-init construction ----------
+
+def _llama_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=128256, theta=500000.0, tie=False,
+               max_pos=8192, llama3_scaling=False, head_dim=None):
+  cfg = {
+    "model_type": "llama", "hidden_size": dim, "num_hidden_layers": n_layers,
+    "num_attention_heads": n_heads, "num_key_value_heads": n_kv,
+    "intermediate_size": inter, "vocab_size": vocab, "rope_theta": theta,
+    "rms_norm_eps": 1e-5, "max_position_embeddings": max_pos,
+    "tie_word_embeddings": tie, "torch_dtype": "bfloat16",
+    "bos_token_id": 128000, "eos_token_id": 128001,
+  }
+  if head_dim:
+    cfg["head_dim"] = head_dim
+  if llama3_scaling:
+    cfg["rope_scaling"] = {
+      "rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+      "high_freq_factor": 4.0, "original_max_position_embeddings": 8192,
+    }
+    cfg["max_position_embeddings"] = 131072
+  return cfg
+
+
+def _qwen_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=151936, tie=False, max_pos=32768):
+  return {
+    "model_type": "qwen2", "hidden_size": dim, "num_hidden_layers": n_layers,
+    "num_attention_heads": n_heads, "num_key_value_heads": n_kv,
+    "intermediate_size": inter, "vocab_size": vocab, "rope_theta": 1000000.0,
+    "rms_norm_eps": 1e-6, "max_position_embeddings": max_pos,
+    "tie_word_embeddings": tie, "torch_dtype": "bfloat16",
+    "bos_token_id": 151643, "eos_token_id": 151645,
+  }
+
+
+def _mixtral_cfg(dim, n_layers, n_heads, n_kv, inter, n_experts=8, top_k=2):
+  return {
+    "model_type": "mixtral", "hidden_size": dim, "num_hidden_layers": n_layers,
+    "num_attention_heads": n_heads, "num_key_value_heads": n_kv,
+    "intermediate_size": inter, "vocab_size": 32000, "rope_theta": 1000000.0,
+    "rms_norm_eps": 1e-5, "max_position_embeddings": 32768,
+    "num_local_experts": n_experts, "num_experts_per_tok": top_k,
+    "torch_dtype": "bfloat16", "bos_token_id": 1, "eos_token_id": 2,
+  }
+
+
+BUILTIN_CONFIGS: Dict[str, dict] = {
+  # llama 3.x family
+  "llama-3.2-1b": _llama_cfg(2048, 16, 32, 8, 8192, tie=True, llama3_scaling=True, head_dim=64),
+  "llama-3.2-3b": _llama_cfg(3072, 28, 24, 8, 8192, tie=True, llama3_scaling=True, head_dim=128),
+  "llama-3.1-8b": _llama_cfg(4096, 32, 32, 8, 14336, llama3_scaling=True),
+  "llama-3-8b": _llama_cfg(4096, 32, 32, 8, 14336),
+  "llama-3-70b": _llama_cfg(8192, 80, 64, 8, 28672),
+  "llama-3.1-70b": _llama_cfg(8192, 80, 64, 8, 28672, llama3_scaling=True),
+  "llama-3.3-70b": _llama_cfg(8192, 80, 64, 8, 28672, llama3_scaling=True),
+  "llama-3.1-405b": _llama_cfg(16384, 126, 128, 8, 53248, llama3_scaling=True),
+  # qwen 2.5 family
+  "qwen-2.5-0.5b": _qwen_cfg(896, 24, 14, 2, 4864, tie=True),
+  "qwen-2.5-1.5b": _qwen_cfg(1536, 28, 12, 2, 8960, tie=True),
+  "qwen-2.5-3b": _qwen_cfg(2048, 36, 16, 2, 11008, tie=True),
+  "qwen-2.5-7b": _qwen_cfg(3584, 28, 28, 4, 18944),
+  "qwen-2.5-14b": _qwen_cfg(5120, 48, 40, 8, 13824),
+  "qwen-2.5-32b": _qwen_cfg(5120, 64, 40, 8, 27648),
+  "qwen-2.5-72b": _qwen_cfg(8192, 80, 64, 8, 29568),
+  # mistral
+  "mistral-nemo": _llama_cfg(5120, 40, 32, 8, 14336, vocab=131072, theta=1000000.0, max_pos=128000, head_dim=128),
+  "mistral-large": _llama_cfg(12288, 88, 96, 8, 28672, vocab=32768, theta=1000000.0, max_pos=32768),
+  # deepseek r1 distills (qwen/llama backbones)
+  "deepseek-r1-distill-qwen-1.5b": _qwen_cfg(1536, 28, 12, 2, 8960, tie=True),
+  "deepseek-r1-distill-qwen-7b": _qwen_cfg(3584, 28, 28, 4, 18944),
+  "deepseek-r1-distill-qwen-32b": _qwen_cfg(5120, 64, 40, 8, 27648),
+  "deepseek-r1-distill-llama-8b": _llama_cfg(4096, 32, 32, 8, 14336, llama3_scaling=True),
+  "deepseek-r1-distill-llama-70b": _llama_cfg(8192, 80, 64, 8, 28672, llama3_scaling=True),
+  # moe
+  "mixtral-8x7b": _mixtral_cfg(4096, 32, 32, 8, 14336),
+  # phi-4-mini (llama-like enough for the generic decoder)
+  "phi-4-mini": _llama_cfg(3072, 32, 24, 8, 8192, vocab=200064, theta=10000.0, max_pos=131072, tie=True),
+  # tiny test model
+  "dummy": _llama_cfg(64, 4, 4, 2, 128, vocab=256, theta=10000.0, max_pos=256, tie=True),
+}
+
+
+# model cards: layers + HF repo per engine (engine names of THIS framework)
+model_cards: Dict[str, dict] = {
+  "llama-3.2-1b": {"layers": 16, "repo": {"TorchEngine": "unsloth/Llama-3.2-1B-Instruct", "HIPEngine": "unsloth/Llama-3.2-1B-Instruct"}},
+  "llama-3.2-3b": {"layers": 28, "repo": {"TorchEngine": "unsloth/Llama-3.2-3B-Instruct", "HIPEngine": "unsloth/Llama-3.2-3B-Instruct"}},
+  "llama-3.1-8b": {"layers": 32, "repo": {"TorchEngine": "mlx-community/Meta-Llama-3.1-8B-Instruct-bf16", "HIPEngine": "mlx-community/Meta-Llama-3.1-8B-Instruct-bf16"}},
+  "llama-3-8b": {"layers": 32, "repo": {"TorchEngine": "NousResearch/Meta-Llama-3-8B-Instruct", "HIPEngine": "NousResearch/Meta-Llama-3-8B-Instruct"}},
+  "llama-3-70b": {"layers": 80, "repo": {"TorchEngine": "NousResearch/Meta-Llama-3-70B-Instruct", "HIPEngine": "NousResearch/Meta-Llama-3-70B-Instruct"}},
+  "llama-3.1-70b": {"layers": 80, "repo": {"TorchEngine": "mlx-community/Meta-Llama-3.1-70B-Instruct", "HIPEngine": "mlx-community/Meta-Llama-3.1-70B-Instruct"}},
+  "llama-3.3-70b": {"layers": 80, "repo": {"TorchEngine": "unsloth/Llama-3.3-70B-Instruct", "HIPEngine": "unsloth/Llama-3.3-70B-Instruct"}},
+  "llama-3.1-405b": {"layers": 126, "repo": {"TorchEngine": "unsloth/Meta-Llama-3.1-405B-Instruct", "HIPEngine": "unsloth/Meta-Llama-3.1-405B-Instruct"}},
+  "qwen-2.5-0.5b": {"layers": 24, "repo": {"TorchEngine": "Qwen/Qwen2.5-0.5B-Instruct", "HIPEngine": "Qwen/Qwen2.5-0.5B-Instruct"}},
+  "qwen-2.5-1.5b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen2.5-1.5B-Instruct", "HIPEngine": "Qwen/Qwen2.5-1.5B-Instruct"}},
+  "qwen-2.5-3b": {"layers": 36, "repo": {"TorchEngine": "Qwen/Qwen2.5-3B-Instruct", "HIPEngine": "Qwen/Qwen2.5-3B-Instruct"}},
+  "qwen-2.5-7b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen2.5-7B-Instruct", "HIPEngine": "Qwen/Qwen2.5-7B-Instruct"}},
+  "qwen-2.5-14b": {"layers": 48, "repo": {"TorchEngine": "Qwen/Qwen2.5-14B-Instruct", "HIPEngine": "Qwen/Qwen2.5-14B-Instruct"}},
+  "qwen-2.5-32b": {"layers": 64, "repo": {"TorchEngine": "Qwen/Qwen2.5-32B-Instruct", "HIPEngine": "Qwen/Qwen2.5-32B-Instruct"}},
+  "qwen-2.5-72b": {"layers": 80, "repo": {"TorchEngine": "Qwen/Qwen2.5-72B-Instruct", "HIPEngine": "Qwen/Qwen2.5-72B-Instruct"}},
+  "mistral-nemo": {"layers": 40, "repo": {"TorchEngine": "unsloth/Mistral-Nemo-Instruct-2407", "HIPEngine": "unsloth/Mistral-Nemo-Instruct-2407"}},
+  "mistral-large": {"layers": 88, "repo": {"TorchEngine": "mistralai/Mistral-Large-Instruct-2407", "HIPEngine": "mistralai/Mistral-Large-Instruct-2407"}},
+  "deepseek-r1-distill-qwen-1.5b": {"layers": 28, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-1.5B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-1.5B"}},
+  "deepseek-r1-distill-qwen-7b": {"layers": 28, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-7B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-7B"}},
+  "deepseek-r1-distill-qwen-32b": {"layers": 64, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-32B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-32B"}},
+  "deepseek-r1-distill-llama-8b": {"layers": 32, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Llama-8B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Llama-8B"}},
+  "deepseek-r1-distill-llama-70b": {"layers": 80, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Llama-70B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Llama-70B"}},
+  "mixtral-8x7b": {"layers": 32, "repo": {"TorchEngine": "mistralai/Mixtral-8x7B-Instruct-v0.1", "HIPEngine": "mistralai/Mixtral-8x7B-Instruct-v0.1"}},
+  "phi-4-mini": {"layers": 32, "repo": {"TorchEngine": "microsoft/Phi-4-mini-instruct", "HIPEngine": "microsoft/Phi-4-mini-instruct"}},
+  "dummy": {"layers": 4, "repo": {"TorchEngine": "dummy", "HIPEngine": "dummy", "DummyEngine": "dummy"}},
+}
+
+pretty_names = {
+  "llama-3.2-1b": "Llama 3.2 1B",
+  "llama-3.2-3b": "Llama 3.2 3B",
+  "llama-3.1-8b": "Llama 3.1 8B",
+  "llama-3-8b": "Llama 3 8B",
+  "llama-3-70b": "Llama 3 70B",
+  "llama-3.1-70b": "Llama 3.1 70B",
+  "llama-3.3-70b": "Llama 3.3 70B",
+  "llama-3.1-405b": "Llama 3.1 405B",
+  "qwen-2.5-0.5b": "Qwen 2.5 0.5B",
+  "qwen-2.5-1.5b": "Qwen 2.5 1.5B",
+  "qwen-2.5-3b": "Qwen 2.5 3B",
+  "qwen-2.5-7b": "Qwen 2.5 7B",
+  "qwen-2.5-14b": "Qwen 2.5 14B",
+  "qwen-2.5-32b": "Qwen 2.5 32B",
+  "qwen-2.5-72b": "Qwen 2.5 72B",
+  "mistral-nemo": "Mistral Nemo 12B",
+  "mistral-large": "Mistral Large 123B",
+  "deepseek-r1-distill-qwen-1.5b": "DeepSeek R1 Distill Qwen 1.5B",
+  "deepseek-r1-distill-qwen-7b": "DeepSeek R1 Distill Qwen 7B",
+  "deepseek-r1-distill-qwen-32b": "DeepSeek R1 Distill Qwen 32B",
+  "deepseek-r1-distill-llama-8b": "DeepSeek R1 Distill Llama 8B",
+  "deepseek-r1-distill-llama-70b": "DeepSeek R1 Distill Llama 70B",
+  "mixtral-8x7b": "Mixtral 8x7B",
+  "phi-4-mini": "Phi-4 Mini",
+  "dummy": "Dummy (test)",
+}
+
+
+def pretty_name(model_id: str) -> str:
+  return pretty_names.get(model_id, model_id)
+
+
+def get_repo(model_id: str, engine_classname: str) -> Optional[str]:
+  return model_cards.get(model_id, {}).get("repo", {}).get(engine_classname)
+
+
+def build_base_shard(model_id: str, engine_classname: str = "HIPEngine") -> Optional[Shard]:
+  """Shard covering layer 0 only — the placeholder routed to the partitioner."""
+  n_layers = model_cards.get(model_id, {}).get("layers", 0)
+  if n_layers < 1:
+    return None
+  return Shard(model_id=model_id, start_layer=0, end_layer=0, n_layers=n_layers)
+
+
+def build_full_shard(model_id: str, engine_classname: str = "HIPEngine") -> Optional[Shard]:
+  base = build_base_shard(model_id, engine_classname)
+  if base is None:
+    return None
+  return Shard(model_id=model_id, start_layer=0, end_layer=base.n_layers - 1, n_layers=base.n_layers)
+
+
+def get_supported_models(supported_engine_lists: Optional[List[List[str]]] = None) -> List[str]:
+  """Model ids runnable by at least one engine in EVERY peer's engine list."""
+  if not supported_engine_lists:
+    return list(model_cards.keys())
+  out = []
+  for model_id, card in model_cards.items():
+    repos = card.get("repo", {})
+    if all(any(e in repos for e in engines) for engines in supported_engine_lists):
+      out.append(model_id)
+  return out
+
+
+def builtin_config(model_id: str) -> Optional[dict]:
+  return BUILTIN_CONFIGS.get(model_id)

@@ -1,0 +1,180 @@
+"""Pure-PyTorch reference implementations of every hot op.
+
+These are the numerics oracle for the HIP kernels (tests compare the HIP path
+against these in fp32) and the CPU execution path for the plumbing tests.
+They implement the same ops the reference delegates to torchtune
+(SURVEY.md §2.2 table), with HF "rotate-half" RoPE convention so HF
+safetensors load without the q/k permute torchtune needs
+(/root/reference/xotorch/inference/torch/llm_utils.py:175-183).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+  """RMSNorm in fp32 math, cast back to input dtype."""
+  xf = x.float()
+  norm = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+  return (norm * weight.float()).to(x.dtype)
+
+
+def rmsnorm_residual(
+  x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+) -> Tuple[torch.Tensor, torch.Tensor]:
+  """Fused residual-add + RMSNorm: returns (norm(x+residual), x+residual)."""
+  s = (x.float() + residual.float())
+  norm = s * torch.rsqrt(s.pow(2).mean(-1, keepdim=True) + eps)
+  return (norm * weight.float()).to(x.dtype), s.to(x.dtype)
+
+
+def rope_cos_sin(
+  head_dim: int,
+  max_seq_len: int,
+  theta: float,
+  scaling=None,
+  dtype: torch.dtype = torch.float32,
+  device="cpu",
+) -> Tuple[torch.Tensor, torch.Tensor]:
+  """Precompute RoPE cos/sin tables [max_seq_len, head_dim//2] (fp32).
+
+  `scaling` is an optional llama3-style RopeScaling (models/config.py).
+  """
+  inv_freq = 1.0 / (theta ** (torch.arange(0, head_dim, 2, dtype=torch.float32, device=device) / head_dim))
+  if scaling is not None and getattr(scaling, "rope_type", "default") == "llama3":
+    # llama3 frequency scaling: scale low-frequency components, smooth the band between.
+    low_freq_wavelen = scaling.original_max_position_embeddings / scaling.low_freq_factor
+    high_freq_wavelen = scaling.original_max_position_embeddings / scaling.high_freq_factor
+    wavelen = 2 * math.pi / inv_freq
+    scaled = torch.where(wavelen > low_freq_wavelen, inv_freq / scaling.factor, inv_freq)
+    smooth = (scaling.original_max_position_embeddings / wavelen - scaling.low_freq_factor) / (
+      scaling.high_freq_factor - scaling.low_freq_factor
+    )
+    # the medium band interpolates from the UNSCALED frequency
+    smoothed = (1 - smooth) * inv_freq / scaling.factor + smooth * inv_freq
+    is_medium = (wavelen >= high_freq_wavelen) & (wavelen <= low_freq_wavelen)
+    inv_freq = torch.where(is_medium, smoothed, scaled)
+  t = torch.arange(max_seq_len, dtype=torch.float32, device=device)
+  freqs = torch.outer(t, inv_freq)
+  return freqs.cos().to(dtype), freqs.sin().to(dtype)
+
+
+def _rotate_half(x: torch.Tensor) -> torch.Tensor:
+  half = x.shape[-1] // 2
+  x1, x2 = x[..., :half], x[..., half:]
+  return torch.cat((-x2, x1), dim=-1)
+
+
+def rope_apply(
+  q: torch.Tensor,
+  k: torch.Tensor,
+  cos: torch.Tensor,
+  sin: torch.Tensor,
+  positions: torch.Tensor,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+  """Apply HF rotate-half RoPE.
+
+  q: [B, S, H, hd], k: [B, S, KVH, hd]; cos/sin: [max_seq, hd//2];
+  positions: [S] or [B, S] absolute position ids.
+  """
+  c = cos[positions].float()  # [S, hd//2] or [B,S,hd//2]
+  s = sin[positions].float()
+  if c.dim() == 2:
+    c = c.unsqueeze(0)
+    s = s.unsqueeze(0)
+  c = torch.cat([c, c], dim=-1).unsqueeze(2)  # [B,S,1,hd]
+  s = torch.cat([s, s], dim=-1).unsqueeze(2)
+  qf, kf = q.float(), k.float()
+  q_out = qf * c + _rotate_half(qf) * s
+  k_out = kf * c + _rotate_half(kf) * s
+  return q_out.to(q.dtype), k_out.to(k.dtype)
+
+
+def kv_append(
+  k_cache: torch.Tensor,
+  v_cache: torch.Tensor,
+  k: torch.Tensor,
+  v: torch.Tensor,
+  start_pos: int,
+) -> None:
+  """Append k,v ([B, S, KVH, hd]) into caches laid out [B, KVH, T, hd] at start_pos."""
+  S = k.shape[1]
+  k_cache[:, :, start_pos:start_pos + S, :] = k.transpose(1, 2).to(k_cache.dtype)
+  v_cache[:, :, start_pos:start_pos + S, :] = v.transpose(1, 2).to(v_cache.dtype)
+
+
+def attn_prefill(
+  q: torch.Tensor,
+  k_cache: torch.Tensor,
+  v_cache: torch.Tensor,
+  start_pos: int,
+  s_len: int,
+) -> torch.Tensor:
+  """Causal GQA attention for a prefill chunk.
+
+  q: [B, S, H, hd] (post-RoPE). Caches [B, KVH, T, hd] already contain keys
+  through start_pos + s_len. Query position i attends to cache[0..start_pos+i].
+  Returns [B, S, H, hd].
+  """
+  B, S, H, hd = q.shape
+  KVH = k_cache.shape[1]
+  total = start_pos + s_len
+  qh = q.transpose(1, 2)  # [B,H,S,hd]
+  keys = k_cache[:, :, :total].repeat_interleave(H // KVH, dim=1)
+  vals = v_cache[:, :, :total].repeat_interleave(H // KVH, dim=1)
+  # mask: query i (absolute pos start_pos+i) sees keys 0..start_pos+i
+  mask = torch.ones(S, total, dtype=torch.bool, device=q.device).tril(diagonal=start_pos)
+  out = F.scaled_dot_product_attention(
+    qh.float(), keys.float(), vals.float(), attn_mask=mask
+  )
+  return out.transpose(1, 2).to(q.dtype)
+
+
+def attn_decode(
+  q: torch.Tensor,
+  k_cache: torch.Tensor,
+  v_cache: torch.Tensor,
+  seq_len: int,
+) -> torch.Tensor:
+  """Single-position GQA attention against the KV cache.
+
+  q: [B, 1, H, hd]; caches [B, KVH, T, hd] valid through seq_len.
+  Returns [B, 1, H, hd].
+  """
+  B, _, H, hd = q.shape
+  KVH = k_cache.shape[1]
+  qh = q.transpose(1, 2).float()  # [B,H,1,hd]
+  keys = k_cache[:, :, :seq_len].repeat_interleave(H // KVH, dim=1).float()
+  vals = v_cache[:, :, :seq_len].repeat_interleave(H // KVH, dim=1).float()
+  out = F.scaled_dot_product_attention(qh, keys, vals)
+  return out.transpose(1, 2).to(q.dtype)
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+  """silu(gate) * up, fp32 internally."""
+  return (F.silu(gate.float()) * up.float()).to(gate.dtype)
+
+
+def softmax_sample(
+  logits: torch.Tensor,
+  temperature: float = 0.0,
+  top_k: int = 0,
+  generator: Optional[torch.Generator] = None,
+) -> torch.Tensor:
+  """Temperature + top-k sampling via the exponential trick; temp 0 → argmax.
+
+  logits: [B, V] → token ids [B].
+  """
+  if temperature <= 0.0:
+    return logits.argmax(dim=-1)
+  logits = logits.float() / temperature
+  if top_k and top_k > 0 and top_k < logits.shape[-1]:
+    kth = torch.topk(logits, top_k, dim=-1).values[..., -1, None]
+    logits = torch.where(logits < kth, torch.full_like(logits, float("-inf")), logits)
+  probs = torch.softmax(logits, dim=-1)
+  q = torch.empty_like(probs).exponential_(1, generator=generator)
+  return (probs / q).argmax(dim=-1)
