@@ -1,0 +1,119 @@
+"""TorchEngine end-to-end on CPU with a random-init tiny model (offline)."""
+import asyncio
+
+import numpy as np
+import pytest
+import torch
+
+from xotorch_amd.engine.dummy import DummyEngine
+from xotorch_amd.engine.torch_engine import TorchEngine
+from xotorch_amd.models.registry import build_full_shard
+from xotorch_amd.shard import Shard
+
+
+def run(coro):
+  return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def test_engine_prefill_decode_loop():
+  async def go():
+    eng = TorchEngine(device="cpu", dtype=torch.float32)
+    shard = build_full_shard("dummy", "TorchEngine")
+    out, state = await eng.infer_prompt("req1", shard, "hello world test prompt")
+    assert out.shape[-1] == 256  # dummy vocab
+    tok = await eng.sample(out, temp=0.0)
+    # 3 decode steps
+    for _ in range(3):
+      out, state = await eng.infer_tensor("req1", shard, tok.reshape(1, 1), state)
+      assert np.isfinite(out).all()
+      tok = await eng.sample(out, temp=0.0)
+    assert state["curr_pos"] > 4
+    return True
+  assert run(go())
+
+
+def test_engine_sharded_ring_in_process():
+  """Two shard engines chained in-process: hidden state hop between them."""
+  async def go():
+    n_layers = 4
+    sh0 = Shard("dummy", 0, 1, n_layers)
+    sh1 = Shard("dummy", 2, 3, n_layers)
+    e0 = TorchEngine(device="cpu", dtype=torch.float32)
+    e1 = TorchEngine(device="cpu", dtype=torch.float32)
+    full = TorchEngine(device="cpu", dtype=torch.float32)
+    prompt = "the quick brown fox"
+    h, s0 = await e0.infer_prompt("r", sh0, prompt)
+    assert h.ndim == 3  # hidden state
+    logits, s1 = await e1.infer_tensor("r", sh1, h, s0)
+    ref, _ = await full.infer_prompt("r", build_full_shard("dummy", "TorchEngine"), prompt)
+    assert np.allclose(logits, ref, atol=1e-4)
+    return True
+  assert run(go())
+
+
+def test_engine_train_last_shard():
+  async def go():
+    eng = TorchEngine(device="cpu", dtype=torch.float32)
+    shard = build_full_shard("dummy", "TorchEngine")
+    B, S = 2, 8
+    inputs = np.random.randint(0, 200, (B, S))
+    targets = np.roll(inputs, -1, axis=1)
+    lengths = np.array([S, S - 2])
+    loss1, _ = await eng.train("t", shard, inputs, targets, lengths)
+    loss2, _ = await eng.evaluate("t", shard, inputs, targets, lengths)
+    assert np.isfinite(loss1) and np.isfinite(loss2)
+    # a few steps reduce loss on a fixed batch
+    for _ in range(8):
+      loss3, _ = await eng.train("t", shard, inputs, targets, lengths)
+    assert loss3 < loss1
+    return True
+  assert run(go())
+
+
+def test_engine_ring_train_back_gradient():
+  """Pipeline training: last shard CE + gradient hop back to the first shard."""
+  async def go():
+    n_layers = 4
+    sh0 = Shard("dummy", 0, 1, n_layers)
+    sh1 = Shard("dummy", 2, 3, n_layers)
+    e0 = TorchEngine(device="cpu", dtype=torch.float32)
+    e1 = TorchEngine(device="cpu", dtype=torch.float32)
+    B, S = 2, 6
+    inputs = np.random.randint(0, 200, (B, S))
+    targets = np.roll(inputs, -1, axis=1)
+    lengths = np.array([S, S])
+    # forward through shard0 (training-mode forward returns hidden)
+    h, _ = await e0.infer_tensor("t", sh0, inputs, {"curr_pos": 0})
+    loss, back_grad = await e1.train("t", sh1, h, targets, lengths)
+    assert np.isfinite(loss) and back_grad is not None
+    assert back_grad.shape == h.shape
+    loss0, _ = await e0.train("t", sh0, inputs, back_grad, lengths, loss="back_gradient")
+    return True
+  assert run(go())
+
+
+def test_engine_checkpoint_roundtrip(tmp_path):
+  async def go():
+    eng = TorchEngine(device="cpu", dtype=torch.float32)
+    shard = build_full_shard("dummy", "TorchEngine")
+    await eng.ensure_shard(shard)
+    p = str(tmp_path / "ckpt.safetensors")
+    await eng.save_checkpoint(shard, p)
+    w_before = eng.model.layers["0"].self_attn.q_proj.weight.clone()
+    with torch.no_grad():
+      eng.model.layers["0"].self_attn.q_proj.weight.add_(1.0)
+    await eng.load_checkpoint(shard, p)
+    assert torch.allclose(eng.model.layers["0"].self_attn.q_proj.weight, w_before)
+    return True
+  assert run(go())
+
+
+def test_dummy_engine_plumbing():
+  async def go():
+    eng = DummyEngine()
+    shard = build_full_shard("dummy", "DummyEngine")
+    out, state = await eng.infer_prompt("r", shard, "hi")
+    tok = await eng.sample(out)
+    assert tok.shape[0] == 1
+    return True
+  assert run(go())

@@ -170,10 +170,10 @@ class TorchEngine(InferenceEngine):
     S = x.shape[1]
     total_len = meta.total_len or min(self.cfg.max_seq_len, S + DEFAULT_MAX_GEN)
     sess = self._session(request_id, B, total_len)
-    if meta.curr_pos == 0 and not is_tokens and inference_state is not None:
-      # mid-ring stage receiving a fresh prefill: state carries positions
-      pass
-    start_pos = meta.curr_pos if inference_state is not None else sess.state.curr_pos
+    # Each stage tracks its OWN position: every stage of the ring processes
+    # the same chunks in the same order, so the local counter is always the
+    # chunk's start. (Position state never travels — SURVEY.md §2.4.)
+    start_pos = sess.state.curr_pos
     x = x.to(self.device)
     if not is_tokens:
       x = x.to(self.dtype)

@@ -79,9 +79,14 @@ def random_init(model: ShardedModel, seed: int = 1234, std: float = 0.02) -> Non
   """
   import zlib
   for name, p in model.named_parameters():
+    # a tied-embedding last shard without the embedding holds a COPY of the
+    # embedding matrix as its head: seed it as the embedding so split == full
+    seed_name = name
+    if name == "lm_head.weight" and model.cfg.tie_word_embeddings:
+      seed_name = "embed_tokens.weight"
     g = torch.Generator(device="cpu")
     # stable across processes (python's str hash is salted per process)
-    g.manual_seed((seed + zlib.crc32(name.encode())) % (2**63))
+    g.manual_seed((seed + zlib.crc32(seed_name.encode())) % (2**63))
     cpu_t = torch.empty(p.shape, dtype=torch.float32)
     if name.endswith("layernorm.weight") or name.endswith("norm.weight"):
       cpu_t.fill_(1.0)

@@ -81,6 +81,8 @@ def rope_apply(
   q: [B, S, H, hd], k: [B, S, KVH, hd]; cos/sin: [max_seq, hd//2];
   positions: [S] or [B, S] absolute position ids.
   """
+  if isinstance(positions, torch.Tensor) and positions.dtype != torch.long:
+    positions = positions.long()
   c = cos[positions].float()  # [S, hd//2] or [B,S,hd//2]
   s = sin[positions].float()
   if c.dim() == 2:
