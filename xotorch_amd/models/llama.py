@@ -139,6 +139,17 @@ class ShardedModel(nn.Module):
     self.register_buffer("rope_cos", cos, persistent=False)
     self.register_buffer("rope_sin", sin, persistent=False)
 
+  def _apply(self, fn, recurse=True):
+    # keep RoPE tables fp32 (HIP kernel contract): a model-wide .to(bf16)
+    # would quantize them; recompute at full precision on the new device.
+    super()._apply(fn, recurse)
+    if self.rope_cos.dtype != torch.float32 or not self.rope_cos.is_contiguous():
+      cos, sin = rope_cos_sin(self.cfg.head_dim, self.cfg.max_seq_len, self.cfg.rope_theta,
+                              self.cfg.rope_scaling, device=self.rope_cos.device)
+      self.rope_cos = cos
+      self.rope_sin = sin
+    return self
+
   @property
   def local_layer_ids(self) -> List[int]:
     return list(range(self.shard.start_layer, self.shard.end_layer + 1))
