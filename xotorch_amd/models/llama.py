@@ -189,7 +189,7 @@ class ShardedModel(nn.Module):
     if not self.shard.is_last_layer:
       return h
     if last_only and h.shape[1] > 1:
-      h = h[:, -1:, :]
+      h = h[:, -1:, :].contiguous()  # kernels require contiguous rows
     h = self.norm(h)
     logits = torch.nn.functional.linear(h, self.head_weight().to(h.dtype))
     if is_decode or last_only:

