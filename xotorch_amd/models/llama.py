@@ -139,15 +139,20 @@ class ShardedModel(nn.Module):
     self.register_buffer("rope_cos", cos, persistent=False)
     self.register_buffer("rope_sin", sin, persistent=False)
 
+  def reset_rope(self):
+    """(Re)compute the fp32 RoPE tables on the current device. Required after
+    meta-device construction + to_empty (tables are uninitialized memory)."""
+    cos, sin = rope_cos_sin(self.cfg.head_dim, self.cfg.max_seq_len, self.cfg.rope_theta,
+                            self.cfg.rope_scaling, device=self.rope_cos.device)
+    self.rope_cos = cos
+    self.rope_sin = sin
+
   def _apply(self, fn, recurse=True):
     # keep RoPE tables fp32 (HIP kernel contract): a model-wide .to(bf16)
     # would quantize them; recompute at full precision on the new device.
     super()._apply(fn, recurse)
-    if self.rope_cos.dtype != torch.float32 or not self.rope_cos.is_contiguous():
-      cos, sin = rope_cos_sin(self.cfg.head_dim, self.cfg.max_seq_len, self.cfg.rope_theta,
-                              self.cfg.rope_scaling, device=self.rope_cos.device)
-      self.rope_cos = cos
-      self.rope_sin = sin
+    if self.rope_cos.device.type != "meta" and self.rope_cos.dtype != torch.float32:
+      self.reset_rope()
     return self
 
   @property

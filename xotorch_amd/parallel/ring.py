@@ -91,7 +91,7 @@ class RingPipeline:
       fast_random_init_gpu(model, seed)
     else:
       random_init(model, seed)
-    # _apply recomputes fp32 rope tables on-device
+    model.reset_rope()  # to_empty left the tables uninitialized
     model.eval()
     self.model = model
 
