@@ -99,11 +99,11 @@ def test_engine_checkpoint_roundtrip(tmp_path):
     await eng.ensure_shard(shard)
     p = str(tmp_path / "ckpt.safetensors")
     await eng.save_checkpoint(shard, p)
-    w_before = eng.model.layers["0"].self_attn.q_proj.weight.clone()
+    w_before = eng.model.layers["0"].self_attn.qkv_proj.weight.clone()
     with torch.no_grad():
-      eng.model.layers["0"].self_attn.q_proj.weight.add_(1.0)
+      eng.model.layers["0"].self_attn.qkv_proj.weight.add_(1.0)
     await eng.load_checkpoint(shard, p)
-    assert torch.allclose(eng.model.layers["0"].self_attn.q_proj.weight, w_before)
+    assert torch.allclose(eng.model.layers["0"].self_attn.qkv_proj.weight, w_before)
     return True
   assert run(go())
 

@@ -40,19 +40,19 @@ def test_rmsnorm_residual(hip):
 
 
 @pytest.mark.parametrize("hd,H,KVH,S", [(128, 8, 2, 5), (64, 4, 4, 1), (128, 64, 8, 3)])
-def test_rope_kv_append(hip, hd, H, KVH, S):
+def test_rope_qkv_append(hip, hd, H, KVH, S):
   from xotorch_amd.ops import torch_ref
   B, T, start = 2, 32, 11
-  q, k, v = bt(B, S, H, hd), bt(B, S, KVH, hd, seed=4), bt(B, S, KVH, hd, seed=5)
+  qkv = bt(B, S, (H + 2 * KVH) * hd)
   cos, sin = torch_ref.rope_cos_sin(hd, T, 500000.0, device="cuda")
   kc = torch.zeros(B, KVH, T, hd, dtype=torch.bfloat16, device="cuda")
   vc = torch.zeros_like(kc)
   kc_ref, vc_ref = kc.clone(), vc.clone()
   positions = torch.arange(start, start + S, dtype=torch.int32, device="cuda")
-  q_ref, k_ref = torch_ref.rope_apply(q.clone(), k, cos, sin, positions.long())
-  torch_ref.kv_append(kc_ref, vc_ref, k_ref, v, start)
-  q_out = hip.rope_kv_append(q.clone(), k, v, cos, sin, positions, kc, vc, start)
-  assert torch.allclose(q_out.float(), q_ref.float(), atol=2e-2, rtol=2e-2)
+  qkv_ref = qkv.clone()
+  torch_ref.rope_qkv_append(qkv_ref, cos, sin, positions, kc_ref, vc_ref, H, KVH, hd)
+  hip.rope_qkv_append(qkv, cos, sin, positions, kc, vc, H, KVH, hd)
+  assert torch.allclose(qkv.float(), qkv_ref.float(), atol=2e-2, rtol=2e-2)
   assert torch.allclose(kc.float(), kc_ref.float(), atol=2e-2, rtol=2e-2)
   assert torch.equal(vc, vc_ref)
 
@@ -97,6 +97,28 @@ def test_swiglu(hip):
   out = hip.swiglu(g, u)
   ref = torch_ref.swiglu(g, u)
   assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_swiglu_packed(hip):
+  from xotorch_amd.ops import torch_ref
+  gu = bt(5, 2, 2 * 4864, seed=11)
+  out = hip.swiglu_packed(gu)
+  ref = torch_ref.swiglu_packed(gu)
+  assert out.shape == (5, 2, 4864)
+  assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_attn_decode_strided_q(hip):
+  """q as a strided view into a packed qkv row (the fused-GEMM layout)."""
+  from xotorch_amd.ops import torch_ref
+  B, H, KVH, hd, T, sl = 2, 16, 4, 128, 256, 200
+  qkv = bt(B, 1, (H + 2 * KVH) * hd, seed=12)
+  q = qkv[:, :, : H * hd].view(B, 1, H, hd)
+  kc, vc = bt(B, KVH, T, hd, seed=13), bt(B, KVH, T, hd, seed=14)
+  lens = torch.full((B,), sl, dtype=torch.int32, device="cuda")
+  out = hip.attn_decode(q, kc, vc, lens)
+  ref = torch_ref.attn_decode(q.contiguous(), kc, vc, sl)
+  assert torch.allclose(out.float(), ref.float(), atol=3e-2, rtol=3e-2)
 
 
 def test_model_gpu_vs_cpu():

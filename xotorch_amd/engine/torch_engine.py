@@ -229,9 +229,13 @@ class TorchEngine(InferenceEngine):
       attn = layer.self_attn
       B = h.shape[0]
       cfgm = attn.cfg
-      q = attn.q_proj(normed).view(B, S, cfgm.n_heads, cfgm.head_dim)
-      k = attn.k_proj(normed).view(B, S, cfgm.n_kv_heads, cfgm.head_dim)
-      v = attn.v_proj(normed).view(B, S, cfgm.n_kv_heads, cfgm.head_dim)
+      qkv = attn.qkv_proj(normed)
+      q_sz = cfgm.n_heads * cfgm.head_dim
+      kv_sz = cfgm.n_kv_heads * cfgm.head_dim
+      q, k, v = torch.split(qkv, [q_sz, kv_sz, kv_sz], dim=-1)
+      q = q.reshape(B, S, cfgm.n_heads, cfgm.head_dim)
+      k = k.reshape(B, S, cfgm.n_kv_heads, cfgm.head_dim)
+      v = v.reshape(B, S, cfgm.n_kv_heads, cfgm.head_dim)
       q, k = tr.rope_apply(q, k, cos, sin, positions)
       rep = cfgm.n_heads // cfgm.n_kv_heads
       out = torch.nn.functional.scaled_dot_product_attention(

@@ -38,39 +38,45 @@ class RMSNorm(nn.Module):
 
 
 class Attention(nn.Module):
+  """GQA attention with a FUSED qkv projection: one [D, (H+2KVH)*hd] GEMM per
+  layer instead of three (decode on MI355X is GEMM-launch bound at small
+  batch); the packed output feeds the fused RoPE+KV-append kernel directly."""
+
   def __init__(self, cfg: ModelConfig):
     super().__init__()
     self.cfg = cfg
     H, KVH, hd, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim, cfg.dim
-    self.q_proj = nn.Linear(D, H * hd, bias=cfg.attn_bias)
-    self.k_proj = nn.Linear(D, KVH * hd, bias=cfg.attn_bias)
-    self.v_proj = nn.Linear(D, KVH * hd, bias=cfg.attn_bias)
+    self.qkv_proj = nn.Linear(D, (H + 2 * KVH) * hd, bias=cfg.attn_bias)
     self.o_proj = nn.Linear(H * hd, D, bias=False)
 
   def forward(self, x, cos, sin, positions, k_cache, v_cache, start_pos: int, is_decode: bool, seq_lens=None):
     B, S, _ = x.shape
     cfg = self.cfg
-    q = self.q_proj(x).view(B, S, cfg.n_heads, cfg.head_dim)
-    k = self.k_proj(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
-    v = self.v_proj(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
-    q = ops.rope_kv_append(q, k, v, cos, sin, positions, k_cache, v_cache, start_pos)
+    H, hd = cfg.n_heads, cfg.head_dim
+    qkv = self.qkv_proj(x)  # [B, S, (H+2KVH)*hd]
+    ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, cfg.n_kv_heads, hd)
+    q = qkv[:, :, : H * hd].view(B, S, H, hd)  # strided view; kernels accept it
     if is_decode:
       sl = seq_lens if seq_lens is not None else start_pos + 1
       out = ops.attn_decode(q, k_cache, v_cache, sl)
     else:
       out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S)
-    return self.o_proj(out.reshape(B, S, cfg.n_heads * cfg.head_dim))
+    return self.o_proj(out.reshape(B, S, H * hd))
 
 
 class MLP(nn.Module):
-  def __init__(self, cfg: ModelConfig):
+  """SwiGLU MLP with a FUSED gate+up projection ([D, 2I] GEMM); the packed
+  [gate | up] output feeds the packed SwiGLU kernel."""
+
+  def __init__(self, cfg: ModelConfig, intermediate: Optional[int] = None):
     super().__init__()
-    self.gate_proj = nn.Linear(cfg.dim, cfg.intermediate_dim, bias=False)
-    self.up_proj = nn.Linear(cfg.dim, cfg.intermediate_dim, bias=False)
-    self.down_proj = nn.Linear(cfg.intermediate_dim, cfg.dim, bias=False)
+    I = intermediate or cfg.intermediate_dim
+    self.intermediate = I
+    self.gate_up_proj = nn.Linear(cfg.dim, 2 * I, bias=False)
+    self.down_proj = nn.Linear(I, cfg.dim, bias=False)
 
   def forward(self, x):
-    return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+    return self.down_proj(ops.swiglu_packed(self.gate_up_proj(x)))
 
 
 class MoEMLP(nn.Module):
@@ -205,9 +211,9 @@ class ShardedModel(nn.Module):
 def hf_key_map(shard: Shard, cfg: ModelConfig):
   """Map HF checkpoint keys → this ShardedModel's state-dict keys (shard-aware).
 
-  Returns a dict {hf_key: our_key}; HF keys outside the shard map to None.
-  Naming mirrors HF so this is near-identity (no q/k permute —
-  we use HF's rotate-half RoPE directly).
+  Packed parameters (fused qkv_proj / gate_up_proj GEMMs) map several HF keys
+  to `our_key#part`; the loader concatenates the parts along dim 0 in part
+  order. No q/k permute is needed (we use HF's rotate-half RoPE directly).
   """
   mapping = {}
   if shard.is_first_layer:
@@ -221,17 +227,21 @@ def hf_key_map(shard: Shard, cfg: ModelConfig):
   for lid in range(shard.start_layer, shard.end_layer + 1):
     hf = f"model.layers.{lid}."
     ours = f"layers.{lid}."
-    for sub in (
-      "self_attn.q_proj.weight", "self_attn.k_proj.weight", "self_attn.v_proj.weight",
-      "self_attn.o_proj.weight", "self_attn.q_proj.bias", "self_attn.k_proj.bias",
-      "self_attn.v_proj.bias", "mlp.gate_proj.weight", "mlp.up_proj.weight",
-      "mlp.down_proj.weight", "input_layernorm.weight", "post_attention_layernorm.weight",
-    ):
-      mapping[hf + sub] = ours + sub
+    for suffix in ("weight",) + (("bias",) if cfg.attn_bias else ()):
+      mapping[hf + f"self_attn.q_proj.{suffix}"] = ours + f"self_attn.qkv_proj.{suffix}#0"
+      mapping[hf + f"self_attn.k_proj.{suffix}"] = ours + f"self_attn.qkv_proj.{suffix}#1"
+      mapping[hf + f"self_attn.v_proj.{suffix}"] = ours + f"self_attn.qkv_proj.{suffix}#2"
+    mapping[hf + "self_attn.o_proj.weight"] = ours + "self_attn.o_proj.weight"
+    mapping[hf + "input_layernorm.weight"] = ours + "input_layernorm.weight"
+    mapping[hf + "post_attention_layernorm.weight"] = ours + "post_attention_layernorm.weight"
     if cfg.n_experts > 0:
       mapping[hf + "block_sparse_moe.gate.weight"] = ours + "mlp.gate.weight"
       for e in range(cfg.n_experts):
-        mapping[hf + f"block_sparse_moe.experts.{e}.w1.weight"] = ours + f"mlp.experts.{e}.gate_proj.weight"
-        mapping[hf + f"block_sparse_moe.experts.{e}.w3.weight"] = ours + f"mlp.experts.{e}.up_proj.weight"
+        mapping[hf + f"block_sparse_moe.experts.{e}.w1.weight"] = ours + f"mlp.experts.{e}.gate_up_proj.weight#0"
+        mapping[hf + f"block_sparse_moe.experts.{e}.w3.weight"] = ours + f"mlp.experts.{e}.gate_up_proj.weight#1"
         mapping[hf + f"block_sparse_moe.experts.{e}.w2.weight"] = ours + f"mlp.experts.{e}.down_proj.weight"
+    else:
+      mapping[hf + "mlp.gate_proj.weight"] = ours + "mlp.gate_up_proj.weight#0"
+      mapping[hf + "mlp.up_proj.weight"] = ours + "mlp.gate_up_proj.weight#1"
+      mapping[hf + "mlp.down_proj.weight"] = ours + "mlp.down_proj.weight"
   return mapping

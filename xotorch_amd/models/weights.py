@@ -49,12 +49,22 @@ def load_shard_weights(model: ShardedModel, model_dir: Path, device: str = "cpu"
   if files is None:
     raise FileNotFoundError(f"no safetensors found under {model_dir}")
   state: Dict[str, torch.Tensor] = {}
+  packed: Dict[str, Dict[int, torch.Tensor]] = {}
   for fn, keys in files.items():
     with safe_open(str(model_dir / fn), framework="pt", device=device) as f:
       names = set(f.keys())
       for hf_key in keys:
-        if hf_key in names:
-          state[mapping[hf_key]] = f.get_tensor(hf_key)
+        if hf_key not in names:
+          continue
+        our = mapping[hf_key]
+        if "#" in our:
+          base, part = our.split("#")
+          packed.setdefault(base, {})[int(part)] = f.get_tensor(hf_key)
+        else:
+          state[our] = f.get_tensor(hf_key)
+  # fused qkv_proj / gate_up_proj: concatenate the HF fragments along dim 0
+  for base, parts in packed.items():
+    state[base] = torch.cat([parts[i] for i in sorted(parts)], dim=0)
   missing, unexpected = model.load_state_dict(state, strict=False)
   # tied-embedding models ship no lm_head tensor; rope buffers are computed
   real_missing = [m for m in missing if not m.startswith("rope_")]

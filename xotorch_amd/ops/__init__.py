@@ -37,12 +37,13 @@ def hip_available() -> bool:
   return _load_hip() is not None
 
 
-def _use_hip(t: torch.Tensor) -> bool:
+def _use_hip(*tensors: torch.Tensor) -> bool:
+  t = tensors[0]
   if not t.is_cuda:
     return False
-  if torch.is_grad_enabled():
+  if torch.is_grad_enabled() and any(x.requires_grad for x in tensors):
     # training path: HIP kernels are inference-only (no autograd); the torch
-    # ops are differentiable. Inference always runs under inference_mode.
+    # ops are differentiable. Inference runs on the HIP kernels.
     return False
   if _load_hip() is not None:
     return True
@@ -57,7 +58,7 @@ def _use_hip(t: torch.Tensor) -> bool:
 
 
 def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
-  if _use_hip(x) and x.dtype == torch.bfloat16:
+  if _use_hip(x, weight) and x.dtype == torch.bfloat16:
     return _hip.rmsnorm(x, weight, eps)
   return torch_ref.rmsnorm(x, weight, eps)
 
@@ -65,27 +66,25 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
 def rmsnorm_residual(
   x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
 ) -> Tuple[torch.Tensor, torch.Tensor]:
-  if _use_hip(x) and x.dtype == torch.bfloat16:
+  if _use_hip(x, residual, weight) and x.dtype == torch.bfloat16:
     return _hip.rmsnorm_residual(x, residual, weight, eps)
   return torch_ref.rmsnorm_residual(x, residual, weight, eps)
 
 
 def rope_apply(q, k, cos, sin, positions):
-  if _use_hip(q) and q.dtype == torch.bfloat16:
+  if _use_hip(q, k) and q.dtype == torch.bfloat16:
     return _hip.rope_apply(q, k, cos, sin, positions)
   return torch_ref.rope_apply(q, k, cos, sin, positions)
 
 
-def rope_kv_append(q, k, v, cos, sin, positions, k_cache, v_cache, start_pos: int):
-  """Fused: RoPE on q,k + append (rotated k, v) into the cache at start_pos.
-
-  Returns rotated q. k/v: [B, S, KVH, hd]; caches [B, KVH, T, hd].
+def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_kv_heads: int, head_dim: int):
+  """Fused on the packed qkv GEMM output [B,S,(H+2KVH)*hd]: RoPE-rotate the q
+  heads in place, rotate k heads into the cache, copy v heads into the cache.
   """
-  if _use_hip(q) and q.dtype == torch.bfloat16:
-    return _hip.rope_kv_append(q, k, v, cos, sin, positions, k_cache, v_cache, start_pos)
-  q_r, k_r = torch_ref.rope_apply(q, k, cos, sin, positions)
-  torch_ref.kv_append(k_cache, v_cache, k_r, v, start_pos)
-  return q_r
+  if _use_hip(qkv) and qkv.dtype == torch.bfloat16:
+    _hip.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim)
+    return
+  torch_ref.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim)
 
 
 def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int):
@@ -111,6 +110,8 @@ def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int):
 def attn_decode(q, k_cache, v_cache, seq_len):
   """seq_len: int or int32 device tensor [B] (per-sequence lengths)."""
   if _use_hip(q) and q.dtype == torch.bfloat16:
+    if not isinstance(seq_len, torch.Tensor):
+      seq_len = torch.full((q.shape[0],), int(seq_len), dtype=torch.int32, device=q.device)
     return _hip.attn_decode(q, k_cache, v_cache, seq_len)
   if isinstance(seq_len, torch.Tensor):
     seq_len = int(seq_len.max().item())
@@ -118,14 +119,18 @@ def attn_decode(q, k_cache, v_cache, seq_len):
 
 
 def swiglu(gate, up):
-  if _use_hip(gate) and gate.dtype == torch.bfloat16:
+  if _use_hip(gate, up) and gate.dtype == torch.bfloat16:
     return _hip.swiglu(gate, up)
   return torch_ref.swiglu(gate, up)
 
 
+def swiglu_packed(gu):
+  """silu(gate)*up on the packed [.., 2I] fused gate_up GEMM output."""
+  if _use_hip(gu) and gu.dtype == torch.bfloat16:
+    return _hip.swiglu_packed(gu)
+  return torch_ref.swiglu_packed(gu)
+
+
 def softmax_sample(logits, temperature: float = 0.0, top_k: int = 0, generator=None):
-  # sampling is tiny; HIP path exists to keep the decode step graph-capturable
-  if _use_hip(logits) and temperature > 0.0 and top_k > 0:
-    q = torch.empty_like(logits, dtype=torch.float32).exponential_(1, generator=generator)
-    return _hip.topk_sample(logits, q, temperature, top_k)
+  # torch ops here are graph-capturable (argmax / exponential+argmax)
   return torch_ref.softmax_sample(logits, temperature, top_k, generator)

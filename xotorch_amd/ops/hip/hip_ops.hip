@@ -107,15 +107,16 @@ __global__ __launch_bounds__(256) void rmsnorm_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Fused RoPE (HF rotate-half) + KV-cache append
-// one wave per (b, s, head-slot); slots: [0,H) rotate q in place,
-// [H, H+KVH) rotate k into the cache, [H+KVH, H+2KVH) copy v into the cache.
-// cache layout [B, KVH, T, hd]; cos/sin tables fp32 [maxT, hd/2].
+// Fused RoPE (HF rotate-half) + KV-cache append, on the PACKED qkv tensor
+// (one fused QKV GEMM writes [B, S, (H+2*KVH)*hd]; this kernel rotates the
+// q heads in place, rotates k heads into the cache, copies v heads into the
+// cache — no separate q/k/v tensors ever materialize).
+// one wave per (b, s, head-slot); cache layout [B, KVH, T, hd];
+// cos/sin tables fp32 [maxT, hd/2].
 // ---------------------------------------------------------------------------
 
-__global__ __launch_bounds__(256) void rope_kv_append_kernel(
-    unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
-    const unsigned short* __restrict__ v, const float* __restrict__ cosb,
+__global__ __launch_bounds__(256) void rope_qkv_append_kernel(
+    unsigned short* __restrict__ qkv, const float* __restrict__ cosb,
     const float* __restrict__ sinb, const int* __restrict__ positions,
     unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
     int B, int S, int H, int KVH, int hd, int T) {
@@ -129,34 +130,32 @@ __global__ __launch_bounds__(256) void rope_kv_append_kernel(
   const int b = bs / S;
   const int pos = positions[s];
   const int hd2 = hd >> 1;
+  unsigned short* row = qkv + ((size_t)bs * slots + slot) * hd;
   if (slot < H) {
-    unsigned short* qp = q + (((size_t)bs * H + slot) * hd);
     if (lane < hd2) {
       const float c = cosb[(size_t)pos * hd2 + lane];
       const float sn = sinb[(size_t)pos * hd2 + lane];
-      const float x1 = b2f(qp[lane]);
-      const float x2 = b2f(qp[lane + hd2]);
-      qp[lane] = f2b(x1 * c - x2 * sn);
-      qp[lane + hd2] = f2b(x2 * c + x1 * sn);
+      const float x1 = b2f(row[lane]);
+      const float x2 = b2f(row[lane + hd2]);
+      row[lane] = f2b(x1 * c - x2 * sn);
+      row[lane + hd2] = f2b(x2 * c + x1 * sn);
     }
   } else if (slot < H + KVH) {
     const int h = slot - H;
-    const unsigned short* kp = k + (((size_t)bs * KVH + h) * hd);
     unsigned short* dst = kc + (((size_t)(b * KVH + h) * T + pos) * hd);
     if (lane < hd2) {
       const float c = cosb[(size_t)pos * hd2 + lane];
       const float sn = sinb[(size_t)pos * hd2 + lane];
-      const float x1 = b2f(kp[lane]);
-      const float x2 = b2f(kp[lane + hd2]);
+      const float x1 = b2f(row[lane]);
+      const float x2 = b2f(row[lane + hd2]);
       dst[lane] = f2b(x1 * c - x2 * sn);
       dst[lane + hd2] = f2b(x2 * c + x1 * sn);
     }
   } else {
     const int h = slot - H - KVH;
-    const unsigned short* vp = v + (((size_t)bs * KVH + h) * hd);
     unsigned short* dst = vc + (((size_t)(b * KVH + h) * T + pos) * hd);
     if (lane * 2 < hd)
-      *(unsigned int*)(dst + lane * 2) = *(const unsigned int*)(vp + lane * 2);
+      *(unsigned int*)(dst + lane * 2) = *(const unsigned int*)(row + lane * 2);
   }
 }
 
@@ -177,7 +176,7 @@ __global__ __launch_bounds__(256) void attn_decode_partial(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ kc,
     const unsigned short* __restrict__ vc, const int* __restrict__ seq_lens,
     float* __restrict__ ws_o, float* __restrict__ ws_ml,
-    int B, int H, int KVH, int T, int nsplit, int qh0, float scale) {
+    int B, int H, int KVH, int T, int nsplit, int qh0, float scale, long long q_stride) {
   constexpr int EPL = HD / 16;  // bf16 elements per lane (8 @128, 4 @64)
   const int blk = blockIdx.x;
   const int split = blk % nsplit;
@@ -196,7 +195,7 @@ __global__ __launch_bounds__(256) void attn_decode_partial(
   float qreg[NQ][EPL];
 #pragma unroll
   for (int n = 0; n < NQ; ++n) {
-    const unsigned short* qp = q + (((size_t)b * H + qh_base + n) * HD + gl * EPL);
+    const unsigned short* qp = q + ((size_t)b * q_stride + (qh_base + n) * HD + gl * EPL);
 #pragma unroll
     for (int e = 0; e < EPL; ++e) qreg[n][e] = b2f(qp[e]) * scale;
   }
@@ -328,7 +327,31 @@ __global__ __launch_bounds__(256) void swiglu_kernel(
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const float gf = b2f(gv[j]);
-      const float s = gf / (1.f + __expf(-gf));
+      const float s = gf * __builtin_amdgcn_rcpf(1.f + __expf(-gf));
+      o[j] = f2b(s * b2f(uv[j]));
+    }
+    *(ushort8*)(out + i * 8) = o;
+  }
+}
+
+// packed variant: gu = [rows, 2*I] from the fused gate_up GEMM ([gate | up])
+__global__ __launch_bounds__(256) void swiglu_packed_kernel(
+    const unsigned short* __restrict__ gu, unsigned short* __restrict__ out,
+    long long rows, int I) {
+  const long long n8 = rows * (I >> 3);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const int i8 = I >> 3;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n8; i += stride) {
+    const long long row = i / i8;
+    const long long col8 = i % i8;
+    const unsigned short* base = gu + row * (2LL * I) + col8 * 8;
+    ushort8 gv = *(const ushort8*)base;
+    ushort8 uv = *(const ushort8*)(base + I);
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = b2f(gv[j]);
+      const float s = gf * __builtin_amdgcn_rcpf(1.f + __expf(-gf));
       o[j] = f2b(s * b2f(uv[j]));
     }
     *(ushort8*)(out + i * 8) = o;
@@ -372,37 +395,35 @@ std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res, 
   return {out, res_out};
 }
 
-torch::Tensor rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                             torch::Tensor cos, torch::Tensor sin, torch::Tensor positions,
-                             torch::Tensor kc, torch::Tensor vc, int64_t start_pos) {
-  (void)start_pos;  // positions tensor is authoritative (graph-safe)
-  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
-  CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
+                     torch::Tensor positions, torch::Tensor kc, torch::Tensor vc,
+                     int64_t n_heads, int64_t n_kv_heads, int64_t head_dim) {
+  CHK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
   CHK(kc.is_contiguous() && vc.is_contiguous());
   CHK(positions.dtype() == torch::kInt32 && positions.is_cuda());
-  const int B = q.size(0), S = q.size(1), H = q.size(2), hd = q.size(3);
-  const int KVH = k.size(2), T = kc.size(2);
+  const int B = qkv.size(0), S = qkv.size(1);
+  const int H = (int)n_heads, KVH = (int)n_kv_heads, hd = (int)head_dim;
+  const int T = kc.size(2);
+  CHK(qkv.size(2) == (H + 2 * KVH) * hd);
   CHK(hd <= 128 && hd % 2 == 0);
   const int waves = B * S * (H + 2 * KVH);
   const int blocks = (waves + 3) / 4;
-  hipLaunchKernelGGL(rope_kv_append_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
-                     (unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),
-                     (const unsigned short*)v.data_ptr(), cos.data_ptr<float>(), sin.data_ptr<float>(),
+  hipLaunchKernelGGL(rope_qkv_append_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (unsigned short*)qkv.data_ptr(), cos.data_ptr<float>(), sin.data_ptr<float>(),
                      positions.data_ptr<int>(), (unsigned short*)kc.data_ptr(),
                      (unsigned short*)vc.data_ptr(), B, S, H, KVH, hd, T);
-  return q;
 }
 
 template <int HD>
 static void launch_attn_partial(int NQ, const unsigned short* q, const unsigned short* kc,
                                 const unsigned short* vc, const int* sl, float* ws_o, float* ws_ml,
                                 int B, int H, int KVH, int T, int nsplit, int qh0, float scale,
-                                hipStream_t stream) {
+                                long long q_stride, hipStream_t stream) {
   const dim3 grid(B * KVH * nsplit), block(256);
 #define CASE(NQV) \
   case NQV: \
     hipLaunchKernelGGL((attn_decode_partial<NQV, HD>), grid, block, 0, stream, q, kc, vc, sl, \
-                       ws_o, ws_ml, B, H, KVH, T, nsplit, qh0, scale); \
+                       ws_o, ws_ml, B, H, KVH, T, nsplit, qh0, scale, q_stride); \
     break;
   switch (NQ) {
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
@@ -412,9 +433,13 @@ static void launch_attn_partial(int NQ, const unsigned short* q, const unsigned 
 }
 
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, torch::Tensor seq_lens) {
-  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  // q may be a strided view into the packed qkv tensor: [B, 1, H, hd] with an
+  // arbitrary batch stride but contiguous (head, dim) rows.
+  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  CHK(q.stride(3) == 1 && q.stride(2) == q.size(3));
   CHK(seq_lens.dtype() == torch::kInt32 && seq_lens.is_cuda());
   const int B = q.size(0), H = q.size(2), hd = q.size(3);
+  const long long q_stride = q.stride(0);
   const int KVH = kc.size(1), T = kc.size(2);
   TORCH_CHECK(hd == 64 || hd == 128, "attn_decode: head_dim must be 64 or 128, got ", hd);
   const int rep = H / KVH;
@@ -424,7 +449,8 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, t
   auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
   auto ws_o = torch::empty({(long)B * H * nsplit * hd}, opts);
   auto ws_ml = torch::empty({(long)B * H * nsplit * 2}, opts);
-  auto out = torch::empty_like(q);
+  auto out = torch::empty({q.size(0), q.size(1), q.size(2), q.size(3)},
+                          torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
   const float scale = 1.0f / sqrtf((float)hd);
   auto stream = cur_stream();
   for (int qh0 = 0; qh0 < rep; qh0 += 8) {
@@ -433,12 +459,12 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, t
       launch_attn_partial<128>(nq, (const unsigned short*)q.data_ptr(), (const unsigned short*)kc.data_ptr(),
                                (const unsigned short*)vc.data_ptr(), seq_lens.data_ptr<int>(),
                                ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T, nsplit, qh0,
-                               scale, stream);
+                               scale, q_stride, stream);
     else
       launch_attn_partial<64>(nq, (const unsigned short*)q.data_ptr(), (const unsigned short*)kc.data_ptr(),
                               (const unsigned short*)vc.data_ptr(), seq_lens.data_ptr<int>(),
                               ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T, nsplit, qh0,
-                              scale, stream);
+                              scale, q_stride, stream);
   }
   if (hd == 128)
     hipLaunchKernelGGL((attn_decode_merge<128>), dim3(B * H), dim3(128), 0, stream,
@@ -461,10 +487,27 @@ torch::Tensor swiglu(torch::Tensor g, torch::Tensor u) {
   return out;
 }
 
+torch::Tensor swiglu_packed(torch::Tensor gu) {
+  CHK(gu.is_cuda() && gu.dtype() == torch::kBFloat16 && gu.is_contiguous());
+  const int twoI = gu.size(-1);
+  CHK(twoI % 16 == 0);
+  const int I = twoI / 2;
+  const long long rows = gu.numel() / twoI;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = torch::empty(sizes, gu.options());
+  const long long n8 = rows * (I / 8);
+  const int blocks = (int)std::min<long long>(2048, (n8 + 255) / 256);
+  hipLaunchKernelGGL(swiglu_packed_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)gu.data_ptr(), (unsigned short*)out.data_ptr(), rows, I);
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("swiglu_packed", &swiglu_packed, "SwiGLU on the packed [gate|up] GEMM output");
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual add + RMSNorm");
-  m.def("rope_kv_append", &rope_kv_append, "fused RoPE + KV-cache append");
+  m.def("rope_qkv_append", &rope_qkv_append, "fused RoPE + KV-cache append on packed qkv");
   m.def("attn_decode", &attn_decode, "GQA decode attention (flash-decoding split-KV)");
   m.def("swiglu", &swiglu, "SwiGLU activation");
 }

@@ -156,6 +156,36 @@ def attn_decode(
   return out.transpose(1, 2).to(q.dtype)
 
 
+def rope_qkv_append(
+  qkv: torch.Tensor,
+  cos: torch.Tensor,
+  sin: torch.Tensor,
+  positions: torch.Tensor,
+  k_cache: torch.Tensor,
+  v_cache: torch.Tensor,
+  n_heads: int,
+  n_kv_heads: int,
+  head_dim: int,
+) -> None:
+  """Reference for the fused packed-qkv kernel: rotate q in place inside qkv,
+  rotate k and append (k, v) into the caches at `positions`."""
+  B, S, _ = qkv.shape
+  H, KVH, hd = n_heads, n_kv_heads, head_dim
+  q = qkv[:, :, : H * hd].view(B, S, H, hd)
+  k = qkv[:, :, H * hd: (H + KVH) * hd].view(B, S, KVH, hd)
+  v = qkv[:, :, (H + KVH) * hd:].view(B, S, KVH, hd)
+  q_r, k_r = rope_apply(q, k, cos, sin, positions)
+  q.copy_(q_r)
+  start_pos = int(positions.reshape(-1)[0])
+  kv_append(k_cache, v_cache, k_r, v, start_pos)
+
+
+def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
+  """silu(gate) * up on the packed [.., 2I] fused gate_up GEMM output."""
+  I = gu.shape[-1] // 2
+  return swiglu(gu[..., :I], gu[..., I:])
+
+
 def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
   """silu(gate) * up, fp32 internally."""
   return (F.silu(gate.float()) * up.float()).to(gate.dtype)
