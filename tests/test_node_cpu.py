@@ -1,0 +1,176 @@
+"""Node orchestration tests: single-node decode, two-node TCP ring (the
+reference's multi-node-without-cluster pattern — SURVEY.md §4), elastic
+topology, ring training over the wire."""
+import asyncio
+import json
+
+import numpy as np
+import pytest
+
+from xotorch_amd.engine.dummy import DummyEngine
+from xotorch_amd.engine.torch_engine import TorchEngine
+from xotorch_amd.helpers import find_available_port
+from xotorch_amd.models.registry import build_base_shard
+from xotorch_amd.orchestration.discovery import ManualDiscovery
+from xotorch_amd.orchestration.node import Node
+from xotorch_amd.orchestration.server import Server
+
+
+def run(coro):
+  return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def test_single_node_decode_loop():
+  async def go():
+    node = Node("solo", None, DummyEngine(), None, max_generate_tokens=4)
+    await node.start()
+    shard = build_base_shard("dummy", "DummyEngine")
+    done = asyncio.Event()
+    got = []
+
+    def on_token(rid, toks, fin):
+      got.extend(toks)
+      if fin:
+        done.set()
+
+    node.on_token.register("t").on_next(on_token)
+    await node.process_prompt(shard, "hello", "req-1")
+    await asyncio.wait_for(done.wait(), 30)
+    assert len(got) >= 4
+    await node.stop()
+    return True
+  assert run(go())
+
+
+async def _make_tcp_pair(tmp_path, engine_cls):
+  ports = [find_available_port("127.0.0.1"), find_available_port("127.0.0.1")]
+  cfg = {
+    "peers": {
+      "node-a": {"address": f"127.0.0.1:{ports[0]}"},
+      "node-b": {"address": f"127.0.0.1:{ports[1]}"},
+    }
+  }
+  cfg_path = tmp_path / "topo.json"
+  cfg_path.write_text(json.dumps(cfg))
+  nodes = []
+  for name, port in zip(("node-a", "node-b"), ports):
+    disc = ManualDiscovery(str(cfg_path), name, interval=0.5)
+    node = Node(name, None, engine_cls(), disc, max_generate_tokens=4)
+    node.server = Server(node, "127.0.0.1", port)
+    nodes.append(node)
+  for n in nodes:
+    await n.server.start()
+  for n in nodes:
+    await n.start(wait_for_peers=1)
+  return nodes
+
+
+@pytest.mark.timeout(120)
+def test_two_node_tcp_ring_decode(tmp_path):
+  async def go():
+    import torch
+    nodes = await _make_tcp_pair(tmp_path, lambda: TorchEngine(device="cpu", dtype=torch.float32))
+    a, b = nodes
+    assert len(a.peers) == 1 and len(b.peers) == 1
+    assert len(a.topology.nodes) == 2
+    shard = build_base_shard("dummy", "TorchEngine")
+    # which node is ring stage 0? partitions sorted by (mem desc, id) — on one
+    # host both report equal memory so node-a leads.
+    done = asyncio.Event()
+    got = []
+
+    def on_token(rid, toks, fin):
+      got[:] = toks  # broadcast sends the buffered tail
+      if fin:
+        done.set()
+
+    a.on_token.register("t").on_next(on_token)
+    b.on_token.register("t2").on_next(lambda rid, toks, fin: None)
+    await a.process_prompt(shard, "hello world", "req-tcp")
+    await asyncio.wait_for(done.wait(), 60)
+    assert len(got) >= 1
+    for n in nodes:
+      await n.stop()
+    return True
+  assert run(go())
+
+
+@pytest.mark.timeout(120)
+def test_two_node_tcp_ring_training(tmp_path):
+  async def go():
+    import torch
+    nodes = await _make_tcp_pair(tmp_path, lambda: TorchEngine(device="cpu", dtype=torch.float32))
+    a, b = nodes
+    shard = build_base_shard("dummy", "TorchEngine")
+    B, S = 2, 6
+    inputs = np.random.randint(0, 200, (B, S))
+    targets = np.roll(inputs, -1, axis=1)
+    lengths = np.array([S, S])
+    loss, grad = await a.enqueue_example(shard, inputs, targets, lengths, train=True)
+    assert np.isfinite(loss)
+    loss2, _ = await a.enqueue_example(shard, inputs, targets, lengths, train=False)
+    assert np.isfinite(loss2)
+    for n in nodes:
+      await n.stop()
+    return True
+  assert run(go())
+
+
+@pytest.mark.timeout(60)
+def test_manual_discovery_hot_reload(tmp_path):
+  async def go():
+    port_a = find_available_port("127.0.0.1")
+    cfg_path = tmp_path / "topo.json"
+    cfg_path.write_text(json.dumps({"peers": {"a": {"address": f"127.0.0.1:{port_a}"}}}))
+    node_a = Node("a", None, DummyEngine(), None)
+    node_a.server = Server(node_a, "127.0.0.1", port_a)
+    await node_a.server.start()
+    disc = ManualDiscovery(str(cfg_path), "watcher", interval=0.2)
+    await disc.start()
+    peers = await disc.discover_peers(wait_for_peers=1)
+    assert [p.id() for p in peers] == ["a"]
+    # hot add a second peer
+    port_b = find_available_port("127.0.0.1")
+    node_b = Node("b", None, DummyEngine(), None)
+    node_b.server = Server(node_b, "127.0.0.1", port_b)
+    await node_b.server.start()
+    cfg_path.write_text(json.dumps({"peers": {
+      "a": {"address": f"127.0.0.1:{port_a}"},
+      "b": {"address": f"127.0.0.1:{port_b}"},
+    }}))
+    for _ in range(50):
+      peers = await disc.discover_peers()
+      if len(peers) == 2:
+        break
+      await asyncio.sleep(0.2)
+    assert len(peers) == 2
+    # kill b: it must be dropped
+    await node_b.server.stop()
+    for _ in range(50):
+      peers = await disc.discover_peers()
+      if len(peers) == 1:
+        break
+      await asyncio.sleep(0.2)
+    assert [p.id() for p in peers] == ["a"]
+    await disc.stop()
+    await node_a.server.stop()
+    return True
+  assert run(go())
+
+
+def test_elastic_repartition():
+  """Partitions recompute from live topology on every routing decision."""
+  async def go():
+    node = Node("n1", None, DummyEngine(), None)
+    await node.start()
+    shard = build_base_shard("dummy", "DummyEngine")
+    full = node.get_current_shard(shard)
+    assert full.start_layer == 0 and full.end_layer == shard.n_layers - 1
+    # fake a second peer appearing in the topology
+    from xotorch_amd.parallel.topology import DeviceCapabilities
+    node.topology.update_node("n2", DeviceCapabilities(model="m", chip="c", memory=node.device_capabilities.memory))
+    mine = node.get_current_shard(shard)
+    assert mine.get_layer_count() < shard.n_layers
+    await node.stop()
+    return True
+  assert run(go())

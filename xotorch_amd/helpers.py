@@ -110,12 +110,18 @@ class PrefixDict(Generic[K, T]):
     return max(matches, key=lambda x: len(x[0]))
 
 
+_handed_out_ports: set = set()
+
+
 def find_available_port(host: str = "", min_port: int = 49152, max_port: int = 65535) -> int:
-  for _ in range(100):
+  for _ in range(200):
     port = random.randint(min_port, max_port)
+    if port in _handed_out_ports:
+      continue
     with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
       try:
         s.bind((host, port))
+        _handed_out_ports.add(port)
         return port
       except OSError:
         continue

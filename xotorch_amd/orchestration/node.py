@@ -107,23 +107,32 @@ class Node:
     return True
 
   async def _periodic_topology_collection(self, interval: float):
+    # re-collect unconditionally (reference node.py:520-531): peers may have
+    # probed their devices after our first pass
     while True:
       try:
-        changed = await self.update_peers()
-        if changed:
-          await self.collect_topology(set())
+        await self.update_peers()
+        await self.collect_topology(set())
       except Exception as e:
         if DEBUG >= 1:
           print(f"topology collection error: {e}")
       await asyncio.sleep(interval)
 
   async def collect_topology(self, visited: set, max_depth: int = 4) -> Topology:
+    if self.device_capabilities.memory == 0:
+      # answering gossip before start() finished: probe now
+      loop = asyncio.get_running_loop()
+      self.device_capabilities = await loop.run_in_executor(None, device_capabilities)
     topo = Topology()
     topo.update_node(self.id, self.device_capabilities)
     topo.active_node_id = self.id
     visited = set(visited) | {self.id}
     for peer in self.peers:
-      topo.update_node(peer.id(), peer.device_capabilities())
+      # prefer what we already learned over the handle's stub caps: answering
+      # a gossip query must not clobber known-good info for visited peers
+      known = self.topology.get_node(peer.id())
+      caps = known if (known is not None and known.memory > 0) else peer.device_capabilities()
+      topo.update_node(peer.id(), caps)
       topo.add_edge(self.id, peer.id(), peer.description())
       if peer.id() in visited or max_depth <= 0:
         continue
@@ -158,7 +167,10 @@ class Node:
       index = self.get_partition_index()
     partitions = self.partitioning_strategy.partition(self.topology)
     shards = map_partitions_to_shards(partitions, base_shard.n_layers, base_shard.model_id)
-    return shards[index]
+    if not shards:
+      return Shard(base_shard.model_id, 0, base_shard.n_layers - 1, base_shard.n_layers)
+    # rounding can leave fewer shards than partitions (empty ranges skipped)
+    return shards[min(index, len(shards) - 1)]
 
   def _peer_by_index(self, index: int) -> Optional[PeerHandle]:
     partitions = self.partitioning_strategy.partition(self.topology)
@@ -192,11 +204,7 @@ class Node:
     }))
 
   def get_first_partition_index(self) -> int:
-    partitions = self.partitioning_strategy.partition(self.topology)
-    shards = None
-    for i in range(len(partitions)):
-      # stage 0 is whichever partition maps to the first layers
-      return 0
+    # partitions are ordered from layer 0, so ring stage 0 is index 0
     return 0
 
   async def process_tensor(self, base_shard: Shard, tensor: np.ndarray, request_id: Optional[str] = None,

@@ -21,6 +21,8 @@ class Server:
     self._server: Optional[asyncio.AbstractServer] = None
 
   async def start(self):
+    if self._server is not None:
+      return  # idempotent: Node.start() also starts the server
     self._server = await asyncio.start_server(self._handle, self.host, self.port)
 
   async def stop(self):

@@ -120,6 +120,10 @@ class Topology:
 
   def merge(self, other: "Topology", merging_peer_id: Optional[str] = None):
     for node_id, cap in other.nodes.items():
+      # never downgrade a known node to a zero-memory placeholder (gossip
+      # replies list already-visited peers with handle-level stub caps)
+      if cap.memory == 0 and node_id in self.nodes and self.nodes[node_id].memory > 0:
+        continue
       self.update_node(node_id, cap)
     for node_id, conns in other.peer_graph.items():
       for conn in conns:
