@@ -1,0 +1,107 @@
+"""ChatGPT-compatible API tests against a dummy-engine Node (no GPU)."""
+import asyncio
+import json
+
+import pytest
+from aiohttp.test_utils import TestClient, TestServer
+
+from xotorch_amd.api.chatgpt import ChatGPTAPI
+from xotorch_amd.engine.dummy import DummyEngine
+from xotorch_amd.orchestration.node import Node
+
+
+def run(coro):
+  return asyncio.new_event_loop().run_until_complete(coro)
+
+
+async def make_client():
+  node = Node("api-node", None, DummyEngine(), None, max_generate_tokens=4)
+  await node.start()
+  api = ChatGPTAPI(node, "DummyEngine", default_model="dummy")
+  client = TestClient(TestServer(api.app))
+  await client.start_server()
+  return node, client
+
+
+def test_models_and_health():
+  async def go():
+    node, client = await make_client()
+    r = await client.get("/healthcheck")
+    assert (await r.json())["status"] == "ok"
+    r = await client.get("/v1/models")
+    data = await r.json()
+    ids = [m["id"] for m in data["data"]]
+    assert "llama-3-70b" in ids and "dummy" in ids
+    r = await client.get("/modelpool")
+    assert "model pool" in await r.json()
+    r = await client.get("/initial_models")
+    assert "llama-3.2-1b" in await r.json()
+    r = await client.get("/v1/topology")
+    assert "api-node" in (await r.json())["nodes"]
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())
+
+
+def test_chat_completion_non_streaming():
+  async def go():
+    node, client = await make_client()
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "messages": [{"role": "user", "content": "hi there"}],
+    })
+    assert r.status == 200, await r.text()
+    data = await r.json()
+    assert data["object"] == "chat.completion"
+    assert data["choices"][0]["message"]["content"]
+    assert data["usage"]["completion_tokens"] >= 1
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())
+
+
+def test_chat_completion_streaming():
+  async def go():
+    node, client = await make_client()
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "stream": True,
+      "messages": [{"role": "user", "content": "stream me"}],
+    })
+    assert r.status == 200
+    body = await r.content.read()
+    text = body.decode()
+    chunks = [l[6:] for l in text.splitlines() if l.startswith("data: ")]
+    assert chunks[-1] == "[DONE]"
+    parsed = [json.loads(c) for c in chunks[:-1]]
+    assert any(p["choices"][0].get("delta", {}).get("content") for p in parsed)
+    assert parsed[-1]["choices"][0]["finish_reason"] == "stop"
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())
+
+
+def test_invalid_model_rejected():
+  async def go():
+    node, client = await make_client()
+    r = await client.post("/v1/chat/completions", json={
+      "model": "not-a-model", "messages": [{"role": "user", "content": "x"}],
+    })
+    assert r.status == 400
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())
+
+
+def test_tinychat_served():
+  async def go():
+    node, client = await make_client()
+    r = await client.get("/")
+    assert r.status == 200
+    assert "xotorch_amd" in await r.text()
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())

@@ -19,6 +19,7 @@ from typing import Dict, List, Optional, Tuple
 import numpy as np
 
 from xotorch_amd.engine.interface import InferenceEngine
+from xotorch_amd.orchestration.tracing import tracer
 from xotorch_amd.helpers import DEBUG, AsyncCallbackSystem
 from xotorch_amd.orchestration.peer import PeerHandle
 from xotorch_amd.parallel.partitioning import PartitioningStrategy, RingMemoryWeightedPartitioningStrategy, map_partitions_to_shards
@@ -191,8 +192,11 @@ class Node:
       "base_shard": base_shard.to_dict(), "shard": shard.to_dict(),
       "prompt": prompt[:100], "request_id": request_id,
     }))
+    tracer.extract((inference_state or {}).get("traceparent"), request_id)
     if not shard.is_first_layer:
       # route the prompt to ring stage 0
+      inference_state = dict(inference_state or {})
+      inference_state["traceparent"] = tracer.inject(tracer.contexts[request_id])
       await self.forward_prompt(base_shard, prompt, request_id, self.get_first_partition_index(), inference_state)
     else:
       self.outstanding_requests[request_id] = "processing"
@@ -233,6 +237,7 @@ class Node:
       eos_id = getattr(getattr(self.inference_engine, "tokenizer", None), "eos_token_id", None)
       is_finished = (eos_id is not None and tok == eos_id) or len(buffered) >= self.max_generate_tokens
       self.buffered_token_output[request_id] = (buffered, is_finished)
+      tracer.handle_token(request_id, is_finished)
       self.trigger_on_token_callbacks(request_id, [tok], is_finished)
       asyncio.create_task(self.broadcast_result(request_id, buffered[-16:], is_finished))
       if is_finished:
