@@ -253,7 +253,19 @@ class TorchEngine(InferenceEngine):
       h = torch.nn.functional.linear(h, model.head_weight().to(h.dtype))
     return h
 
+  _lora_applied = False
+
+  def _maybe_apply_lora(self):
+    """XOT_LORA_RANK>0 switches training to LoRA adapters (lazy, once)."""
+    rank = int(os.getenv("XOT_LORA_RANK", "0"))
+    if rank > 0 and not self._lora_applied:
+      from xotorch_amd.train.lora import apply_lora
+      apply_lora(self.model, rank=rank, alpha=float(os.getenv("XOT_LORA_ALPHA", str(2 * rank))))
+      self._lora_applied = True
+      self._opt = None  # rebuild over the adapter params
+
   def _train_sync(self, request_id, inputs, targets, lengths, loss, opt, do_step):
+    self._maybe_apply_lora()
     model = self.model
     was_training = model.training
     model.train(do_step)
@@ -297,7 +309,8 @@ class TorchEngine(InferenceEngine):
 
   def _default_optimizer(self):
     if self._opt is None:
-      self._opt = torch.optim.AdamW(self.model.parameters(), lr=float(os.getenv("XOT_LR", "1e-5")))
+      params = [p for p in self.model.parameters() if p.requires_grad]
+      self._opt = torch.optim.AdamW(params, lr=float(os.getenv("XOT_LR", "1e-5")))
     return self._opt
 
   # ---------- checkpointing ----------
@@ -306,7 +319,11 @@ class TorchEngine(InferenceEngine):
     await self.ensure_shard(shard)
     from safetensors.torch import save_file
     Path(path).parent.mkdir(parents=True, exist_ok=True)
-    sd = {k: v.detach().contiguous().cpu() for k, v in self.model.state_dict().items() if not k.startswith("rope_")}
+    if self._lora_applied:
+      from xotorch_amd.train.lora import lora_state_dict
+      sd = {k: v.contiguous() for k, v in lora_state_dict(self.model).items()}
+    else:
+      sd = {k: v.detach().contiguous().cpu() for k, v in self.model.state_dict().items() if not k.startswith("rope_")}
     save_file(sd, path)
 
   async def load_checkpoint(self, shard: Shard, path: str):
