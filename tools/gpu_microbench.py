@@ -1,0 +1,64 @@
+"""Microbenchmarks for the custom kernels at decode shapes (B=64, 70B dims)."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+import torch  # noqa: E402
+
+from xotorch_amd.ops import _hip_ops as hip  # noqa: E402
+
+
+def timeit(fn, iters=200):
+  for _ in range(10):
+    fn()
+  torch.cuda.synchronize()
+  start = torch.cuda.Event(enable_timing=True)
+  end = torch.cuda.Event(enable_timing=True)
+  start.record()
+  for _ in range(iters):
+    fn()
+  end.record()
+  torch.cuda.synchronize()
+  return start.elapsed_time(end) / iters * 1000  # us
+
+
+def main():
+  B, D, I, H, KVH, hd, T = 64, 8192, 28672, 64, 8, 128, 640
+  gu = torch.randn(B, 1, 2 * I, device="cuda").to(torch.bfloat16)
+  x = torch.randn(B, 1, D, device="cuda").to(torch.bfloat16)
+  r = torch.randn_like(x)
+  w = torch.randn(D, device="cuda").to(torch.bfloat16)
+  us = timeit(lambda: hip.swiglu_packed(gu))
+  bytes_io = gu.numel() * 2 * 1.5
+  print(f"swiglu_packed [{B},2x{I}]: {us:.1f} us  {bytes_io/us/1e3:.2f} TB/s")
+  us = timeit(lambda: (torch.nn.functional.silu(gu[..., :I]) * gu[..., I:]))
+  print(f"torch silu*mul          : {us:.1f} us")
+  us = timeit(lambda: hip.rmsnorm(x, w, 1e-5))
+  print(f"rmsnorm [{B},{D}]       : {us:.1f} us  {x.numel()*2*2/us/1e3:.2f} TB/s")
+  us = timeit(lambda: hip.rmsnorm_residual(x, r, w, 1e-5))
+  print(f"rmsnorm_residual        : {us:.1f} us")
+  qkv = torch.randn(B, 1, (H + 2 * KVH) * hd, device="cuda").to(torch.bfloat16)
+  cos = torch.randn(T, hd // 2, device="cuda")
+  sin = torch.randn(T, hd // 2, device="cuda")
+  kc = torch.zeros(B, KVH, T, hd, dtype=torch.bfloat16, device="cuda")
+  vc = torch.zeros_like(kc)
+  pos = torch.full((1,), 500, dtype=torch.int32, device="cuda")
+  us = timeit(lambda: hip.rope_qkv_append(qkv, cos, sin, pos, kc, vc, H, KVH, hd))
+  print(f"rope_qkv_append         : {us:.1f} us")
+  q = qkv[:, :, : H * hd].view(B, 1, H, hd)
+  sl = torch.full((B,), 576, dtype=torch.int32, device="cuda")
+  us = timeit(lambda: hip.attn_decode(q, kc, vc, sl))
+  kv_bytes = B * KVH * 576 * hd * 2 * 2
+  print(f"attn_decode B{B} sl576  : {us:.1f} us  {kv_bytes/us/1e3:.2f} TB/s")
+  # GEMM reference points (hipBLASLt via torch)
+  for (m, k, n, tag) in ((B, D, (H + 2 * KVH) * hd, "qkv"), (B, D, 2 * I, "gate_up"),
+                         (B, I, D, "down"), (B, H * hd, D, "o")):
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
+    wt = torch.randn(n, k, device="cuda").to(torch.bfloat16)
+    us = timeit(lambda: torch.nn.functional.linear(a, wt))
+    wbytes = n * k * 2
+    print(f"linear {tag:8s} [{m},{k}]x[{k},{n}]: {us:7.1f} us  {wbytes/us/1e3:.2f} TB/s(w)")
+
+
+if __name__ == "__main__":
+  main()
