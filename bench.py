@@ -27,6 +27,15 @@ from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parent))
 
+# hipBLASLt/rocBLAS algorithm selection pre-tuned on MI355X (gfx950) for the
+# bench GEMM shapes; read-only (tuning off) so TTFT stays clean. Must be set
+# before torch initializes.
+_tuned = Path(__file__).resolve().parent / "profiles" / "tunableop_gfx950.csv"
+if _tuned.exists() and os.getenv("XOT_TUNABLEOP", "1") == "1":
+  os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+  os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+  os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", str(_tuned))
+
 import torch  # noqa: E402
 
 
@@ -36,7 +45,7 @@ def main():
   p.add_argument("--steps", type=int, default=32)
   p.add_argument("--warmup", type=int, default=8)
   p.add_argument("--model", type=str, default="llama-3-70b")
-  p.add_argument("--mb-batch", type=int, default=32, help="sequences per micro-batch (per pipeline slot)")
+  p.add_argument("--mb-batch", type=int, default=64, help="sequences per micro-batch (per pipeline slot)")
   p.add_argument("--prompt-len", type=int, default=512)
   p.add_argument("--no-graphs", action="store_true")
   p.add_argument("--device", type=str, default=None)
