@@ -161,3 +161,34 @@ def test_model_gpu_vs_cpu():
     lg2 = m_gpu(nxt.cuda(), caches=cache_gpu.caches, positions=torch.tensor([S], dtype=torch.int32, device="cuda"),
                 start_pos=S, is_decode=True, seq_lens=sl)
     assert torch.allclose(lc2, lg2.float().cpu(), atol=0.15, rtol=0.15)
+
+
+@pytest.mark.parametrize("M,K,N,bias", [
+  (32, 2048, 256, False),
+  (64, 8192, 10240, False),
+  (64, 28672, 8192, False),
+  (128, 8192, 1152, True),   # qwen-7b qkv shape class (bias)
+  (256, 4096, 14336, False),
+  (96, 1024, 128, True),     # single n-tile, K not pow2-split friendly
+  (64, 8192, 128256, False), # lm_head (large N, splitk=1 path)
+])
+def test_skinny_gemm(hip, M, K, N, bias):
+  from xotorch_amd.ops import _hip_ops
+  x = bt(M, K, scale=0.5, seed=M + N)
+  w = bt(N, K, scale=0.02, seed=K)
+  b = bt(N, seed=5) if bias else None
+  got = _hip_ops.skinny_gemm(x, w, b).float()
+  ref = torch.nn.functional.linear(x.float(), w.float(), b.float() if bias else None)
+  denom = ref.abs().max().item() + 1e-9
+  assert (got - ref).abs().max().item() / denom < 2e-2, \
+    f"max abs err {(got - ref).abs().max().item()} denom {denom}"
+
+
+def test_ops_linear_dispatch_matches(hip):
+  """ops.linear must agree with F.linear on an eligible decode shape."""
+  from xotorch_amd import ops
+  x = bt(64, 1, 4096, scale=0.5).view(64, 1, 4096)
+  w = bt(512, 4096, scale=0.02, seed=9)
+  got = ops.linear(x, w).float()
+  ref = torch.nn.functional.linear(x.float(), w.float())
+  assert (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-9) < 2e-2

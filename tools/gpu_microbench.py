@@ -50,14 +50,21 @@ def main():
   us = timeit(lambda: hip.attn_decode(q, kc, vc, sl))
   kv_bytes = B * KVH * 576 * hd * 2 * 2
   print(f"attn_decode B{B} sl576  : {us:.1f} us  {kv_bytes/us/1e3:.2f} TB/s")
-  # GEMM reference points (hipBLASLt via torch)
-  for (m, k, n, tag) in ((B, D, (H + 2 * KVH) * hd, "qkv"), (B, D, 2 * I, "gate_up"),
-                         (B, I, D, "down"), (B, H * hd, D, "o")):
-    a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
-    wt = torch.randn(n, k, device="cuda").to(torch.bfloat16)
-    us = timeit(lambda: torch.nn.functional.linear(a, wt))
-    wbytes = n * k * 2
-    print(f"linear {tag:8s} [{m},{k}]x[{k},{n}]: {us:7.1f} us  {wbytes/us/1e3:.2f} TB/s(w)")
+  # GEMM A/B: hipBLASLt via torch vs the skinny weight-streaming MFMA kernel
+  for M in (64, 128, 256):
+    print(f"-- decode GEMMs at M={M} (70B dims) --")
+    for (m, k, n, tag) in ((M, D, (H + 2 * KVH) * hd, "qkv"), (M, D, 2 * I, "gate_up"),
+                           (M, I, D, "down"), (M, H * hd, D, "o"), (M, D, 128256, "lm_head")):
+      a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
+      wt = torch.randn(n, k, device="cuda").to(torch.bfloat16) * 0.02
+      us = timeit(lambda: torch.nn.functional.linear(a, wt), iters=50)
+      wbytes = n * k * 2
+      us2 = timeit(lambda: hip.skinny_gemm(a, wt, None), iters=50)
+      ref = torch.nn.functional.linear(a.float(), wt.float())
+      got = hip.skinny_gemm(a, wt, None).float()
+      err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-9)
+      print(f"linear {tag:8s} [{m},{k}]x[{k},{n}]: blaslt {us:7.1f} us {wbytes/us/1e3:5.2f} TB/s"
+            f" | skinny {us2:7.1f} us {wbytes/us2/1e3:5.2f} TB/s  relerr {err:.2e}")
 
 
 if __name__ == "__main__":

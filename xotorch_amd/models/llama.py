@@ -27,6 +27,15 @@ from xotorch_amd.ops.torch_ref import rope_cos_sin
 from xotorch_amd.shard import Shard
 
 
+class XotLinear(nn.Linear):
+  """nn.Linear routed through ops.linear: decode-shaped (rows <= 256) bf16
+  GEMMs run the hand-written CDNA4 weight-streaming MFMA kernel, everything
+  else hipBLASLt. Storage/state-dict identical to nn.Linear."""
+
+  def forward(self, x):
+    return ops.linear(x, self.weight, self.bias)
+
+
 class RMSNorm(nn.Module):
   def __init__(self, dim: int, eps: float):
     super().__init__()
@@ -46,8 +55,8 @@ class Attention(nn.Module):
     super().__init__()
     self.cfg = cfg
     H, KVH, hd, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim, cfg.dim
-    self.qkv_proj = nn.Linear(D, (H + 2 * KVH) * hd, bias=cfg.attn_bias)
-    self.o_proj = nn.Linear(H * hd, D, bias=False)
+    self.qkv_proj = XotLinear(D, (H + 2 * KVH) * hd, bias=cfg.attn_bias)
+    self.o_proj = XotLinear(H * hd, D, bias=False)
 
   def forward(self, x, cos, sin, positions, k_cache, v_cache, start_pos: int, is_decode: bool, seq_lens=None):
     B, S, _ = x.shape
@@ -72,8 +81,8 @@ class MLP(nn.Module):
     super().__init__()
     I = intermediate or cfg.intermediate_dim
     self.intermediate = I
-    self.gate_up_proj = nn.Linear(cfg.dim, 2 * I, bias=False)
-    self.down_proj = nn.Linear(I, cfg.dim, bias=False)
+    self.gate_up_proj = XotLinear(cfg.dim, 2 * I, bias=False)
+    self.down_proj = XotLinear(I, cfg.dim, bias=False)
 
   def forward(self, x):
     return self.down_proj(ops.swiglu_packed(self.gate_up_proj(x)))
@@ -136,11 +145,11 @@ class ShardedModel(nn.Module):
     if shard.is_last_layer:
       self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
       if not cfg.tie_word_embeddings:
-        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        self.lm_head = XotLinear(cfg.dim, cfg.vocab_size, bias=False)
       elif not shard.is_first_layer:
         # tied embeddings but the first shard (owner of embed_tokens) is
         # elsewhere: keep a local copy of the embedding matrix as the head.
-        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        self.lm_head = XotLinear(cfg.dim, cfg.vocab_size, bias=False)
     cos, sin = rope_cos_sin(cfg.head_dim, cfg.max_seq_len, cfg.rope_theta, cfg.rope_scaling)
     self.register_buffer("rope_cos", cos, persistent=False)
     self.register_buffer("rope_sin", sin, persistent=False)
@@ -202,7 +211,7 @@ class ShardedModel(nn.Module):
     if last_only and h.shape[1] > 1:
       h = h[:, -1:, :].contiguous()  # kernels require contiguous rows
     h = self.norm(h)
-    logits = torch.nn.functional.linear(h, self.head_weight().to(h.dtype))
+    logits = ops.linear(h, self.head_weight().to(h.dtype))
     if is_decode or last_only:
       return logits[:, -1, :]
     return logits
