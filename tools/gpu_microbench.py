@@ -60,11 +60,18 @@ def main():
       us = timeit(lambda: torch.nn.functional.linear(a, wt), iters=50)
       wbytes = n * k * 2
       us2 = timeit(lambda: hip.skinny_gemm(a, wt, None), iters=50)
+      from xotorch_amd.ops import pack_decode_weight
+      wp = pack_decode_weight(wt)
+      us3 = timeit(lambda: hip.skinny_gemm_packed(a, wp, n, None), iters=50)
       ref = torch.nn.functional.linear(a.float(), wt.float())
       got = hip.skinny_gemm(a, wt, None).float()
       err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-9)
+      gotp = hip.skinny_gemm_packed(a, wp, n, None).float()
+      errp = (gotp - ref).abs().max().item() / (ref.abs().max().item() + 1e-9)
+      del wp
       print(f"linear {tag:8s} [{m},{k}]x[{k},{n}]: blaslt {us:7.1f} us {wbytes/us/1e3:5.2f} TB/s"
-            f" | skinny {us2:7.1f} us {wbytes/us2/1e3:5.2f} TB/s  relerr {err:.2e}")
+            f" | skinny {us2:7.1f} us {wbytes/us2/1e3:5.2f} TB/s e={err:.1e}"
+            f" | packed {us3:7.1f} us {wbytes/us3/1e3:5.2f} TB/s e={errp:.1e}")
 
 
 if __name__ == "__main__":

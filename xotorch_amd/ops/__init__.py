@@ -134,6 +134,16 @@ def swiglu_packed(gu):
 _SKINNY = os.getenv("XOT_SKINNY", "1") == "1"
 
 
+def pack_decode_weight(w: torch.Tensor) -> torch.Tensor:
+  """Pre-shuffle a [N, K] bf16 weight into MFMA A-fragment order for the
+  packed skinny decode GEMM: [N/32, K/16, 64 lanes, 8 bf16] with lane =
+  (k-half)*32 + n-row. One-time at weight load; coalesced 1 KB wave streams
+  at decode."""
+  N, K = w.shape
+  assert N % 32 == 0 and K % 64 == 0, (N, K)
+  return w.view(N // 32, 32, K // 16, 2, 8).permute(0, 2, 3, 1, 4).contiguous()
+
+
 def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None) -> torch.Tensor:
   """y = x @ weight^T (+bias).
 

@@ -478,6 +478,124 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v2: prepacked-weight variant.  W is pre-shuffled ONCE at load time into
+// MFMA A-fragment order: [N/32][K/16][lane 0..63][8 bf16], lane = (k-half)*32
+// + n-row — so each A-fragment load is one fully-coalesced wave-wide 1 KB
+// stream (v1's fragment-shaped 16 B gathers are TA/load-path-bound: measured
+// 2.0 vs 5.3 TB/s).  X is staged cooperatively per 64-k step into a padded
+// LDS image (row stride 72 elems = 144 B -> ds_read_b128 bank-conflict-free)
+// and read back as B fragments.  K % 64 == 0 required.
+// ---------------------------------------------------------------------------
+
+template <int MT, bool SPLIT>
+__global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
+    const unsigned short* __restrict__ Wp, const unsigned short* __restrict__ X,
+    unsigned short* __restrict__ Y, float* __restrict__ P,
+    const unsigned short* __restrict__ bias,
+    int N, long long K, int kc, int nsplit) {
+  const int ntiles = N >> 7;
+  const int tile = blockIdx.x % ntiles;
+  const int split = blockIdx.x / ntiles;
+  const long long k0 = (long long)split * kc;
+  const long long k1 = min(k0 + (long long)kc, K);
+  if (k0 >= k1) return;
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+
+  __shared__ unsigned short xs[MT * 32 * 72];
+
+  floatx16 acc[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) acc[t] = (floatx16)(0.f);
+
+  // A stream: packed row for this wave's 32-n tile
+  const int n32 = tile * 4 + wv;
+  const unsigned short* wp = Wp + ((size_t)n32 * (K >> 4) + (k0 >> 4)) * 512 + (size_t)lane * 8;
+  // X stage: thread tid covers chunk (row r = c/8, 16B piece q = c%8) of each
+  // 32-row m-tile; global [M,K] row-major.
+  const int xr = tid >> 3, xq = tid & 7;
+  const unsigned short* xp = X + (size_t)xr * K + k0 + xq * 8;
+  const int xs_off = xr * 72 + xq * 8;  // padded LDS image
+
+  const int nsteps = (int)((k1 - k0) >> 6);
+  ushort8 xv[MT];
+  bf16x8 a_cur[4], a_nxt[4];
+
+  // prologue: stage step 0, preload A(0)
+#pragma unroll
+  for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) a_cur[u] = *reinterpret_cast<const bf16x8*>(wp + u * 512);
+  wp += 4 * 512;
+#pragma unroll
+  for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];
+  __syncthreads();
+
+  for (int s = 0; s < nsteps; ++s) {
+    const bool last = (s == nsteps - 1);
+    if (!last) {
+      // prefetch next step's X chunks and A fragments (plain loads stay in
+      // flight through the compute below)
+#pragma unroll
+      for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K + (s + 1) * 64);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) a_nxt[u] = *reinterpret_cast<const bf16x8*>(wp + u * 512);
+      wp += 4 * 512;
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        // B fragment: lane reads X[m = lane&31][u*16 + (lane>>5)*8 ..+8] of tile t
+        const bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            xs + t * 32 * 72 + (lane & 31) * 72 + u * 16 + (lane >> 5) * 8);
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_cur[u], b, acc[t], 0, 0, 0);
+      }
+    }
+    if (!last) {
+      __syncthreads();
+#pragma unroll
+      for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];
+      __syncthreads();
+#pragma unroll
+      for (int u = 0; u < 4; ++u) a_cur[u] = a_nxt[u];
+    }
+  }
+
+  const int m_local = lane & 31;
+  const int nbase = tile * 128 + wv * 32 + 4 * (lane >> 5);
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    const int m = t * 32 + m_local;
+    if (SPLIT) {
+      float* prow = P + ((size_t)split * (MT * 32) + m) * N;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        floatx4 v4;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) v4[j] = acc[t][g * 4 + j];
+        *reinterpret_cast<floatx4*>(prow + nbase + g * 8) = v4;
+      }
+    } else {
+      unsigned short* yrow = Y + (size_t)m * N;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        unsigned short o[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float v = acc[t][g * 4 + j];
+          if (bias) v += b2f(bias[nbase + g * 8 + j]);
+          o[j] = f2b(v);
+        }
+        *reinterpret_cast<unsigned long long*>(yrow + nbase + g * 8) =
+            *reinterpret_cast<unsigned long long*>(o);
+      }
+    }
+  }
+}
+
 // combine fp32 split-K partials [S, M, N] -> bf16 [M, N] (+bias)
 __global__ __launch_bounds__(256) void skinny_combine_kernel(
     const float* __restrict__ P, unsigned short* __restrict__ Y,
@@ -705,9 +823,66 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
+// Packed variant. wp: the [N/32, K/16, 64, 8] prepack of w (see
+// ops.pack_decode_weight), x: [M, K] bf16, M % 32 == 0, K % 64 == 0.
+torch::Tensor skinny_gemm_packed(torch::Tensor x, torch::Tensor wp, int64_t N,
+                                 c10::optional<torch::Tensor> bias) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  CHK(wp.is_cuda() && wp.dtype() == torch::kBFloat16 && wp.is_contiguous());
+  const long long K = wp.numel() / N;
+  const long long M = x.numel() / K;
+  CHK(M >= 32 && M <= 256 && M % 32 == 0);
+  CHK(N % 128 == 0 && K % 64 == 0);
+  const unsigned short* bptr = nullptr;
+  if (bias.has_value()) {
+    CHK(bias->is_contiguous() && bias->dtype() == torch::kBFloat16 && bias->numel() == N);
+    bptr = (const unsigned short*)bias->data_ptr();
+  }
+  auto sizes = x.sizes().vec();
+  sizes.back() = (long)N;
+  auto y = torch::empty(sizes, x.options());
+  const int ntiles = (int)(N / 128);
+  int nsplit = 1;
+  while (ntiles * nsplit * 2 < 1024 && (K / (nsplit * 2)) >= 1024 && nsplit < 16) nsplit *= 2;
+  int kc = (int)((K / nsplit + 63) / 64 * 64);
+  while ((long long)kc * (nsplit - 1) >= K) nsplit--;
+  auto stream = cur_stream();
+  const int MT = (int)(M / 32);
+  const dim3 grid(ntiles * nsplit), block(256);
+#define SGP_CASE(MTV) \
+  case MTV: \
+    if (nsplit == 1) { \
+      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false>), grid, block, 0, stream, \
+                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                         (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit); \
+    } else { \
+      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true>), grid, block, 0, stream, \
+                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                         nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+    } \
+    break;
+  if (nsplit == 1) {
+    torch::Tensor P;
+    switch (MT) { SGP_CASE(1) SGP_CASE(2) SGP_CASE(3) SGP_CASE(4) SGP_CASE(5) SGP_CASE(6) SGP_CASE(7) SGP_CASE(8) }
+  } else {
+    auto P = torch::empty({nsplit, M, (long long)N},
+                          torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+    switch (MT) { SGP_CASE(1) SGP_CASE(2) SGP_CASE(3) SGP_CASE(4) SGP_CASE(5) SGP_CASE(6) SGP_CASE(7) SGP_CASE(8) }
+    const long long MN = M * (long long)N;
+    const int blocks = (int)std::min<long long>(2048, (MN / 4 + 255) / 256);
+    hipLaunchKernelGGL(skinny_combine_kernel, dim3(blocks), dim3(256), 0, stream,
+                       P.data_ptr<float>(), (unsigned short*)y.data_ptr(), bptr, MN, N, nsplit);
+  }
+#undef SGP_CASE
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm, "decode GEMM y = x @ w^T (+bias), bf16 MFMA weight-streaming",
         py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+  m.def("skinny_gemm_packed", &skinny_gemm_packed,
+        "decode GEMM on prepacked weights (MFMA fragment order)",
+        py::arg("x"), py::arg("wp"), py::arg("n"), py::arg("bias") = py::none());
   m.def("swiglu_packed", &swiglu_packed, "SwiGLU on the packed [gate|up] GEMM output");
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual add + RMSNorm");
