@@ -192,3 +192,34 @@ def test_ops_linear_dispatch_matches(hip):
   got = ops.linear(x, w).float()
   ref = torch.nn.functional.linear(x.float(), w.float())
   assert (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-9) < 2e-2
+
+
+@pytest.mark.parametrize("M,K,N,bias", [
+  (64, 8192, 1280, True),
+  (64, 28672, 8192, False),
+  (256, 4096, 14336, False),
+  (32, 1024, 128, False),
+])
+def test_skinny_gemm_packed(hip, M, K, N, bias):
+  from xotorch_amd import ops as xops
+  from xotorch_amd.ops import _hip_ops
+  x = bt(M, K, scale=0.5, seed=M + N + 1)
+  w = bt(N, K, scale=0.02, seed=K + 1)
+  b = bt(N, seed=7) if bias else None
+  wp = xops.pack_decode_weight(w)
+  got = _hip_ops.skinny_gemm_packed(x, wp, N, b).float()
+  ref = torch.nn.functional.linear(x.float(), w.float(), b.float() if bias else None)
+  denom = ref.abs().max().item() + 1e-9
+  assert (got - ref).abs().max().item() / denom < 2e-2
+
+
+def test_xotlinear_packed_matches_unpacked(hip):
+  from xotorch_amd.models.llama import XotLinear
+  m = XotLinear(4096, 512, bias=False).to("cuda").to(torch.bfloat16)
+  x = bt(64, 1, 4096, scale=0.5)
+  with torch.inference_mode():
+    y0 = m(x).float()
+    m.pack_decode()
+    assert m.weight_packed is not None
+    y1 = m(x).float()
+  assert (y0 - y1).abs().max().item() / (y0.abs().max().item() + 1e-9) < 2e-2

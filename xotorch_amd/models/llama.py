@@ -16,6 +16,7 @@ and llm_utils.py:335-489), built for MI355X execution:
 from __future__ import annotations
 
 import math
+import os
 from typing import List, Optional, Tuple
 
 import torch
@@ -28,11 +29,40 @@ from xotorch_amd.shard import Shard
 
 
 class XotLinear(nn.Linear):
-  """nn.Linear routed through ops.linear: decode-shaped (rows <= 256) bf16
-  GEMMs run the hand-written CDNA4 weight-streaming MFMA kernel, everything
-  else hipBLASLt. Storage/state-dict identical to nn.Linear."""
+  """nn.Linear with an optional decode-path prepack: `pack_decode()` stores a
+  second copy of the weight in MFMA A-fragment order ([N/32,K/16,64,8], see
+  ops.pack_decode_weight); decode-shaped (rows 32..256) bf16 GEMMs then run
+  the hand-written CDNA4 weight-streaming kernel (measured 5.2-6.0 TB/s vs
+  hipBLASLt 3.0-5.4 at L3-cold decode shapes), everything else (prefill,
+  training, CPU) hipBLASLt/aten. Storage/state-dict identical to nn.Linear."""
+
+  weight_packed: Optional[torch.Tensor]
+
+  def __init__(self, *a, **kw):
+    super().__init__(*a, **kw)
+    self.weight_packed = None
+
+  def packable(self) -> bool:
+    N, K = self.weight.shape
+    return N % 128 == 0 and K % 64 == 0
+
+  def pack_decode(self):
+    if self.weight_packed is None and self.packable() and self.weight.is_cuda \
+       and self.weight.dtype == torch.bfloat16:
+      self.weight_packed = ops.pack_decode_weight(self.weight.detach())
+
+  def unpack_decode(self):
+    self.weight_packed = None
 
   def forward(self, x):
+    if self.weight_packed is not None and x.is_cuda and x.dtype == torch.bfloat16:
+      K = self.weight.shape[1]
+      M = x.numel() // K
+      if 32 <= M <= 256 and M % 32 == 0 and x.is_contiguous() and not torch.is_grad_enabled():
+        from xotorch_amd.ops import _load_hip
+        hip = _load_hip()
+        if hip is not None:
+          return hip.skinny_gemm_packed(x, self.weight_packed, self.weight.shape[0], self.bias)
     return ops.linear(x, self.weight, self.bias)
 
 
@@ -174,6 +204,43 @@ class ShardedModel(nn.Module):
   def local_layer_ids(self) -> List[int]:
     return list(range(self.shard.start_layer, self.shard.end_layer + 1))
 
+  def pack_decode_weights(self, reserve_bytes: int = 8 << 30) -> int:
+    """Prepack decode-GEMM weights (second copy in MFMA fragment order) in
+    descending measured-win order — down_proj (hipBLASLt ~3 TB/s at K=28672
+    vs 5.4 packed), lm_head, gate_up, qkv/o — greedily while at least
+    `reserve_bytes` of HBM stay free (KV caches and activations are usually
+    allocated before this is called). Returns bytes packed. Policy override:
+    XOT_PACK=none|down|all."""
+    mode = os.getenv("XOT_PACK", "auto")
+    if mode == "none" or not torch.cuda.is_available():
+      return 0
+    groups: List[List[XotLinear]] = [[], [], [], []]
+    for name, mod in self.named_modules():
+      if not isinstance(mod, XotLinear) or not mod.packable():
+        continue
+      if "down_proj" in name:
+        groups[0].append(mod)
+      elif "lm_head" in name:
+        groups[1].append(mod)
+      elif "gate_up_proj" in name:
+        groups[2].append(mod)
+      else:  # qkv_proj / o_proj / experts' projections
+        groups[3].append(mod)
+    if self.cfg.tie_word_embeddings and not hasattr(self, "lm_head"):
+      pass  # tied head uses embed_tokens.weight via ops.linear (unpacked)
+    if mode == "down":
+      groups = groups[:2]
+    packed = 0
+    for grp in groups:
+      need = sum(m.weight.numel() * 2 for m in grp)
+      free, _ = torch.cuda.mem_get_info()
+      if need + reserve_bytes > free:
+        continue
+      for m in grp:
+        m.pack_decode()
+      packed += need
+    return packed
+
   def head_weight(self):
     if self.cfg.tie_word_embeddings and hasattr(self, "embed_tokens"):
       return self.embed_tokens.weight
@@ -211,7 +278,10 @@ class ShardedModel(nn.Module):
     if last_only and h.shape[1] > 1:
       h = h[:, -1:, :].contiguous()  # kernels require contiguous rows
     h = self.norm(h)
-    logits = ops.linear(h, self.head_weight().to(h.dtype))
+    if hasattr(self, "lm_head"):
+      logits = self.lm_head(h)  # XotLinear: packed decode kernel when prepacked
+    else:
+      logits = ops.linear(h, self.head_weight().to(h.dtype))
     if is_decode or last_only:
       return logits[:, -1, :]
     return logits

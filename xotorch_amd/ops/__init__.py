@@ -145,26 +145,12 @@ def pack_decode_weight(w: torch.Tensor) -> torch.Tensor:
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-  """y = x @ weight^T (+bias).
+  """y = x @ weight^T (+bias) on hipBLASLt/aten.
 
-  Decode shapes (rows <= 256) go to the hand-written CDNA4 weight-streaming
-  MFMA kernel (`skinny_gemm`): at decode the GEMM is bandwidth-bound on the
-  weight matrix and hipBLASLt measures well under the HBM3E roofline at large
-  K (see profiles/). Prefill/compute-bound shapes stay on hipBLASLt.
+  The decode hot path uses XotLinear.pack_decode() + skinny_gemm_packed (the
+  hand-written CDNA4 weight-streaming MFMA kernel) instead; this is the
+  library-GEMM path for prefill/training/CPU and unpacked weights.
   """
-  if (
-    _SKINNY
-    and x.is_cuda
-    and x.dtype == torch.bfloat16
-    and weight.dtype == torch.bfloat16
-    and not (torch.is_grad_enabled() and (x.requires_grad or weight.requires_grad))
-    and x.is_contiguous()
-  ):
-    K = weight.shape[1]
-    N = weight.shape[0]
-    M = x.numel() // K
-    if 32 <= M <= 256 and M % 32 == 0 and K % 16 == 0 and N % 128 == 0 and _use_hip(x):
-      return _hip.skinny_gemm(x, weight, bias)
   return torch.nn.functional.linear(x, weight, bias)
 
 

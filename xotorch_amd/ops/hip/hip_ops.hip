@@ -523,11 +523,13 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   ushort8 xv[MT];
   bf16x8 a_cur[4], a_nxt[4];
 
-  // prologue: stage step 0, preload A(0)
+  // prologue: stage step 0, preload A(0).  The W stream is read exactly once
+  // per launch -> non-temporal (L1-bypass) loads.
 #pragma unroll
   for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K);
 #pragma unroll
-  for (int u = 0; u < 4; ++u) a_cur[u] = *reinterpret_cast<const bf16x8*>(wp + u * 512);
+  for (int u = 0; u < 4; ++u)
+    a_cur[u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
   wp += 4 * 512;
 #pragma unroll
   for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];
@@ -541,7 +543,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
 #pragma unroll
       for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K + (s + 1) * 64);
 #pragma unroll
-      for (int u = 0; u < 4; ++u) a_nxt[u] = *reinterpret_cast<const bf16x8*>(wp + u * 512);
+      for (int u = 0; u < 4; ++u)
+        a_nxt[u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
       wp += 4 * 512;
     }
 #pragma unroll

@@ -94,6 +94,14 @@ class RingPipeline:
     model.reset_rope()  # to_empty left the tables uninitialized
     model.eval()
     self.model = model
+    if device == "cuda":
+      # decode-GEMM weight prepack (memory permitting; KV caches below still
+      # need room — reserve their size + headroom before packing greedily)
+      kv_bytes = (
+        self.M * 2 * shards[rank].get_layer_count() * mb_batch * self.cfg.n_kv_heads
+        * self.total_len * self.cfg.head_dim * (2 if dtype == torch.bfloat16 else 4)
+      )
+      model.pack_decode_weights(reserve_bytes=kv_bytes + (16 << 30))
 
     # --- per-micro-batch state ---
     B = mb_batch
