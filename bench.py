@@ -32,9 +32,17 @@ sys.path.insert(0, str(Path(__file__).resolve().parent))
 # before torch initializes.
 _tuned = Path(__file__).resolve().parent / "profiles" / "tunableop_gfx950.csv"
 if _tuned.exists() and os.getenv("XOT_TUNABLEOP", "1") == "1":
+  # TunableOp inserts the device ordinal before ".csv" when resolving the
+  # filename -- stage per-ordinal copies it will actually find.
+  import shutil
+  base = Path("/tmp/xot_tunableop.csv")
+  for i in list(range(8)) + [""]:
+    dst = Path(f"/tmp/xot_tunableop{i}.csv")
+    if not dst.exists():
+      shutil.copyfile(_tuned, dst)
   os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
   os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
-  os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", str(_tuned))
+  os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", str(base))
 
 import torch  # noqa: E402
 
