@@ -219,16 +219,20 @@ class ShardedModel(nn.Module):
       if not isinstance(mod, XotLinear) or not mod.packable():
         continue
       if "down_proj" in name:
-        groups[0].append(mod)
+        # the packed kernel wins when the split-K grid still fills the chip
+        # (>= ~1.5 blocks/CU); small-N down projections stay on hipBLASLt
+        if mod.weight.shape[0] // 128 * 8 >= 384 or mode == "all":
+          groups[0].append(mod)
       elif "lm_head" in name:
         groups[1].append(mod)
       elif "gate_up_proj" in name:
         groups[2].append(mod)
       else:  # qkv_proj / o_proj / experts' projections
         groups[3].append(mod)
-    if self.cfg.tie_word_embeddings and not hasattr(self, "lm_head"):
-      pass  # tied head uses embed_tokens.weight via ops.linear (unpacked)
-    if mode == "down":
+    if mode != "all":
+      # default: only the measured winners — down_proj (hipBLASLt ~3 TB/s at
+      # long K vs 5.4 packed) and lm_head (5.6 -> 6.0 TB/s); qkv/o/gate_up
+      # measure at or above the packed kernel on hipBLASLt.
       groups = groups[:2]
     packed = 0
     for grp in groups:
