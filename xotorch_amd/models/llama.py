@@ -235,14 +235,22 @@ class ShardedModel(nn.Module):
       # measure at or above the packed kernel on hipBLASLt.
       groups = groups[:2]
     packed = 0
-    for grp in groups:
+    debug = os.getenv("XOT_DEBUG", "0") != "0"
+    for gi, grp in enumerate(groups):
+      if not grp:
+        continue
       need = sum(m.weight.numel() * 2 for m in grp)
       free, _ = torch.cuda.mem_get_info()
       if need + reserve_bytes > free:
+        if debug:
+          print(f"[xot] pack group {gi}: skip (need {need>>20} MiB + reserve "
+                f"{reserve_bytes>>20} MiB > free {free>>20} MiB)", flush=True)
         continue
       for m in grp:
         m.pack_decode()
       packed += need
+      if debug:
+        print(f"[xot] pack group {gi}: packed {len(grp)} modules, {need>>20} MiB", flush=True)
     return packed
 
   def head_weight(self):

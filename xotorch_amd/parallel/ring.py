@@ -83,8 +83,16 @@ class RingPipeline:
     self.prev_rank = (rank - 1) % world
 
     # --- model ---
-    with torch.device("meta"):
-      model = ShardedModel(self.cfg, self.shard)
+    # construct at the target dtype so to_empty materializes bf16 directly
+    # (fp32-then-cast would peak at 2x the weight bytes and leave the
+    # allocator's reserved pool at that level)
+    prev_dtype = torch.get_default_dtype()
+    torch.set_default_dtype(dtype)
+    try:
+      with torch.device("meta"):
+        model = ShardedModel(self.cfg, self.shard)
+    finally:
+      torch.set_default_dtype(prev_dtype)
     model = model.to_empty(device=device)
     model = model.to(dtype)
     if device == "cuda" and self.cfg.dim >= 2048:
@@ -101,6 +109,8 @@ class RingPipeline:
         self.M * 2 * shards[rank].get_layer_count() * mb_batch * self.cfg.n_kv_heads
         * self.total_len * self.cfg.head_dim * (2 if dtype == torch.bfloat16 else 4)
       )
+      torch.cuda.empty_cache()  # release init-time cached blocks so the
+      # pack policy's mem_get_info reflects actually-usable HBM
       model.pack_decode_weights(reserve_bytes=kv_bytes + (16 << 30))
 
     # --- per-micro-batch state ---
