@@ -230,10 +230,11 @@ class ShardedModel(nn.Module):
       else:  # qkv_proj / o_proj / experts' projections
         groups[3].append(mod)
     if mode != "all":
-      # default: only the measured winners — down_proj (hipBLASLt ~3 TB/s at
-      # long K vs 5.4 packed) and lm_head (5.6 -> 6.0 TB/s); qkv/o/gate_up
-      # measure at or above the packed kernel on hipBLASLt.
-      groups = groups[:2]
+      # default: the measured winners — down_proj (tuned hipBLASLt 110 us vs
+      # 84.6 packed at 70B decode shapes), lm_head (5.3 -> 6.2 TB/s), gate_up
+      # (175.5 vs 178.9 us); qkv/o measure tied on hipBLASLt — not worth the
+      # second weight copy.
+      groups = groups[:3]
     packed = 0
     debug = os.getenv("XOT_DEBUG", "0") != "0"
     for gi, grp in enumerate(groups):

@@ -521,51 +521,65 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
 
   const int nsteps = (int)((k1 - k0) >> 6);
   ushort8 xv[MT];
-  bf16x8 a_cur[4], a_nxt[4];
+  bf16x8 a_buf[2][4];
 
-  // prologue: stage step 0, preload A(0).  The W stream is read exactly once
-  // per launch -> non-temporal (L1-bypass) loads.
+  // prologue: stage step 0, preload A(0) and A(1).  The W stream is read
+  // exactly once per launch -> non-temporal (L1-bypass) loads; depth-2
+  // prefetch keeps ~8 KB per wave in flight across the staging barriers.
 #pragma unroll
   for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K);
+  const unsigned short* wp1 = (nsteps > 1) ? wp + 4 * 512 : wp;  // clamp: no OOB at nsteps==1
 #pragma unroll
-  for (int u = 0; u < 4; ++u)
-    a_cur[u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
-  wp += 4 * 512;
+  for (int u = 0; u < 4; ++u) {
+    a_buf[0][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
+    a_buf[1][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp1 + u * 512));
+  }
+  wp += 8 * 512;
 #pragma unroll
   for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];
   __syncthreads();
 
-  for (int s = 0; s < nsteps; ++s) {
-    const bool last = (s == nsteps - 1);
-    if (!last) {
-      // prefetch next step's X chunks and A fragments (plain loads stay in
-      // flight through the compute below)
-#pragma unroll
-      for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K + (s + 1) * 64);
-#pragma unroll
-      for (int u = 0; u < 4; ++u)
-        a_nxt[u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
-      wp += 4 * 512;
-    }
-#pragma unroll
-    for (int u = 0; u < 4; ++u) {
-#pragma unroll
-      for (int t = 0; t < MT; ++t) {
-        // B fragment: lane reads X[m = lane&31][u*16 + (lane>>5)*8 ..+8] of tile t
-        const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            xs + t * 32 * 72 + (lane & 31) * 72 + u * 16 + (lane >> 5) * 8);
-        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_cur[u], b, acc[t], 0, 0, 0);
-      }
-    }
-    if (!last) {
-      __syncthreads();
-#pragma unroll
-      for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];
-      __syncthreads();
-#pragma unroll
-      for (int u = 0; u < 4; ++u) a_cur[u] = a_nxt[u];
-    }
+  // NOTE: the A double-buffer index must be a compile-time constant — a
+  // dynamic a_buf[s & 1] spills the whole array to scratch (measured 25-40x).
+#define SGP_STEP(BUF)                                                                          \
+  {                                                                                            \
+    const bool last = (s == nsteps - 1);                                                       \
+    if (!last) {                                                                               \
+      _Pragma("unroll")                                                                        \
+      for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K + (s + 1) * 64); \
+    }                                                                                          \
+    _Pragma("unroll")                                                                          \
+    for (int u = 0; u < 4; ++u) {                                                              \
+      _Pragma("unroll")                                                                        \
+      for (int t = 0; t < MT; ++t) {                                                           \
+        const bf16x8 b = *reinterpret_cast<const bf16x8*>(                                     \
+            xs + t * 32 * 72 + (lane & 31) * 72 + u * 16 + (lane >> 5) * 8);                   \
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_buf[BUF][u], b, acc[t], 0, 0, 0);   \
+      }                                                                                        \
+    }                                                                                          \
+    if (s + 2 < nsteps) {                                                                      \
+      _Pragma("unroll")                                                                        \
+      for (int u = 0; u < 4; ++u)                                                              \
+        a_buf[BUF][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512)); \
+      wp += 4 * 512;                                                                           \
+    }                                                                                          \
+    if (!last) {                                                                               \
+      __syncthreads();                                                                         \
+      _Pragma("unroll")                                                                        \
+      for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];             \
+      __syncthreads();                                                                         \
+    }                                                                                          \
   }
+
+  int s = 0;
+  while (s + 2 <= nsteps) {
+    SGP_STEP(0);
+    ++s;
+    SGP_STEP(1);
+    ++s;
+  }
+  if (s < nsteps) SGP_STEP(0);
+#undef SGP_STEP
 
   const int m_local = lane & 31;
   const int nbase = tile * 128 + wv * 32 + 4 * (lane >> 5);
