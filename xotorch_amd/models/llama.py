@@ -28,13 +28,32 @@ from xotorch_amd.ops.torch_ref import rope_cos_sin
 from xotorch_amd.shard import Shard
 
 
+# (N, K, M) -> True when the packed skinny kernel beat hipBLASLt when timed
+# in-situ (first eligible call at that M, before hipGraph capture).
+_PACKED_WINS: dict = {}
+
+
+def _time_us(fn, reps=3) -> float:
+  start = torch.cuda.Event(enable_timing=True)
+  end = torch.cuda.Event(enable_timing=True)
+  fn()  # warm
+  start.record()
+  for _ in range(reps):
+    fn()
+  end.record()
+  end.synchronize()
+  return start.elapsed_time(end) * 1000.0 / reps
+
+
 class XotLinear(nn.Linear):
   """nn.Linear with an optional decode-path prepack: `pack_decode()` stores a
   second copy of the weight in MFMA A-fragment order ([N/32,K/16,64,8], see
   ops.pack_decode_weight); decode-shaped (rows 32..256) bf16 GEMMs then run
-  the hand-written CDNA4 weight-streaming kernel (measured 5.2-6.0 TB/s vs
-  hipBLASLt 3.0-5.4 at L3-cold decode shapes), everything else (prefill,
-  training, CPU) hipBLASLt/aten. Storage/state-dict identical to nn.Linear."""
+  the hand-written CDNA4 weight-streaming kernel where it measures faster
+  than hipBLASLt (auto-picked per (N,K,M) at first call: e.g. 70B down_proj
+  110 -> 84.6 us, lm_head 5.3 -> 6.2 TB/s; hipBLASLt keeps the shapes it
+  wins). Prefill/training/CPU stay on hipBLASLt/aten. Storage/state-dict
+  identical to nn.Linear."""
 
   weight_packed: Optional[torch.Tensor]
 
@@ -56,13 +75,28 @@ class XotLinear(nn.Linear):
 
   def forward(self, x):
     if self.weight_packed is not None and x.is_cuda and x.dtype == torch.bfloat16:
-      K = self.weight.shape[1]
+      N, K = self.weight.shape
       M = x.numel() // K
       if 32 <= M <= 256 and M % 32 == 0 and x.is_contiguous() and not torch.is_grad_enabled():
         from xotorch_amd.ops import _load_hip
         hip = _load_hip()
         if hip is not None:
-          return hip.skinny_gemm_packed(x, self.weight_packed, self.weight.shape[0], self.bias)
+          key = (N, K, M)
+          use = _PACKED_WINS.get(key)
+          if use is None:
+            if torch.cuda.is_current_stream_capturing():
+              use = True  # no timing under capture; normal warmup decides first
+            else:
+              wp, w, b = self.weight_packed, self.weight, self.bias
+              t_packed = _time_us(lambda: hip.skinny_gemm_packed(x, wp, N, b))
+              t_blaslt = _time_us(lambda: torch.nn.functional.linear(x, w, b))
+              use = t_packed < t_blaslt
+              _PACKED_WINS[key] = use
+              if os.getenv("XOT_DEBUG", "0") != "0":
+                print(f"[xot] gemm auto-pick N={N} K={K} M={M}: packed {t_packed:.1f} us "
+                      f"vs blaslt {t_blaslt:.1f} us -> {'packed' if use else 'blaslt'}", flush=True)
+          if use:
+            return hip.skinny_gemm_packed(x, self.weight_packed, N, self.bias)
     return ops.linear(x, self.weight, self.bias)
 
 
