@@ -223,3 +223,40 @@ def test_xotlinear_packed_matches_unpacked(hip):
     assert m.weight_packed is not None
     y1 = m(x).float()
   assert (y0 - y1).abs().max().item() / (y0.abs().max().item() + 1e-9) < 2e-2
+
+
+def test_skinny_gemm_grouped(hip):
+  from xotorch_amd import ops as xops
+  from xotorch_amd.ops import _hip_ops
+  E, C, K, N = 8, 64, 512, 256
+  x = bt(E, C, K, scale=0.5, seed=11)
+  ws = [bt(N, K, scale=0.05, seed=100 + e) for e in range(E)]
+  wp = torch.stack([xops.pack_decode_weight(w) for w in ws]).contiguous()
+  got = _hip_ops.skinny_gemm_grouped(x, wp, E, N).float()
+  for e in range(E):
+    ref = torch.nn.functional.linear(x[e].float(), ws[e].float())
+    err = (got[e] - ref).abs().max().item() / (ref.abs().max().item() + 1e-9)
+    assert err < 2e-2, (e, err)
+
+
+def test_moe_grouped_gpu_matches_loop():
+  """Mixtral MoE block: grouped MFMA kernel path vs the dynamic loop."""
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.registry import builtin_config
+  from xotorch_amd.models.llama import MoEMLP
+  raw = dict(builtin_config("mixtral-8x7b"))
+  raw.update(hidden_size=256, intermediate_size=512)
+  cfg = config_from_hf(raw, "mixtral-gpu-tiny")
+  torch.manual_seed(2)
+  moe = MoEMLP(cfg).to("cuda").to(torch.bfloat16).eval()
+  x = (torch.randn(64, 1, 256, device="cuda") * 0.5).to(torch.bfloat16)
+  with torch.inference_mode():
+    y_unpacked = moe(x).float()  # routed path, per-expert XotLinear (unpacked)
+    for e in moe.experts:
+      e.gate_up_proj.pack_decode()
+      e.down_proj.pack_decode()
+    moe.pack_grouped()
+    assert moe.wp_gate_up is not None
+    y_grouped = moe(x).float()
+  err = (y_unpacked - y_grouped).abs().max().item() / (y_unpacked.abs().max().item() + 1e-9)
+  assert err < 3e-2, err

@@ -138,3 +138,47 @@ def test_qwen_bias_and_moe_construct():
     out = m(torch.randint(0, 199, (1, 5)), caches=cache.caches,
             positions=torch.arange(5), start_pos=0)
   assert out.shape == (1, 199)
+
+
+def test_moe_routed_decode_matches_loop():
+  """The static-capacity routed MoE path (hipGraph-capturable) must match the
+  dynamic per-expert gather loop (reference semantics, llm_utils.py:502-590)."""
+  import torch
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.registry import builtin_config
+  from xotorch_amd.models.llama import MoEMLP
+
+  raw = dict(builtin_config("mixtral-8x7b"))
+  raw.update(hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+             num_attention_heads=4, num_key_value_heads=2)
+  cfg = config_from_hf(raw, "mixtral-tiny")
+  torch.manual_seed(0)
+  moe = MoEMLP(cfg).eval()
+  x = torch.randn(5, 1, 64)
+  with torch.no_grad():
+    ref = moe(x)  # CPU -> dynamic loop
+    routed = moe._forward_decode(x.view(-1, 64)).view(5, 1, 64).to(x.dtype)
+  assert torch.allclose(ref, routed, atol=1e-4, rtol=1e-4), (ref - routed).abs().max()
+
+
+def test_moe_capacity_is_lossless():
+  """Every (token, expert) assignment must land in a slot (top-k experts are
+  distinct per token, so per-expert load <= T <= capacity)."""
+  import torch
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.registry import builtin_config
+  from xotorch_amd.models.llama import MoEMLP
+
+  raw = dict(builtin_config("mixtral-8x7b"))
+  raw.update(hidden_size=32, intermediate_size=64)
+  cfg = config_from_hf(raw, "mixtral-tiny2")
+  torch.manual_seed(1)
+  moe = MoEMLP(cfg).eval()
+  # adversarial: bias the router so one expert gets every token
+  with torch.no_grad():
+    moe.gate.weight.zero_()
+    moe.gate.weight[3].fill_(5.0)
+    x = torch.randn(64, 1, 32)
+    ref = moe(x)
+    routed = moe._forward_decode(x.view(-1, 32)).view(64, 1, 32).to(x.dtype)
+  assert torch.allclose(ref, routed, atol=1e-4, rtol=1e-4)

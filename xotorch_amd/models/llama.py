@@ -153,21 +153,52 @@ class MLP(nn.Module):
 
 
 class MoEMLP(nn.Module):
-  """Mixtral-class sparse MoE block (top-k routed SwiGLU experts)."""
+  """Mixtral-class sparse MoE block (top-k routed SwiGLU experts).
+
+  Decode (<=256 tokens, GPU inference) runs a STATIC-SHAPE routed path that
+  is hipGraph-capturable: assignments are argsorted by expert, every expert
+  gets a fixed token capacity C = round_up(T, 32) (top-k experts are distinct
+  per token so a single expert can receive at most T assignments — lossless),
+  padded slots carry weight 0, and the per-expert GEMMs run as ONE grouped
+  MFMA launch on stacked prepacked weights (`skinny_gemm_grouped`) when
+  packed, else per-expert XotLinear. Prefill/CPU/training use the dynamic
+  gather loop (the reference's semantics, llm_utils.py:502-590)."""
 
   def __init__(self, cfg: ModelConfig):
     super().__init__()
     self.n_experts = cfg.n_experts
     self.top_k = cfg.n_experts_per_tok
+    self.dim = cfg.dim
+    self.intermediate = cfg.intermediate_dim
     self.gate = nn.Linear(cfg.dim, cfg.n_experts, bias=False)
     self.experts = nn.ModuleList([MLP(cfg) for _ in range(cfg.n_experts)])
+    self.wp_gate_up: Optional[torch.Tensor] = None
+    self.wp_down: Optional[torch.Tensor] = None
+
+  def pack_grouped(self):
+    """Stack the experts' prepacked weights for the grouped decode GEMM."""
+    if self.wp_gate_up is None and all(e.gate_up_proj.weight_packed is not None for e in self.experts) \
+       and all(e.down_proj.weight_packed is not None for e in self.experts):
+      self.wp_gate_up = torch.stack([e.gate_up_proj.weight_packed for e in self.experts]).contiguous()
+      self.wp_down = torch.stack([e.down_proj.weight_packed for e in self.experts]).contiguous()
+      for e in self.experts:  # the grouped stack supersedes the per-expert packs
+        e.gate_up_proj.weight_packed = None
+        e.down_proj.weight_packed = None
+
+  def _route(self, flat):
+    """Top-k routing -> normalized weights [T, k] + expert ids [T, k]."""
+    router = self.gate(flat).float()
+    weights, selected = torch.topk(torch.softmax(router, dim=-1), self.top_k, dim=-1)
+    weights = weights / weights.sum(dim=-1, keepdim=True)
+    return weights, selected
 
   def forward(self, x):
     B, S, D = x.shape
     flat = x.view(-1, D)
-    router = self.gate(flat).float()
-    weights, selected = torch.topk(torch.softmax(router, dim=-1), self.top_k, dim=-1)
-    weights = weights / weights.sum(dim=-1, keepdim=True)
+    T = flat.shape[0]
+    if x.is_cuda and not torch.is_grad_enabled() and T <= 256:
+      return self._forward_decode(flat).view(B, S, D).to(x.dtype)
+    weights, selected = self._route(flat)
     out = torch.zeros_like(flat, dtype=torch.float32)
     for e in range(self.n_experts):
       token_idx, k_idx = torch.where(selected == e)
@@ -176,6 +207,45 @@ class MoEMLP(nn.Module):
       expert_out = self.experts[e](flat[token_idx]).float()
       out.index_add_(0, token_idx, expert_out * weights[token_idx, k_idx, None])
     return out.view(B, S, D).to(x.dtype)
+
+  def _forward_decode(self, flat):
+    T, D = flat.shape
+    E, k = self.n_experts, self.top_k
+    dev = flat.device
+    weights, selected = self._route(flat)                     # [T, k]
+    A = T * k
+    expert_of = selected.reshape(A)                           # [A]
+    token_of = torch.arange(T, device=dev).repeat_interleave(k)
+    w_of = weights.reshape(A)
+    order = torch.argsort(expert_of)                          # static shape [A]
+    sorted_token = token_of[order]
+    sorted_w = w_of[order]
+    counts = (expert_of.unsqueeze(0) == torch.arange(E, device=dev).unsqueeze(1)).sum(1)  # [E]
+    offsets = torch.cumsum(counts, 0) - counts                # exclusive prefix
+    C = max(32, -(-T // 32) * 32)                             # capacity (lossless: count_e <= T)
+    c_idx = torch.arange(C, device=dev)
+    pos = offsets.unsqueeze(1) + c_idx.unsqueeze(0)           # [E, C]
+    valid = c_idx.unsqueeze(0) < counts.unsqueeze(1)
+    pos_c = pos.clamp(max=A - 1)
+    gather_tok = sorted_token[pos_c.reshape(-1)]              # [E*C]
+    scale = torch.where(valid, sorted_w[pos_c], torch.zeros((), dtype=sorted_w.dtype, device=dev))
+    xg = flat[gather_tok]                                     # [E*C, D]
+    if self.wp_gate_up is not None and not torch.is_grad_enabled():
+      from xotorch_amd.ops import _load_hip
+      hip = _load_hip()
+      gu = hip.skinny_gemm_grouped(xg.view(E, C, D), self.wp_gate_up, E, 2 * self.intermediate)
+      h = ops.swiglu_packed(gu.view(E * C, 2 * self.intermediate))
+      y = hip.skinny_gemm_grouped(h.view(E, C, self.intermediate), self.wp_down, E, D)
+      y = y.view(E * C, D)
+    else:
+      xe = xg.view(E, C, D)
+      outs = []
+      for e in range(E):
+        outs.append(self.experts[e](xe[e]))
+      y = torch.cat(outs, dim=0)
+    out = torch.zeros(T, D, dtype=torch.float32, device=dev)
+    out.index_add_(0, gather_tok, y.float() * scale.reshape(-1, 1))
+    return out
 
 
 class DecoderLayer(nn.Module):
@@ -254,8 +324,9 @@ class ShardedModel(nn.Module):
         continue
       if "down_proj" in name:
         # the packed kernel wins when the split-K grid still fills the chip
-        # (>= ~1.5 blocks/CU); small-N down projections stay on hipBLASLt
-        if mod.weight.shape[0] // 128 * 8 >= 384 or mode == "all":
+        # (>= ~1.5 blocks/CU); small-N down projections stay on hipBLASLt.
+        # MoE experts always pack: the grouped launch fills via blockIdx.y.
+        if "experts." in name or mod.weight.shape[0] // 128 * 8 >= 384 or mode == "all":
           groups[0].append(mod)
       elif "lm_head" in name:
         groups[1].append(mod)
@@ -286,6 +357,9 @@ class ShardedModel(nn.Module):
       packed += need
       if debug:
         print(f"[xot] pack group {gi}: packed {len(grp)} modules, {need>>20} MiB", flush=True)
+    for mod in self.modules():
+      if isinstance(mod, MoEMLP):
+        mod.pack_grouped()
     return packed
 
   def head_weight(self):

@@ -488,6 +488,11 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 // and read back as B fragments.  K % 64 == 0 required.
 // ---------------------------------------------------------------------------
 
+// Also the MoE grouped-GEMM kernel (the reference's latent
+// MOEExpert/MOELayer scaffolding, llm_utils.py:502-590, never executed): for
+// Mixtral-class decode each expert e (blockIdx.y) computes its own
+// [C, N] = [C, K] @ Wp_e^T over a fixed per-expert token capacity C — one
+// launch covers every expert, weights stream from the stacked prepack.
 template <int MT, bool SPLIT>
 __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     const unsigned short* __restrict__ Wp, const unsigned short* __restrict__ X,
@@ -497,6 +502,11 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   const int ntiles = N >> 7;
   const int tile = blockIdx.x % ntiles;
   const int split = blockIdx.x / ntiles;
+  // grouped mode: expert blockIdx.y owns rows [e*MT*32, (e+1)*MT*32) of X/Y
+  // and its own weight block; gridDim.y == 1 degenerates to the plain GEMM.
+  const int e = blockIdx.y;
+  Wp += (size_t)e * (size_t)N * (size_t)K;
+  X += (size_t)e * (size_t)(MT * 32) * (size_t)K;
   const long long k0 = (long long)split * kc;
   const long long k1 = min(k0 + (long long)kc, K);
   if (k0 >= k1) return;
@@ -582,12 +592,13 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
 #undef SGP_STEP
 
   const int m_local = lane & 31;
+  const int rows_total = gridDim.y * MT * 32;  // across all experts
   const int nbase = tile * 128 + wv * 32 + 4 * (lane >> 5);
 #pragma unroll
   for (int t = 0; t < MT; ++t) {
-    const int m = t * 32 + m_local;
+    const int m = e * MT * 32 + t * 32 + m_local;
     if (SPLIT) {
-      float* prow = P + ((size_t)split * (MT * 32) + m) * N;
+      float* prow = P + ((size_t)split * rows_total + m) * N;
 #pragma unroll
       for (int g = 0; g < 4; ++g) {
         floatx4 v4;
@@ -894,12 +905,62 @@ torch::Tensor skinny_gemm_packed(torch::Tensor x, torch::Tensor wp, int64_t N,
   return y;
 }
 
+// MoE grouped decode GEMM: x [E, C, K] (C = per-expert token capacity,
+// 32..256, %32), wp: stacked prepacks [E, N/32, K/16, 64, 8] -> y [E, C, N].
+// One launch for all experts (blockIdx.y = expert).
+torch::Tensor skinny_gemm_grouped(torch::Tensor x, torch::Tensor wp, int64_t E, int64_t N) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  CHK(wp.is_cuda() && wp.dtype() == torch::kBFloat16 && wp.is_contiguous());
+  const long long K = wp.numel() / (E * N);
+  const long long C = x.numel() / (K * E);
+  CHK(C >= 32 && C <= 256 && C % 32 == 0);
+  CHK(N % 128 == 0 && K % 64 == 0 && E >= 1 && E <= 256);
+  auto y = torch::empty({(long)E, (long)C, (long)N}, x.options());
+  const int ntiles = (int)(N / 128);
+  int nsplit = 1;
+  while (ntiles * nsplit * E * 2 < 1024 && (K / (nsplit * 2)) >= 1024 && nsplit < 16) nsplit *= 2;
+  int kc = (int)((K / nsplit + 63) / 64 * 64);
+  while ((long long)kc * (nsplit - 1) >= K) nsplit--;
+  auto stream = cur_stream();
+  const int MT = (int)(C / 32);
+  const dim3 grid(ntiles * nsplit, (unsigned)E), block(256);
+#define SGG_CASE(MTV) \
+  case MTV: \
+    if (nsplit == 1) { \
+      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false>), grid, block, 0, stream, \
+                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                         (unsigned short*)y.data_ptr(), nullptr, nullptr, (int)N, K, kc, nsplit); \
+    } else { \
+      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true>), grid, block, 0, stream, \
+                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                         nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+    } \
+    break;
+  if (nsplit == 1) {
+    torch::Tensor P;
+    switch (MT) { SGG_CASE(1) SGG_CASE(2) SGG_CASE(3) SGG_CASE(4) SGG_CASE(5) SGG_CASE(6) SGG_CASE(7) SGG_CASE(8) }
+  } else {
+    auto P = torch::empty({nsplit, (long)(E * C), (long)N},
+                          torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+    switch (MT) { SGG_CASE(1) SGG_CASE(2) SGG_CASE(3) SGG_CASE(4) SGG_CASE(5) SGG_CASE(6) SGG_CASE(7) SGG_CASE(8) }
+    const long long MN = E * C * N;
+    const int blocks = (int)std::min<long long>(2048, (MN / 4 + 255) / 256);
+    hipLaunchKernelGGL(skinny_combine_kernel, dim3(blocks), dim3(256), 0, stream,
+                       P.data_ptr<float>(), (unsigned short*)y.data_ptr(), nullptr, MN, N, nsplit);
+  }
+#undef SGG_CASE
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm, "decode GEMM y = x @ w^T (+bias), bf16 MFMA weight-streaming",
         py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
   m.def("skinny_gemm_packed", &skinny_gemm_packed,
         "decode GEMM on prepacked weights (MFMA fragment order)",
         py::arg("x"), py::arg("wp"), py::arg("n"), py::arg("bias") = py::none());
+  m.def("skinny_gemm_grouped", &skinny_gemm_grouped,
+        "MoE grouped decode GEMM on stacked prepacked expert weights",
+        py::arg("x"), py::arg("wp"), py::arg("n_experts"), py::arg("n"));
   m.def("swiglu_packed", &swiglu_packed, "SwiGLU on the packed [gate|up] GEMM output");
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual add + RMSNorm");
