@@ -260,3 +260,81 @@ def test_moe_grouped_gpu_matches_loop():
     y_grouped = moe(x).float()
   err = (y_unpacked - y_grouped).abs().max().item() / (y_unpacked.abs().max().item() + 1e-9)
   assert err < 3e-2, err
+
+
+def _pack_k(k, t32):
+  """Python reference packer for the MFMA K cache layout (tests only)."""
+  B, KVH, T, hd = k.shape
+  kk = torch.zeros(B, KVH, t32, hd, dtype=k.dtype, device=k.device)
+  kk[:, :, :T] = k
+  # hd = c*32 + qt*8 + j ; lane = qt*16 + pos16
+  v = kk.view(B, KVH, t32 // 16, 16, 4, 4, 8)
+  return v.permute(0, 1, 2, 4, 5, 3, 6).contiguous().view(B, KVH, t32 // 16, 4, 64, 8)
+
+
+def _pack_v(v, t32):
+  B, KVH, T, hd = v.shape
+  vv = torch.zeros(B, KVH, t32, hd, dtype=v.dtype, device=v.device)
+  vv[:, :, :T] = v
+  # pos = tp*32 + qt*8 + j ; hd = g*16 + c16 ; lane = qt*16 + c16
+  w = vv.view(B, KVH, t32 // 32, 4, 8, 8, 16)
+  return w.permute(0, 1, 5, 2, 3, 6, 4).contiguous().view(B, KVH, 8, t32 // 32, 64, 8)
+
+
+def test_mfma16_probe(hip):
+  """Validate the assumed v_mfma_f32_16x16x32_bf16 fragment layout (asymmetric B)."""
+  from xotorch_amd.ops import _hip_ops
+  a = bt(16, 32, scale=0.5, seed=31)
+  b = bt(32, 16, scale=0.5, seed=32)
+  d = _hip_ops.mfma16_probe(a, b)
+  ref = a.float() @ b.float()
+  assert torch.allclose(d, ref, atol=5e-2, rtol=5e-2), (d - ref).abs().max()
+
+
+@pytest.mark.parametrize("B,H,KVH,T,sl", [(2, 8, 2, 128, 100), (3, 32, 8, 640, 576), (1, 16, 16, 64, 33)])
+def test_attn_decode_mfma(hip, B, H, KVH, T, sl):
+  from xotorch_amd.ops import _hip_ops, torch_ref
+  hd = 128
+  q = bt(B, 1, H, hd, seed=41)
+  k = bt(B, KVH, T, hd, seed=42)
+  v = bt(B, KVH, T, hd, seed=43)
+  t32 = (T + 31) // 32 * 32
+  kp, vp = _pack_k(k, t32), _pack_v(v, t32)
+  sl_t = torch.full((B,), sl, dtype=torch.int32, device="cuda")
+  out = _hip_ops.attn_decode_mfma(q, kp, vp, sl_t, T).float()
+  ref = torch_ref.attn_decode(q, k, v, sl).float()
+  assert torch.allclose(out, ref, atol=3e-2, rtol=3e-2), (out - ref).abs().max()
+
+
+def test_attn_decode_mfma_ragged(hip):
+  from xotorch_amd.ops import _hip_ops, torch_ref
+  B, H, KVH, T, hd = 4, 8, 4, 256, 128
+  q = bt(B, 1, H, hd, seed=51)
+  k = bt(B, KVH, T, hd, seed=52)
+  v = bt(B, KVH, T, hd, seed=53)
+  t32 = (T + 31) // 32 * 32
+  kp, vp = _pack_k(k, t32), _pack_v(v, t32)
+  sls = [7, 64, 130, 256]
+  sl_t = torch.tensor(sls, dtype=torch.int32, device="cuda")
+  out = _hip_ops.attn_decode_mfma(q, kp, vp, sl_t, T).float()
+  for b, s in enumerate(sls):
+    ref = torch_ref.attn_decode(q[b:b + 1], k[b:b + 1, :, :s], v[b:b + 1, :, :s], s).float()
+    assert torch.allclose(out[b:b + 1], ref, atol=3e-2, rtol=3e-2), (b, (out[b:b+1] - ref).abs().max())
+
+
+def test_rope_append_writes_packed(hip):
+  """The fused append kernel's packed-cache writes must match the reference packer."""
+  from xotorch_amd.ops import _hip_ops
+  B, S, H, KVH, hd, T = 2, 48, 8, 2, 128, 64
+  t32 = 64
+  qkv = bt(B, S, (H + 2 * KVH) * hd, seed=61)
+  cos = torch.randn(T, hd // 2, device="cuda")
+  sin = torch.randn(T, hd // 2, device="cuda")
+  kc = torch.zeros(B, KVH, T, hd, dtype=torch.bfloat16, device="cuda")
+  vc = torch.zeros_like(kc)
+  kp = torch.zeros(B, KVH, t32 // 16, 4, 64, 8, dtype=torch.bfloat16, device="cuda")
+  vp = torch.zeros(B, KVH, 8, t32 // 32, 64, 8, dtype=torch.bfloat16, device="cuda")
+  pos = torch.arange(S, dtype=torch.int32, device="cuda")
+  _hip_ops.rope_qkv_append(qkv, cos, sin, pos, kc, vc, H, KVH, hd, kp, vp)
+  assert torch.equal(kp, _pack_k(kc, t32))
+  assert torch.equal(vp, _pack_v(vc, t32))

@@ -1,30 +1,61 @@
-"""Device-resident KV caches, laid out for the CDNA4 decode-attention kernel.
+"""Device-resident KV caches, laid out for the CDNA4 decode-attention kernels.
 
-Layout [B, KVH, T, hd]: each (b, kv_head) has T contiguous rows of hd
-elements (256 B at hd=128 bf16) — the decode kernel streams rows with
-coalesced 16 B/lane loads. Capacity is sized per request (prompt + max gen),
-allocated from the 288 GB HBM pool; the reference's equivalent is torchtune's
-per-layer cache setup (/root/reference/xotorch/inference/torch/sharded_inference_engine.py:71-82).
+Standard layout [B, KVH, T, hd]: each (b, kv_head) has T contiguous rows of
+hd elements (256 B at hd=128 bf16) — prefill attention (sdpa) and the VALU
+decode kernel stream rows with coalesced 16 B/lane loads.
+
+MFMA layout (GPU, hd=128): a second copy of K and V pre-shuffled into
+v_mfma_f32_16x16x32_bf16 B-fragment order so the MFMA flash-decoding kernel
+streams the cache with fully-coalesced 1 KB wave loads:
+  K_packed [B, KVH, T32/16, 4, 64, 8]   (16-position x 32-hd-chunk tiles)
+  V_packed [B, KVH, 8, T32/32, 64, 8]   (32-position x 16-hd-column tiles)
+(T32 = capacity rounded up to 32; both copies are appended by the fused
+rope_qkv_append kernel.)  The reference's equivalent is torchtune's per-layer
+cache setup (/root/reference/xotorch/inference/torch/sharded_inference_engine.py:71-82).
 """
 from __future__ import annotations
 
-from typing import List, Tuple
+import os
+from typing import List, NamedTuple, Optional
 
 import torch
 
 
+class LayerKV(NamedTuple):
+  k: torch.Tensor
+  v: torch.Tensor
+  kp: Optional[torch.Tensor] = None  # MFMA-packed K (cuda, hd=128 only)
+  vp: Optional[torch.Tensor] = None  # MFMA-packed V
+
+
+def _want_packed(device: str, head_dim: int, dtype: torch.dtype) -> bool:
+  return (
+    str(device).startswith("cuda")
+    and head_dim == 128
+    and dtype == torch.bfloat16
+    and os.getenv("XOT_MFMA_ATTN", "1") == "1"
+  )
+
+
 class ShardKVCache:
-  """One (k, v) cache pair per local layer of a shard."""
+  """One LayerKV per local layer of a shard."""
 
   def __init__(self, n_layers: int, batch: int, n_kv_heads: int, capacity: int, head_dim: int,
                dtype: torch.dtype = torch.bfloat16, device: str = "cpu"):
     self.capacity = capacity
     self.batch = batch
-    self.caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
+    self.caches: List[LayerKV] = []
+    packed = _want_packed(device, head_dim, dtype)
+    t32 = (capacity + 31) // 32 * 32
     for _ in range(n_layers):
       k = torch.zeros(batch, n_kv_heads, capacity, head_dim, dtype=dtype, device=device)
       v = torch.zeros(batch, n_kv_heads, capacity, head_dim, dtype=dtype, device=device)
-      self.caches.append((k, v))
+      if packed:
+        kp = torch.zeros(batch, n_kv_heads, t32 // 16, 4, 64, 8, dtype=dtype, device=device)
+        vp = torch.zeros(batch, n_kv_heads, 8, t32 // 32, 64, 8, dtype=dtype, device=device)
+        self.caches.append(LayerKV(k, v, kp, vp))
+      else:
+        self.caches.append(LayerKV(k, v))
 
   def __getitem__(self, i):
     return self.caches[i]
@@ -33,9 +64,17 @@ class ShardKVCache:
     return len(self.caches)
 
   def reset(self):
-    for k, v in self.caches:
-      k.zero_()
-      v.zero_()
+    for c in self.caches:
+      c.k.zero_()
+      c.v.zero_()
+      if c.kp is not None:
+        c.kp.zero_()
+        c.vp.zero_()
 
   def nbytes(self) -> int:
-    return sum(k.numel() * k.element_size() + v.numel() * v.element_size() for k, v in self.caches)
+    total = 0
+    for c in self.caches:
+      total += c.k.numel() * c.k.element_size() + c.v.numel() * c.v.element_size()
+      if c.kp is not None:
+        total += c.kp.numel() * c.kp.element_size() + c.vp.numel() * c.vp.element_size()
+    return total

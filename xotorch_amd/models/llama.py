@@ -122,16 +122,19 @@ class Attention(nn.Module):
     self.qkv_proj = XotLinear(D, (H + 2 * KVH) * hd, bias=cfg.attn_bias)
     self.o_proj = XotLinear(H * hd, D, bias=False)
 
-  def forward(self, x, cos, sin, positions, k_cache, v_cache, start_pos: int, is_decode: bool, seq_lens=None):
+  def forward(self, x, cos, sin, positions, kv, start_pos: int, is_decode: bool, seq_lens=None):
     B, S, _ = x.shape
     cfg = self.cfg
     H, hd = cfg.n_heads, cfg.head_dim
+    k_cache, v_cache = kv[0], kv[1]
+    kp = kv[2] if len(kv) > 2 else None  # MFMA-packed cache copies (GPU, hd=128)
+    vp = kv[3] if len(kv) > 3 else None
     qkv = self.qkv_proj(x)  # [B, S, (H+2KVH)*hd]
-    ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, cfg.n_kv_heads, hd)
+    ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, cfg.n_kv_heads, hd, kp, vp)
     q = qkv[:, :, : H * hd].view(B, S, H, hd)  # strided view; kernels accept it
     if is_decode:
       sl = seq_lens if seq_lens is not None else start_pos + 1
-      out = ops.attn_decode(q, k_cache, v_cache, sl)
+      out = ops.attn_decode(q, k_cache, v_cache, sl, kp, vp)
     else:
       out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S)
     return self.o_proj(out.reshape(B, S, H * hd))
@@ -257,9 +260,9 @@ class DecoderLayer(nn.Module):
     self.mlp = MoEMLP(cfg) if cfg.n_experts > 0 else MLP(cfg)
     self.eps = cfg.norm_eps
 
-  def forward(self, x, cos, sin, positions, k_cache, v_cache, start_pos, is_decode, seq_lens=None):
+  def forward(self, x, cos, sin, positions, kv, start_pos, is_decode, seq_lens=None):
     attn_out = self.self_attn(
-      self.input_layernorm(x), cos, sin, positions, k_cache, v_cache, start_pos, is_decode, seq_lens
+      self.input_layernorm(x), cos, sin, positions, kv, start_pos, is_decode, seq_lens
     )
     # fused residual-add + norm: h = x + attn_out; normed = rmsnorm(h)
     normed, h = ops.rmsnorm_residual(attn_out, x, self.post_attention_layernorm.weight, self.eps)
@@ -392,8 +395,7 @@ class ShardedModel(nn.Module):
       h = x
     cos, sin = self.rope_cos, self.rope_sin
     for idx, lid in enumerate(self.local_layer_ids):
-      k_cache, v_cache = caches[idx]
-      h = self.layers[str(lid)](h, cos, sin, positions, k_cache, v_cache, start_pos, is_decode, seq_lens)
+      h = self.layers[str(lid)](h, cos, sin, positions, caches[idx], start_pos, is_decode, seq_lens)
     if not self.shard.is_last_layer:
       return h
     if last_only and h.shape[1] > 1:

@@ -77,12 +77,15 @@ def rope_apply(q, k, cos, sin, positions):
   return torch_ref.rope_apply(q, k, cos, sin, positions)
 
 
-def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_kv_heads: int, head_dim: int):
+def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_kv_heads: int,
+                    head_dim: int, kp=None, vp=None):
   """Fused on the packed qkv GEMM output [B,S,(H+2KVH)*hd]: RoPE-rotate the q
   heads in place, rotate k heads into the cache, copy v heads into the cache.
+  When the MFMA-packed cache copies (kp, vp) exist they are appended too.
   """
   if _use_hip(qkv) and qkv.dtype == torch.bfloat16:
-    _hip.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim)
+    _hip.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim,
+                         kp, vp)
     return
   torch_ref.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim)
 
@@ -107,11 +110,16 @@ def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int):
   return torch_ref.attn_prefill(q, k_cache, v_cache, start_pos, s_len)
 
 
-def attn_decode(q, k_cache, v_cache, seq_len):
-  """seq_len: int or int32 device tensor [B] (per-sequence lengths)."""
+def attn_decode(q, k_cache, v_cache, seq_len, kp=None, vp=None):
+  """seq_len: int or int32 device tensor [B] (per-sequence lengths).
+  With MFMA-packed cache copies (kp, vp) the flash-decoding kernel scores on
+  matrix cores (v_mfma_f32_16x16x32_bf16, coalesced 1 KB cache streams)."""
   if _use_hip(q) and q.dtype == torch.bfloat16:
     if not isinstance(seq_len, torch.Tensor):
       seq_len = torch.full((q.shape[0],), int(seq_len), dtype=torch.int32, device=q.device)
+    rep = q.shape[2] // k_cache.shape[1]
+    if kp is not None and rep <= 16 and os.getenv("XOT_MFMA_ATTN", "1") == "1":
+      return _hip.attn_decode_mfma(q, kp, vp, seq_len, k_cache.shape[2])
     return _hip.attn_decode(q, k_cache, v_cache, seq_len)
   if isinstance(seq_len, torch.Tensor):
     seq_len = int(seq_len.max().item())

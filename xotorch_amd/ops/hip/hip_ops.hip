@@ -28,6 +28,10 @@
 
 #include <vector>
 
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(16))) float floatx16;
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+
 typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
 typedef __attribute__((ext_vector_type(4))) unsigned short ushort4_t;
 
@@ -115,11 +119,17 @@ __global__ __launch_bounds__(256) void rmsnorm_kernel(
 // cos/sin tables fp32 [maxT, hd/2].
 // ---------------------------------------------------------------------------
 
+// When the MFMA-packed cache copies exist (kpc/vpc non-null, hd == 128) the
+// same kernel also appends into them (decode attention then streams the
+// cache as coalesced v_mfma_f32_16x16x32_bf16 B-fragments):
+//   K_packed [B,KVH][T32/16 tile][4 hd-chunk][64 lane][8]  lane=(hd&31)/8*16 + (pos&15)
+//   V_packed [B,KVH][8 hd-group][T32/32 tile][64 lane][8]  lane=((pos&31)>>3)*16 + (hd&15)
 __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
     unsigned short* __restrict__ qkv, const float* __restrict__ cosb,
     const float* __restrict__ sinb, const int* __restrict__ positions,
     unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
-    int B, int S, int H, int KVH, int hd, int T) {
+    int B, int S, int H, int KVH, int hd, int T,
+    unsigned short* __restrict__ kpc, unsigned short* __restrict__ vpc, int T32) {
   const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   const int slots = H + 2 * KVH;
@@ -148,14 +158,35 @@ __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
       const float sn = sinb[(size_t)pos * hd2 + lane];
       const float x1 = b2f(row[lane]);
       const float x2 = b2f(row[lane + hd2]);
-      dst[lane] = f2b(x1 * c - x2 * sn);
-      dst[lane + hd2] = f2b(x2 * c + x1 * sn);
+      const unsigned short r1 = f2b(x1 * c - x2 * sn);
+      const unsigned short r2 = f2b(x2 * c + x1 * sn);
+      dst[lane] = r1;
+      dst[lane + hd2] = r2;
+      if (kpc) {
+        // packed K: element for hd-dim d at
+        //   [(b*KVH+h)][pos>>4][d>>5][((d&31)>>3)*16 + (pos&15)][d&7]
+        const size_t base = ((size_t)(b * KVH + h) * (T32 >> 4) + (pos >> 4)) * 2048;
+        const int d1 = lane, d2 = lane + hd2;
+        kpc[base + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = r1;
+        kpc[base + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = r2;
+      }
     }
   } else {
     const int h = slot - H - KVH;
     unsigned short* dst = vc + (((size_t)(b * KVH + h) * T + pos) * hd);
     if (lane * 2 < hd)
       *(unsigned int*)(dst + lane * 2) = *(const unsigned int*)(row + lane * 2);
+    if (vpc && lane * 2 < hd) {
+      // packed V: element for hd-dim d at
+      //   [(b*KVH+h)][d>>4][pos>>5][((pos&31)>>3)*16 + (d&15)][pos&7]
+      const int tp = pos >> 5, qt = (pos & 31) >> 3, j = pos & 7;
+      const size_t gbase = (size_t)(b * KVH + h) * 8;
+#pragma unroll
+      for (int e = 0; e < 2; ++e) {
+        const int d = lane * 2 + e;
+        vpc[((gbase + (d >> 4)) * (T32 >> 5) + tp) * 512 + (qt * 16 + (d & 15)) * 8 + j] = row[d];
+      }
+    }
   }
 }
 
@@ -293,6 +324,166 @@ __global__ __launch_bounds__(256) void attn_decode_partial(
   }
 }
 
+// ---------------------------------------------------------------------------
+// MFMA flash-decoding attention (hd = 128): scores and PV on matrix cores.
+//
+// One WAVE per (b, kv-head, kv-split); the wave's all NQ (<=16) query heads
+// are scored together: per 32-position tile,
+//   scores[16q][32p] = 2 x (4 x v_mfma_f32_16x16x32_bf16)  (Q fragments resident,
+//                      K streamed from the packed cache as coalesced 1 KB loads)
+//   online softmax row stats via 4 shfl_xor over the 16-lane column groups
+//   P -> bf16 A-fragments through a 96 B-stride LDS image (conflict-free)
+//   out[16q][128] += 8 x v_mfma_f32_16x16x32_bf16 (V streamed packed)
+// No cross-wave barriers (per-wave LDS slice), so ragged per-batch lengths
+// cannot deadlock. Partials go to the same (m, l, o) workspace as the VALU
+// kernel and are merged by attn_decode_merge<128>.
+// Replaces the torchtune cached-decode step (reference general_mha.py:211-215);
+// the VALU split-KV kernel above stays as the hd=64 / unpacked-cache path.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
+    const unsigned short* __restrict__ vp, const int* __restrict__ seq_lens,
+    float* __restrict__ ws_o, float* __restrict__ ws_ml,
+    int B, int H, int KVH, int T32, int nsplit, float scale, long long q_stride) {
+  const int wv = threadIdx.x >> 6;
+  const int wid = blockIdx.x * 4 + wv;
+  __shared__ unsigned short plds_all[4][16 * 48];  // 96 B row stride: bank-conflict-free b128 reads
+  if (wid >= B * KVH * nsplit) return;
+  const int split = wid % nsplit;
+  const int kvh = (wid / nsplit) % KVH;
+  const int b = wid / (nsplit * KVH);
+  const int lane = threadIdx.x & 63;
+  const int NQ = H / KVH;
+  const int sl = seq_lens[b];
+  const int chunk = ((T32 / nsplit + 31) >> 5) << 5;
+  const int c0 = split * chunk;
+  const int c1 = min(c0 + chunk, sl);
+  unsigned short* plds = plds_all[wv];
+
+  // Q fragments: A[i = lane&15 (query head, clamped), k = (lane>>4)*8 + j]
+  const int qh_base = kvh * NQ;
+  const int qi = min(lane & 15, NQ - 1);
+  bf16x8 qf[4];
+  {
+    const unsigned short* qrow = q + (size_t)b * q_stride + (size_t)(qh_base + qi) * 128;
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+      qf[c] = *reinterpret_cast<const bf16x8*>(qrow + c * 32 + (lane >> 4) * 8);
+  }
+
+  float m[4], lsum[4];
+  floatx4 acco[8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m[r] = -INFINITY; lsum[r] = 0.f; }
+#pragma unroll
+  for (int g = 0; g < 8; ++g) acco[g] = (floatx4)(0.f);
+
+  const size_t kbase = (size_t)(b * KVH + kvh) * (T32 >> 4) * 2048;
+  const size_t vbase = (size_t)(b * KVH + kvh) * 8 * (T32 >> 5) * 512;
+  const int col = lane & 15;
+
+  for (int t = c0; t < c1; t += 32) {
+    // ---- scores: two 16-position half-tiles ----
+    floatx4 sc[2];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      sc[h] = (floatx4)(0.f);
+      const unsigned short* kt = kp + kbase + ((size_t)((t >> 4) + h)) * 2048 + (size_t)lane * 8;
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        const bf16x8 kb = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+        sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kb, sc[h], 0, 0, 0);
+      }
+    }
+    // ---- online softmax (rows r are this lane's 4 query heads) ----
+    float rmax[4];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int pos = t + h * 16 + col;
+      const bool ok = pos < c1;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = ok ? sc[h][r] * scale : -INFINITY;
+        sc[h][r] = s;
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) rmax[r] = fmaxf(sc[0][r], sc[1][r]);
+#pragma unroll
+    for (int mm = 1; mm < 16; mm <<= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) rmax[r] = fmaxf(rmax[r], __shfl_xor(rmax[r], mm));
+    float alpha[4], rsum[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float mn = fmaxf(m[r], rmax[r]);
+      alpha[r] = (lsum[r] > 0.f) ? __expf(m[r] - mn) : 0.f;
+      m[r] = mn;
+      float p0 = __expf(sc[0][r] - mn);
+      float p1 = __expf(sc[1][r] - mn);
+      sc[0][r] = p0;
+      sc[1][r] = p1;
+      rsum[r] = p0 + p1;
+    }
+#pragma unroll
+    for (int mm = 1; mm < 16; mm <<= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) rsum[r] += __shfl_xor(rsum[r], mm);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) lsum[r] = lsum[r] * alpha[r] + rsum[r];
+#pragma unroll
+    for (int g = 0; g < 8; ++g)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acco[g][r] *= alpha[r];
+    // ---- P -> LDS (bf16, 96 B row stride) and back as A fragments ----
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        plds[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = f2b(sc[h][r]);
+    const bf16x8 pf = *reinterpret_cast<const bf16x8*>(plds + (lane & 15) * 48 + (lane >> 4) * 8);
+    // ---- PV: out[16q][g*16..] += P x V ----
+    const unsigned short* vt = vp + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
+#pragma unroll
+    for (int g = 0; g < 8; ++g) {
+      const bf16x8 vb = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16x8*>(vt + (size_t)g * (T32 >> 5) * 512));
+      acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vb, acco[g], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: one (m, l, o[128]) partial per (b, qh, split) ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = (lane >> 4) * 4 + r;
+    if (qrow < NQ) {
+      const size_t pidx = ((size_t)(b * H + qh_base + qrow) * nsplit + split);
+#pragma unroll
+      for (int g = 0; g < 8; ++g) ws_o[pidx * 128 + g * 16 + col] = acco[g][r];
+      if (col == 0) {
+        ws_ml[pidx * 2 + 0] = m[r];
+        ws_ml[pidx * 2 + 1] = lsum[r];
+      }
+    }
+  }
+}
+
+// layout probe for tests: one v_mfma_f32_16x16x32_bf16, row-major inputs
+__global__ void mfma16_probe_kernel(const unsigned short* __restrict__ A,
+                                    const unsigned short* __restrict__ Bm,
+                                    float* __restrict__ D) {
+  const int lane = threadIdx.x;
+  bf16x8 a = *reinterpret_cast<const bf16x8*>(A + (lane & 15) * 32 + (lane >> 4) * 8);
+  bf16x8 b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) b[j] = *reinterpret_cast<const __bf16*>(Bm + ((lane >> 4) * 8 + j) * 16 + (lane & 15));
+  floatx4 d = (floatx4)(0.f);
+  d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, d, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = d[r];
+}
+
 template <int HD>
 __global__ __launch_bounds__(128) void attn_decode_merge(
     const float* __restrict__ ws_o, const float* __restrict__ ws_ml,
@@ -380,10 +571,6 @@ __global__ __launch_bounds__(256) void swiglu_packed_kernel(
 //   * fp32 split-K partials + a tiny vectorized combine kernel (bias fused);
 //     SPLITK == 1 writes bf16 directly.
 // ---------------------------------------------------------------------------
-
-typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
-typedef __attribute__((ext_vector_type(16))) float floatx16;
-typedef __attribute__((ext_vector_type(4))) float floatx4;
 
 DEVINL bf16x8 load_bf16x8(const unsigned short* p) {
   return *reinterpret_cast<const bf16x8*>(p);
@@ -688,7 +875,8 @@ std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res, 
 
 void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
                      torch::Tensor positions, torch::Tensor kc, torch::Tensor vc,
-                     int64_t n_heads, int64_t n_kv_heads, int64_t head_dim) {
+                     int64_t n_heads, int64_t n_kv_heads, int64_t head_dim,
+                     c10::optional<torch::Tensor> kp, c10::optional<torch::Tensor> vp) {
   CHK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
   CHK(kc.is_contiguous() && vc.is_contiguous());
   CHK(positions.dtype() == torch::kInt32 && positions.is_cuda());
@@ -697,12 +885,21 @@ void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
   const int T = kc.size(2);
   CHK(qkv.size(2) == (H + 2 * KVH) * hd);
   CHK(hd <= 128 && hd % 2 == 0);
+  unsigned short* kpc = nullptr;
+  unsigned short* vpc = nullptr;
+  int T32 = 0;
+  if (kp.has_value() && vp.has_value()) {
+    CHK(hd == 128 && kp->is_contiguous() && vp->is_contiguous());
+    kpc = (unsigned short*)kp->data_ptr();
+    vpc = (unsigned short*)vp->data_ptr();
+    T32 = (int)kp->size(2) * 16;
+  }
   const int waves = B * S * (H + 2 * KVH);
   const int blocks = (waves + 3) / 4;
   hipLaunchKernelGGL(rope_qkv_append_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
                      (unsigned short*)qkv.data_ptr(), cos.data_ptr<float>(), sin.data_ptr<float>(),
                      positions.data_ptr<int>(), (unsigned short*)kc.data_ptr(),
-                     (unsigned short*)vc.data_ptr(), B, S, H, KVH, hd, T);
+                     (unsigned short*)vc.data_ptr(), B, S, H, KVH, hd, T, kpc, vpc, T32);
 }
 
 template <int HD>
@@ -764,6 +961,51 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, t
     hipLaunchKernelGGL((attn_decode_merge<64>), dim3(B * H), dim3(128), 0, stream,
                        ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), (unsigned short*)out.data_ptr(), nsplit);
   return out;
+}
+
+// q: [B, 1, H, 128] (strided batch OK); kp/vp: the packed cache copies.
+torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor vp,
+                               torch::Tensor seq_lens, int64_t t_capacity) {
+  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  CHK(q.stride(3) == 1 && q.stride(2) == q.size(3));
+  CHK(seq_lens.dtype() == torch::kInt32 && seq_lens.is_cuda());
+  CHK(kp.is_contiguous() && vp.is_contiguous());
+  const int B = q.size(0), H = q.size(2), hd = q.size(3);
+  TORCH_CHECK(hd == 128, "attn_decode_mfma requires head_dim 128");
+  const int KVH = kp.size(1);
+  const int T32 = kp.size(2) * 16;
+  const long long q_stride = q.stride(0);
+  // one wave per (b, kvh, split); target >= ~2048 waves so 16 waves/CU keep
+  // the packed-cache stream deep while chunks stay >= 32 positions
+  int nsplit = 1;
+  while (B * KVH * nsplit * 2 < 4096 && (T32 / (nsplit * 2)) >= 32 && nsplit < 32) nsplit *= 2;
+  auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
+  auto ws_o = torch::empty({(long)B * H * nsplit * hd}, opts);
+  auto ws_ml = torch::empty({(long)B * H * nsplit * 2}, opts);
+  auto out = torch::empty({q.size(0), q.size(1), q.size(2), q.size(3)},
+                          torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
+  const float scale = 1.0f / sqrtf((float)hd);
+  auto stream = cur_stream();
+  const int waves = B * KVH * nsplit;
+  hipLaunchKernelGGL(attn_decode_mfma_kernel, dim3((waves + 3) / 4), dim3(256), 0, stream,
+                     (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
+                     (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
+                     ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,
+                     scale, q_stride);
+  hipLaunchKernelGGL((attn_decode_merge<128>), dim3(B * H), dim3(128), 0, stream,
+                     ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
+                     (unsigned short*)out.data_ptr(), nsplit);
+  return out;
+}
+
+torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
+  CHK(a.is_cuda() && a.dtype() == torch::kBFloat16 && a.is_contiguous() && a.numel() == 16 * 32);
+  CHK(b.is_cuda() && b.dtype() == torch::kBFloat16 && b.is_contiguous() && b.numel() == 32 * 16);
+  auto d = torch::empty({16, 16}, torch::TensorOptions().dtype(torch::kFloat32).device(a.device()));
+  hipLaunchKernelGGL(mfma16_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     (const unsigned short*)a.data_ptr(), (const unsigned short*)b.data_ptr(),
+                     d.data_ptr<float>());
+  return d;
 }
 
 torch::Tensor swiglu(torch::Tensor g, torch::Tensor u) {
@@ -964,7 +1206,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_packed", &swiglu_packed, "SwiGLU on the packed [gate|up] GEMM output");
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual add + RMSNorm");
-  m.def("rope_qkv_append", &rope_qkv_append, "fused RoPE + KV-cache append on packed qkv");
+  m.def("rope_qkv_append", &rope_qkv_append, "fused RoPE + KV-cache append on packed qkv",
+        py::arg("qkv"), py::arg("cos"), py::arg("sin"), py::arg("positions"), py::arg("kc"),
+        py::arg("vc"), py::arg("n_heads"), py::arg("n_kv_heads"), py::arg("head_dim"),
+        py::arg("kp") = py::none(), py::arg("vp") = py::none());
   m.def("attn_decode", &attn_decode, "GQA decode attention (flash-decoding split-KV)");
+  m.def("attn_decode_mfma", &attn_decode_mfma,
+        "GQA decode attention on matrix cores (packed cache, hd=128)");
+  m.def("mfma16_probe", &mfma16_probe, "v_mfma_f32_16x16x32_bf16 layout probe (tests)");
   m.def("swiglu", &swiglu, "SwiGLU activation");
 }
