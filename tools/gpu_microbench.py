@@ -74,5 +74,27 @@ def main():
             f" | packed {us3:7.1f} us {wbytes/us3/1e3:5.2f} TB/s e={errp:.1e}")
 
 
+def attn_mfma_bench():
+  import sys
+  sys.path.insert(0, ".")
+  from tests.test_kernels_gpu import _pack_k, _pack_v
+  from xotorch_amd.ops import torch_ref
+  for (B, H, KVH, T, sl) in ((64, 64, 8, 640, 576), (256, 32, 8, 640, 576)):
+    hd = 128
+    q = torch.randn(B, 1, H, hd, device="cuda").to(torch.bfloat16)
+    k = torch.randn(B, KVH, T, hd, device="cuda").to(torch.bfloat16)
+    v = torch.randn(B, KVH, T, hd, device="cuda").to(torch.bfloat16)
+    t32 = (T + 31) // 32 * 32
+    kp, vp = _pack_k(k, t32), _pack_v(v, t32)
+    sl_t = torch.full((B,), sl, dtype=torch.int32, device="cuda")
+    us_old = timeit(lambda: hip.attn_decode(q, k, v, sl_t))
+    us_new = timeit(lambda: hip.attn_decode_mfma(q, kp, vp, sl_t, T))
+    kv_bytes = B * KVH * sl * hd * 2 * 2
+    err = (hip.attn_decode_mfma(q, kp, vp, sl_t, T).float() - hip.attn_decode(q, k, v, sl_t).float()).abs().max().item()
+    print(f"attn B{B} H{H} sl{sl}: valu {us_old:6.1f} us {kv_bytes/us_old/1e3:5.2f} TB/s"
+          f" | mfma {us_new:6.1f} us {kv_bytes/us_new/1e3:5.2f} TB/s  maxdiff {err:.3e}")
+
+
 if __name__ == "__main__":
   main()
+  attn_mfma_bench()
