@@ -99,8 +99,15 @@ class TorchEngine(InferenceEngine):
     if cfg_dict is None:
       raise ValueError(f"no config available for model {shard.model_id}")
     cfg = config_from_hf(cfg_dict, model_id=shard.model_id)
-    with torch.device("meta"):
-      model = ShardedModel(cfg, shard)
+    # construct at the target dtype so to_empty materializes it directly
+    # (fp32-then-cast would peak at 2x the weight bytes)
+    prev_dtype = torch.get_default_dtype()
+    torch.set_default_dtype(self.dtype)
+    try:
+      with torch.device("meta"):
+        model = ShardedModel(cfg, shard)
+    finally:
+      torch.set_default_dtype(prev_dtype)
     model = model.to_empty(device=self.device)
     model = model.to(self.dtype)
     # recompute rope tables (to_empty leaves buffers uninitialized)
@@ -116,6 +123,11 @@ class TorchEngine(InferenceEngine):
     else:
       random_init(model)
     model.eval()
+    if self.device == "cuda":
+      # decode-GEMM weight prepack (auto-picked vs hipBLASLt per shape at
+      # first decode); keep headroom for per-request KV caches
+      torch.cuda.empty_cache()
+      model.pack_decode_weights(reserve_bytes=24 << 30)
     self.model = model
     self.cfg = cfg
     self.shard = shard

@@ -41,8 +41,13 @@ class RingWorker:
     self.device, self.dtype = device, dtype
     self.cfg = config_from_hf(builtin_config(model_id) or {}, model_id)
     self.shard = equal_ring_shards(model_id, self.cfg.n_layers, world)[rank]
-    with torch.device("meta"):
-      model = ShardedModel(self.cfg, self.shard)
+    prev_dtype = torch.get_default_dtype()
+    torch.set_default_dtype(dtype)
+    try:
+      with torch.device("meta"):
+        model = ShardedModel(self.cfg, self.shard)
+    finally:
+      torch.set_default_dtype(prev_dtype)
     model = model.to_empty(device=device).to(dtype)
     if model_dir:
       load_shard_weights(model, model_dir, device="cpu")
@@ -53,6 +58,9 @@ class RingWorker:
       random_init(model)
     model.reset_rope()
     model.eval()
+    if device == "cuda":
+      torch.cuda.empty_cache()
+      model.pack_decode_weights(reserve_bytes=24 << 30)
     self.model = model
     self.cache = ShardKVCache(self.shard.get_layer_count(), 1, self.cfg.n_kv_heads,
                               MAX_SEQ, self.cfg.head_dim, dtype, device)
