@@ -21,8 +21,11 @@ class LoRALinear(nn.Module):
     self.base = base
     self.rank = rank
     self.scaling = alpha / rank
-    self.lora_a = nn.Parameter(torch.zeros(rank, base.in_features))
-    self.lora_b = nn.Parameter(torch.zeros(base.out_features, rank))
+    # adapters live on the base weight's device, in fp32 (cast per-forward):
+    # fp32 master adapters keep LoRA updates stable under bf16 bases
+    dev = base.weight.device
+    self.lora_a = nn.Parameter(torch.zeros(rank, base.in_features, device=dev, dtype=torch.float32))
+    self.lora_b = nn.Parameter(torch.zeros(base.out_features, rank, device=dev, dtype=torch.float32))
     nn.init.kaiming_uniform_(self.lora_a, a=math.sqrt(5))
     self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
     self.merged = False
