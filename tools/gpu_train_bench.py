@@ -47,5 +47,5 @@ for _ in range(N):
   loss = trainer.step(fwd, inputs, targets, lengths)
 torch.cuda.synchronize()
 dt = (time.perf_counter() - t0) / N
-print(f"lora train {model_id}: {n_lora} adapters, loss {float(loss):.3f}, "
+print(f"lora train {model_id}: {len(n_lora)} adapters, loss {float(loss):.3f}, "
       f"{dt*1e3:.1f} ms/step, {B*S/dt:.0f} tokens/s/gpu")
