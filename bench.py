@@ -72,6 +72,7 @@ def main():
   if device == "cpu" and args.model == "llama-3-70b":
     # CPU smoke path (no GPU in dev container): tiny model, same code path
     args.model = "dummy"
+    args.prompt_len = min(args.prompt_len, 128)  # dummy max_seq_len is 256
 
   from xotorch_amd.parallel.ring import RingPipeline
 
