@@ -109,9 +109,12 @@ class RingPipeline:
         self.M * 2 * shards[rank].get_layer_count() * mb_batch * self.cfg.n_kv_heads
         * self.total_len * self.cfg.head_dim * (2 if dtype == torch.bfloat16 else 4)
       )
+      if self.cfg.head_dim == 128 and dtype == torch.bfloat16 \
+         and os.getenv("XOT_MFMA_ATTN", "1") == "1":
+        kv_bytes *= 2  # the MFMA-packed cache copies double KV residency
       torch.cuda.empty_cache()  # release init-time cached blocks so the
       # pack policy's mem_get_info reflects actually-usable HBM
-      model.pack_decode_weights(reserve_bytes=kv_bytes + (16 << 30))
+      model.pack_decode_weights(reserve_bytes=kv_bytes + (24 << 30))
 
     # --- per-micro-batch state ---
     B = mb_batch
