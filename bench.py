@@ -53,7 +53,7 @@ def main():
   p.add_argument("--steps", type=int, default=32)
   p.add_argument("--warmup", type=int, default=8)
   p.add_argument("--model", type=str, default="llama-3-70b")
-  p.add_argument("--mb-batch", type=int, default=64, help="sequences per micro-batch (per pipeline slot)")
+  p.add_argument("--mb-batch", type=int, default=128, help="sequences per micro-batch (per pipeline slot)")
   p.add_argument("--prompt-len", type=int, default=512)
   p.add_argument("--no-graphs", action="store_true")
   p.add_argument("--device", type=str, default=None)

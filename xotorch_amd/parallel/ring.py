@@ -189,6 +189,26 @@ class RingPipeline:
 
   # ---------- prefill ----------
 
+  def _prefill_forward(self, mb: int, x: torch.Tensor, pos: torch.Tensor) -> torch.Tensor:
+    """Prefill one micro-batch through this shard in batch chunks: peak
+    activation memory (the fused gate_up output is B*S*2I bf16 — 15 GB at
+    B=256, S=512 on 70B) stays bounded at the chunk size regardless of B."""
+    B = x.shape[0]
+    bc = min(B, int(os.getenv("XOT_PREFILL_CHUNK", "64")))
+    if bc >= B:
+      return self.model(x, caches=self.caches[mb].caches, positions=pos, start_pos=0)
+    outs = []
+    for c0 in range(0, B, bc):
+      c1 = min(c0 + bc, B)
+      sliced = []
+      for layer in self.caches[mb].caches:
+        if len(layer) > 2 and layer[2] is not None:
+          sliced.append((layer[0][c0:c1], layer[1][c0:c1], layer[2][c0:c1], layer[3][c0:c1]))
+        else:
+          sliced.append((layer[0][c0:c1], layer[1][c0:c1]))
+      outs.append(self.model(x[c0:c1], caches=sliced, positions=pos, start_pos=0))
+    return torch.cat(outs, dim=0)
+
   def prefill(self, prompts: Optional[List[torch.Tensor]] = None) -> RingStats:
     """Prefill all micro-batches through the ring; returns per-mb TTFT (ms)
     measured on the last stage from the common post-barrier start."""
@@ -218,11 +238,11 @@ class RingPipeline:
           else:
             g = torch.Generator(device="cpu").manual_seed(1000 + mb)
             tokens = torch.randint(0, self.cfg.vocab_size, (B, S), generator=g).to(self.device)
-          h = self.model(tokens, caches=self.caches[mb].caches, positions=pos, start_pos=0)
+          h = self._prefill_forward(mb, tokens, pos)
         else:
           hbuf = torch.empty(B, S, self.cfg.dim, dtype=self.dtype, device=self.device)
           dist.recv(hbuf, self.prev_rank)
-          h = self.model(hbuf, caches=self.caches[mb].caches, positions=pos, start_pos=0)
+          h = self._prefill_forward(mb, hbuf, pos)
         if self.is_last:
           tok = h.argmax(dim=-1, keepdim=True)  # h is [B, V] logits
           if self.device == "cuda":
