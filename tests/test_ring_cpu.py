@@ -68,3 +68,28 @@ def test_two_stage_ring_matches_single():
     assert ref == two, "2-stage ring tokens diverged from single-process decode"
     # sanity: we actually generated STEPS+1 tokens per micro-batch
     assert len(ref[0]) == STEPS + 1
+
+
+def test_chunked_prefill_matches_unchunked(tmp_path, monkeypatch):
+  """Batch-chunked prefill (bounded activation peak) must produce the same
+  tokens as single-shot prefill."""
+  import torch
+  from xotorch_amd.parallel.ring import RingPipeline
+
+  def run(chunk):
+    if chunk:
+      monkeypatch.setenv("XOT_PREFILL_CHUNK", "4")
+    else:
+      monkeypatch.delenv("XOT_PREFILL_CHUNK", raising=False)
+    ring = RingPipeline(model_id="dummy", rank=0, world=1, device="cpu",
+                        dtype=torch.float32, mb_batch=12, prompt_len=48, max_gen=8,
+                        use_graphs=False, seed=7)
+    ring.capture_tokens = True
+    ring.prefill()
+    for _ in range(6):
+      ring.decode_step()
+    return torch.cat([t for t in ring.generated[0]], dim=1)
+
+  toks_chunked = run(True)
+  toks_full = run(False)
+  assert torch.equal(toks_chunked, toks_full)
