@@ -194,7 +194,7 @@ class RingPipeline:
     activation memory (the fused gate_up output is B*S*2I bf16 — 15 GB at
     B=256, S=512 on 70B) stays bounded at the chunk size regardless of B."""
     B = x.shape[0]
-    bc = min(B, int(os.getenv("XOT_PREFILL_CHUNK", "64")))
+    bc = min(B, int(os.getenv("XOT_PREFILL_CHUNK", "128")))
     if bc >= B:
       return self.model(x, caches=self.caches[mb].caches, positions=pos, start_pos=0)
     outs = []
