@@ -338,3 +338,50 @@ def test_rope_append_writes_packed(hip):
   _hip_ops.rope_qkv_append(qkv, cos, sin, pos, kc, vc, H, KVH, hd, kp, vp)
   assert torch.equal(kp, _pack_k(kc, t32))
   assert torch.equal(vp, _pack_v(vc, t32))
+
+
+def test_quant_fp8_rows(hip):
+  from xotorch_amd.ops import _hip_ops
+  x = bt(16, 512, scale=3.0, seed=71)
+  x8, sx = _hip_ops.quant_fp8_rows(x)
+  deq = x8.view(torch.float8_e4m3fn).float() * sx[:, None]
+  err = (deq - x.float()).abs().max().item() / x.float().abs().max().item()
+  assert err < 0.08, err  # e4m3 relative step ~ 2^-3 near max
+
+
+@pytest.mark.parametrize("M,K,N,bias", [(64, 8192, 1280, True), (64, 28672, 8192, False),
+                                        (128, 4096, 14336, False)])
+def test_skinny_gemm_fp8(hip, M, K, N, bias):
+  """W8A8 kernel vs the quantize-dequantize fp32 reference (tight tolerance:
+  the quantization error itself is excluded by dequantizing the same values)."""
+  from xotorch_amd import ops as xops
+  from xotorch_amd.ops import _hip_ops
+  x = bt(M, K, scale=0.5, seed=M + N + 3)
+  w = bt(N, K, scale=0.02, seed=K + 3)
+  b = bt(N, seed=8) if bias else None
+  wp8, sw = xops.pack_decode_weight_fp8(w)
+  x8, sx = _hip_ops.quant_fp8_rows(x)
+  got = _hip_ops.skinny_gemm_fp8(x8, sx, wp8, sw, 1, N, b).float()
+  w_deq = (w.float() / sw[:, None]).clamp(-448, 448).to(torch.float8_e4m3fn).float() * sw[:, None]
+  x_deq = x8.view(torch.float8_e4m3fn).float() * sx[:, None]
+  ref = torch.nn.functional.linear(x_deq, w_deq, b.float() if bias else None)
+  err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-9)
+  assert err < 2e-2, err
+  # and end-to-end (including quantization error) stays in the W8A8 class
+  ref_full = torch.nn.functional.linear(x.float(), w.float(), b.float() if bias else None)
+  e2e = (got - ref_full).abs().max().item() / (ref_full.abs().max().item() + 1e-9)
+  assert e2e < 0.12, e2e
+
+
+def test_xotlinear_fp8_mode(hip, monkeypatch):
+  from xotorch_amd.models.llama import XotLinear
+  monkeypatch.setenv("XOT_FP8_GEMM", "1")
+  m = XotLinear(4096, 512, bias=True).to("cuda").to(torch.bfloat16)
+  x = bt(64, 1, 4096, scale=0.5)
+  with torch.inference_mode():
+    y_bf16 = m(x).float()
+    m.pack_decode()
+    assert m.weight_packed_fp8 is not None
+    y_fp8 = m(x).float()
+  err = (y_bf16 - y_fp8).abs().max().item() / (y_bf16.abs().max().item() + 1e-9)
+  assert err < 0.12, err

@@ -65,20 +65,38 @@ class XotLinear(nn.Linear):
   def __init__(self, *a, **kw):
     super().__init__(*a, **kw)
     self.weight_packed = None
+    self.weight_packed_fp8 = None
+    self.weight_scale = None
 
   def packable(self) -> bool:
     N, K = self.weight.shape
     return N % 128 == 0 and K % 64 == 0
 
   def pack_decode(self):
-    if self.weight_packed is None and self.packable() and self.weight.is_cuda \
-       and self.weight.dtype == torch.bfloat16:
-      self.weight_packed = ops.pack_decode_weight(self.weight.detach())
+    if self.packable() and self.weight.is_cuda and self.weight.dtype == torch.bfloat16:
+      if ops.fp8_gemm_enabled():
+        if getattr(self, "weight_packed_fp8", None) is None:
+          self.weight_packed_fp8, self.weight_scale = ops.pack_decode_weight_fp8(self.weight.detach())
+      elif self.weight_packed is None:
+        self.weight_packed = ops.pack_decode_weight(self.weight.detach())
 
   def unpack_decode(self):
     self.weight_packed = None
+    self.weight_packed_fp8 = None
 
   def forward(self, x):
+    wp8 = getattr(self, "weight_packed_fp8", None)
+    if wp8 is not None and x.is_cuda and x.dtype == torch.bfloat16 and not torch.is_grad_enabled():
+      N, K = self.weight.shape
+      M = x.numel() // K
+      if 32 <= M <= 256 and M % 32 == 0 and x.is_contiguous():
+        from xotorch_amd.ops import _load_hip
+        hip = _load_hip()
+        if hip is not None:
+          x8, sx = hip.quant_fp8_rows(x.view(M, K))
+          y = hip.skinny_gemm_fp8(x8, sx, wp8, self.weight_scale, 1, N, self.bias)
+          sizes = list(x.shape[:-1]) + [N]
+          return y.view(sizes)
     if self.weight_packed is not None and x.is_cuda and x.dtype == torch.bfloat16:
       N, K = self.weight.shape
       M = x.numel() // K
@@ -182,9 +200,25 @@ class MoEMLP(nn.Module):
     self.experts = nn.ModuleList([MLP(cfg) for _ in range(cfg.n_experts)])
     self.wp_gate_up: Optional[torch.Tensor] = None
     self.wp_down: Optional[torch.Tensor] = None
+    self.wp_gate_up_fp8: Optional[torch.Tensor] = None
+    self.wp_down_fp8: Optional[torch.Tensor] = None
+    self.sw_gate_up: Optional[torch.Tensor] = None
+    self.sw_down: Optional[torch.Tensor] = None
 
   def pack_grouped(self):
     """Stack the experts' prepacked weights for the grouped decode GEMM."""
+    if ops.fp8_gemm_enabled():
+      if self.wp_gate_up_fp8 is None \
+         and all(getattr(e.gate_up_proj, "weight_packed_fp8", None) is not None for e in self.experts) \
+         and all(getattr(e.down_proj, "weight_packed_fp8", None) is not None for e in self.experts):
+        self.wp_gate_up_fp8 = torch.stack([e.gate_up_proj.weight_packed_fp8 for e in self.experts]).contiguous()
+        self.sw_gate_up = torch.stack([e.gate_up_proj.weight_scale for e in self.experts]).contiguous()
+        self.wp_down_fp8 = torch.stack([e.down_proj.weight_packed_fp8 for e in self.experts]).contiguous()
+        self.sw_down = torch.stack([e.down_proj.weight_scale for e in self.experts]).contiguous()
+        for e in self.experts:
+          e.gate_up_proj.weight_packed_fp8 = None
+          e.down_proj.weight_packed_fp8 = None
+      return
     if self.wp_gate_up is None and all(e.gate_up_proj.weight_packed is not None for e in self.experts) \
        and all(e.down_proj.weight_packed is not None for e in self.experts):
       self.wp_gate_up = torch.stack([e.gate_up_proj.weight_packed for e in self.experts]).contiguous()
@@ -238,7 +272,17 @@ class MoEMLP(nn.Module):
     gather_tok = sorted_token[pos_c.reshape(-1)]              # [E*C]
     scale = torch.where(valid, sorted_w[pos_c], torch.zeros((), dtype=sorted_w.dtype, device=dev))
     xg = flat[gather_tok]                                     # [E*C, D]
-    if self.wp_gate_up is not None and not torch.is_grad_enabled():
+    if self.wp_gate_up_fp8 is not None and not torch.is_grad_enabled():
+      from xotorch_amd.ops import _load_hip
+      hip = _load_hip()
+      I = self.intermediate
+      x8, sx = hip.quant_fp8_rows(xg)
+      gu = hip.skinny_gemm_fp8(x8, sx, self.wp_gate_up_fp8, self.sw_gate_up.view(-1), E, 2 * I)
+      h = ops.swiglu_packed(gu.view(E * C, 2 * I))
+      h8, sh = hip.quant_fp8_rows(h)
+      y = hip.skinny_gemm_fp8(h8, sh, self.wp_down_fp8, self.sw_down.view(-1), E, D)
+      y = y.view(E * C, D)
+    elif self.wp_gate_up is not None and not torch.is_grad_enabled():
       from xotorch_amd.ops import _load_hip
       hip = _load_hip()
       gu = hip.skinny_gemm_grouped(xg.view(E, C, D), self.wp_gate_up, E, 2 * self.intermediate)
@@ -353,7 +397,8 @@ class ShardedModel(nn.Module):
     for gi, grp in enumerate(groups):
       if not grp:
         continue
-      need = sum(m.weight.numel() * 2 for m in grp)
+      bytes_per = 1 if ops.fp8_gemm_enabled() else 2
+      need = sum(m.weight.numel() * bytes_per for m in grp)
       free, _ = torch.cuda.mem_get_info()
       if need + reserve_bytes > free:
         if debug:

@@ -152,6 +152,23 @@ def pack_decode_weight(w: torch.Tensor) -> torch.Tensor:
   return w.view(N // 32, 32, K // 16, 2, 8).permute(0, 2, 3, 1, 4).contiguous()
 
 
+def pack_decode_weight_fp8(w: torch.Tensor):
+  """Quantize a [N, K] weight to OCP e4m3 with per-output-channel scales and
+  pre-shuffle into the MFMA fragment order (bytes): returns (packed_uint8,
+  s_w fp32 [N]). Half the stream bytes of the bf16 prepack — the W8A8 decode
+  GEMM path (XOT_FP8_GEMM=1)."""
+  N, K = w.shape
+  assert N % 32 == 0 and K % 64 == 0, (N, K)
+  s_w = (w.float().abs().amax(dim=1).clamp(min=1e-12) / 448.0)
+  w8 = (w.float() / s_w[:, None]).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+  packed = w8.view(torch.uint8).view(N // 32, 32, K // 16, 2, 8).permute(0, 2, 3, 1, 4).contiguous()
+  return packed, s_w.contiguous()
+
+
+def fp8_gemm_enabled() -> bool:
+  return os.getenv("XOT_FP8_GEMM", "0") == "1"
+
+
 def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None) -> torch.Tensor:
   """y = x @ weight^T (+bias) on hipBLASLt/aten.
 

@@ -811,6 +811,184 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// fp8 (OCP e4m3) W8A8 variant: same structure as the bf16 packed kernel but
+// operands are e4m3 with per-output-channel weight scales s_w[n] and
+// per-token activation scales s_x[m] (dynamic, computed by quant_fp8_rows).
+// v_mfma_f32_32x32x16_fp8_fp8 has the SAME fragment geometry as the bf16
+// form (8 elements/lane), so the prepack layout carries over at half the
+// bytes — decode GEMMs are weight-bandwidth-bound, so fp8 approaches 2x.
+// Opt-in serving mode (XOT_FP8_GEMM=1); the benchmark path stays bf16.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(8))) char fp8x8;
+
+DEVINL long as_i64(fp8x8 v) { return *reinterpret_cast<long*>(&v); }
+
+template <int MT, bool SPLIT>
+__global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
+    const unsigned char* __restrict__ Wp, const unsigned char* __restrict__ X8,
+    const float* __restrict__ sw, const float* __restrict__ sx,
+    unsigned short* __restrict__ Y, float* __restrict__ P,
+    const unsigned short* __restrict__ bias,
+    int N, long long K, int kc, int nsplit) {
+  const int ntiles = N >> 7;
+  const int tile = blockIdx.x % ntiles;
+  const int split = blockIdx.x / ntiles;
+  const int e = blockIdx.y;  // grouped mode (MoE): expert index
+  Wp += (size_t)e * (size_t)N * (size_t)K;
+  X8 += (size_t)e * (size_t)(MT * 32) * (size_t)K;
+  const long long k0 = (long long)split * kc;
+  const long long k1 = min(k0 + (long long)kc, K);
+  if (k0 >= k1) return;
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+
+  __shared__ unsigned char xs8[MT * 32 * 72];  // fp8 rows, 72 B stride (64 + 8 pad)
+
+  floatx16 acc[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) acc[t] = (floatx16)(0.f);
+
+  const int n32 = tile * 4 + wv;
+  const unsigned char* wp = Wp + ((size_t)n32 * (K >> 4) + (k0 >> 4)) * 512 + (size_t)lane * 8;
+  // X8 staging: thread tid covers 8-byte piece q (8 of them per 64-k row)
+  const int xr = tid >> 3, xq = tid & 7;
+  const unsigned char* xp = X8 + (size_t)xr * K + k0 + xq * 8;
+  const int xs_off = xr * 72 + xq * 8;
+
+  const int nsteps = (int)((k1 - k0) >> 6);
+  unsigned long long xv[MT];
+  fp8x8 a_buf[2][4];
+
+  const unsigned char* wp1 = (nsteps > 1) ? wp + 4 * 512 : wp;
+#pragma unroll
+  for (int t = 0; t < MT; ++t) xv[t] = *(const unsigned long long*)(xp + (size_t)t * 32 * K);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    a_buf[0][u] = __builtin_nontemporal_load(reinterpret_cast<const fp8x8*>(wp + u * 512));
+    a_buf[1][u] = __builtin_nontemporal_load(reinterpret_cast<const fp8x8*>(wp1 + u * 512));
+  }
+  wp += 8 * 512;
+#pragma unroll
+  for (int t = 0; t < MT; ++t) *(unsigned long long*)(xs8 + t * 32 * 72 + xs_off) = xv[t];
+  __syncthreads();
+
+#define SGF_STEP(BUF)                                                                          \
+  {                                                                                            \
+    const bool last = (s == nsteps - 1);                                                       \
+    if (!last) {                                                                               \
+      _Pragma("unroll")                                                                        \
+      for (int t = 0; t < MT; ++t)                                                             \
+        xv[t] = *(const unsigned long long*)(xp + (size_t)t * 32 * K + (s + 1) * 64);          \
+    }                                                                                          \
+    _Pragma("unroll")                                                                          \
+    for (int u = 0; u < 4; ++u) {                                                              \
+      _Pragma("unroll")                                                                        \
+      for (int t = 0; t < MT; ++t) {                                                           \
+        const fp8x8 b = *reinterpret_cast<const fp8x8*>(                                       \
+            xs8 + t * 32 * 72 + (lane & 31) * 72 + u * 16 + (lane >> 5) * 8);                  \
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_fp8_fp8(                                   \
+            as_i64(a_buf[BUF][u]), as_i64(b), acc[t], 0, 0, 0);                                \
+      }                                                                                        \
+    }                                                                                          \
+    if (s + 2 < nsteps) {                                                                      \
+      _Pragma("unroll")                                                                        \
+      for (int u = 0; u < 4; ++u)                                                              \
+        a_buf[BUF][u] = __builtin_nontemporal_load(reinterpret_cast<const fp8x8*>(wp + u * 512)); \
+      wp += 4 * 512;                                                                           \
+    }                                                                                          \
+    if (!last) {                                                                               \
+      __syncthreads();                                                                         \
+      _Pragma("unroll")                                                                        \
+      for (int t = 0; t < MT; ++t) *(unsigned long long*)(xs8 + t * 32 * 72 + xs_off) = xv[t]; \
+      __syncthreads();                                                                         \
+    }                                                                                          \
+  }
+
+  int s = 0;
+  while (s + 2 <= nsteps) {
+    SGF_STEP(0);
+    ++s;
+    SGF_STEP(1);
+    ++s;
+  }
+  if (s < nsteps) SGF_STEP(0);
+#undef SGF_STEP
+
+  // epilogue: dequantize acc * s_x[m] * s_w[n]
+  const int m_local = lane & 31;
+  const int rows_total = gridDim.y * MT * 32;
+  const int nbase = tile * 128 + wv * 32 + 4 * (lane >> 5);
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    const int m = e * MT * 32 + t * 32 + m_local;
+    const float sxm = sx[m];
+    if (SPLIT) {
+      float* prow = P + ((size_t)split * rows_total + m) * N;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        floatx4 v4;
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          v4[j] = acc[t][g * 4 + j] * sxm * sw[(size_t)e * N + nbase + g * 8 + j];
+        *reinterpret_cast<floatx4*>(prow + nbase + g * 8) = v4;
+      }
+    } else {
+      unsigned short* yrow = Y + (size_t)m * N;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        unsigned short o[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float v = acc[t][g * 4 + j] * sxm * sw[(size_t)e * N + nbase + g * 8 + j];
+          if (bias) v += b2f(bias[nbase + g * 8 + j]);
+          o[j] = f2b(v);
+        }
+        *reinterpret_cast<unsigned long long*>(yrow + nbase + g * 8) =
+            *reinterpret_cast<unsigned long long*>(o);
+      }
+    }
+  }
+}
+
+// per-row dynamic e4m3 quantization: x [M, K] bf16 -> x8 [M, K] + s_x [M]
+// (s_x = rowmax/448; one 256-thread workgroup per row, K <= 64K)
+__global__ __launch_bounds__(256) void quant_fp8_rows_kernel(
+    const unsigned short* __restrict__ x, unsigned char* __restrict__ x8,
+    float* __restrict__ sx, long long K) {
+  const long long row = blockIdx.x;
+  const unsigned short* xr = x + row * K;
+  unsigned char* qr = x8 + row * K;
+  float mx = 0.f;
+  for (long long i = threadIdx.x * 8; i < K; i += 256 * 8) {
+    ushort8 v = *(const ushort8*)(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) mx = fmaxf(mx, fabsf(b2f(v[j])));
+  }
+#pragma unroll
+  for (int m = 32; m > 0; m >>= 1) mx = fmaxf(mx, __shfl_xor(mx, m));
+  __shared__ float red[4];
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = mx;
+  __syncthreads();
+  const float scale = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3])) / 448.f;
+  const float inv = (scale > 0.f) ? 1.f / scale : 0.f;
+  if (threadIdx.x == 0) sx[row] = (scale > 0.f) ? scale : 1.f;
+  for (long long i = threadIdx.x * 8; i < K; i += 256 * 8) {
+    ushort8 v = *(const ushort8*)(xr + i);
+    unsigned char o[8];
+#pragma unroll
+    for (int j = 0; j < 8; j += 2) {
+      const float a = b2f(v[j]) * inv, b = b2f(v[j + 1]) * inv;
+      const unsigned short pk = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false);
+      o[j] = (unsigned char)(pk & 0xff);
+      o[j + 1] = (unsigned char)(pk >> 8);
+    }
+    *(unsigned long long*)(qr + i) = *(unsigned long long*)o;
+  }
+}
+
 // combine fp32 split-K partials [S, M, N] -> bf16 [M, N] (+bias)
 __global__ __launch_bounds__(256) void skinny_combine_kernel(
     const float* __restrict__ P, unsigned short* __restrict__ Y,
@@ -1194,7 +1372,85 @@ torch::Tensor skinny_gemm_grouped(torch::Tensor x, torch::Tensor wp, int64_t E, 
   return y;
 }
 
+// x: [M, K] bf16 -> (x8 uint8 [M, K], s_x fp32 [M])
+std::vector<torch::Tensor> quant_fp8_rows(torch::Tensor x) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  const long long K = x.size(-1);
+  const long long M = x.numel() / K;
+  CHK(K % 8 == 0);
+  auto x8 = torch::empty({M, K}, torch::TensorOptions().dtype(torch::kUInt8).device(x.device()));
+  auto sx = torch::empty({M}, torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+  hipLaunchKernelGGL(quant_fp8_rows_kernel, dim3((unsigned)M), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)x.data_ptr(), (unsigned char*)x8.data_ptr(),
+                     sx.data_ptr<float>(), K);
+  return {x8, sx};
+}
+
+// W8A8 e4m3 decode GEMM. x8/sx from quant_fp8_rows; wp8: fp8 prepack
+// [E][N/32][K/16][64][8] bytes; sw: [E, N] fp32 per-channel scales. E == 1
+// for plain GEMMs; E > 1 = MoE grouped mode (x8 is [E, C, K]).
+torch::Tensor skinny_gemm_fp8(torch::Tensor x8, torch::Tensor sx, torch::Tensor wp8,
+                              torch::Tensor sw, int64_t E, int64_t N,
+                              c10::optional<torch::Tensor> bias) {
+  CHK(x8.is_cuda() && x8.dtype() == torch::kUInt8 && x8.is_contiguous());
+  CHK(wp8.is_cuda() && wp8.dtype() == torch::kUInt8 && wp8.is_contiguous());
+  CHK(sx.dtype() == torch::kFloat32 && sw.dtype() == torch::kFloat32);
+  CHK(sw.is_contiguous() && sx.is_contiguous());
+  const long long K = wp8.numel() / (E * N);
+  const long long M = x8.numel() / (K * E);  // rows per expert
+  CHK(M >= 32 && M <= 256 && M % 32 == 0);
+  CHK(N % 128 == 0 && K % 64 == 0 && sw.numel() == E * N && sx.numel() == E * M);
+  const unsigned short* bptr = nullptr;
+  if (bias.has_value()) {
+    CHK(bias->is_contiguous() && bias->dtype() == torch::kBFloat16 && bias->numel() == N);
+    bptr = (const unsigned short*)bias->data_ptr();
+  }
+  auto y = torch::empty({(long)(E * M), (long)N},
+                        torch::TensorOptions().dtype(torch::kBFloat16).device(x8.device()));
+  const int ntiles = (int)(N / 128);
+  int nsplit = 1;
+  while (ntiles * nsplit * E * 2 < 1024 && (K / (nsplit * 2)) >= 1024 && nsplit < 16) nsplit *= 2;
+  int kc = (int)((K / nsplit + 63) / 64 * 64);
+  while ((long long)kc * (nsplit - 1) >= K) nsplit--;
+  auto stream = cur_stream();
+  const int MT = (int)(M / 32);
+  const dim3 grid(ntiles * nsplit, (unsigned)E), block(256);
+#define SGF_CASE(MTV) \
+  case MTV: \
+    if (nsplit == 1) { \
+      hipLaunchKernelGGL((skinny_gemm_fp8_kernel<MTV, false>), grid, block, 0, stream, \
+                         (const unsigned char*)wp8.data_ptr(), (const unsigned char*)x8.data_ptr(), \
+                         sw.data_ptr<float>(), sx.data_ptr<float>(), \
+                         (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit); \
+    } else { \
+      hipLaunchKernelGGL((skinny_gemm_fp8_kernel<MTV, true>), grid, block, 0, stream, \
+                         (const unsigned char*)wp8.data_ptr(), (const unsigned char*)x8.data_ptr(), \
+                         sw.data_ptr<float>(), sx.data_ptr<float>(), \
+                         nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+    } \
+    break;
+  if (nsplit == 1) {
+    torch::Tensor P;
+    switch (MT) { SGF_CASE(1) SGF_CASE(2) SGF_CASE(3) SGF_CASE(4) SGF_CASE(5) SGF_CASE(6) SGF_CASE(7) SGF_CASE(8) }
+  } else {
+    auto P = torch::empty({nsplit, (long)(E * M), (long)N},
+                          torch::TensorOptions().dtype(torch::kFloat32).device(x8.device()));
+    switch (MT) { SGF_CASE(1) SGF_CASE(2) SGF_CASE(3) SGF_CASE(4) SGF_CASE(5) SGF_CASE(6) SGF_CASE(7) SGF_CASE(8) }
+    const long long MN = E * M * N;
+    const int blocks = (int)std::min<long long>(2048, (MN / 4 + 255) / 256);
+    hipLaunchKernelGGL(skinny_combine_kernel, dim3(blocks), dim3(256), 0, stream,
+                       P.data_ptr<float>(), (unsigned short*)y.data_ptr(), bptr, MN, N, nsplit);
+  }
+#undef SGF_CASE
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("quant_fp8_rows", &quant_fp8_rows, "per-row dynamic e4m3 quantization");
+  m.def("skinny_gemm_fp8", &skinny_gemm_fp8,
+        "W8A8 e4m3 decode GEMM on prepacked weights (per-channel/per-token scales)",
+        py::arg("x8"), py::arg("sx"), py::arg("wp8"), py::arg("sw"), py::arg("n_experts"),
+        py::arg("n"), py::arg("bias") = py::none());
   m.def("skinny_gemm", &skinny_gemm, "decode GEMM y = x @ w^T (+bias), bf16 MFMA weight-streaming",
         py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
   m.def("skinny_gemm_packed", &skinny_gemm_packed,
