@@ -135,7 +135,7 @@ def main():
       "higher_is_better": True,
       "scaling": "weak",
       "vs_baseline": None,
-      "dtype": args.dtype,
+      "dtype": (args.dtype + "+fp8-w8a8" if os.getenv("XOT_FP8_GEMM", "0") == "1" else args.dtype),
       "data": "synthetic",
       "config": {
         "model": args.model,

@@ -386,11 +386,12 @@ class ShardedModel(nn.Module):
         groups[2].append(mod)
       else:  # qkv_proj / o_proj / experts' projections
         groups[3].append(mod)
-    if mode != "all":
-      # default: the measured winners — down_proj (tuned hipBLASLt 110 us vs
-      # 84.6 packed at 70B decode shapes), lm_head (5.3 -> 6.2 TB/s), gate_up
-      # (175.5 vs 178.9 us); qkv/o measure tied on hipBLASLt — not worth the
-      # second weight copy.
+    if mode != "all" and not ops.fp8_gemm_enabled():
+      # default (bf16): the measured winners — down_proj (tuned hipBLASLt
+      # 110 us vs 84.6 packed at 70B decode shapes), lm_head (5.3 -> 6.2
+      # TB/s), gate_up (175.5 vs 178.9 us); qkv/o measure tied on hipBLASLt
+      # — not worth the second weight copy. In fp8 mode every group wins
+      # (half the stream bytes), so all are packed.
       groups = groups[:3]
     packed = 0
     debug = os.getenv("XOT_DEBUG", "0") != "0"
