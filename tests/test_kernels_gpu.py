@@ -385,3 +385,36 @@ def test_xotlinear_fp8_mode(hip, monkeypatch):
     y_fp8 = m(x).float()
   err = (y_bf16 - y_fp8).abs().max().item() / (y_bf16.abs().max().item() + 1e-9)
   assert err < 0.12, err
+
+
+@pytest.mark.parametrize("B,H,KVH,S,start", [(2, 8, 2, 128, 0), (1, 32, 8, 512, 0),
+                                             (2, 16, 4, 96, 0), (1, 8, 8, 64, 40)])
+def test_attn_prefill_mfma(hip, B, H, KVH, S, start):
+  """Causal MFMA prefill vs the fp32 torch reference (full prefix + causal
+  block; start > 0 = chat continuation with an existing prefix)."""
+  from xotorch_amd.ops import _hip_ops, torch_ref
+  hd = 128
+  total = start + S
+  t32 = (total + 31) // 32 * 32
+  q = bt(B, S, H, hd, seed=81)
+  k = bt(B, KVH, total, hd, seed=82)
+  v = bt(B, KVH, total, hd, seed=83)
+  kp, vp = _pack_k(k, t32), _pack_v(v, t32)
+  out = _hip_ops.attn_prefill_mfma(q, kp, vp, start).float()
+  ref = torch_ref.attn_prefill(q, k, v, start, S).float()
+  assert torch.allclose(out, ref, atol=3e-2, rtol=3e-2), (out - ref).abs().max()
+
+
+def test_attn_prefill_mfma_strided_q(hip):
+  """q as the strided view into the packed qkv row (the fused-GEMM layout)."""
+  from xotorch_amd.ops import _hip_ops, torch_ref
+  B, S, H, KVH, hd = 2, 80, 16, 4, 128
+  t32 = (S + 31) // 32 * 32
+  qkv = bt(B, S, (H + 2 * KVH) * hd, seed=91)
+  q = qkv[:, :, : H * hd].view(B, S, H, hd)
+  k = bt(B, KVH, S, hd, seed=92)
+  v = bt(B, KVH, S, hd, seed=93)
+  kp, vp = _pack_k(k, t32), _pack_v(v, t32)
+  out = _hip_ops.attn_prefill_mfma(q, kp, vp, 0).float()
+  ref = torch_ref.attn_prefill(q.contiguous(), k, v, 0, S).float()
+  assert torch.allclose(out, ref, atol=3e-2, rtol=3e-2), (out - ref).abs().max()

@@ -159,7 +159,7 @@ class Attention(nn.Module):
       sl = seq_lens if seq_lens is not None else start_pos + 1
       out = ops.attn_decode(q, k_cache, v_cache, sl, kp, vp)
     else:
-      out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S)
+      out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S, kp, vp)
     return self.o_proj(out.reshape(B, S, H * hd))
 
 

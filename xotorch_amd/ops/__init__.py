@@ -90,11 +90,14 @@ def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_
   torch_ref.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim)
 
 
-def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int):
+def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int, kp=None, vp=None):
+  if q.is_cuda and q.dtype == torch.bfloat16 and kp is not None and q.shape[3] == 128 \
+     and os.getenv("XOT_MFMA_ATTN", "1") == "1" and _use_hip(q):
+    # causal flash-forward on matrix cores, streaming the MFMA-packed cache
+    return _hip.attn_prefill_mfma(q, kp, vp, start_pos)
   if q.is_cuda:
-    # Prefill goes through sdpa (rocm flash/mem-efficient backends) in model
-    # dtype with native GQA — a hand-written MFMA flash-prefill kernel is a
-    # planned replacement; decode (the headline metric) is the HIP kernel.
+    # fallback: sdpa (rocm flash/mem-efficient backends) in model dtype with
+    # native GQA — hd != 128 or unpacked-cache path
     import torch.nn.functional as F
     B, S, H, hd = q.shape
     total = start_pos + s_len

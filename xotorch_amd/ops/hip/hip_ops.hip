@@ -469,6 +469,195 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// MFMA flash-forward prefill attention (hd = 128, causal, GQA).
+//
+// Replaces torch sdpa for the prefill step (reference: torchtune
+// MultiHeadAttention prefill, general_mha.py:218-226; sdpa measures ~1.8 ms
+// per 70B layer at B=64xS=512 — VALU/flash at ~300 TF).  Grid: one 4-wave
+// workgroup per (b, q-head, 128-row q block); the block's waves share K/V
+// tiles staged in LDS from the MFMA-packed cache (fragment layout is
+// lane-linear, so staging is 1 KB coalesced loads and conflict-free b128
+// reads).  Each wave owns 32 q rows (2 MFMA sub-tiles), walks kv position
+// tiles of 32 with online softmax, masks the causal boundary, writes final
+// bf16 rows (no split-K workspace).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
+    const unsigned short* __restrict__ vp, unsigned short* __restrict__ out,
+    int B, int S, int H, int KVH, int T32, int start_pos,
+    long long q_bstride, long long q_sstride, float scale) {
+  // block -> (b, h, q-block)
+  const int qblocks = (S + 127) >> 7;
+  const int h = blockIdx.x % H;
+  const int t1 = blockIdx.x / H;
+  const int qb = t1 % qblocks;
+  const int b = t1 / qblocks;
+  const int wv = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int kvh = h / (H / KVH);
+
+  // this wave's 32 q rows: [r0, r0+32)
+  const int r0 = qb * 128 + wv * 32;
+  const int col = lane & 15;
+
+  __shared__ unsigned short kv_lds[2][(4 + 8) * 512];  // K tile 4 KB + V tile 8 KB, dbuf
+
+  // Q fragments for 2 sub-tiles x 4 hd-chunks; rows clamped to S-1 (writes
+  // are predicated, extra rows only waste compute)
+  bf16x8 qf[2][4];
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    const int row = min(r0 + t * 16 + col, S - 1);
+    const unsigned short* qrow = q + (size_t)b * q_bstride + (size_t)row * q_sstride + (size_t)h * 128;
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+      qf[t][c] = *reinterpret_cast<const bf16x8*>(qrow + c * 32 + (lane >> 4) * 8);
+  }
+
+  float m[2][4], lsum[2][4];
+  floatx4 acco[2][8];
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) { m[t][r] = -INFINITY; lsum[t][r] = 0.f; }
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int g = 0; g < 8; ++g) acco[t][g] = (floatx4)(0.f);
+
+  const size_t kbase = (size_t)(b * KVH + kvh) * (T32 >> 4) * 2048;
+  const size_t vbase = (size_t)(b * KVH + kvh) * 8 * (T32 >> 5) * 512;
+  // causal end for the BLOCK (max row): positions [0, start_pos + block_hi]
+  const int block_hi = min(qb * 128 + 127, S - 1);
+  const int kv_end = start_pos + block_hi + 1;  // exclusive
+
+  __shared__ unsigned short plds_all[4][16 * 48 * 2];
+
+  // cooperative stage of kv tile `tp` (32 positions) into kv_lds[buf]
+  auto stage = [&](int tp, int buf) {
+    // K: two 16-pos tiles x 4 chunks x 512 elems; V: 8 groups x 512 elems
+    // 256 threads x 24 pieces of 256 B... simpler: each thread copies 16 B x3
+    const unsigned short* ksrc = kp + kbase + (size_t)(tp * 2) * 2048;
+    const unsigned short* vsrc = vp + vbase + (size_t)tp * 512;
+    unsigned short* dk = kv_lds[buf];
+    unsigned short* dv = kv_lds[buf] + 4 * 512;
+    const int tid = threadIdx.x;
+    // K: 4096 elems = 8192 B: 256 threads x 2 x 16 B
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int e = (tid + i * 256) * 8;
+      *(ushort8*)(dk + e) = *(const ushort8*)(ksrc + e);
+    }
+    // V: 8 groups x 512 elems, group stride (T32>>5)*512 in global
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int e = tid + i * 256;           // 0..511 -> (group, piece)
+      const int g = e >> 6, piece = e & 63;  // 64 pieces of 8 elems per group
+      *(ushort4_t*)(dv + g * 512 + piece * 8) =
+          *(const ushort4_t*)(vsrc + (size_t)g * (T32 >> 5) * 512 + piece * 8);
+      *(ushort4_t*)(dv + g * 512 + piece * 8 + 4) =
+          *(const ushort4_t*)(vsrc + (size_t)g * (T32 >> 5) * 512 + piece * 8 + 4);
+    }
+  };
+
+  const int ntiles = (kv_end + 31) >> 5;
+  stage(0, 0);
+  __syncthreads();
+
+  for (int tp = 0; tp < ntiles; ++tp) {
+    const int buf = tp & 1;
+    if (tp + 1 < ntiles) stage(tp + 1, buf ^ 1);  // plain loads overlap compute
+    const unsigned short* dk = kv_lds[buf];
+    const unsigned short* dv = kv_lds[buf] + 4 * 512;
+    unsigned short* plds = plds_all[wv];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      // scores for two 16-pos halves of this kv tile
+      floatx4 sc[2];
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+        sc[hh] = (floatx4)(0.f);
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const bf16x8 kb = *reinterpret_cast<const bf16x8*>(dk + (hh * 4 + c) * 512 + lane * 8);
+          sc[hh] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[t][c], kb, sc[hh], 0, 0, 0);
+        }
+      }
+      // causal mask: kv pos must be < start_pos + row + 1
+      float rmax[4];
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+        const int pos = tp * 32 + hh * 16 + col;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = r0 + t * 16 + (lane >> 4) * 4 + r;
+          const bool ok = (pos <= start_pos + row) && (pos < kv_end);
+          sc[hh][r] = ok ? sc[hh][r] * scale : -INFINITY;
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) rmax[r] = fmaxf(sc[0][r], sc[1][r]);
+#pragma unroll
+      for (int mm = 1; mm < 16; mm <<= 1)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) rmax[r] = fmaxf(rmax[r], __shfl_xor(rmax[r], mm));
+      float alpha[4], rsum[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float mn = fmaxf(m[t][r], rmax[r]);
+        if (mn == -INFINITY) { alpha[r] = 1.f; rsum[r] = 0.f; sc[0][r] = 0.f; sc[1][r] = 0.f; continue; }
+        alpha[r] = (lsum[t][r] > 0.f) ? __expf(m[t][r] - mn) : 0.f;
+        m[t][r] = mn;
+        const float p0 = __expf(sc[0][r] - mn);
+        const float p1 = __expf(sc[1][r] - mn);
+        sc[0][r] = p0;
+        sc[1][r] = p1;
+        rsum[r] = p0 + p1;
+      }
+#pragma unroll
+      for (int mm = 1; mm < 16; mm <<= 1)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) rsum[r] += __shfl_xor(rsum[r], mm);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) lsum[t][r] = lsum[t][r] * alpha[r] + rsum[r];
+#pragma unroll
+      for (int g = 0; g < 8; ++g)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) acco[t][g][r] *= alpha[r];
+      unsigned short* pl = plds + t * 16 * 48;  // reuse the 2x buffer per sub-tile
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          pl[((lane >> 4) * 4 + r) * 48 + hh * 16 + col] = f2b(sc[hh][r]);
+      const bf16x8 pf = *reinterpret_cast<const bf16x8*>(pl + (lane & 15) * 48 + (lane >> 4) * 8);
+#pragma unroll
+      for (int g = 0; g < 8; ++g) {
+        const bf16x8 vb = *reinterpret_cast<const bf16x8*>(dv + g * 512 + lane * 8);
+        acco[t][g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vb, acco[t][g], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // all waves done with kv_lds[buf] before restaging
+  }
+
+  // epilogue: out[b, row, h, g*16 + col] = acc / lsum
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r0 + t * 16 + (lane >> 4) * 4 + r;
+      if (row < S) {
+        const float inv = (lsum[t][r] > 0.f) ? 1.f / lsum[t][r] : 0.f;
+        unsigned short* orow = out + (((size_t)b * S + row) * H + h) * 128;
+#pragma unroll
+        for (int g = 0; g < 8; ++g) orow[g * 16 + col] = f2b(acco[t][g][r] * inv);
+      }
+    }
+  }
+}
+
 // layout probe for tests: one v_mfma_f32_16x16x32_bf16, row-major inputs
 __global__ void mfma16_probe_kernel(const unsigned short* __restrict__ A,
                                     const unsigned short* __restrict__ Bm,
@@ -1176,6 +1365,29 @@ torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor 
   return out;
 }
 
+// q: [B, S, H, 128] (strided over b/s OK, head rows contiguous); kp/vp: the
+// packed cache copies already holding positions [0, start_pos + S).
+torch::Tensor attn_prefill_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor vp,
+                                int64_t start_pos) {
+  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  CHK(q.stride(3) == 1 && q.stride(2) == q.size(3));
+  CHK(kp.is_contiguous() && vp.is_contiguous());
+  const int B = q.size(0), S = q.size(1), H = q.size(2), hd = q.size(3);
+  TORCH_CHECK(hd == 128, "attn_prefill_mfma requires head_dim 128");
+  const int KVH = kp.size(1);
+  const int T32 = kp.size(2) * 16;
+  CHK(start_pos + S <= T32);
+  auto out = torch::empty({B, S, H, hd},
+                          torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
+  const float scale = 1.0f / sqrtf((float)hd);
+  const int qblocks = (S + 127) / 128;
+  hipLaunchKernelGGL(attn_prefill_mfma_kernel, dim3(B * qblocks * H), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
+                     (const unsigned short*)vp.data_ptr(), (unsigned short*)out.data_ptr(),
+                     B, S, H, KVH, T32, (int)start_pos, q.stride(0), q.stride(1), scale);
+  return out;
+}
+
 torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
   CHK(a.is_cuda() && a.dtype() == torch::kBFloat16 && a.is_contiguous() && a.numel() == 16 * 32);
   CHK(b.is_cuda() && b.dtype() == torch::kBFloat16 && b.is_contiguous() && b.numel() == 32 * 16);
@@ -1469,6 +1681,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode, "GQA decode attention (flash-decoding split-KV)");
   m.def("attn_decode_mfma", &attn_decode_mfma,
         "GQA decode attention on matrix cores (packed cache, hd=128)");
+  m.def("attn_prefill_mfma", &attn_prefill_mfma,
+        "causal GQA prefill flash attention on matrix cores (packed cache, hd=128)");
   m.def("mfma16_probe", &mfma16_probe, "v_mfma_f32_16x16x32_bf16 layout probe (tests)");
   m.def("swiglu", &swiglu, "SwiGLU activation");
 }
