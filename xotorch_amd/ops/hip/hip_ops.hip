@@ -502,7 +502,7 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
   const int r0 = qb * 128 + wv * 32;
   const int col = lane & 15;
 
-  __shared__ unsigned short kv_lds[2][(4 + 8) * 512];  // K tile 4 KB + V tile 8 KB, dbuf
+  __shared__ unsigned short kv_lds[2][8192];  // K tile 8 KB + V tile 8 KB, dbuf
 
   // Q fragments for 2 sub-tiles x 4 hd-chunks; rows clamped to S-1 (writes
   // are predicated, extra rows only waste compute)
@@ -542,7 +542,7 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
     const unsigned short* ksrc = kp + kbase + (size_t)(tp * 2) * 2048;
     const unsigned short* vsrc = vp + vbase + (size_t)tp * 512;
     unsigned short* dk = kv_lds[buf];
-    unsigned short* dv = kv_lds[buf] + 4 * 512;
+    unsigned short* dv = kv_lds[buf] + 4096;
     const int tid = threadIdx.x;
     // K: 4096 elems = 8192 B: 256 threads x 2 x 16 B
 #pragma unroll
@@ -570,7 +570,7 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
     const int buf = tp & 1;
     if (tp + 1 < ntiles) stage(tp + 1, buf ^ 1);  // plain loads overlap compute
     const unsigned short* dk = kv_lds[buf];
-    const unsigned short* dv = kv_lds[buf] + 4 * 512;
+    const unsigned short* dv = kv_lds[buf] + 4096;
     unsigned short* plds = plds_all[wv];
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
