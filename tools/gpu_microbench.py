@@ -98,3 +98,27 @@ def attn_mfma_bench():
 if __name__ == "__main__":
   main()
   attn_mfma_bench()
+  prefill_attn_bench()
+
+
+def prefill_attn_bench():
+  import sys
+  sys.path.insert(0, ".")
+  from tests.test_kernels_gpu import _pack_k, _pack_v
+  import torch.nn.functional as F
+  B, S, H, KVH, hd = 64, 512, 64, 8, 128
+  t32 = (S + 31) // 32 * 32
+  q = torch.randn(B, S, H, hd, device="cuda").to(torch.bfloat16)
+  k = torch.randn(B, KVH, S, hd, device="cuda").to(torch.bfloat16)
+  v = torch.randn(B, KVH, S, hd, device="cuda").to(torch.bfloat16)
+  kp, vp = _pack_k(k, t32), _pack_v(v, t32)
+  def sdpa():
+    return F.scaled_dot_product_attention(q.transpose(1, 2), k, v, is_causal=True,
+                                          enable_gqa=True).transpose(1, 2).contiguous()
+  us_sdpa = timeit(sdpa, iters=20)
+  us_mfma = timeit(lambda: hip._load_hip().attn_prefill_mfma(q, kp, vp, 0) if hasattr(hip, "_load_hip") else None, iters=1) if False else 0
+  from xotorch_amd.ops import _hip_ops
+  us_mfma = timeit(lambda: _hip_ops.attn_prefill_mfma(q, kp, vp, 0), iters=20)
+  flops = B * H * S * (S / 2) * 4 * hd
+  print(f"prefill attn B{B} S{S}: sdpa {us_sdpa:7.1f} us {flops/us_sdpa/1e6:6.0f} TF"
+        f" | mfma {us_mfma:7.1f} us {flops/us_mfma/1e6:6.0f} TF")
