@@ -174,3 +174,39 @@ def test_elastic_repartition():
     await node.stop()
     return True
   assert run(go())
+
+
+def test_request_finishes_at_context_exhaustion():
+  """Generation hitting the model context must FINISH the request (deliver
+  what was produced) rather than silently dropping it — a dropped request
+  leaves every waiter hanging (reference behavior: crash past max_seq_len)."""
+  import asyncio
+  asyncio.run(_ctx_exhaustion())
+
+
+async def _ctx_exhaustion():
+  import asyncio
+  from xotorch_amd.engine.torch_engine import TorchEngine
+  from xotorch_amd.models.registry import build_base_shard
+  from xotorch_amd.orchestration.node import Node
+  from xotorch_amd.parallel.partitioning import RingMemoryWeightedPartitioningStrategy
+
+  eng = TorchEngine()
+  node = Node("ctx-test", None, eng, None, RingMemoryWeightedPartitioningStrategy(),
+              max_generate_tokens=100000)
+  await node.start(0)
+  try:
+    shard = build_base_shard("dummy", "TorchEngine")
+    done = asyncio.Event()
+    got = []
+    def on_token(rid, toks, fin):
+      got.extend(toks)
+      if fin:
+        done.set()
+    node.on_token.register("ctx").on_next(on_token)
+    await node.process_prompt(shard, "hi there", "ctx-req")
+    await asyncio.wait_for(done.wait(), timeout=120)
+    # dummy max_seq_len is 256; prompt ~8 tokens -> ~248 generated, never 100000
+    assert 0 < len(got) < 300
+  finally:
+    await node.stop()

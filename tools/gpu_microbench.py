@@ -116,7 +116,6 @@ def prefill_attn_bench():
     return F.scaled_dot_product_attention(q.transpose(1, 2), k, v, is_causal=True,
                                           enable_gqa=True).transpose(1, 2).contiguous()
   us_sdpa = timeit(sdpa, iters=20)
-  us_mfma = timeit(lambda: hip._load_hip().attn_prefill_mfma(q, kp, vp, 0) if hasattr(hip, "_load_hip") else None, iters=1) if False else 0
   from xotorch_amd.ops import _hip_ops
   us_mfma = timeit(lambda: _hip_ops.attn_prefill_mfma(q, kp, vp, 0), iters=20)
   flops = B * H * S * (S / 2) * 4 * hd

@@ -31,10 +31,27 @@ class DummyTokenizer:
     return "dummy " * max(1, len(tokens))
 
 
-async def resolve_tokenizer(model_path_or_repo: Union[str, Path, None]):
+async def resolve_tokenizer(model_path_or_repo: Union[str, Path, None], timeout: float = 20.0):
+  """Resolve a tokenizer without ever wedging the caller: local paths load
+  directly; hub repos are resolved in a worker thread with a timeout (an
+  air-gapped box can block for minutes inside DNS/TCP otherwise) and fall
+  back to the deterministic DummyTokenizer."""
+  import asyncio
+  import os
   if model_path_or_repo in (None, "dummy"):
     return DummyTokenizer()
-  return _resolve_tokenizer(model_path_or_repo)
+  p = Path(str(model_path_or_repo))
+  local = p.exists()
+  if not local:
+    os.environ.setdefault("HF_HUB_ETAG_TIMEOUT", "5")
+    os.environ.setdefault("HF_HUB_DOWNLOAD_TIMEOUT", "10")
+  try:
+    return await asyncio.wait_for(
+      asyncio.get_running_loop().run_in_executor(None, _resolve_tokenizer, model_path_or_repo),
+      timeout=None if local else timeout,
+    )
+  except (asyncio.TimeoutError, RuntimeError, Exception):
+    return DummyTokenizer()
 
 
 def _resolve_tokenizer(model_path_or_repo: Union[str, Path]):

@@ -40,6 +40,11 @@ TOP_K = int(os.getenv("XOT_TOP_K", "35"))
 DEFAULT_MAX_GEN = int(os.getenv("XOT_MAX_GEN", "512"))
 
 
+class ContextExhausted(RuntimeError):
+  """Generation reached the model context / KV capacity; the request must be
+  finished gracefully (the reference crashes past max_seq_len)."""
+
+
 @dataclass
 class Session:
   cache: ShardKVCache
@@ -186,6 +191,11 @@ class TorchEngine(InferenceEngine):
     # the same chunks in the same order, so the local counter is always the
     # chunk's start. (Position state never travels — SURVEY.md §2.4.)
     start_pos = sess.state.curr_pos
+    if start_pos + S > min(self.cfg.max_seq_len, sess.cache.capacity):
+      raise ContextExhausted(
+        f"request {request_id}: position {start_pos + S} exceeds context "
+        f"(max_seq_len {self.cfg.max_seq_len}, cache capacity {sess.cache.capacity})"
+      )
     x = x.to(self.device)
     if not is_tokens:
       x = x.to(self.dtype)

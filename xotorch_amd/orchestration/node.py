@@ -220,10 +220,17 @@ class Node:
       result, state = await self.inference_engine.infer_tensor(request_id, shard, tensor, inference_state)
       await self.process_inference_result(base_shard, result, request_id, state)
     except Exception as e:
+      # Fail LOUD and finish the request: a silently dropped request leaves
+      # every waiter (API stream, CLI) hanging until its own timeout.
       self.outstanding_requests.pop(request_id, None)
       if DEBUG >= 1:
         import traceback
         traceback.print_exc()
+      buffered, _ = self.buffered_token_output.get(request_id, ([], False))
+      self.buffered_token_output[request_id] = (buffered, True)
+      self.trigger_on_token_callbacks(request_id, [], True)
+      asyncio.create_task(self.broadcast_result(request_id, buffered[-16:], True))
+      await self.inference_engine.clear_session(request_id)
 
   async def process_inference_result(self, base_shard: Shard, result: np.ndarray, request_id: str,
                                      inference_state: Optional[dict] = None) -> None:
