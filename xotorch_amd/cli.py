@@ -185,11 +185,15 @@ async def daemon(args):
       loop.add_signal_handler(sig, stop.set)
     except NotImplementedError:
       pass
-  await stop.wait()
-  if viz:
-    viz.stop()
-  await api.stop()
-  await node.stop()
+  try:
+    await stop.wait()
+  except asyncio.CancelledError:
+    pass  # /quit cancels the main task; shut down cleanly below
+  finally:
+    if viz:
+      viz.stop()
+    await api.stop()
+    await node.stop()
 
 
 def serve_ring(args):
@@ -224,7 +228,10 @@ def run():
   elif args.command == "serve":
     serve_ring(args)
   else:
-    asyncio.run(daemon(args))
+    try:
+      asyncio.run(daemon(args))
+    except (KeyboardInterrupt, asyncio.CancelledError):
+      pass
 
 
 if __name__ == "__main__":
