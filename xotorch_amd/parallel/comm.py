@@ -33,7 +33,8 @@ def init_distributed(backend: Optional[str] = None) -> tuple[int, int]:
   dist.init_process_group(backend=backend, rank=rank, world_size=world,
                           timeout=datetime.timedelta(seconds=300))
   if backend == "nccl":
-    torch.cuda.set_device(int(os.getenv("LOCAL_RANK", str(rank % max(1, torch.cuda.device_count())))))
+    local = int(os.getenv("LOCAL_RANK", str(rank)))
+    torch.cuda.set_device(local % max(1, torch.cuda.device_count()))
   return rank, world
 
 
