@@ -555,10 +555,8 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
     for (int i = 0; i < 2; ++i) {
       const int e = tid + i * 256;           // 0..511 -> (group, piece)
       const int g = e >> 6, piece = e & 63;  // 64 pieces of 8 elems per group
-      *(ushort4_t*)(dv + g * 512 + piece * 8) =
-          *(const ushort4_t*)(vsrc + (size_t)g * (T32 >> 5) * 512 + piece * 8);
-      *(ushort4_t*)(dv + g * 512 + piece * 8 + 4) =
-          *(const ushort4_t*)(vsrc + (size_t)g * (T32 >> 5) * 512 + piece * 8 + 4);
+      *(ushort8*)(dv + g * 512 + piece * 8) =
+          *(const ushort8*)(vsrc + (size_t)g * (T32 >> 5) * 512 + piece * 8);
     }
   };
 
