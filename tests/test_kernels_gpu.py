@@ -418,3 +418,27 @@ def test_attn_prefill_mfma_strided_q(hip):
   out = _hip_ops.attn_prefill_mfma(q, kp, vp, 0).float()
   ref = torch_ref.attn_prefill(q.contiguous(), k, v, 0, S).float()
   assert torch.allclose(out, ref, atol=3e-2, rtol=3e-2), (out - ref).abs().max()
+
+
+def test_moe_fp8_matches_bf16(hip, monkeypatch):
+  """fp8 grouped expert path vs the bf16 routed path (W8A8 tolerance)."""
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.registry import builtin_config
+  from xotorch_amd.models.llama import MoEMLP
+  raw = dict(builtin_config("mixtral-8x7b"))
+  raw.update(hidden_size=256, intermediate_size=512)
+  cfg = config_from_hf(raw, "mixtral-fp8-tiny")
+  torch.manual_seed(3)
+  moe = MoEMLP(cfg).to("cuda").to(torch.bfloat16).eval()
+  x = (torch.randn(64, 1, 256, device="cuda") * 0.5).to(torch.bfloat16)
+  with torch.inference_mode():
+    y_bf16 = moe(x).float()
+    monkeypatch.setenv("XOT_FP8_GEMM", "1")
+    for e in moe.experts:
+      e.gate_up_proj.pack_decode()
+      e.down_proj.pack_decode()
+    moe.pack_grouped()
+    assert moe.wp_gate_up_fp8 is not None
+    y_fp8 = moe(x).float()
+  err = (y_bf16 - y_fp8).abs().max().item() / (y_bf16.abs().max().item() + 1e-9)
+  assert err < 0.15, err

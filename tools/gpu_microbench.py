@@ -103,7 +103,8 @@ if __name__ == "__main__":
 
 def prefill_attn_bench():
   import sys
-  sys.path.insert(0, ".")
+  from pathlib import Path
+  sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
   from tests.test_kernels_gpu import _pack_k, _pack_v
   import torch.nn.functional as F
   B, S, H, KVH, hd = 64, 512, 64, 8, 128
