@@ -50,8 +50,8 @@ async def resolve_tokenizer(model_path_or_repo: Union[str, Path, None], timeout:
       asyncio.get_running_loop().run_in_executor(None, _resolve_tokenizer, model_path_or_repo),
       timeout=None if local else timeout,
     )
-  except (asyncio.TimeoutError, RuntimeError, Exception):
-    return DummyTokenizer()
+  except Exception:  # timeout, no network, missing repo, bad config — all
+    return DummyTokenizer()  # degrade to the deterministic fallback
 
 
 def _resolve_tokenizer(model_path_or_repo: Union[str, Path]):
