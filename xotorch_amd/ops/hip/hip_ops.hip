@@ -1,25 +1,32 @@
-// xotorch_amd CDNA4 (gfx950 / MI355X) kernels for the per-token decode hot path.
+// xotorch_amd CDNA4 (gfx950 / MI355X) kernels for the transformer hot path.
 //
 // These replace the ops the reference delegates to torchtune
 // (SURVEY.md §2.2 op table; reference call sites cited per kernel below):
 //   - fused residual-add + RMSNorm      (llm_utils.py:465-476)
-//   - fused RoPE rotate + KV-cache append (general_mha.py:78-106, llm_utils.py:369-414)
-//   - GQA decode attention, flash-decoding style split-KV with online softmax
-//     (general_mha.py:211-215 — one query position vs the KV cache)
+//   - fused RoPE rotate + KV-cache append, dual layout (general_mha.py:78-106)
+//   - GQA attention, prefill AND decode, on matrix cores
+//     (general_mha.py:211-226): flash-forward / flash-decoding with online
+//     softmax, v_mfma_f32_16x16x32_bf16 for scores and PV, streaming an
+//     MFMA-fragment-packed KV cache with coalesced 1 KB wave loads
+//   - decode projections (qkv/o/gate_up/down/lm_head, general_mha.py:83-102,
+//     llm_utils.py:491-500): weight-streaming split-K GEMMs on
+//     v_mfma_f32_32x32x16_bf16 over prepacked weight fragments (non-temporal
+//     loads; a grouped variant runs every MoE expert in one launch, and a
+//     W8A8 e4m3 variant halves the stream) — auto-picked per shape against
+//     TunableOp-tuned hipBLASLt at runtime
 //   - SwiGLU activation                  (llm_utils.py:491-500)
 //
 // Design notes (per /opt/skills/guides/cdna_hip_programming.md):
 //   * wave = 64 lanes; all block sizes are multiples of 64
-//   * bf16 is loaded vectorized (ushort4/ushort8 = 8/16 B per lane — G13)
-//   * decode attention reads each KV row (hd=128 -> 256 B) with 16-lane
-//     groups x 16 B coalesced loads; one workgroup per (batch, kv-head,
-//     kv-split) computes ALL of that kv-head's query heads so KV traffic is
-//     paid once per kv-head, not once per q-head
+//   * bf16 is loaded vectorized (ushort8 = 16 B per lane); the big streamed
+//     operands (weights, KV) are pre-shuffled ONCE into MFMA fragment order
+//     so every hot-loop load is a coalesced wave-wide 1 KB stream
+//     (fragment-shaped 16 B gathers measured TA-bound at ~2 TB/s vs 5.5)
 //   * everything is launch-shape-static so the whole decode step can be
 //     captured in a hipGraph (lengths come from device tensors)
 //
-// Projections (QKV / O / MLP / LM head) stay on hipBLASLt through torch —
-// plain library GEMMs per the MI355X build rules.
+// Prefill projections (compute-bound) stay on hipBLASLt through torch —
+// plain library GEMMs per the MI355X build rules; measured ~1.6 PF there.
 
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
