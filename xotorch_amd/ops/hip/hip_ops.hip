@@ -874,12 +874,14 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 // Mixtral-class decode each expert e (blockIdx.y) computes its own
 // [C, N] = [C, K] @ Wp_e^T over a fixed per-expert token capacity C — one
 // launch covers every expert, weights stream from the stacked prepack.
-template <int MT, bool SPLIT>
+template <int MT, bool SPLIT, int BK>
 __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     const unsigned short* __restrict__ Wp, const unsigned short* __restrict__ X,
     unsigned short* __restrict__ Y, float* __restrict__ P,
     const unsigned short* __restrict__ bias,
     int N, long long K, int kc, int nsplit) {
+  constexpr int BKC = BK / 16;   // MFMA k-chunks (A fragments) per K-step
+  constexpr int XROW = BK + 8;   // padded LDS row stride (elems): conflict-free b128
   const int ntiles = N >> 7;
   const int tile = blockIdx.x % ntiles;
   const int split = blockIdx.x / ntiles;
@@ -895,7 +897,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   const int wv = threadIdx.x >> 6;
   const int tid = threadIdx.x;
 
-  __shared__ unsigned short xs[MT * 32 * 72];
+  __shared__ unsigned short xs[MT * 32 * XROW];
 
   floatx16 acc[MT];
 #pragma unroll
@@ -904,30 +906,36 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   // A stream: packed row for this wave's 32-n tile
   const int n32 = tile * 4 + wv;
   const unsigned short* wp = Wp + ((size_t)n32 * (K >> 4) + (k0 >> 4)) * 512 + (size_t)lane * 8;
-  // X stage: thread tid covers chunk (row r = c/8, 16B piece q = c%8) of each
-  // 32-row m-tile; global [M,K] row-major.
-  const int xr = tid >> 3, xq = tid & 7;
-  const unsigned short* xp = X + (size_t)xr * K + k0 + xq * 8;
-  const int xs_off = xr * 72 + xq * 8;  // padded LDS image
-
-  const int nsteps = (int)((k1 - k0) >> 6);
-  ushort8 xv[MT];
-  bf16x8 a_buf[2][4];
+  // X stage: thread tid covers 16 B pieces p = tid + i*256 (coalesced);
+  // row = p / (BK/8), piece-in-row q = p % (BK/8); global [M, K] row-major.
+  constexpr int PPT = MT * BK / 64;  // pieces per thread per step
+  const unsigned short* xp = X + k0;
+  const int nsteps = (int)((k1 - k0) >> (BK == 64 ? 6 : 7));
+  ushort8 xv[PPT];
+  bf16x8 a_buf[2][BKC];
 
   // prologue: stage step 0, preload A(0) and A(1).  The W stream is read
   // exactly once per launch -> non-temporal (L1-bypass) loads; depth-2
-  // prefetch keeps ~8 KB per wave in flight across the staging barriers.
+  // prefetch keeps 2*BK*32n*2B per wave in flight across staging barriers.
+  const unsigned short* wp1 = (nsteps > 1) ? wp + BKC * 512 : wp;  // clamp: no OOB at nsteps==1
 #pragma unroll
-  for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K);
-  const unsigned short* wp1 = (nsteps > 1) ? wp + 4 * 512 : wp;  // clamp: no OOB at nsteps==1
+  for (int i = 0; i < PPT; ++i) {
+    const int p = tid + i * 256;
+    const int row = p / (BK / 8), q = p % (BK / 8);
+    xv[i] = *(const ushort8*)(xp + (size_t)row * K + q * 8);
+  }
 #pragma unroll
-  for (int u = 0; u < 4; ++u) {
+  for (int u = 0; u < BKC; ++u) {
     a_buf[0][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
     a_buf[1][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp1 + u * 512));
   }
-  wp += 8 * 512;
+  wp += 2 * BKC * 512;
 #pragma unroll
-  for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];
+  for (int i = 0; i < PPT; ++i) {
+    const int p = tid + i * 256;
+    const int row = p / (BK / 8), q = p % (BK / 8);
+    *(ushort8*)(xs + row * XROW + q * 8) = xv[i];
+  }
   __syncthreads();
 
   // NOTE: the A double-buffer index must be a compile-time constant — a
@@ -937,27 +945,35 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     const bool last = (s == nsteps - 1);                                                       \
     if (!last) {                                                                               \
       _Pragma("unroll")                                                                        \
-      for (int t = 0; t < MT; ++t) xv[t] = *(const ushort8*)(xp + (size_t)t * 32 * K + (s + 1) * 64); \
+      for (int i = 0; i < PPT; ++i) {                                                          \
+        const int p = tid + i * 256;                                                           \
+        const int row = p / (BK / 8), q = p % (BK / 8);                                        \
+        xv[i] = *(const ushort8*)(xp + (size_t)row * K + (size_t)(s + 1) * BK + q * 8);        \
+      }                                                                                        \
     }                                                                                          \
     _Pragma("unroll")                                                                          \
-    for (int u = 0; u < 4; ++u) {                                                              \
+    for (int u = 0; u < BKC; ++u) {                                                            \
       _Pragma("unroll")                                                                        \
       for (int t = 0; t < MT; ++t) {                                                           \
         const bf16x8 b = *reinterpret_cast<const bf16x8*>(                                     \
-            xs + t * 32 * 72 + (lane & 31) * 72 + u * 16 + (lane >> 5) * 8);                   \
+            xs + t * 32 * XROW + (lane & 31) * XROW + u * 16 + (lane >> 5) * 8);               \
         acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_buf[BUF][u], b, acc[t], 0, 0, 0);   \
       }                                                                                        \
     }                                                                                          \
     if (s + 2 < nsteps) {                                                                      \
       _Pragma("unroll")                                                                        \
-      for (int u = 0; u < 4; ++u)                                                              \
+      for (int u = 0; u < BKC; ++u)                                                            \
         a_buf[BUF][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512)); \
-      wp += 4 * 512;                                                                           \
+      wp += BKC * 512;                                                                         \
     }                                                                                          \
     if (!last) {                                                                               \
       __syncthreads();                                                                         \
       _Pragma("unroll")                                                                        \
-      for (int t = 0; t < MT; ++t) *(ushort8*)(xs + t * 32 * 72 + xs_off) = xv[t];             \
+      for (int i = 0; i < PPT; ++i) {                                                          \
+        const int p = tid + i * 256;                                                           \
+        const int row = p / (BK / 8), q = p % (BK / 8);                                        \
+        *(ushort8*)(xs + row * XROW + q * 8) = xv[i];                                          \
+      }                                                                                        \
       __syncthreads();                                                                         \
     }                                                                                          \
   }
@@ -1005,7 +1021,6 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   }
 }
 
-// ---------------------------------------------------------------------------
 // fp8 (OCP e4m3) W8A8 variant: same structure as the bf16 packed kernel but
 // operands are e4m3 with per-output-channel weight scales s_w[n] and
 // per-token activation scales s_x[m] (dynamic, computed by quant_fp8_rows).
@@ -1525,17 +1540,30 @@ torch::Tensor skinny_gemm_packed(torch::Tensor x, torch::Tensor wp, int64_t N,
   while ((long long)kc * (nsplit - 1) >= K) nsplit--;
   auto stream = cur_stream();
   const int MT = (int)(M / 32);
+  // BK=128 halves the staging-barrier count per byte; needs 128-aligned splits
+  const bool bk128 = (K % 128 == 0) && (kc % 128 == 0) && kc >= 2048
+                     && getenv("XOT_SKINNY_BK64") == nullptr;
   const dim3 grid(ntiles * nsplit), block(256);
 #define SGP_CASE(MTV) \
   case MTV: \
     if (nsplit == 1) { \
-      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false>), grid, block, 0, stream, \
-                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
-                         (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit); \
+      if (bk128) \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false, 128>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit); \
+      else \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false, 64>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit); \
     } else { \
-      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true>), grid, block, 0, stream, \
-                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
-                         nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+      if (bk128) \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true, 128>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+      else \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true, 64>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
     } \
     break;
   if (nsplit == 1) {
@@ -1573,16 +1601,28 @@ torch::Tensor skinny_gemm_grouped(torch::Tensor x, torch::Tensor wp, int64_t E, 
   auto stream = cur_stream();
   const int MT = (int)(C / 32);
   const dim3 grid(ntiles * nsplit, (unsigned)E), block(256);
+  const bool bk128 = (K % 128 == 0) && (kc % 128 == 0) && kc >= 2048
+                     && getenv("XOT_SKINNY_BK64") == nullptr;
 #define SGG_CASE(MTV) \
   case MTV: \
     if (nsplit == 1) { \
-      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false>), grid, block, 0, stream, \
-                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
-                         (unsigned short*)y.data_ptr(), nullptr, nullptr, (int)N, K, kc, nsplit); \
+      if (bk128) \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false, 128>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           (unsigned short*)y.data_ptr(), nullptr, nullptr, (int)N, K, kc, nsplit); \
+      else \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false, 64>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           (unsigned short*)y.data_ptr(), nullptr, nullptr, (int)N, K, kc, nsplit); \
     } else { \
-      hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true>), grid, block, 0, stream, \
-                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
-                         nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+      if (bk128) \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true, 128>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+      else \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true, 64>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                           nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
     } \
     break;
   if (nsplit == 1) {
