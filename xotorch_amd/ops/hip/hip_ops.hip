@@ -909,21 +909,27 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   // X stage: thread tid covers 16 B pieces p = tid + i*256 (coalesced);
   // row = p / (BK/8), piece-in-row q = p % (BK/8); global [M, K] row-major.
   constexpr int PPT = MT * BK / 64;  // pieces per thread per step
-  const unsigned short* xp = X + k0;
   const int nsteps = (int)((k1 - k0) >> (BK == 64 ? 6 : 7));
   ushort8 xv[PPT];
   bf16x8 a_buf[2][BKC];
+  // per-thread staging addresses, strength-reduced (recomputing row/q per
+  // step costs ~10% on short-kc shapes)
+  const unsigned short* xpp[PPT];
+  int xso[PPT];
+#pragma unroll
+  for (int i = 0; i < PPT; ++i) {
+    const int p = tid + i * 256;
+    const int row = p / (BK / 8), q = p % (BK / 8);
+    xpp[i] = X + k0 + (size_t)row * K + q * 8;
+    xso[i] = row * XROW + q * 8;
+  }
 
   // prologue: stage step 0, preload A(0) and A(1).  The W stream is read
   // exactly once per launch -> non-temporal (L1-bypass) loads; depth-2
   // prefetch keeps 2*BK*32n*2B per wave in flight across staging barriers.
   const unsigned short* wp1 = (nsteps > 1) ? wp + BKC * 512 : wp;  // clamp: no OOB at nsteps==1
 #pragma unroll
-  for (int i = 0; i < PPT; ++i) {
-    const int p = tid + i * 256;
-    const int row = p / (BK / 8), q = p % (BK / 8);
-    xv[i] = *(const ushort8*)(xp + (size_t)row * K + q * 8);
-  }
+  for (int i = 0; i < PPT; ++i) xv[i] = *(const ushort8*)(xpp[i]);
 #pragma unroll
   for (int u = 0; u < BKC; ++u) {
     a_buf[0][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
@@ -931,11 +937,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   }
   wp += 2 * BKC * 512;
 #pragma unroll
-  for (int i = 0; i < PPT; ++i) {
-    const int p = tid + i * 256;
-    const int row = p / (BK / 8), q = p % (BK / 8);
-    *(ushort8*)(xs + row * XROW + q * 8) = xv[i];
-  }
+  for (int i = 0; i < PPT; ++i) *(ushort8*)(xs + xso[i]) = xv[i];
   __syncthreads();
 
   // NOTE: the A double-buffer index must be a compile-time constant — a
@@ -945,11 +947,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     const bool last = (s == nsteps - 1);                                                       \
     if (!last) {                                                                               \
       _Pragma("unroll")                                                                        \
-      for (int i = 0; i < PPT; ++i) {                                                          \
-        const int p = tid + i * 256;                                                           \
-        const int row = p / (BK / 8), q = p % (BK / 8);                                        \
-        xv[i] = *(const ushort8*)(xp + (size_t)row * K + (size_t)(s + 1) * BK + q * 8);        \
-      }                                                                                        \
+      for (int i = 0; i < PPT; ++i)                                                            \
+        xv[i] = *(const ushort8*)(xpp[i] + (size_t)(s + 1) * BK);                              \
     }                                                                                          \
     _Pragma("unroll")                                                                          \
     for (int u = 0; u < BKC; ++u) {                                                            \
@@ -969,11 +968,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     if (!last) {                                                                               \
       __syncthreads();                                                                         \
       _Pragma("unroll")                                                                        \
-      for (int i = 0; i < PPT; ++i) {                                                          \
-        const int p = tid + i * 256;                                                           \
-        const int row = p / (BK / 8), q = p % (BK / 8);                                        \
-        *(ushort8*)(xs + row * XROW + q * 8) = xv[i];                                          \
-      }                                                                                        \
+      for (int i = 0; i < PPT; ++i) *(ushort8*)(xs + xso[i]) = xv[i];                          \
       __syncthreads();                                                                         \
     }                                                                                          \
   }
