@@ -105,3 +105,24 @@ def test_tinychat_served():
     await node.stop()
     return True
   assert run(go())
+
+
+def test_max_tokens_enforced():
+  """`max_tokens` must cap generation per request (it travels in
+  inference_state and survives engine hops — engines merge, not replace,
+  the state dict)."""
+  async def go():
+    node, client = await make_client()
+    try:
+      resp = await client.post("/v1/chat/completions", json={
+        "model": "dummy", "messages": [{"role": "user", "content": "hello"}],
+        "max_tokens": 3,
+      })
+      j = await resp.json()
+      text = j["choices"][0]["message"]["content"]
+      # DummyTokenizer decodes each token as "dummy "
+      assert len(text.split()) <= 3, text
+    finally:
+      await client.close()
+      await node.stop()
+  run(go())

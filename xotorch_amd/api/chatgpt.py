@@ -204,7 +204,8 @@ class ChatGPTAPI:
     request_id = str(uuid.uuid4())
     self.token_queues[request_id] = asyncio.Queue()
     try:
-      await self.node.process_prompt(shard, prompt, request_id)
+      state = {"max_tokens": int(chat_request.max_tokens)} if chat_request.max_tokens else None
+      await self.node.process_prompt(shard, prompt, request_id, state)
       if stream:
         return await self._stream_response(request, request_id, model_id, tokenizer)
       return await self._full_response(request_id, model_id, tokenizer)

@@ -216,7 +216,11 @@ class TorchEngine(InferenceEngine):
     sess.state.curr_pos = new_pos
     sess.state.total_len = total_len
     out_np = out.float().cpu().numpy()
-    return out_np, {"curr_pos": new_pos, "total_len": total_len, "batch": B}
+    # merge pass-through metadata (max_tokens, traceparent, ...) so request
+    # attributes survive the hop — engines own only their position keys
+    out_state = dict(inference_state or {})
+    out_state.update({"curr_pos": new_pos, "total_len": total_len, "batch": B})
+    return out_np, out_state
 
   # ---------- training (capability the reference declared but never built) ----------
 

@@ -242,7 +242,11 @@ class Node:
       buffered, _ = self.buffered_token_output.setdefault(request_id, ([], False))
       buffered.append(tok)
       eos_id = getattr(getattr(self.inference_engine, "tokenizer", None), "eos_token_id", None)
-      is_finished = (eos_id is not None and tok == eos_id) or len(buffered) >= self.max_generate_tokens
+      limit = self.max_generate_tokens
+      req_max = (inference_state or {}).get("max_tokens")
+      if req_max:
+        limit = min(limit, int(req_max))  # per-request cap (API max_tokens)
+      is_finished = (eos_id is not None and tok == eos_id) or len(buffered) >= limit
       self.buffered_token_output[request_id] = (buffered, is_finished)
       tracer.handle_token(request_id, is_finished)
       self.trigger_on_token_callbacks(request_id, [tok], is_finished)
