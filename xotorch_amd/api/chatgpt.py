@@ -204,8 +204,12 @@ class ChatGPTAPI:
     request_id = str(uuid.uuid4())
     self.token_queues[request_id] = asyncio.Queue()
     try:
-      state = {"max_tokens": int(chat_request.max_tokens)} if chat_request.max_tokens else None
-      await self.node.process_prompt(shard, prompt, request_id, state)
+      state = {}
+      if chat_request.max_tokens:
+        state["max_tokens"] = int(chat_request.max_tokens)
+      if chat_request.temperature:
+        state["temperature"] = float(chat_request.temperature)
+      await self.node.process_prompt(shard, prompt, request_id, state or None)
       if stream:
         return await self._stream_response(request, request_id, model_id, tokenizer)
       return await self._full_response(request_id, model_id, tokenizer)

@@ -237,7 +237,8 @@ class Node:
     shard = self.get_current_shard(base_shard)
     if shard.is_last_layer:
       # sample a token from the logits
-      token = (await self.inference_engine.sample(result, temp=self.default_sample_temperature)).reshape(-1)
+      temp = (inference_state or {}).get("temperature", self.default_sample_temperature)
+      token = (await self.inference_engine.sample(result, temp=float(temp))).reshape(-1)
       tok = int(token[0])
       buffered, _ = self.buffered_token_output.setdefault(request_id, ([], False))
       buffered.append(tok)
