@@ -219,18 +219,22 @@ class Node:
     try:
       result, state = await self.inference_engine.infer_tensor(request_id, shard, tensor, inference_state)
       await self.process_inference_result(base_shard, result, request_id, state)
-    except Exception as e:
-      # Fail LOUD and finish the request: a silently dropped request leaves
-      # every waiter (API stream, CLI) hanging until its own timeout.
-      self.outstanding_requests.pop(request_id, None)
-      if DEBUG >= 1:
-        import traceback
-        traceback.print_exc()
-      buffered, _ = self.buffered_token_output.get(request_id, ([], False))
-      self.buffered_token_output[request_id] = (buffered, True)
-      self.trigger_on_token_callbacks(request_id, [], True)
-      asyncio.create_task(self.broadcast_result(request_id, buffered[-16:], True))
-      await self.inference_engine.clear_session(request_id)
+    except Exception:
+      await self._fail_request(request_id)
+
+  async def _fail_request(self, request_id: str) -> None:
+    """Fail LOUD and FINISH the request: a silently dropped request leaves
+    every waiter (API stream, CLI) hanging until its own timeout. Delivers
+    whatever was generated with is_finished=True."""
+    self.outstanding_requests.pop(request_id, None)
+    if DEBUG >= 1:
+      import traceback
+      traceback.print_exc()
+    buffered, _ = self.buffered_token_output.get(request_id, ([], False))
+    self.buffered_token_output[request_id] = (buffered, True)
+    self.trigger_on_token_callbacks(request_id, [], True)
+    asyncio.create_task(self.broadcast_result(request_id, buffered[-16:], True))
+    await self.inference_engine.clear_session(request_id)
 
   async def process_inference_result(self, base_shard: Shard, result: np.ndarray, request_id: str,
                                      inference_state: Optional[dict] = None) -> None:
@@ -280,8 +284,13 @@ class Node:
     peer = self._peer_by_index(target_index)
     if peer is None:
       await self.process_tensor(base_shard, tensor, request_id, inference_state)
-    else:
+      return
+    try:
       await peer.send_tensor(self.get_current_shard(base_shard, target_index), tensor, request_id, inference_state)
+    except Exception:
+      # peer died mid-request (this runs in a fire-and-forget task whose
+      # exception would otherwise vanish): finish the request loudly
+      await self._fail_request(request_id)
 
   # ---------------- training (ring forward/backward) ----------------
 
