@@ -210,3 +210,47 @@ async def _ctx_exhaustion():
     assert 0 < len(got) < 300
   finally:
     await node.stop()
+
+
+def test_concurrent_requests_interleave():
+  """Two in-flight requests on one node must both complete with isolated
+  per-request KV sessions (the reference serializes GPU work through a
+  1-thread executor but interleaves requests as async tasks)."""
+  import asyncio
+  asyncio.run(_concurrent())
+
+
+async def _concurrent():
+  import asyncio
+  from xotorch_amd.engine.torch_engine import TorchEngine
+  from xotorch_amd.models.registry import build_base_shard
+  from xotorch_amd.orchestration.node import Node
+  from xotorch_amd.parallel.partitioning import RingMemoryWeightedPartitioningStrategy
+
+  eng = TorchEngine()
+  node = Node("conc-test", None, eng, None, RingMemoryWeightedPartitioningStrategy(),
+              max_generate_tokens=24)
+  await node.start(0)
+  try:
+    shard = build_base_shard("dummy", "TorchEngine")
+    done = {"a": asyncio.Event(), "b": asyncio.Event()}
+    got = {"a": [], "b": []}
+
+    def on_token(rid, toks, fin):
+      if rid in got:
+        got[rid].extend(toks)
+        if fin:
+          done[rid].set()
+
+    node.on_token.register("conc").on_next(on_token)
+    await asyncio.gather(
+      node.process_prompt(shard, "first prompt", "a"),
+      node.process_prompt(shard, "a different second prompt", "b"),
+    )
+    await asyncio.wait_for(asyncio.gather(done["a"].wait(), done["b"].wait()), timeout=120)
+    assert len(got["a"]) == 24 and len(got["b"]) == 24
+    # isolated sessions: same greedy model, different prompts may share a
+    # prefix of tokens but the streams must be internally consistent
+    assert all(isinstance(t, int) for t in got["a"] + got["b"])
+  finally:
+    await node.stop()
