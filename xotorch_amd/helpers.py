@@ -171,3 +171,20 @@ async def shutdown(signal_name, loop, server=None):
     task.cancel()
   await asyncio.gather(*tasks, return_exceptions=True)
   loop.stop()
+
+
+def interface_priority(ifname: str) -> int:
+  """Rank a NIC for peer connections (reference helpers.py:284-315):
+  Thunderbolt/USB4 > Ethernet > WiFi > cellular > virtual/loopback."""
+  n = (ifname or "").lower()
+  if n.startswith(("tb", "thunderbolt", "usb4")):
+    return 5
+  if n.startswith(("en", "eth", "enp", "eno", "ens")):
+    return 4
+  if n.startswith(("wl", "wifi", "wlan")):
+    return 3
+  if n.startswith(("ww", "cell")):
+    return 2
+  if n.startswith(("lo", "docker", "veth", "br-", "virbr", "tun", "tap", "utun")):
+    return 1
+  return 2

@@ -159,9 +159,15 @@ class UDPDiscovery(Discovery):
     return [h for h, _, _ in self.known_peers.values()]
 
   async def _broadcast_loop(self):
+    from xotorch_amd.helpers import interface_priority
+    try:
+      ifname = next((n for n in __import__("os").listdir("/sys/class/net") if not n.startswith("lo")), "eth0")
+    except Exception:
+      ifname = "eth0"
     msg = json.dumps({
       "type": "discovery", "node_id": self.node_id, "node_port": self.node_port,
       "device_capabilities": self.device_caps.to_dict(),
+      "interface_name": ifname, "interface_priority": interface_priority(ifname),
     }).encode()
     sock = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
     sock.setsockopt(socket.SOL_SOCKET, socket.SO_BROADCAST, 1)
@@ -208,13 +214,25 @@ class UDPDiscovery(Discovery):
     if self.allowed_node_ids and pid not in self.allowed_node_ids:
       return
     now = time.time()
+    prio = int(msg.get("interface_priority", 2))
     if pid in self.known_peers:
       handle, first, _ = self.known_peers[pid]
+      # a higher-priority interface replaces the stored connection
+      # (reference udp_discovery.py:180-186)
+      new_addr = f"{addr[0]}:{msg.get('node_port')}"
+      if prio > getattr(handle, "_iface_priority", 2) and new_addr != handle.addr():
+        caps = DeviceCapabilities.from_dict(msg.get("device_capabilities", {}))
+        replacement = self.create_peer_handle(pid, new_addr, "udp", caps)
+        replacement._iface_priority = prio
+        if await replacement.health_check():
+          self.known_peers[pid] = (replacement, first, now)
+          return
       self.known_peers[pid] = (handle, first, now)
       return
     caps = DeviceCapabilities.from_dict(msg.get("device_capabilities", {}))
     peer_addr = f"{addr[0]}:{msg.get('node_port')}"
     handle = self.create_peer_handle(pid, peer_addr, "udp", caps)
+    handle._iface_priority = prio
     if await handle.health_check():
       if DEBUG_DISCOVERY >= 1:
         print(f"discovered peer {pid} at {peer_addr}")

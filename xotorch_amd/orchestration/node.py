@@ -378,6 +378,25 @@ class Node:
   def _on_opaque_status(self, request_id: str, status: str) -> None:
     if DEBUG >= 2:
       print(f"[status] {request_id}: {status[:120]}")
+    # Preemptive shard load (reference main.py:201-212): when a peer gossips
+    # start_process_prompt, begin loading OUR shard so the ring doesn't stall
+    # on model load when the first hidden state arrives.
+    try:
+      msg = json.loads(status)
+    except Exception:
+      return
+    if msg.get("type") == "node_status" and msg.get("status") == "start_process_prompt" \
+       and msg.get("node_id") != self.id and msg.get("base_shard"):
+      base = Shard.from_dict(msg["base_shard"])
+      my_shard = self.get_current_shard(base)
+      asyncio.create_task(self._preload_shard(my_shard))
+
+  async def _preload_shard(self, shard: Shard) -> None:
+    try:
+      await self.inference_engine.ensure_shard(shard)
+    except Exception as e:
+      if DEBUG >= 1:
+        print(f"preemptive shard load failed: {e}")
 
   @property
   def current_topology(self) -> Topology:
