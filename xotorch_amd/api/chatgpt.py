@@ -105,10 +105,19 @@ class ChatGPTAPI:
       self.app.router.add_get("/", self.handle_root)
       self.app.router.add_static("/static/", static_dir, name="static")
     self.app.middlewares.append(self._timeout_middleware)
+    self.app.middlewares.append(self._log_middleware)
     # node token plumbing
     if node is not None:
       node.on_token.register("chatgpt-api-token-handler").on_next(self._on_token)
     self._runner = None
+
+  @web.middleware
+  async def _log_middleware(self, request, handler):
+    # request log (reference chatgpt_api.py:255-260), DEBUG-gated
+    from xotorch_amd.helpers import DEBUG
+    if DEBUG >= 2:
+      print(f"[api] {request.method} {request.path}")
+    return await handler(request)
 
   @web.middleware
   async def _timeout_middleware(self, request, handler):
