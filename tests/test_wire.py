@@ -1,0 +1,71 @@
+"""Wire-protocol unit tests: framing, tensor round-trip, server dispatch
+edges (reference surface: networking/grpc/node_service.proto + grpc_server)."""
+import asyncio
+
+import numpy as np
+import pytest
+
+from xotorch_amd.orchestration import wire
+
+
+def test_tensor_roundtrip():
+  for arr in (np.random.randn(3, 5).astype(np.float32),
+              np.random.randn(2, 1, 4).astype(np.float16),
+              np.arange(7, dtype=np.int64)):
+    d = wire.pack_tensor(arr)
+    back = wire.unpack_tensor(d)
+    assert back.dtype == arr.dtype and back.shape == arr.shape
+    assert np.array_equal(back, arr)
+  assert wire.pack_tensor(None) is None
+  assert wire.unpack_tensor(None) is None
+
+
+def test_frame_roundtrip_and_limits():
+  async def go():
+    server_got = []
+
+    async def handle(reader, writer):
+      msg = await wire.read_frame(reader)
+      server_got.append(msg)
+      wire.write_frame(writer, {"ok": True, "echo": msg["x"]})
+      await writer.drain()
+      writer.close()
+
+    srv = await asyncio.start_server(handle, "127.0.0.1", 0)
+    port = srv.sockets[0].getsockname()[1]
+    reply = await wire.request("127.0.0.1", port, {"x": 41, "blob": b"\x00" * 1024})
+    assert reply == {"ok": True, "echo": 41}
+    assert server_got[0]["blob"] == b"\x00" * 1024
+    srv.close()
+    await srv.wait_closed()
+  asyncio.run(go())
+
+
+def test_oversize_frame_rejected():
+  async def go():
+    async def handle(reader, writer):
+      # advertise an over-cap frame length; client must refuse to read it
+      import struct
+      writer.write(struct.pack("!I", wire.MAX_FRAME + 1))
+      await writer.drain()
+      await asyncio.sleep(0.2)
+      writer.close()
+
+    srv = await asyncio.start_server(handle, "127.0.0.1", 0)
+    port = srv.sockets[0].getsockname()[1]
+    reader, writer = await asyncio.open_connection("127.0.0.1", port)
+    with pytest.raises(ValueError):
+      await wire.read_frame(reader)
+    writer.close()
+    srv.close()
+    await srv.wait_closed()
+  asyncio.run(go())
+
+
+def test_server_unknown_type():
+  async def go():
+    from xotorch_amd.orchestration.server import Server
+    srv = Server(node=None, host="127.0.0.1", port=0)
+    reply = await srv._dispatch({"type": "nonsense"})
+    assert reply["ok"] is False and "unknown" in reply["error"]
+  asyncio.run(go())
