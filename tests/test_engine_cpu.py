@@ -117,3 +117,36 @@ def test_dummy_engine_plumbing():
     assert tok.shape[0] == 1
     return True
   assert run(go())
+
+
+def test_sessions_cleaned_after_finish_and_error():
+  """Per-request KV sessions must not leak: cleared on normal finish and on
+  context exhaustion (288 GB fills fast if sessions accumulate)."""
+  import asyncio
+
+  async def go():
+    from xotorch_amd.engine.torch_engine import TorchEngine
+    from xotorch_amd.models.registry import build_base_shard
+    from xotorch_amd.orchestration.node import Node
+    eng = TorchEngine()
+    node = Node("sess-test", None, eng, None, max_generate_tokens=6)
+    await node.start(0)
+    try:
+      shard = build_base_shard("dummy", "TorchEngine")
+      done = asyncio.Event()
+      node.on_token.register("s").on_next(lambda rid, t, fin: done.set() if fin else None)
+      await node.process_prompt(shard, "hello", "sess-1")
+      await asyncio.wait_for(done.wait(), 60)
+      await asyncio.sleep(0.1)  # let the cleanup task run
+      assert "sess-1" not in eng.sessions, list(eng.sessions)
+      # context-exhaustion path also cleans up
+      node.max_generate_tokens = 100000
+      done2 = asyncio.Event()
+      node.on_token.register("s2").on_next(lambda rid, t, fin: done2.set() if fin else None)
+      await node.process_prompt(shard, "again", "sess-2")
+      await asyncio.wait_for(done2.wait(), 120)
+      await asyncio.sleep(0.1)
+      assert "sess-2" not in eng.sessions, list(eng.sessions)
+    finally:
+      await node.stop()
+  asyncio.run(go())
