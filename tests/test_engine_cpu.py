@@ -150,3 +150,23 @@ def test_sessions_cleaned_after_finish_and_error():
     finally:
       await node.stop()
   asyncio.run(go())
+
+
+def test_engine_preserves_passthrough_state():
+  """Engines must MERGE (not replace) inference_state: request metadata like
+  max_tokens / traceparent has to survive every ring hop."""
+  import asyncio
+
+  async def go():
+    from xotorch_amd.engine.torch_engine import TorchEngine
+    from xotorch_amd.models.registry import build_full_shard
+    import numpy as np
+    eng = TorchEngine()
+    shard = build_full_shard("dummy", "TorchEngine")
+    toks = np.array([[1, 2, 3]], dtype=np.int64)
+    state_in = {"max_tokens": 7, "traceparent": "00-abc-def-01"}
+    _, state_out = await eng.infer_tensor("pt-1", shard, toks, state_in)
+    assert state_out["max_tokens"] == 7
+    assert state_out["traceparent"] == "00-abc-def-01"
+    assert state_out["curr_pos"] == 3
+  asyncio.run(go())
