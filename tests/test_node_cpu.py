@@ -254,3 +254,35 @@ async def _concurrent():
     assert all(isinstance(t, int) for t in got["a"] + got["b"])
   finally:
     await node.stop()
+
+
+def test_coordinate_save_roundtrip(tmp_path):
+  """Node.coordinate_save writes the shard checkpoint; load restores weights
+  (reference declares coordinate_save but its engines' save/load are no-ops)."""
+  import asyncio
+
+  async def go():
+    import torch
+    from xotorch_amd.engine.torch_engine import TorchEngine
+    from xotorch_amd.models.registry import build_base_shard
+    from xotorch_amd.orchestration.node import Node
+    eng = TorchEngine()
+    node = Node("ckpt-test", None, eng, None)
+    await node.start(0)
+    try:
+      shard = build_base_shard("dummy", "TorchEngine")
+      await eng.ensure_shard(node.get_current_shard(shard))
+      await node.coordinate_save(shard, iteration=3, destination=str(tmp_path))
+      files = list(tmp_path.rglob("*.safetensors"))
+      assert len(files) == 1 and "-3.safetensors" in files[0].name
+      # perturb a weight, then load restores it
+      w = eng.model.layers["0"].self_attn.qkv_proj.weight
+      orig = w.detach().clone()
+      with torch.no_grad():
+        w.add_(1.0)
+      await eng.load_checkpoint(node.get_current_shard(shard), str(files[0]))
+      assert torch.allclose(eng.model.layers["0"].self_attn.qkv_proj.weight, orig)
+      assert node.checkpoint_iters["dummy"] == 3
+    finally:
+      await node.stop()
+  asyncio.run(go())
