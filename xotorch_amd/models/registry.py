@@ -86,6 +86,17 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
   "deepseek-r1-distill-qwen-32b": _qwen_cfg(5120, 64, 40, 8, 27648),
   "deepseek-r1-distill-llama-8b": _llama_cfg(4096, 32, 32, 8, 14336, llama3_scaling=True),
   "deepseek-r1-distill-llama-70b": _llama_cfg(8192, 80, 64, 8, 28672, llama3_scaling=True),
+  # qwen 2.5 coder / math (identical architectures to the base sizes)
+  "qwen-2.5-coder-1.5b": _qwen_cfg(1536, 28, 12, 2, 8960, tie=True),
+  "qwen-2.5-coder-3b": _qwen_cfg(2048, 36, 16, 2, 11008, tie=True),
+  "qwen-2.5-coder-7b": _qwen_cfg(3584, 28, 28, 4, 18944),
+  "qwen-2.5-coder-14b": _qwen_cfg(5120, 48, 40, 8, 13824),
+  "qwen-2.5-coder-32b": _qwen_cfg(5120, 64, 40, 8, 27648),
+  "qwen-2.5-math-7b": _qwen_cfg(3584, 28, 28, 4, 18944),
+  "qwen-2.5-math-72b": _qwen_cfg(8192, 80, 64, 8, 29568),
+  "deepseek-r1-distill-qwen-14b": _qwen_cfg(5120, 48, 40, 8, 13824),
+  # nemotron (llama-3.1-70b architecture)
+  "nemotron-70b": _llama_cfg(8192, 80, 64, 8, 28672, llama3_scaling=True),
   # moe
   "mixtral-8x7b": _mixtral_cfg(4096, 32, 32, 8, 14336),
   # phi-4-mini (llama-like enough for the generic decoder)
@@ -121,6 +132,27 @@ model_cards: Dict[str, dict] = {
   "deepseek-r1-distill-llama-70b": {"layers": 80, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Llama-70B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Llama-70B"}},
   "mixtral-8x7b": {"layers": 32, "repo": {"TorchEngine": "mistralai/Mixtral-8x7B-Instruct-v0.1", "HIPEngine": "mistralai/Mixtral-8x7B-Instruct-v0.1"}},
   "phi-4-mini": {"layers": 32, "repo": {"TorchEngine": "microsoft/Phi-4-mini-instruct", "HIPEngine": "microsoft/Phi-4-mini-instruct"}},
+  "qwen-2.5-coder-1.5b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen2.5-Coder-1.5B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Coder-1.5B-Instruct"}},
+  "qwen-2.5-coder-3b": {"layers": 36, "repo": {"TorchEngine": "Qwen/Qwen2.5-Coder-3B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Coder-3B-Instruct"}},
+  "qwen-2.5-coder-7b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen2.5-Coder-7B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Coder-7B-Instruct"}},
+  "qwen-2.5-coder-14b": {"layers": 48, "repo": {"TorchEngine": "Qwen/Qwen2.5-Coder-14B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Coder-14B-Instruct"}},
+  "qwen-2.5-coder-32b": {"layers": 64, "repo": {"TorchEngine": "Qwen/Qwen2.5-Coder-32B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Coder-32B-Instruct"}},
+  "qwen-2.5-math-7b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen2.5-Math-7B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Math-7B-Instruct"}},
+  "qwen-2.5-math-72b": {"layers": 80, "repo": {"TorchEngine": "Qwen/Qwen2.5-Math-72B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Math-72B-Instruct"}},
+  "deepseek-r1-distill-qwen-14b": {"layers": 48, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-14B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-14B"}},
+  "nemotron-70b": {"layers": 80, "repo": {"TorchEngine": "nvidia/Llama-3.1-Nemotron-70B-Instruct-HF", "HIPEngine": "nvidia/Llama-3.1-Nemotron-70B-Instruct-HF"}},
+  # Listed for registry parity with the reference but with NO supported engine
+  # here: these architectures (qwen3 qk-norm, gemma2 softcap/SWA, deepseek
+  # MLA+MoE, llava vision) are not implemented by this decoder — the
+  # reference lists them too, but its torchtune GQA assembly cannot run them
+  # either. get_supported_models() filters them out.
+  "qwen-3-32b": {"layers": 64, "repo": {}},
+  "qwen-3-30b-a3b": {"layers": 48, "repo": {}},
+  "gemma2-9b": {"layers": 42, "repo": {}},
+  "gemma2-27b": {"layers": 46, "repo": {}},
+  "deepseek-r1": {"layers": 61, "repo": {}},
+  "deepseek-v3": {"layers": 61, "repo": {}},
+  "llava-1.5-7b-hf": {"layers": 32, "repo": {}},
   "dummy": {"layers": 4, "repo": {"TorchEngine": "dummy", "HIPEngine": "dummy", "DummyEngine": "dummy"}},
 }
 
@@ -148,6 +180,22 @@ pretty_names = {
   "deepseek-r1-distill-llama-8b": "DeepSeek R1 Distill Llama 8B",
   "deepseek-r1-distill-llama-70b": "DeepSeek R1 Distill Llama 70B",
   "mixtral-8x7b": "Mixtral 8x7B",
+  "qwen-2.5-coder-1.5b": "Qwen 2.5 Coder 1.5B",
+  "qwen-2.5-coder-3b": "Qwen 2.5 Coder 3B",
+  "qwen-2.5-coder-7b": "Qwen 2.5 Coder 7B",
+  "qwen-2.5-coder-14b": "Qwen 2.5 Coder 14B",
+  "qwen-2.5-coder-32b": "Qwen 2.5 Coder 32B",
+  "qwen-2.5-math-7b": "Qwen 2.5 Math 7B",
+  "qwen-2.5-math-72b": "Qwen 2.5 Math 72B",
+  "deepseek-r1-distill-qwen-14b": "DeepSeek R1 Distill Qwen 14B",
+  "nemotron-70b": "Nemotron 70B",
+  "qwen-3-32b": "Qwen 3 32B (unsupported arch)",
+  "qwen-3-30b-a3b": "Qwen 3 30B A3B (unsupported arch)",
+  "gemma2-9b": "Gemma2 9B (unsupported arch)",
+  "gemma2-27b": "Gemma2 27B (unsupported arch)",
+  "deepseek-r1": "DeepSeek R1 (unsupported arch)",
+  "deepseek-v3": "DeepSeek V3 (unsupported arch)",
+  "llava-1.5-7b-hf": "LLaVa 1.5 7B (unsupported arch)",
   "phi-4-mini": "Phi-4 Mini",
   "dummy": "Dummy (test)",
 }
