@@ -442,3 +442,25 @@ def test_moe_fp8_matches_bf16(hip, monkeypatch):
     y_fp8 = moe(x).float()
   err = (y_bf16 - y_fp8).abs().max().item() / (y_bf16.abs().max().item() + 1e-9)
   assert err < 0.15, err
+
+
+@pytest.mark.parametrize("hd,H,KVH,S", [(128, 8, 2, 4), (64, 4, 2, 1)])
+def test_rope_qkv_append_qk_norm(hip, hd, H, KVH, S):
+  """Fused qwen3 per-head q/k RMSNorm inside the rope+append kernel."""
+  from xotorch_amd.ops import torch_ref
+  B, T, start = 2, 32, 5
+  qkv = bt(B, S, (H + 2 * KVH) * hd, seed=101)
+  cos, sin = torch_ref.rope_cos_sin(hd, T, 1000000.0, device="cuda")
+  kc = torch.zeros(B, KVH, T, hd, dtype=torch.bfloat16, device="cuda")
+  vc = torch.zeros_like(kc)
+  kc_ref, vc_ref = kc.clone(), vc.clone()
+  qn = (torch.randn(hd, device="cuda") * 0.2 + 1.0).to(torch.bfloat16)
+  kn = (torch.randn(hd, device="cuda") * 0.2 + 1.0).to(torch.bfloat16)
+  positions = torch.arange(start, start + S, dtype=torch.int32, device="cuda")
+  qkv_ref = qkv.clone()
+  torch_ref.rope_qkv_append(qkv_ref, cos, sin, positions, kc_ref, vc_ref, H, KVH, hd, qn, kn, 1e-6)
+  hip.rope_qkv_append(qkv, cos, sin, positions, kc, vc, H, KVH, hd, None, None, qn, kn, 1e-6)
+  assert torch.allclose(qkv.float(), qkv_ref.float(), atol=3e-2, rtol=3e-2), \
+    (qkv.float() - qkv_ref.float()).abs().max()
+  assert torch.allclose(kc.float(), kc_ref.float(), atol=3e-2, rtol=3e-2)
+  assert torch.equal(vc, vc_ref)

@@ -182,3 +182,53 @@ def test_moe_capacity_is_lossless():
     ref = moe(x)
     routed = moe._forward_decode(x.view(-1, 32)).view(64, 1, 32).to(x.dtype)
   assert torch.allclose(ref, routed, atol=1e-4, rtol=1e-4)
+
+
+def test_qwen3_qk_norm_model():
+  """qwen3 family: per-head q/k RMSNorm before RoPE, no attn bias, explicit
+  head_dim; full-vs-split logits equality must hold with qk-norm active."""
+  import torch
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.llama import ShardedModel, hf_key_map
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+
+  tiny = {
+    "model_type": "qwen3", "hidden_size": 128, "num_hidden_layers": 4,
+    "num_attention_heads": 4, "num_key_value_heads": 2, "head_dim": 32,
+    "intermediate_size": 256, "vocab_size": 151, "rope_theta": 1000000.0,
+    "rms_norm_eps": 1e-6, "max_position_embeddings": 64, "torch_dtype": "float32",
+  }
+  cfg = config_from_hf(tiny, "qwen3-tiny")
+  assert cfg.qk_norm and not cfg.attn_bias and cfg.head_dim == 32
+  full = Shard("qwen3-tiny", 0, 3, 4)
+  m = ShardedModel(cfg, full).float()
+  random_init(m)
+  # perturb the norms so they actually do something
+  with torch.no_grad():
+    for lid in m.local_layer_ids:
+      m.layers[str(lid)].self_attn.q_norm.mul_(1.5)
+      m.layers[str(lid)].self_attn.k_norm.mul_(0.8)
+  m.eval()
+  mapping = hf_key_map(full, cfg)
+  assert "model.layers.0.self_attn.q_norm.weight" in mapping
+  B, S = 2, 10
+  toks = torch.randint(0, 151, (B, S))
+  cache_f = ShardKVCache(4, B, 2, S + 2, 32, torch.float32, "cpu")
+  with torch.inference_mode():
+    pos = torch.arange(S)
+    lf = m(toks, caches=cache_f.caches, positions=pos, start_pos=0)
+    assert torch.isfinite(lf).all()
+    # split in two shards, run sequentially, compare logits
+    s0, s1 = Shard("qwen3-tiny", 0, 1, 4), Shard("qwen3-tiny", 2, 3, 4)
+    m0, m1 = ShardedModel(cfg, s0).float(), ShardedModel(cfg, s1).float()
+    sd = m.state_dict()
+    m0.load_state_dict({k: v for k, v in sd.items() if k in m0.state_dict()})
+    m1.load_state_dict({k: v for k, v in sd.items() if k in m1.state_dict()})
+    m0.eval(); m1.eval()
+    c0 = ShardKVCache(2, B, 2, S + 2, 32, torch.float32, "cpu")
+    c1 = ShardKVCache(2, B, 2, S + 2, 32, torch.float32, "cpu")
+    h = m0(toks, caches=c0.caches, positions=pos, start_pos=0)
+    ls = m1(h, caches=c1.caches, positions=pos, start_pos=0)
+    assert torch.equal(lf, ls), (lf - ls).abs().max()

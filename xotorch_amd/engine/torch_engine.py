@@ -262,6 +262,9 @@ class TorchEngine(InferenceEngine):
       q = q.reshape(B, S, cfgm.n_heads, cfgm.head_dim)
       k = k.reshape(B, S, cfgm.n_kv_heads, cfgm.head_dim)
       v = v.reshape(B, S, cfgm.n_kv_heads, cfgm.head_dim)
+      if cfgm.qk_norm:  # qwen3 per-head q/k RMSNorm (training path)
+        q = tr.rmsnorm(q.float(), attn.q_norm.float(), cfgm.norm_eps).to(q.dtype)
+        k = tr.rmsnorm(k.float(), attn.k_norm.float(), cfgm.norm_eps).to(k.dtype)
       q, k = tr.rope_apply(q, k, cos, sin, positions)
       rep = cfgm.n_heads // cfgm.n_kv_heads
       out = torch.nn.functional.scaled_dot_product_attention(

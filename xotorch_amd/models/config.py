@@ -50,6 +50,7 @@ class ModelConfig:
   max_seq_len: int = 8192
   tie_word_embeddings: bool = False
   attn_bias: bool = False  # qwen2 q/k/v bias
+  qk_norm: bool = False    # qwen3 per-head RMSNorm on q/k before RoPE
   torch_dtype: torch.dtype = torch.bfloat16
   bos_token_id: Optional[int] = None
   eos_token_id: Optional[int] = None
@@ -120,6 +121,7 @@ def config_from_hf(config_path: Path | str | dict, model_id: str = "unknown") ->
     max_seq_len=max_seq_len,
     tie_word_embeddings=raw.get("tie_word_embeddings", False),
     attn_bias=mtype in ("qwen2",) or raw.get("attention_bias", False),
+    qk_norm=mtype in ("qwen3", "qwen3_moe"),
     torch_dtype=_DTYPE_MAP.get(raw.get("torch_dtype", "bfloat16"), torch.bfloat16),
     bos_token_id=raw.get("bos_token_id"),
     eos_token_id=eos,

@@ -146,6 +146,9 @@ class Attention(nn.Module):
     H, KVH, hd, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim, cfg.dim
     self.qkv_proj = XotLinear(D, (H + 2 * KVH) * hd, bias=cfg.attn_bias)
     self.o_proj = XotLinear(H * hd, D, bias=False)
+    if cfg.qk_norm:  # qwen3: per-head RMSNorm on q/k before RoPE
+      self.q_norm = nn.Parameter(torch.ones(hd))
+      self.k_norm = nn.Parameter(torch.ones(hd))
 
   def forward(self, x, cos, sin, positions, kv, start_pos: int, is_decode: bool, seq_lens=None):
     B, S, _ = x.shape
@@ -155,7 +158,10 @@ class Attention(nn.Module):
     kp = kv[2] if len(kv) > 2 else None  # MFMA-packed cache copies (GPU, hd=128)
     vp = kv[3] if len(kv) > 3 else None
     qkv = self.qkv_proj(x)  # [B, S, (H+2KVH)*hd]
-    ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, cfg.n_kv_heads, hd, kp, vp)
+    qn = self.q_norm if cfg.qk_norm else None
+    kn = self.k_norm if cfg.qk_norm else None
+    ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, cfg.n_kv_heads, hd, kp, vp,
+                        qn, kn, cfg.norm_eps)
     q = qkv[:, :, : H * hd].view(B, S, H, hd)  # strided view; kernels accept it
     if is_decode:
       sl = seq_lens if seq_lens is not None else start_pos + 1
@@ -487,6 +493,9 @@ def hf_key_map(shard: Shard, cfg: ModelConfig):
       mapping[hf + f"self_attn.k_proj.{suffix}"] = ours + f"self_attn.qkv_proj.{suffix}#1"
       mapping[hf + f"self_attn.v_proj.{suffix}"] = ours + f"self_attn.qkv_proj.{suffix}#2"
     mapping[hf + "self_attn.o_proj.weight"] = ours + "self_attn.o_proj.weight"
+    if cfg.qk_norm:
+      mapping[hf + "self_attn.q_norm.weight"] = ours + "self_attn.q_norm"
+      mapping[hf + "self_attn.k_norm.weight"] = ours + "self_attn.k_norm"
     mapping[hf + "input_layernorm.weight"] = ours + "input_layernorm.weight"
     mapping[hf + "post_attention_layernorm.weight"] = ours + "post_attention_layernorm.weight"
     if cfg.n_experts > 0:

@@ -48,6 +48,17 @@ def _qwen_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=151936, tie=False, max_
   }
 
 
+def _qwen3_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=151936, tie=False, max_pos=40960, head_dim=128):
+  return {
+    "model_type": "qwen3", "hidden_size": dim, "num_hidden_layers": n_layers,
+    "num_attention_heads": n_heads, "num_key_value_heads": n_kv,
+    "head_dim": head_dim, "intermediate_size": inter, "vocab_size": vocab,
+    "rope_theta": 1000000.0, "rms_norm_eps": 1e-6,
+    "max_position_embeddings": max_pos, "tie_word_embeddings": tie,
+    "torch_dtype": "bfloat16", "bos_token_id": 151643, "eos_token_id": 151645,
+  }
+
+
 def _mixtral_cfg(dim, n_layers, n_heads, n_kv, inter, n_experts=8, top_k=2):
   return {
     "model_type": "mixtral", "hidden_size": dim, "num_hidden_layers": n_layers,
@@ -86,6 +97,10 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
   "deepseek-r1-distill-qwen-32b": _qwen_cfg(5120, 64, 40, 8, 27648),
   "deepseek-r1-distill-llama-8b": _llama_cfg(4096, 32, 32, 8, 14336, llama3_scaling=True),
   "deepseek-r1-distill-llama-70b": _llama_cfg(8192, 80, 64, 8, 28672, llama3_scaling=True),
+  # qwen 3 dense (per-head q/k RMSNorm, no attn bias, explicit head_dim)
+  "qwen-3-0.6b": _qwen3_cfg(1024, 28, 16, 8, 3072, tie=True),
+  "qwen-3-8b": _qwen3_cfg(4096, 36, 32, 8, 12288),
+  "qwen-3-32b": _qwen3_cfg(5120, 64, 64, 8, 25600),
   # qwen 2.5 coder / math (identical architectures to the base sizes)
   "qwen-2.5-coder-1.5b": _qwen_cfg(1536, 28, 12, 2, 8960, tie=True),
   "qwen-2.5-coder-3b": _qwen_cfg(2048, 36, 16, 2, 11008, tie=True),
@@ -146,7 +161,9 @@ model_cards: Dict[str, dict] = {
   # MLA+MoE, llava vision) are not implemented by this decoder — the
   # reference lists them too, but its torchtune GQA assembly cannot run them
   # either. get_supported_models() filters them out.
-  "qwen-3-32b": {"layers": 64, "repo": {}},
+  "qwen-3-0.6b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen3-0.6B", "HIPEngine": "Qwen/Qwen3-0.6B"}},
+  "qwen-3-8b": {"layers": 36, "repo": {"TorchEngine": "Qwen/Qwen3-8B", "HIPEngine": "Qwen/Qwen3-8B"}},
+  "qwen-3-32b": {"layers": 64, "repo": {"TorchEngine": "Qwen/Qwen3-32B", "HIPEngine": "Qwen/Qwen3-32B"}},
   "qwen-3-30b-a3b": {"layers": 48, "repo": {}},
   "gemma2-9b": {"layers": 42, "repo": {}},
   "gemma2-27b": {"layers": 46, "repo": {}},
@@ -189,7 +206,9 @@ pretty_names = {
   "qwen-2.5-math-72b": "Qwen 2.5 Math 72B",
   "deepseek-r1-distill-qwen-14b": "DeepSeek R1 Distill Qwen 14B",
   "nemotron-70b": "Nemotron 70B",
-  "qwen-3-32b": "Qwen 3 32B (unsupported arch)",
+  "qwen-3-0.6b": "Qwen 3 0.6B",
+  "qwen-3-8b": "Qwen 3 8B",
+  "qwen-3-32b": "Qwen 3 32B",
   "qwen-3-30b-a3b": "Qwen 3 30B A3B (unsupported arch)",
   "gemma2-9b": "Gemma2 9B (unsupported arch)",
   "gemma2-27b": "Gemma2 27B (unsupported arch)",

@@ -78,16 +78,19 @@ def rope_apply(q, k, cos, sin, positions):
 
 
 def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_kv_heads: int,
-                    head_dim: int, kp=None, vp=None):
-  """Fused on the packed qkv GEMM output [B,S,(H+2KVH)*hd]: RoPE-rotate the q
-  heads in place, rotate k heads into the cache, copy v heads into the cache.
-  When the MFMA-packed cache copies (kp, vp) exist they are appended too.
+                    head_dim: int, kp=None, vp=None, q_norm=None, k_norm=None,
+                    norm_eps: float = 1e-6):
+  """Fused on the packed qkv GEMM output [B,S,(H+2KVH)*hd]: (optionally
+  per-head-RMSNorm q/k — qwen3), RoPE-rotate the q heads in place, rotate k
+  heads into the cache, copy v heads into the cache. When the MFMA-packed
+  cache copies (kp, vp) exist they are appended too.
   """
   if _use_hip(qkv) and qkv.dtype == torch.bfloat16:
     _hip.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim,
-                         kp, vp)
+                         kp, vp, q_norm, k_norm, norm_eps)
     return
-  torch_ref.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim)
+  torch_ref.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads,
+                            head_dim, q_norm, k_norm, norm_eps)
 
 
 def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int, kp=None, vp=None):

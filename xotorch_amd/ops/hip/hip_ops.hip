@@ -131,12 +131,17 @@ __global__ __launch_bounds__(256) void rmsnorm_kernel(
 // cache as coalesced v_mfma_f32_16x16x32_bf16 B-fragments):
 //   K_packed [B,KVH][T32/16 tile][4 hd-chunk][64 lane][8]  lane=(hd&31)/8*16 + (pos&15)
 //   V_packed [B,KVH][8 hd-group][T32/32 tile][64 lane][8]  lane=((pos&31)>>3)*16 + (hd&15)
+// Optional qwen3 per-head q/k RMSNorm (qn/kn weights [hd], fused before the
+// rotation): the slot's wave reduces sum-of-squares over the head row with
+// shfl_xor, then scales while rotating.
 __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
     unsigned short* __restrict__ qkv, const float* __restrict__ cosb,
     const float* __restrict__ sinb, const int* __restrict__ positions,
     unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
     int B, int S, int H, int KVH, int hd, int T,
-    unsigned short* __restrict__ kpc, unsigned short* __restrict__ vpc, int T32) {
+    unsigned short* __restrict__ kpc, unsigned short* __restrict__ vpc, int T32,
+    const unsigned short* __restrict__ qn, const unsigned short* __restrict__ kn,
+    float norm_eps) {
   const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   const int slots = H + 2 * KVH;
@@ -149,22 +154,50 @@ __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
   const int hd2 = hd >> 1;
   unsigned short* row = qkv + ((size_t)bs * slots + slot) * hd;
   if (slot < H) {
+    float x1 = 0.f, x2 = 0.f;
+    if (lane < hd2) {
+      x1 = b2f(row[lane]);
+      x2 = b2f(row[lane + hd2]);
+    }
+    if (qn) {
+      float ss = x1 * x1 + x2 * x2;
+#pragma unroll
+      for (int m = 32; m > 0; m >>= 1) ss += __shfl_xor(ss, m);
+      const float inv = rsqrtf(ss / (float)hd + norm_eps);
+      if (lane < hd2) {
+        x1 *= inv * b2f(qn[lane]);
+        x2 *= inv * b2f(qn[lane + hd2]);
+      }
+    }
     if (lane < hd2) {
       const float c = cosb[(size_t)pos * hd2 + lane];
       const float sn = sinb[(size_t)pos * hd2 + lane];
-      const float x1 = b2f(row[lane]);
-      const float x2 = b2f(row[lane + hd2]);
       row[lane] = f2b(x1 * c - x2 * sn);
       row[lane + hd2] = f2b(x2 * c + x1 * sn);
     }
   } else if (slot < H + KVH) {
     const int h = slot - H;
     unsigned short* dst = kc + (((size_t)(b * KVH + h) * T + pos) * hd);
+    float kx1 = 0.f, kx2 = 0.f;
+    if (lane < hd2) {
+      kx1 = b2f(row[lane]);
+      kx2 = b2f(row[lane + hd2]);
+    }
+    if (kn) {
+      float ss = kx1 * kx1 + kx2 * kx2;
+#pragma unroll
+      for (int m = 32; m > 0; m >>= 1) ss += __shfl_xor(ss, m);
+      const float inv = rsqrtf(ss / (float)hd + norm_eps);
+      if (lane < hd2) {
+        kx1 *= inv * b2f(kn[lane]);
+        kx2 *= inv * b2f(kn[lane + hd2]);
+      }
+    }
     if (lane < hd2) {
       const float c = cosb[(size_t)pos * hd2 + lane];
       const float sn = sinb[(size_t)pos * hd2 + lane];
-      const float x1 = b2f(row[lane]);
-      const float x2 = b2f(row[lane + hd2]);
+      const float x1 = kx1;
+      const float x2 = kx2;
       const unsigned short r1 = f2b(x1 * c - x2 * sn);
       const unsigned short r2 = f2b(x2 * c + x1 * sn);
       dst[lane] = r1;
@@ -1270,7 +1303,9 @@ std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res, 
 void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
                      torch::Tensor positions, torch::Tensor kc, torch::Tensor vc,
                      int64_t n_heads, int64_t n_kv_heads, int64_t head_dim,
-                     c10::optional<torch::Tensor> kp, c10::optional<torch::Tensor> vp) {
+                     c10::optional<torch::Tensor> kp, c10::optional<torch::Tensor> vp,
+                     c10::optional<torch::Tensor> q_norm, c10::optional<torch::Tensor> k_norm,
+                     double norm_eps) {
   CHK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
   CHK(kc.is_contiguous() && vc.is_contiguous());
   CHK(positions.dtype() == torch::kInt32 && positions.is_cuda());
@@ -1288,12 +1323,23 @@ void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
     vpc = (unsigned short*)vp->data_ptr();
     T32 = (int)kp->size(2) * 16;
   }
+  const unsigned short* qnp = nullptr;
+  const unsigned short* knp = nullptr;
+  if (q_norm.has_value()) {
+    CHK(q_norm->is_contiguous() && q_norm->dtype() == torch::kBFloat16 && q_norm->numel() == hd);
+    qnp = (const unsigned short*)q_norm->data_ptr();
+  }
+  if (k_norm.has_value()) {
+    CHK(k_norm->is_contiguous() && k_norm->dtype() == torch::kBFloat16 && k_norm->numel() == hd);
+    knp = (const unsigned short*)k_norm->data_ptr();
+  }
   const int waves = B * S * (H + 2 * KVH);
   const int blocks = (waves + 3) / 4;
   hipLaunchKernelGGL(rope_qkv_append_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
                      (unsigned short*)qkv.data_ptr(), cos.data_ptr<float>(), sin.data_ptr<float>(),
                      positions.data_ptr<int>(), (unsigned short*)kc.data_ptr(),
-                     (unsigned short*)vc.data_ptr(), B, S, H, KVH, hd, T, kpc, vpc, T32);
+                     (unsigned short*)vc.data_ptr(), B, S, H, KVH, hd, T, kpc, vpc, T32,
+                     qnp, knp, (float)norm_eps);
 }
 
 template <int HD>
@@ -1726,10 +1772,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_packed", &swiglu_packed, "SwiGLU on the packed [gate|up] GEMM output");
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual add + RMSNorm");
-  m.def("rope_qkv_append", &rope_qkv_append, "fused RoPE + KV-cache append on packed qkv",
+  m.def("rope_qkv_append", &rope_qkv_append,
+        "fused (qk-RMSNorm +) RoPE + KV-cache append on packed qkv",
         py::arg("qkv"), py::arg("cos"), py::arg("sin"), py::arg("positions"), py::arg("kc"),
         py::arg("vc"), py::arg("n_heads"), py::arg("n_kv_heads"), py::arg("head_dim"),
-        py::arg("kp") = py::none(), py::arg("vp") = py::none());
+        py::arg("kp") = py::none(), py::arg("vp") = py::none(),
+        py::arg("q_norm") = py::none(), py::arg("k_norm") = py::none(),
+        py::arg("norm_eps") = 1e-6);
   m.def("attn_decode", &attn_decode, "GQA decode attention (flash-decoding split-KV)");
   m.def("attn_decode_mfma", &attn_decode_mfma,
         "GQA decode attention on matrix cores (packed cache, hd=128)");

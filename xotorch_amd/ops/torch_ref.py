@@ -166,15 +166,24 @@ def rope_qkv_append(
   n_heads: int,
   n_kv_heads: int,
   head_dim: int,
+  q_norm: torch.Tensor = None,
+  k_norm: torch.Tensor = None,
+  norm_eps: float = 1e-6,
 ) -> None:
-  """Reference for the fused packed-qkv kernel: rotate q in place inside qkv,
-  rotate k and append (k, v) into the caches at `positions`."""
+  """Reference for the fused packed-qkv kernel: (optionally per-head-RMSNorm
+  q/k — qwen3), rotate q in place inside qkv, rotate k and append (k, v)
+  into the caches at `positions`."""
   B, S, _ = qkv.shape
   H, KVH, hd = n_heads, n_kv_heads, head_dim
   q = qkv[:, :, : H * hd].view(B, S, H, hd)
   k = qkv[:, :, H * hd: (H + KVH) * hd].view(B, S, KVH, hd)
   v = qkv[:, :, (H + KVH) * hd:].view(B, S, KVH, hd)
-  q_r, k_r = rope_apply(q, k, cos, sin, positions)
+  qq, kk = q, k
+  if q_norm is not None:
+    qq = rmsnorm(q.float(), q_norm.float(), norm_eps).to(q.dtype)
+  if k_norm is not None:
+    kk = rmsnorm(k.float(), k_norm.float(), norm_eps).to(k.dtype)
+  q_r, k_r = rope_apply(qq, kk, cos, sin, positions)
   q.copy_(q_r)
   start_pos = int(positions.reshape(-1)[0])
   kv_append(k_cache, v_cache, k_r, v, start_pos)
