@@ -232,3 +232,44 @@ def test_qwen3_qk_norm_model():
     h = m0(toks, caches=c0.caches, positions=pos, start_pos=0)
     ls = m1(h, caches=c1.caches, positions=pos, start_pos=0)
     assert torch.equal(lf, ls), (lf - ls).abs().max()
+
+
+def test_qwen3_moe_tiny_forward():
+  """qwen3-moe: qk-norm + per-expert moe_intermediate + qwen3 expert key map;
+  routed decode path must match the dynamic loop."""
+  import torch
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.llama import MoEMLP, ShardedModel, hf_key_map
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.shard import Shard
+
+  tiny = {
+    "model_type": "qwen3_moe", "hidden_size": 64, "num_hidden_layers": 2,
+    "num_attention_heads": 4, "num_key_value_heads": 2, "head_dim": 16,
+    "intermediate_size": 128, "moe_intermediate_size": 48, "num_experts": 8,
+    "num_experts_per_tok": 3, "vocab_size": 131, "rope_theta": 1000000.0,
+    "rms_norm_eps": 1e-6, "max_position_embeddings": 64, "torch_dtype": "float32",
+  }
+  cfg = config_from_hf(tiny, "qwen3moe-tiny")
+  assert cfg.qk_norm and cfg.moe_style == "qwen3" and cfg.moe_intermediate_dim == 48
+  shard = Shard("qwen3moe-tiny", 0, 1, 2)
+  m = ShardedModel(cfg, shard).float()
+  random_init(m)
+  m.eval()
+  # expert shapes use moe_intermediate
+  assert m.layers["0"].mlp.experts[0].gate_up_proj.weight.shape == (96, 64)
+  mapping = hf_key_map(shard, cfg)
+  assert "model.layers.0.mlp.experts.0.gate_proj.weight" in mapping
+  assert "model.layers.0.mlp.gate.weight" in mapping
+  cache = ShardKVCache(2, 3, 2, 20, 16, torch.float32, "cpu")
+  with torch.inference_mode():
+    logits = m(torch.randint(0, 131, (3, 8)), caches=cache.caches,
+               positions=torch.arange(8), start_pos=0)
+    assert torch.isfinite(logits).all()
+    # routed static-capacity path vs dynamic loop
+    moe = m.layers["0"].mlp
+    x = torch.randn(5, 1, 64)
+    ref = moe(x)
+    routed = moe._forward_decode(x.view(-1, 64)).view(5, 1, 64).to(x.dtype)
+    assert torch.allclose(ref, routed, atol=1e-4, rtol=1e-4)

@@ -54,9 +54,11 @@ class ModelConfig:
   torch_dtype: torch.dtype = torch.bfloat16
   bos_token_id: Optional[int] = None
   eos_token_id: Optional[int] = None
-  # MoE (Mixtral-class); n_experts == 0 → dense MLP
+  # MoE (Mixtral / qwen3-moe); n_experts == 0 → dense MLP
   n_experts: int = 0
   n_experts_per_tok: int = 2
+  moe_intermediate_dim: int = 0   # per-expert I (qwen3moe); 0 → intermediate_dim
+  moe_style: str = "mixtral"      # HF checkpoint key layout: mixtral | qwen3
 
   @property
   def kv_mult(self) -> int:
@@ -125,6 +127,8 @@ def config_from_hf(config_path: Path | str | dict, model_id: str = "unknown") ->
     torch_dtype=_DTYPE_MAP.get(raw.get("torch_dtype", "bfloat16"), torch.bfloat16),
     bos_token_id=raw.get("bos_token_id"),
     eos_token_id=eos,
-    n_experts=raw.get("num_local_experts", 0) or 0,
+    n_experts=raw.get("num_local_experts", 0) or raw.get("num_experts", 0) or 0,
     n_experts_per_tok=raw.get("num_experts_per_tok", 2),
+    moe_intermediate_dim=raw.get("moe_intermediate_size", 0) or 0,
+    moe_style="qwen3" if mtype == "qwen3_moe" else "mixtral",
   )

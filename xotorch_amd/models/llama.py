@@ -203,9 +203,11 @@ class MoEMLP(nn.Module):
     self.n_experts = cfg.n_experts
     self.top_k = cfg.n_experts_per_tok
     self.dim = cfg.dim
-    self.intermediate = cfg.intermediate_dim
+    self.intermediate = cfg.moe_intermediate_dim or cfg.intermediate_dim
     self.gate = nn.Linear(cfg.dim, cfg.n_experts, bias=False)
-    self.experts = nn.ModuleList([MLP(cfg) for _ in range(cfg.n_experts)])
+    self.experts = nn.ModuleList(
+      [MLP(cfg, intermediate=self.intermediate) for _ in range(cfg.n_experts)]
+    )
     self.wp_gate_up: Optional[torch.Tensor] = None
     self.wp_down: Optional[torch.Tensor] = None
     self.wp_gate_up_fp8: Optional[torch.Tensor] = None
@@ -498,7 +500,13 @@ def hf_key_map(shard: Shard, cfg: ModelConfig):
       mapping[hf + "self_attn.k_norm.weight"] = ours + "self_attn.k_norm"
     mapping[hf + "input_layernorm.weight"] = ours + "input_layernorm.weight"
     mapping[hf + "post_attention_layernorm.weight"] = ours + "post_attention_layernorm.weight"
-    if cfg.n_experts > 0:
+    if cfg.n_experts > 0 and cfg.moe_style == "qwen3":
+      mapping[hf + "mlp.gate.weight"] = ours + "mlp.gate.weight"
+      for e in range(cfg.n_experts):
+        mapping[hf + f"mlp.experts.{e}.gate_proj.weight"] = ours + f"mlp.experts.{e}.gate_up_proj.weight#0"
+        mapping[hf + f"mlp.experts.{e}.up_proj.weight"] = ours + f"mlp.experts.{e}.gate_up_proj.weight#1"
+        mapping[hf + f"mlp.experts.{e}.down_proj.weight"] = ours + f"mlp.experts.{e}.down_proj.weight"
+    elif cfg.n_experts > 0:
       mapping[hf + "block_sparse_moe.gate.weight"] = ours + "mlp.gate.weight"
       for e in range(cfg.n_experts):
         mapping[hf + f"block_sparse_moe.experts.{e}.w1.weight"] = ours + f"mlp.experts.{e}.gate_up_proj.weight#0"

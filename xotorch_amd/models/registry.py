@@ -59,6 +59,20 @@ def _qwen3_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=151936, tie=False, max
   }
 
 
+def _qwen3_moe_cfg(dim, n_layers, n_heads, n_kv, inter, moe_inter, n_experts, top_k,
+                   vocab=151936, max_pos=40960, head_dim=128):
+  return {
+    "model_type": "qwen3_moe", "hidden_size": dim, "num_hidden_layers": n_layers,
+    "num_attention_heads": n_heads, "num_key_value_heads": n_kv,
+    "head_dim": head_dim, "intermediate_size": inter,
+    "moe_intermediate_size": moe_inter, "num_experts": n_experts,
+    "num_experts_per_tok": top_k, "vocab_size": vocab,
+    "rope_theta": 1000000.0, "rms_norm_eps": 1e-6,
+    "max_position_embeddings": max_pos, "torch_dtype": "bfloat16",
+    "bos_token_id": 151643, "eos_token_id": 151645,
+  }
+
+
 def _mixtral_cfg(dim, n_layers, n_heads, n_kv, inter, n_experts=8, top_k=2):
   return {
     "model_type": "mixtral", "hidden_size": dim, "num_hidden_layers": n_layers,
@@ -114,6 +128,7 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
   "nemotron-70b": _llama_cfg(8192, 80, 64, 8, 28672, llama3_scaling=True),
   # moe
   "mixtral-8x7b": _mixtral_cfg(4096, 32, 32, 8, 14336),
+  "qwen-3-30b-a3b": _qwen3_moe_cfg(2048, 48, 32, 4, 6144, 768, 128, 8),
   # phi-4-mini (llama-like enough for the generic decoder)
   "phi-4-mini": _llama_cfg(3072, 32, 24, 8, 8192, vocab=200064, theta=10000.0, max_pos=131072, tie=True),
   # tiny test model
@@ -164,7 +179,7 @@ model_cards: Dict[str, dict] = {
   "qwen-3-0.6b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen3-0.6B", "HIPEngine": "Qwen/Qwen3-0.6B"}},
   "qwen-3-8b": {"layers": 36, "repo": {"TorchEngine": "Qwen/Qwen3-8B", "HIPEngine": "Qwen/Qwen3-8B"}},
   "qwen-3-32b": {"layers": 64, "repo": {"TorchEngine": "Qwen/Qwen3-32B", "HIPEngine": "Qwen/Qwen3-32B"}},
-  "qwen-3-30b-a3b": {"layers": 48, "repo": {}},
+  "qwen-3-30b-a3b": {"layers": 48, "repo": {"TorchEngine": "Qwen/Qwen3-30B-A3B", "HIPEngine": "Qwen/Qwen3-30B-A3B"}},
   "gemma2-9b": {"layers": 42, "repo": {}},
   "gemma2-27b": {"layers": 46, "repo": {}},
   "deepseek-r1": {"layers": 61, "repo": {}},
@@ -209,7 +224,7 @@ pretty_names = {
   "qwen-3-0.6b": "Qwen 3 0.6B",
   "qwen-3-8b": "Qwen 3 8B",
   "qwen-3-32b": "Qwen 3 32B",
-  "qwen-3-30b-a3b": "Qwen 3 30B A3B (unsupported arch)",
+  "qwen-3-30b-a3b": "Qwen 3 30B A3B (MoE)",
   "gemma2-9b": "Gemma2 9B (unsupported arch)",
   "gemma2-27b": "Gemma2 27B (unsupported arch)",
   "deepseek-r1": "DeepSeek R1 (unsupported arch)",
