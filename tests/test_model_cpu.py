@@ -273,3 +273,23 @@ def test_qwen3_moe_tiny_forward():
     ref = moe(x)
     routed = moe._forward_decode(x.view(-1, 64)).view(5, 1, 64).to(x.dtype)
     assert torch.allclose(ref, routed, atol=1e-4, rtol=1e-4)
+
+
+def test_registry_consistency():
+  """Every model card: pretty name present; when a builtin config exists its
+  layer count must match the card; supported models resolve base shards."""
+  from xotorch_amd.models.registry import (
+    BUILTIN_CONFIGS, build_base_shard, builtin_config, get_supported_models,
+    model_cards, pretty_name)
+  for mid, card in model_cards.items():
+    assert pretty_name(mid), mid
+    assert card.get("layers", 0) > 0, mid
+    cfg = builtin_config(mid)
+    if cfg is not None:
+      assert cfg["num_hidden_layers"] == card["layers"], \
+        f"{mid}: card layers {card['layers']} != builtin {cfg['num_hidden_layers']}"
+  # every builtin config belongs to a card
+  for mid in BUILTIN_CONFIGS:
+    assert mid in model_cards or mid in ("dummy",), mid
+  for mid in get_supported_models([["HIPEngine"]]):
+    assert build_base_shard(mid, "HIPEngine") is not None, mid
