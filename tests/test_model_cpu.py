@@ -293,3 +293,27 @@ def test_registry_consistency():
     assert mid in model_cards or mid in ("dummy",), mid
   for mid in get_supported_models([["HIPEngine"]]):
     assert build_base_shard(mid, "HIPEngine") is not None, mid
+
+
+def test_pack_decode_weight_is_a_permutation():
+  """The MFMA prepack is a pure permutation of W: inverting the documented
+  index map (docs/KERNELS.md) must recover W exactly, and every element must
+  land where the kernel's A-fragment load expects it:
+  packed[n32][kt][half][n][e] == W[n32*32+n][kt*16+half*8+e]."""
+  import torch
+  from xotorch_amd.ops import pack_decode_weight
+  N, K = 96, 128
+  w = torch.arange(N * K, dtype=torch.float32).reshape(N, K).to(torch.bfloat16)
+  p = pack_decode_weight(w)
+  assert p.shape == (N // 32, K // 16, 2, 32, 8)
+  # invert
+  back = p.permute(0, 3, 1, 2, 4).reshape(N, K)
+  assert torch.equal(back, w)
+  # spot-check the lane address math the kernel uses: element loaded by lane
+  # l at (n32, kt) from flat offset (n32*(K/16)+kt)*512 + l*8 + j must be
+  # W[n32*32 + (l&31)][kt*16 + (l>>5)*8 + j]
+  flat = p.reshape(-1)
+  for (n32, kt, l, j) in [(0, 0, 0, 0), (1, 3, 45, 7), (2, 7, 31, 4), (0, 1, 63, 0)]:
+    got = flat[(n32 * (K // 16) + kt) * 512 + l * 8 + j]
+    want = w[n32 * 32 + (l & 31)][kt * 16 + (l >> 5) * 8 + j]
+    assert torch.equal(got, want), (n32, kt, l, j)
