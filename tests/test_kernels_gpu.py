@@ -225,10 +225,10 @@ def test_xotlinear_packed_matches_unpacked(hip):
   assert (y0 - y1).abs().max().item() / (y0.abs().max().item() + 1e-9) < 2e-2
 
 
-def test_skinny_gemm_grouped(hip):
+@pytest.mark.parametrize("E,C,K,N", [(8, 64, 512, 256), (128, 32, 256, 128)])
+def test_skinny_gemm_grouped(hip, E, C, K, N):
   from xotorch_amd import ops as xops
   from xotorch_amd.ops import _hip_ops
-  E, C, K, N = 8, 64, 512, 256
   x = bt(E, C, K, scale=0.5, seed=11)
   ws = [bt(N, K, scale=0.05, seed=100 + e) for e in range(E)]
   wp = torch.stack([xops.pack_decode_weight(w) for w in ws]).contiguous()
