@@ -104,13 +104,18 @@ class TorchEngine(InferenceEngine):
     if cfg_dict is None:
       raise ValueError(f"no config available for model {shard.model_id}")
     cfg = config_from_hf(cfg_dict, model_id=shard.model_id)
+    if cfg.model_type == "gemma2":
+      from xotorch_amd.models.gemma2 import Gemma2Model
+      model_cls = Gemma2Model
+    else:
+      model_cls = ShardedModel
     # construct at the target dtype so to_empty materializes it directly
     # (fp32-then-cast would peak at 2x the weight bytes)
     prev_dtype = torch.get_default_dtype()
     torch.set_default_dtype(self.dtype)
     try:
       with torch.device("meta"):
-        model = ShardedModel(cfg, shard)
+        model = model_cls(cfg, shard)
     finally:
       torch.set_default_dtype(prev_dtype)
     model = model.to_empty(device=self.device)

@@ -51,6 +51,12 @@ class ModelConfig:
   tie_word_embeddings: bool = False
   attn_bias: bool = False  # qwen2 q/k/v bias
   qk_norm: bool = False    # qwen3 per-head RMSNorm on q/k before RoPE
+  model_type: str = "llama"
+  # gemma2 (separate decoder, models/gemma2.py):
+  query_pre_attn_scalar: float = 0.0
+  attn_logit_softcapping: float = 0.0
+  final_logit_softcapping: float = 0.0
+  sliding_window: int = 0
   torch_dtype: torch.dtype = torch.bfloat16
   bos_token_id: Optional[int] = None
   eos_token_id: Optional[int] = None
@@ -124,6 +130,11 @@ def config_from_hf(config_path: Path | str | dict, model_id: str = "unknown") ->
     tie_word_embeddings=raw.get("tie_word_embeddings", False),
     attn_bias=mtype in ("qwen2",) or raw.get("attention_bias", False),
     qk_norm=mtype in ("qwen3", "qwen3_moe"),
+    model_type=mtype,
+    query_pre_attn_scalar=float(raw.get("query_pre_attn_scalar", 0) or 0),
+    attn_logit_softcapping=float(raw.get("attn_logit_softcapping", 0) or 0),
+    final_logit_softcapping=float(raw.get("final_logit_softcapping", 0) or 0),
+    sliding_window=int(raw.get("sliding_window", 0) or 0),
     torch_dtype=_DTYPE_MAP.get(raw.get("torch_dtype", "bfloat16"), torch.bfloat16),
     bos_token_id=raw.get("bos_token_id"),
     eos_token_id=eos,

@@ -1,0 +1,204 @@
+"""Sharded Gemma2 decoder (gemma2-9b / gemma2-27b).
+
+A SEPARATE decoder from ShardedModel (models/llama.py): gemma2 differs in
+too many places to share the hot path — scaled embeddings, gemma-style
+RMSNorm (x_norm * (1 + w), fp32), FOUR norms per block (post-attn and
+post-ffn sandwich norms), attention logit soft-capping, alternating
+sliding-window/global layers, query_pre_attn_scalar scaling, GeGLU MLP and
+final-logit soft-capping. The reference lists gemma2 model cards but its
+torchtune GQA assembly implements none of this
+(/root/reference/xotorch/inference/torch/models/general_mha.py:23-254 — no
+softcap/sliding-window/post-norms), so those cards cannot produce correct
+outputs there; here the family runs correctly on a plain-torch compute path
+(CPU or GPU) validated against transformers' Gemma2ForCausalLM. The
+hand-written CDNA4 kernels are NOT used for gemma2 (soft-capping is not in
+the MFMA attention kernels yet — docs/ROADMAP.md); this is a
+capability-completeness path, not a speed path.
+
+Shard semantics match ShardedModel: first shard owns the (scaled) embedding,
+last owns the final norm + tied head; caches are the engine's LayerKV pairs.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from xotorch_amd.models.config import ModelConfig
+from xotorch_amd.ops.torch_ref import rope_cos_sin
+from xotorch_amd.shard import Shard
+
+
+def _rms(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+  xf = x.float()
+  out = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+  return (out * (1.0 + w.float())).to(x.dtype)
+
+
+def _softcap(x: torch.Tensor, cap: float) -> torch.Tensor:
+  return torch.tanh(x / cap) * cap if cap else x
+
+
+class Gemma2Attention(nn.Module):
+  def __init__(self, cfg: ModelConfig, layer_idx: int):
+    super().__init__()
+    self.cfg = cfg
+    H, KVH, hd, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim, cfg.dim
+    self.q_proj = nn.Linear(D, H * hd, bias=False)
+    self.k_proj = nn.Linear(D, KVH * hd, bias=False)
+    self.v_proj = nn.Linear(D, KVH * hd, bias=False)
+    self.o_proj = nn.Linear(H * hd, D, bias=False)
+    self.scale = (cfg.query_pre_attn_scalar or cfg.head_dim) ** -0.5
+    # even layers use the sliding window, odd layers are global (transformers
+    # Gemma2DecoderLayer: is_sliding = not bool(layer_idx % 2))
+    self.window = cfg.sliding_window if (layer_idx % 2 == 0) else 0
+
+  def forward(self, x, cos, sin, positions, kv, start_pos: int):
+    cfg = self.cfg
+    B, S, _ = x.shape
+    H, KVH, hd = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
+    from xotorch_amd.ops import torch_ref as tr
+    q = self.q_proj(x).view(B, S, H, hd)
+    k = self.k_proj(x).view(B, S, KVH, hd)
+    v = self.v_proj(x).view(B, S, KVH, hd)
+    q, k = tr.rope_apply(q, k, cos, sin, positions)
+    k_cache, v_cache = kv[0], kv[1]
+    tr.kv_append(k_cache, v_cache, k, v, start_pos)
+    total = start_pos + S
+    rep = H // KVH
+    kk = k_cache[:, :, :total].repeat_interleave(rep, dim=1)  # [B, H, T, hd]
+    vv = v_cache[:, :, :total].repeat_interleave(rep, dim=1)
+    scores = torch.einsum("bshd,bhtd->bhst", q.float(), kk.float()) * self.scale
+    scores = _softcap(scores, cfg.attn_logit_softcapping)
+    qpos = torch.arange(start_pos, total, device=x.device)[:, None]
+    kpos = torch.arange(0, total, device=x.device)[None, :]
+    mask = kpos <= qpos
+    if self.window:
+      mask &= kpos > qpos - self.window
+    scores = scores.masked_fill(~mask[None, None], float("-inf"))
+    probs = torch.softmax(scores, dim=-1).to(vv.dtype)
+    out = torch.einsum("bhst,bhtd->bshd", probs, vv).reshape(B, S, H * hd)
+    return self.o_proj(out)
+
+
+class Gemma2MLP(nn.Module):
+  def __init__(self, cfg: ModelConfig):
+    super().__init__()
+    I, D = cfg.intermediate_dim, cfg.dim
+    self.gate_proj = nn.Linear(D, I, bias=False)
+    self.up_proj = nn.Linear(D, I, bias=False)
+    self.down_proj = nn.Linear(I, D, bias=False)
+
+  def forward(self, x):
+    return self.down_proj(
+      nn.functional.gelu(self.gate_proj(x), approximate="tanh") * self.up_proj(x)
+    )
+
+
+class Gemma2Layer(nn.Module):
+  def __init__(self, cfg: ModelConfig, layer_idx: int):
+    super().__init__()
+    self.eps = cfg.norm_eps
+    self.self_attn = Gemma2Attention(cfg, layer_idx)
+    self.mlp = Gemma2MLP(cfg)
+    D = cfg.dim
+    self.input_layernorm = nn.Parameter(torch.zeros(D))
+    self.post_attention_layernorm = nn.Parameter(torch.zeros(D))
+    self.pre_feedforward_layernorm = nn.Parameter(torch.zeros(D))
+    self.post_feedforward_layernorm = nn.Parameter(torch.zeros(D))
+
+  def forward(self, h, cos, sin, positions, kv, start_pos):
+    res = h
+    hs = _rms(h, self.input_layernorm, self.eps)
+    hs = self.self_attn(hs, cos, sin, positions, kv, start_pos)
+    h = res + _rms(hs, self.post_attention_layernorm, self.eps)
+    res = h
+    hs = _rms(h, self.pre_feedforward_layernorm, self.eps)
+    hs = self.mlp(hs)
+    return res + _rms(hs, self.post_feedforward_layernorm, self.eps)
+
+
+class Gemma2Model(nn.Module):
+  """Layer range [shard.start_layer .. shard.end_layer] of a Gemma2 model.
+  Same forward contract as ShardedModel (engine-compatible)."""
+
+  def __init__(self, cfg: ModelConfig, shard: Shard):
+    super().__init__()
+    self.cfg = cfg
+    self.shard = shard
+    if shard.is_first_layer:
+      self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.dim)
+    self.layers = nn.ModuleDict(
+      {str(i): Gemma2Layer(cfg, i) for i in range(shard.start_layer, shard.end_layer + 1)}
+    )
+    if shard.is_last_layer:
+      self.norm = nn.Parameter(torch.zeros(cfg.dim))
+      if not shard.is_first_layer:
+        # tied embeddings, but the owner shard is elsewhere: local head copy
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+    cos, sin = rope_cos_sin(cfg.head_dim, cfg.max_seq_len, cfg.rope_theta, cfg.rope_scaling)
+    self.register_buffer("rope_cos", cos, persistent=False)
+    self.register_buffer("rope_sin", sin, persistent=False)
+
+  def reset_rope(self):
+    cos, sin = rope_cos_sin(self.cfg.head_dim, self.cfg.max_seq_len, self.cfg.rope_theta,
+                            self.cfg.rope_scaling, device=self.rope_cos.device)
+    self.rope_cos, self.rope_sin = cos, sin
+
+  @property
+  def local_layer_ids(self) -> List[int]:
+    return list(range(self.shard.start_layer, self.shard.end_layer + 1))
+
+  def pack_decode_weights(self, reserve_bytes: int = 0) -> int:
+    return 0  # gemma2 runs the plain-torch path (no MFMA prepack yet)
+
+  def head_weight(self):
+    if hasattr(self, "embed_tokens"):
+      return self.embed_tokens.weight
+    return self.lm_head.weight
+
+  def forward(self, x, caches, positions, start_pos: int, is_decode: bool = False,
+              seq_lens=None, last_only: bool = True):
+    cfg = self.cfg
+    if x.dtype in (torch.int32, torch.int64):
+      h = self.embed_tokens(x) * torch.tensor(cfg.dim ** 0.5, dtype=self.embed_tokens.weight.dtype)
+    else:
+      h = x
+    if positions.dim() == 0:
+      positions = positions.reshape(1)
+    for idx, lid in enumerate(self.local_layer_ids):
+      h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, caches[idx], start_pos)
+    if not self.shard.is_last_layer:
+      return h
+    if last_only and h.shape[1] > 1:
+      h = h[:, -1:, :]
+    h = _rms(h, self.norm, cfg.norm_eps)
+    logits = torch.nn.functional.linear(h, self.head_weight().to(h.dtype))
+    logits = _softcap(logits.float(), cfg.final_logit_softcapping).to(logits.dtype)
+    if is_decode or last_only:
+      return logits[:, -1, :]
+    return logits
+
+
+def hf_key_map_gemma2(shard: Shard, cfg: ModelConfig):
+  """HF Gemma2ForCausalLM checkpoint keys → Gemma2Model state-dict keys."""
+  mapping = {}
+  if shard.is_first_layer:
+    mapping["model.embed_tokens.weight"] = "embed_tokens.weight"
+  if shard.is_last_layer:
+    mapping["model.norm.weight"] = "norm"
+    if not shard.is_first_layer:
+      mapping["model.embed_tokens.weight"] = "lm_head.weight"
+  for lid in range(shard.start_layer, shard.end_layer + 1):
+    hf = f"model.layers.{lid}."
+    ours = f"layers.{lid}."
+    for p in ("q_proj", "k_proj", "v_proj", "o_proj"):
+      mapping[hf + f"self_attn.{p}.weight"] = ours + f"self_attn.{p}.weight"
+    for p in ("gate_proj", "up_proj", "down_proj"):
+      mapping[hf + f"mlp.{p}.weight"] = ours + f"mlp.{p}.weight"
+    for p in ("input_layernorm", "post_attention_layernorm",
+              "pre_feedforward_layernorm", "post_feedforward_layernorm"):
+      mapping[hf + f"{p}.weight"] = ours + p
+  return mapping

@@ -59,6 +59,19 @@ def _qwen3_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=151936, tie=False, max
   }
 
 
+def _gemma2_cfg(dim, n_layers, n_heads, n_kv, inter, head_dim, qpas, window=4096):
+  return {
+    "model_type": "gemma2", "hidden_size": dim, "num_hidden_layers": n_layers,
+    "num_attention_heads": n_heads, "num_key_value_heads": n_kv,
+    "head_dim": head_dim, "intermediate_size": inter, "vocab_size": 256000,
+    "rope_theta": 10000.0, "rms_norm_eps": 1e-6, "max_position_embeddings": 8192,
+    "tie_word_embeddings": True, "query_pre_attn_scalar": qpas,
+    "attn_logit_softcapping": 50.0, "final_logit_softcapping": 30.0,
+    "sliding_window": window, "torch_dtype": "bfloat16",
+    "bos_token_id": 2, "eos_token_id": 1,
+  }
+
+
 def _qwen3_moe_cfg(dim, n_layers, n_heads, n_kv, inter, moe_inter, n_experts, top_k,
                    vocab=151936, max_pos=40960, head_dim=128):
   return {
@@ -129,6 +142,9 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
   # moe
   "mixtral-8x7b": _mixtral_cfg(4096, 32, 32, 8, 14336),
   "qwen-3-30b-a3b": _qwen3_moe_cfg(2048, 48, 32, 4, 6144, 768, 128, 8),
+  # gemma2 (separate decoder, models/gemma2.py)
+  "gemma2-9b": _gemma2_cfg(3584, 42, 16, 8, 14336, 256, 256),
+  "gemma2-27b": _gemma2_cfg(4608, 46, 32, 16, 36864, 128, 144),
   # phi-4-mini (llama-like enough for the generic decoder)
   "phi-4-mini": _llama_cfg(3072, 32, 24, 8, 8192, vocab=200064, theta=10000.0, max_pos=131072, tie=True),
   # tiny test model
@@ -180,8 +196,8 @@ model_cards: Dict[str, dict] = {
   "qwen-3-8b": {"layers": 36, "repo": {"TorchEngine": "Qwen/Qwen3-8B", "HIPEngine": "Qwen/Qwen3-8B"}},
   "qwen-3-32b": {"layers": 64, "repo": {"TorchEngine": "Qwen/Qwen3-32B", "HIPEngine": "Qwen/Qwen3-32B"}},
   "qwen-3-30b-a3b": {"layers": 48, "repo": {"TorchEngine": "Qwen/Qwen3-30B-A3B", "HIPEngine": "Qwen/Qwen3-30B-A3B"}},
-  "gemma2-9b": {"layers": 42, "repo": {}},
-  "gemma2-27b": {"layers": 46, "repo": {}},
+  "gemma2-9b": {"layers": 42, "repo": {"TorchEngine": "google/gemma-2-9b-it", "HIPEngine": "google/gemma-2-9b-it"}},
+  "gemma2-27b": {"layers": 46, "repo": {"TorchEngine": "google/gemma-2-27b-it", "HIPEngine": "google/gemma-2-27b-it"}},
   "deepseek-r1": {"layers": 61, "repo": {}},
   "deepseek-v3": {"layers": 61, "repo": {}},
   "llava-1.5-7b-hf": {"layers": 32, "repo": {}},
@@ -225,8 +241,8 @@ pretty_names = {
   "qwen-3-8b": "Qwen 3 8B",
   "qwen-3-32b": "Qwen 3 32B",
   "qwen-3-30b-a3b": "Qwen 3 30B A3B (MoE)",
-  "gemma2-9b": "Gemma2 9B (unsupported arch)",
-  "gemma2-27b": "Gemma2 27B (unsupported arch)",
+  "gemma2-9b": "Gemma2 9B",
+  "gemma2-27b": "Gemma2 27B",
   "deepseek-r1": "DeepSeek R1 (unsupported arch)",
   "deepseek-v3": "DeepSeek V3 (unsupported arch)",
   "llava-1.5-7b-hf": "LLaVa 1.5 7B (unsupported arch)",

@@ -15,7 +15,15 @@ from typing import Dict, Optional
 import torch
 
 from xotorch_amd.models.config import ModelConfig
-from xotorch_amd.models.llama import ShardedModel, hf_key_map
+from xotorch_amd.models.llama import ShardedModel, hf_key_map as _hf_key_map_llama
+
+
+def hf_key_map(shard, cfg):
+  """Checkpoint key map, routed by architecture."""
+  if getattr(cfg, "model_type", "llama") == "gemma2":
+    from xotorch_amd.models.gemma2 import hf_key_map_gemma2
+    return hf_key_map_gemma2(shard, cfg)
+  return _hf_key_map_llama(shard, cfg)
 from xotorch_amd.shard import Shard
 
 
