@@ -84,6 +84,7 @@ def test_gemma2_through_engine_and_node(monkeypatch):
 
   tiny = dict(TINY)
   tiny["model_type"] = "gemma2"
+  tiny["vocab_size"] = 256  # DummyTokenizer emits ids up to 200
   monkeypatch.setitem(registry.BUILTIN_CONFIGS, "gemma2-tiny", tiny)
   monkeypatch.setitem(registry.model_cards, "gemma2-tiny",
                       {"layers": 4, "repo": {"TorchEngine": "dummy"}})
