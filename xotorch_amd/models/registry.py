@@ -188,10 +188,10 @@ model_cards: Dict[str, dict] = {
   "deepseek-r1-distill-qwen-14b": {"layers": 48, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-14B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-14B"}},
   "nemotron-70b": {"layers": 80, "repo": {"TorchEngine": "nvidia/Llama-3.1-Nemotron-70B-Instruct-HF", "HIPEngine": "nvidia/Llama-3.1-Nemotron-70B-Instruct-HF"}},
   # Listed for registry parity with the reference but with NO supported engine
-  # here: these architectures (qwen3 qk-norm, gemma2 softcap/SWA, deepseek
-  # MLA+MoE, llava vision) are not implemented by this decoder — the
-  # reference lists them too, but its torchtune GQA assembly cannot run them
-  # either. get_supported_models() filters them out.
+  # here: deepseek v3/r1 (MLA + shared experts) and llava (vision tower) are
+  # not implemented by this decoder — the reference lists them too, but its
+  # torchtune GQA assembly cannot run them either.
+  # get_supported_models() filters them out.
   "qwen-3-0.6b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen3-0.6B", "HIPEngine": "Qwen/Qwen3-0.6B"}},
   "qwen-3-8b": {"layers": 36, "repo": {"TorchEngine": "Qwen/Qwen3-8B", "HIPEngine": "Qwen/Qwen3-8B"}},
   "qwen-3-32b": {"layers": 64, "repo": {"TorchEngine": "Qwen/Qwen3-32B", "HIPEngine": "Qwen/Qwen3-32B"}},
