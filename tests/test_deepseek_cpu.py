@@ -1,0 +1,167 @@
+"""DeepSeek-V3 MLA decoder vs transformers' DeepseekV3ForCausalLM oracle
+(random weights, CPU fp32): prefill, cached decode with the LATENT cache,
+shard splitting. The reference lists deepseek cards its engine cannot run."""
+import pytest
+import torch
+
+ORACLE_KW = dict(vocab_size=101, hidden_size=64, intermediate_size=128, moe_intermediate_size=48,
+                 num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=4,
+                 n_routed_experts=8, num_experts_per_tok=2, n_shared_experts=1,
+                 n_group=2, topk_group=2, q_lora_rank=32, kv_lora_rank=16,
+                 qk_rope_head_dim=8, qk_nope_head_dim=16, v_head_dim=16,
+                 first_k_dense_replace=1, rms_norm_eps=1e-6,
+                 max_position_embeddings=64, tie_word_embeddings=False)
+
+
+def _load_ours_from_oracle(ref, cfg, shard):
+  """Map the oracle's state dict (3D fused expert params in this transformers
+  version) into DeepseekV3Model's per-expert linears."""
+  from xotorch_amd.models.deepseek_v3 import DeepseekV3Model
+  ours = DeepseekV3Model(cfg, shard).float()
+  sd = ref.state_dict()
+  new = {}
+  for lid in range(shard.start_layer, shard.end_layer + 1):
+    hf = f"model.layers.{lid}."
+    ours_p = f"layers.{lid}."
+    for k in ("q_a_proj.weight", "q_b_proj.weight", "kv_a_proj_with_mqa.weight",
+              "kv_b_proj.weight", "o_proj.weight"):
+      new[ours_p + "self_attn." + k] = sd[hf + "self_attn." + k]
+    new[ours_p + "self_attn.q_a_layernorm"] = sd[hf + "self_attn.q_a_layernorm.weight"]
+    new[ours_p + "self_attn.kv_a_layernorm"] = sd[hf + "self_attn.kv_a_layernorm.weight"]
+    new[ours_p + "input_layernorm"] = sd[hf + "input_layernorm.weight"]
+    new[ours_p + "post_attention_layernorm"] = sd[hf + "post_attention_layernorm.weight"]
+    if hf + "mlp.gate_proj.weight" in sd:  # dense layer
+      for p in ("gate_proj", "up_proj", "down_proj"):
+        new[ours_p + f"mlp.{p}.weight"] = sd[hf + f"mlp.{p}.weight"]
+    else:
+      new[ours_p + "mlp.gate_weight"] = sd[hf + "mlp.gate.weight"]
+      new[ours_p + "mlp.e_score_correction_bias"] = sd[hf + "mlp.gate.e_score_correction_bias"]
+      if hf + "mlp.experts.gate_up_proj" in sd:  # fused 3D expert storage
+        gu = sd[hf + "mlp.experts.gate_up_proj"]   # [E, 2I, D]
+        dn = sd[hf + "mlp.experts.down_proj"]      # [E, D, I]
+        I = dn.shape[2]
+        for e in range(gu.shape[0]):
+          new[ours_p + f"mlp.experts.{e}.gate_proj.weight"] = gu[e, :I]
+          new[ours_p + f"mlp.experts.{e}.up_proj.weight"] = gu[e, I:]
+          new[ours_p + f"mlp.experts.{e}.down_proj.weight"] = dn[e]
+      else:
+        for e in range(8):
+          for p in ("gate_proj", "up_proj", "down_proj"):
+            new[ours_p + f"mlp.experts.{e}.{p}.weight"] = sd[hf + f"mlp.experts.{e}.{p}.weight"]
+      for p in ("gate_proj", "up_proj", "down_proj"):
+        new[ours_p + f"mlp.shared_experts.{p}.weight"] = sd[hf + f"mlp.shared_experts.{p}.weight"]
+  if shard.is_first_layer:
+    new["embed_tokens.weight"] = sd["model.embed_tokens.weight"]
+  if shard.is_last_layer:
+    new["norm"] = sd["model.norm.weight"]
+    new["lm_head.weight"] = sd["lm_head.weight"]
+  missing, unexpected = ours.load_state_dict(new, strict=False)
+  missing = [m for m in missing if "rope" not in m]
+  assert not missing, missing
+  ours.eval()
+  return ours
+
+
+def _build(seed=0):
+  from transformers import DeepseekV3Config, DeepseekV3ForCausalLM
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.shard import Shard
+  torch.manual_seed(seed)
+  hf_cfg = DeepseekV3Config(**ORACLE_KW)
+  ref = DeepseekV3ForCausalLM(hf_cfg).eval().float()
+  raw = {**ORACLE_KW, "model_type": "deepseek_v3",
+         "rope_theta": hf_cfg.rope_parameters["rope_theta"] if hasattr(hf_cfg, "rope_parameters") else getattr(hf_cfg, "rope_theta", 10000.0),
+         "routed_scaling_factor": hf_cfg.routed_scaling_factor,
+         "norm_topk_prob": hf_cfg.norm_topk_prob,
+         "rope_interleave": getattr(hf_cfg, "rope_interleave", True)}
+  cfg = config_from_hf(raw, "ds-tiny")
+  shard = Shard("ds-tiny", 0, cfg.n_layers - 1, cfg.n_layers)
+  ours = _load_ours_from_oracle(ref, cfg, shard)
+  return ref, ours, cfg, shard
+
+
+def _latent_caches(cfg, n_layers, B, T):
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  heads, kdim, vdim = cfg.kv_cache_dims()
+  return ShardKVCache(n_layers, B, heads, T, kdim, torch.float32, "cpu", v_dim=vdim)
+
+
+def test_mla_matches_transformers_prefill_and_decode():
+  ref, ours, cfg, shard = _build()
+  B, S = 2, 11
+  x = torch.randint(0, cfg.vocab_size, (B, S))
+  cache = _latent_caches(cfg, cfg.n_layers, B, S + 4)
+  with torch.no_grad():
+    lref = ref(x).logits
+    lours = ours(x, caches=cache.caches, positions=torch.arange(S), start_pos=0, last_only=False)
+    assert torch.allclose(lours, lref, atol=3e-4, rtol=3e-4), (lours - lref).abs().max()
+    nxt = lref[:, -1].argmax(-1, keepdim=True)
+    lref2 = ref(torch.cat([x, nxt], dim=1)).logits[:, -1]
+    lours2 = ours(nxt, caches=cache.caches, positions=torch.tensor([S]), start_pos=S, is_decode=True)
+    assert torch.allclose(lours2, lref2, atol=3e-4, rtol=3e-4), (lours2 - lref2).abs().max()
+
+
+def test_mla_split_equals_full():
+  ref, ours, cfg, _ = _build(seed=2)
+  from xotorch_amd.shard import Shard
+  B, S = 2, 7
+  x = torch.randint(0, cfg.vocab_size, (B, S))
+  cache = _latent_caches(cfg, cfg.n_layers, B, S + 2)
+  with torch.no_grad():
+    full = ours(x, caches=cache.caches, positions=torch.arange(S), start_pos=0, last_only=False)
+    s0, s1 = Shard("ds-tiny", 0, 1, 4), Shard("ds-tiny", 2, 3, 4)
+    m0 = _load_ours_from_oracle(ref, cfg, s0)
+    m1 = _load_ours_from_oracle(ref, cfg, s1)
+    c0, c1 = _latent_caches(cfg, 2, B, S + 2), _latent_caches(cfg, 2, B, S + 2)
+    h = m0(x, caches=c0.caches, positions=torch.arange(S), start_pos=0)
+    split = m1(h, caches=c1.caches, positions=torch.arange(S), start_pos=0, last_only=False)
+    assert torch.allclose(full, split, atol=1e-5), (full - split).abs().max()
+
+
+def test_latent_cache_is_small():
+  """MLA's point: the cache holds the latent, not expanded KV."""
+  ref, ours, cfg, _ = _build(seed=3)
+  heads, kdim, vdim = cfg.kv_cache_dims()
+  assert heads == 1 and kdim == cfg.kv_lora_rank and vdim == cfg.qk_rope_head_dim
+
+
+def test_deepseek_through_engine_and_node(monkeypatch):
+  """Serving path: TorchEngine routes deepseek to the MLA decoder with the
+  latent cache; a node decode loop completes."""
+  import asyncio
+  from xotorch_amd.models import registry
+
+  tiny = {**ORACLE_KW, "model_type": "deepseek_v3", "vocab_size": 256}
+  monkeypatch.setitem(registry.BUILTIN_CONFIGS, "ds-tiny", tiny)
+  monkeypatch.setitem(registry.model_cards, "ds-tiny",
+                      {"layers": 4, "repo": {"TorchEngine": "dummy"}})
+
+  async def go():
+    from xotorch_amd.engine.torch_engine import TorchEngine
+    from xotorch_amd.models.registry import build_base_shard
+    from xotorch_amd.orchestration.node import Node
+    eng = TorchEngine()
+    node = Node("ds-test", None, eng, None, max_generate_tokens=6)
+    await node.start(0)
+    try:
+      shard = build_base_shard("ds-tiny", "TorchEngine")
+      done = asyncio.Event()
+      got = []
+      def on_token(rid, toks, fin):
+        got.extend(toks)
+        if fin:
+          done.set()
+      node.on_token.register("ds").on_next(on_token)
+      await node.process_prompt(shard, "hello deepseek", "ds-req")
+      await asyncio.wait_for(done.wait(), 120)
+      from xotorch_amd.models.deepseek_v3 import DeepseekV3Model
+      assert isinstance(eng.model, DeepseekV3Model)
+      # latent cache shape check on the live session's first layer
+      sess = next(iter(eng.sessions.values()), None)
+      if sess is not None:
+        k, v = sess.cache.caches[0][0], sess.cache.caches[0][1]
+        assert k.shape[1] == 1 and k.shape[3] == 16 and v.shape[3] == 8
+      assert 0 < len(got) <= 6
+    finally:
+      await node.stop()
+  asyncio.run(go())

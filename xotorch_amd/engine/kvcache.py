@@ -41,15 +41,20 @@ class ShardKVCache:
   """One LayerKV per local layer of a shard."""
 
   def __init__(self, n_layers: int, batch: int, n_kv_heads: int, capacity: int, head_dim: int,
-               dtype: torch.dtype = torch.bfloat16, device: str = "cpu"):
+               dtype: torch.dtype = torch.bfloat16, device: str = "cpu",
+               v_dim: Optional[int] = None):
+    """v_dim: per-position width of the v tensor when it differs from
+    head_dim (MLA latent caches: k stores the kv_lora latent, v the shared
+    roped key — see ModelConfig.kv_cache_dims())."""
     self.capacity = capacity
     self.batch = batch
     self.caches: List[LayerKV] = []
-    packed = _want_packed(device, head_dim, dtype)
+    vd = v_dim if v_dim is not None else head_dim
+    packed = _want_packed(device, head_dim, dtype) and vd == head_dim
     t32 = (capacity + 31) // 32 * 32
     for _ in range(n_layers):
       k = torch.zeros(batch, n_kv_heads, capacity, head_dim, dtype=dtype, device=device)
-      v = torch.zeros(batch, n_kv_heads, capacity, head_dim, dtype=dtype, device=device)
+      v = torch.zeros(batch, n_kv_heads, capacity, vd, dtype=dtype, device=device)
       if packed:
         kp = torch.zeros(batch, n_kv_heads, t32 // 16, 4, 64, 8, dtype=dtype, device=device)
         vp = torch.zeros(batch, n_kv_heads, 8, t32 // 32, 64, 8, dtype=dtype, device=device)

@@ -107,6 +107,9 @@ class TorchEngine(InferenceEngine):
     if cfg.model_type == "gemma2":
       from xotorch_amd.models.gemma2 import Gemma2Model
       model_cls = Gemma2Model
+    elif cfg.model_type in ("deepseek_v3", "deepseek_v2"):
+      from xotorch_amd.models.deepseek_v3 import DeepseekV3Model
+      model_cls = DeepseekV3Model
     else:
       model_cls = ShardedModel
     # construct at the target dtype so to_empty materializes it directly
@@ -120,11 +123,9 @@ class TorchEngine(InferenceEngine):
       torch.set_default_dtype(prev_dtype)
     model = model.to_empty(device=self.device)
     model = model.to(self.dtype)
-    # recompute rope tables (to_empty leaves buffers uninitialized)
-    from xotorch_amd.ops.torch_ref import rope_cos_sin
-    cos, sin = rope_cos_sin(cfg.head_dim, cfg.max_seq_len, cfg.rope_theta, cfg.rope_scaling, device=self.device)
-    model.rope_cos.copy_(cos)
-    model.rope_sin.copy_(sin)
+    # recompute rope tables (to_empty leaves buffers uninitialized); each
+    # decoder derives its own table shape (MLA uses qk_rope_head_dim)
+    model.reset_rope()
     if model_dir is not None and (
       (Path(model_dir) / "model.safetensors.index.json").exists() or (Path(model_dir) / "model.safetensors").exists()
     ):
@@ -146,9 +147,10 @@ class TorchEngine(InferenceEngine):
   def _session(self, request_id: str, batch: int, total_len: int) -> Session:
     sess = self.sessions.get(request_id)
     if sess is None or sess.cache.batch != batch or sess.cache.capacity < total_len:
+      heads, k_dim, v_dim = self.cfg.kv_cache_dims()
       cache = ShardKVCache(
-        n_layers=self.shard.get_layer_count(), batch=batch, n_kv_heads=self.cfg.n_kv_heads,
-        capacity=total_len, head_dim=self.cfg.head_dim, dtype=self.dtype, device=self.device,
+        n_layers=self.shard.get_layer_count(), batch=batch, n_kv_heads=heads,
+        capacity=total_len, head_dim=k_dim, dtype=self.dtype, device=self.device, v_dim=v_dim,
       )
       sess = Session(cache=cache, state=ShardInferenceState(total_len=total_len, batch=batch))
       self.sessions[request_id] = sess

@@ -57,6 +57,27 @@ class ModelConfig:
   attn_logit_softcapping: float = 0.0
   final_logit_softcapping: float = 0.0
   sliding_window: int = 0
+  # deepseek v3/r1 MLA + MoE (separate decoder, models/deepseek_v3.py):
+  q_lora_rank: int = 0
+  kv_lora_rank: int = 0
+  qk_rope_head_dim: int = 0
+  qk_nope_head_dim: int = 0
+  v_head_dim: int = 0
+  n_shared_experts: int = 0
+  n_group: int = 0
+  topk_group: int = 0
+  routed_scaling_factor: float = 1.0
+  norm_topk_prob: bool = True
+  first_k_dense_replace: int = 0
+  rope_interleave: bool = True
+
+  def kv_cache_dims(self):
+    """(heads, k_dim, v_dim) for per-layer cache allocation. MLA caches the
+    LATENT (kv_lora_rank) + the shared roped key (qk_rope_head_dim) instead
+    of expanded per-head K/V."""
+    if self.model_type in ("deepseek_v3", "deepseek_v2"):
+      return 1, self.kv_lora_rank, self.qk_rope_head_dim
+    return self.n_kv_heads, self.head_dim, self.head_dim
   torch_dtype: torch.dtype = torch.bfloat16
   bos_token_id: Optional[int] = None
   eos_token_id: Optional[int] = None
@@ -138,8 +159,21 @@ def config_from_hf(config_path: Path | str | dict, model_id: str = "unknown") ->
     torch_dtype=_DTYPE_MAP.get(raw.get("torch_dtype", "bfloat16"), torch.bfloat16),
     bos_token_id=raw.get("bos_token_id"),
     eos_token_id=eos,
-    n_experts=raw.get("num_local_experts", 0) or raw.get("num_experts", 0) or 0,
+    n_experts=raw.get("num_local_experts", 0) or raw.get("num_experts", 0)
+    or raw.get("n_routed_experts", 0) or 0,
     n_experts_per_tok=raw.get("num_experts_per_tok", 2),
     moe_intermediate_dim=raw.get("moe_intermediate_size", 0) or 0,
     moe_style="qwen3" if mtype == "qwen3_moe" else "mixtral",
+    q_lora_rank=raw.get("q_lora_rank") or 0,
+    kv_lora_rank=raw.get("kv_lora_rank") or 0,
+    qk_rope_head_dim=raw.get("qk_rope_head_dim") or 0,
+    qk_nope_head_dim=raw.get("qk_nope_head_dim") or 0,
+    v_head_dim=raw.get("v_head_dim") or 0,
+    n_shared_experts=raw.get("n_shared_experts") or 0,
+    n_group=raw.get("n_group") or 0,
+    topk_group=raw.get("topk_group") or 0,
+    routed_scaling_factor=float(raw.get("routed_scaling_factor", 1.0) or 1.0),
+    norm_topk_prob=bool(raw.get("norm_topk_prob", True)),
+    first_k_dense_replace=raw.get("first_k_dense_replace") or 0,
+    rope_interleave=bool(raw.get("rope_interleave", True)),
   )

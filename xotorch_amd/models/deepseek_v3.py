@@ -1,0 +1,275 @@
+"""Sharded DeepSeek-V3 / R1 decoder: Multi-head Latent Attention + MoE.
+
+A separate decoder (like models/gemma2.py) — MLA shares nothing with the GQA
+hot path: queries go through a LoRA bottleneck (q_a → norm → q_b), keys and
+values are EXPANDED from a compressed latent (kv_lora_rank + a single shared
+rope head), and only the LATENT is cached — the KV cache is
+(kv_lora_rank + qk_rope_head_dim) elements per token (576 for the real V3)
+instead of n_kv_heads*head_dim, MLA's whole point. MoE adds sigmoid routing
+with e-score correction bias, group-limited top-k (n_group/topk_group),
+routed scaling and always-on shared experts; the first k layers are dense.
+
+The reference lists deepseek-v3 / r1 cards its torchtune GQA assembly cannot
+run; here the family runs correctly on a plain-torch
+compute path validated against transformers' DeepseekV3ForCausalLM (tiny
+random configs — the full 671B needs the 8-GPU ring). The CDNA4 kernels are
+not used (MLA attention kernel: docs/ROADMAP.md). YaRN rope scaling for the
+full-length context is not implemented (default rope only).
+
+Cache contract: caches[idx] is a LayerKV whose k tensor stores the kv_nope
+latent [B, 1, T, kv_lora_rank] and v tensor stores the roped shared key
+[B, 1, T, qk_rope_head_dim] (see ModelConfig.kv_cache_dims()).
+"""
+from __future__ import annotations
+
+from typing import List
+
+import torch
+import torch.nn as nn
+
+from xotorch_amd.models.config import ModelConfig
+from xotorch_amd.ops.torch_ref import rope_cos_sin
+from xotorch_amd.shard import Shard
+
+
+def _rms(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+  xf = x.float()
+  out = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+  return (w.float() * out).to(x.dtype)
+
+
+def _rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, interleave: bool) -> torch.Tensor:
+  """x: [B, S, H, D]; cos/sin: [S, D/2] (per-pair angles)."""
+  c = cos[None, :, None, :].float()
+  s = sin[None, :, None, :].float()
+  xf = x.float()
+  if interleave:
+    x1, x2 = xf[..., 0::2], xf[..., 1::2]
+  else:
+    half = x.shape[-1] // 2
+    x1, x2 = xf[..., :half], xf[..., half:]
+  return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
+
+
+class MLAttention(nn.Module):
+  def __init__(self, cfg: ModelConfig):
+    super().__init__()
+    self.cfg = cfg
+    D, H = cfg.dim, cfg.n_heads
+    self.qk_head_dim = cfg.qk_nope_head_dim + cfg.qk_rope_head_dim
+    if cfg.q_lora_rank:
+      self.q_a_proj = nn.Linear(D, cfg.q_lora_rank, bias=cfg.attn_bias)
+      self.q_a_layernorm = nn.Parameter(torch.ones(cfg.q_lora_rank))
+      self.q_b_proj = nn.Linear(cfg.q_lora_rank, H * self.qk_head_dim, bias=False)
+    else:
+      self.q_proj = nn.Linear(D, H * self.qk_head_dim, bias=False)
+    self.kv_a_proj_with_mqa = nn.Linear(D, cfg.kv_lora_rank + cfg.qk_rope_head_dim, bias=cfg.attn_bias)
+    self.kv_a_layernorm = nn.Parameter(torch.ones(cfg.kv_lora_rank))
+    self.kv_b_proj = nn.Linear(cfg.kv_lora_rank, H * (cfg.qk_nope_head_dim + cfg.v_head_dim), bias=False)
+    self.o_proj = nn.Linear(H * cfg.v_head_dim, D, bias=cfg.attn_bias)
+    self.scale = self.qk_head_dim ** -0.5  # default rope (no yarn mscale)
+
+  def forward(self, x, cos, sin, positions, kv, start_pos: int):
+    cfg = self.cfg
+    B, S, _ = x.shape
+    H = cfg.n_heads
+    nope, rope_d, vd = cfg.qk_nope_head_dim, cfg.qk_rope_head_dim, cfg.v_head_dim
+    if cfg.q_lora_rank:
+      q = self.q_b_proj(_rms(self.q_a_proj(x), self.q_a_layernorm, cfg.norm_eps))
+    else:
+      q = self.q_proj(x)
+    q = q.view(B, S, H, self.qk_head_dim)
+    q_pass, q_rot = q[..., :nope], q[..., nope:]
+
+    ckv = self.kv_a_proj_with_mqa(x)
+    kv_nope, k_rot = ckv[..., : cfg.kv_lora_rank], ckv[..., cfg.kv_lora_rank:]
+    kv_nope = _rms(kv_nope, self.kv_a_layernorm, cfg.norm_eps)
+
+    cs, sn = cos[positions.reshape(-1)], sin[positions.reshape(-1)]
+    q_rot = _rope(q_rot, cs, sn, cfg.rope_interleave)
+    k_rot = _rope(k_rot.view(B, S, 1, rope_d), cs, sn, cfg.rope_interleave)
+
+    # latent cache: k tensor <- kv_nope [B,1,T,kv_lora], v tensor <- roped
+    # shared key [B,1,T,rope_d]
+    lat_c, rot_c = kv[0], kv[1]
+    lat_c[:, 0, start_pos: start_pos + S] = kv_nope
+    rot_c[:, 0, start_pos: start_pos + S] = k_rot[:, :, 0, :]
+    total = start_pos + S
+    lat = lat_c[:, 0, :total]                                   # [B, T, kv_lora]
+    krot = rot_c[:, 0, :total]                                  # [B, T, rope_d]
+
+    # expand latent -> per-head k_nope & v
+    kvx = self.kv_b_proj(lat).view(B, total, H, nope + vd)
+    k_nope, v = kvx[..., :nope], kvx[..., nope:]
+
+    qf = torch.cat([q_pass, q_rot], dim=-1).float()             # [B, S, H, qk]
+    kf = torch.cat([k_nope.float(), krot[:, :, None, :].expand(B, total, H, rope_d).float()], dim=-1)
+    scores = torch.einsum("bshd,bthd->bhst", qf, kf) * self.scale
+    qpos = torch.arange(start_pos, total, device=x.device)[:, None]
+    kpos = torch.arange(0, total, device=x.device)[None, :]
+    scores = scores.masked_fill((kpos > qpos)[None, None], float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    out = torch.einsum("bhst,bthd->bshd", probs, v.float()).to(x.dtype)
+    return self.o_proj(out.reshape(B, S, H * vd))
+
+
+class DsMLP(nn.Module):
+  def __init__(self, cfg: ModelConfig, intermediate: int):
+    super().__init__()
+    D = cfg.dim
+    self.gate_proj = nn.Linear(D, intermediate, bias=False)
+    self.up_proj = nn.Linear(D, intermediate, bias=False)
+    self.down_proj = nn.Linear(intermediate, D, bias=False)
+
+  def forward(self, x):
+    return self.down_proj(nn.functional.silu(self.gate_proj(x)) * self.up_proj(x))
+
+
+class DsMoE(nn.Module):
+  """Sigmoid-routed MoE with e-score correction bias, group-limited top-k,
+  routed scaling and shared experts (transformers DeepseekV3MoE semantics)."""
+
+  def __init__(self, cfg: ModelConfig):
+    super().__init__()
+    self.cfg = cfg
+    E = cfg.n_experts
+    self.gate_weight = nn.Parameter(torch.zeros(E, cfg.dim))
+    self.register_buffer("e_score_correction_bias", torch.zeros(E))
+    self.experts = nn.ModuleList([DsMLP(cfg, cfg.moe_intermediate_dim) for _ in range(E)])
+    self.shared_experts = DsMLP(cfg, cfg.moe_intermediate_dim * max(1, cfg.n_shared_experts))
+
+  def route(self, flat):
+    cfg = self.cfg
+    E, k = cfg.n_experts, cfg.n_experts_per_tok
+    logits = nn.functional.linear(flat.float(), self.gate_weight.float())
+    scores = logits.sigmoid()
+    choice = scores + self.e_score_correction_bias.float()
+    ng = max(1, cfg.n_group)
+    group_scores = choice.view(-1, ng, E // ng).topk(min(2, E // ng), dim=-1)[0].sum(-1)
+    gidx = torch.topk(group_scores, k=max(1, cfg.topk_group), dim=-1, sorted=False)[1]
+    gmask = torch.zeros_like(group_scores).scatter_(1, gidx, 1)
+    smask = gmask[:, :, None].expand(-1, ng, E // ng).reshape(-1, E)
+    choice = choice.masked_fill(~smask.bool(), float("-inf"))
+    idx = torch.topk(choice, k=k, dim=-1, sorted=False)[1]
+    w = scores.gather(1, idx)
+    if cfg.norm_topk_prob:
+      w = w / (w.sum(dim=-1, keepdim=True) + 1e-20)
+    return idx, w * cfg.routed_scaling_factor
+
+  def forward(self, x):
+    B, S, D = x.shape
+    flat = x.view(-1, D)
+    idx, w = self.route(flat)
+    out = torch.zeros_like(flat, dtype=torch.float32)
+    for e in range(self.cfg.n_experts):
+      tok, kk = torch.where(idx == e)
+      if tok.numel() == 0:
+        continue
+      out.index_add_(0, tok, self.experts[e](flat[tok]).float() * w[tok, kk, None])
+    out = out.to(x.dtype) + self.shared_experts(flat)
+    return out.view(B, S, D)
+
+
+class DsLayer(nn.Module):
+  def __init__(self, cfg: ModelConfig, layer_idx: int):
+    super().__init__()
+    self.eps = cfg.norm_eps
+    self.self_attn = MLAttention(cfg)
+    dense = layer_idx < cfg.first_k_dense_replace or cfg.n_experts == 0
+    self.mlp = DsMLP(cfg, cfg.intermediate_dim) if dense else DsMoE(cfg)
+    self.input_layernorm = nn.Parameter(torch.ones(cfg.dim))
+    self.post_attention_layernorm = nn.Parameter(torch.ones(cfg.dim))
+
+  def forward(self, h, cos, sin, positions, kv, start_pos):
+    h = h + self.self_attn(_rms(h, self.input_layernorm, self.eps), cos, sin, positions, kv, start_pos)
+    return h + self.mlp(_rms(h, self.post_attention_layernorm, self.eps))
+
+
+class DeepseekV3Model(nn.Module):
+  """Layer range [shard.start_layer .. shard.end_layer]; engine-contract
+  forward (same as ShardedModel / Gemma2Model)."""
+
+  def __init__(self, cfg: ModelConfig, shard: Shard):
+    super().__init__()
+    self.cfg = cfg
+    self.shard = shard
+    if shard.is_first_layer:
+      self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.dim)
+    self.layers = nn.ModuleDict(
+      {str(i): DsLayer(cfg, i) for i in range(shard.start_layer, shard.end_layer + 1)}
+    )
+    if shard.is_last_layer:
+      self.norm = nn.Parameter(torch.ones(cfg.dim))
+      self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+    # per-pair rope tables at the ROPE head dim
+    cos, sin = rope_cos_sin(cfg.qk_rope_head_dim, cfg.max_seq_len, cfg.rope_theta, None)
+    self.register_buffer("rope_cos", cos, persistent=False)
+    self.register_buffer("rope_sin", sin, persistent=False)
+
+  def reset_rope(self):
+    cos, sin = rope_cos_sin(self.cfg.qk_rope_head_dim, self.cfg.max_seq_len, self.cfg.rope_theta,
+                            None, device=self.rope_cos.device)
+    self.rope_cos, self.rope_sin = cos, sin
+
+  @property
+  def local_layer_ids(self) -> List[int]:
+    return list(range(self.shard.start_layer, self.shard.end_layer + 1))
+
+  def pack_decode_weights(self, reserve_bytes: int = 0) -> int:
+    return 0  # plain-torch compute path (no MFMA prepack yet)
+
+  def head_weight(self):
+    return self.lm_head.weight
+
+  def forward(self, x, caches, positions, start_pos: int, is_decode: bool = False,
+              seq_lens=None, last_only: bool = True):
+    cfg = self.cfg
+    h = self.embed_tokens(x) if x.dtype in (torch.int32, torch.int64) else x
+    if positions.dim() == 0:
+      positions = positions.reshape(1)
+    for idx, lid in enumerate(self.local_layer_ids):
+      h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, caches[idx], start_pos)
+    if not self.shard.is_last_layer:
+      return h
+    if last_only and h.shape[1] > 1:
+      h = h[:, -1:, :]
+    logits = torch.nn.functional.linear(_rms(h, self.norm, cfg.norm_eps),
+                                        self.head_weight().to(h.dtype))
+    if is_decode or last_only:
+      return logits[:, -1, :]
+    return logits
+
+
+def hf_key_map_deepseek(shard: Shard, cfg: ModelConfig):
+  """HF DeepseekV3ForCausalLM checkpoint keys → DeepseekV3Model keys."""
+  mapping = {}
+  if shard.is_first_layer:
+    mapping["model.embed_tokens.weight"] = "embed_tokens.weight"
+  if shard.is_last_layer:
+    mapping["model.norm.weight"] = "norm"
+    mapping["lm_head.weight"] = "lm_head.weight"
+  for lid in range(shard.start_layer, shard.end_layer + 1):
+    hf = f"model.layers.{lid}."
+    ours = f"layers.{lid}."
+    att = [("q_a_proj.weight", "q_a_proj.weight"), ("q_a_layernorm.weight", "q_a_layernorm"),
+           ("q_b_proj.weight", "q_b_proj.weight")] if cfg.q_lora_rank else [("q_proj.weight", "q_proj.weight")]
+    att += [("kv_a_proj_with_mqa.weight", "kv_a_proj_with_mqa.weight"),
+            ("kv_a_layernorm.weight", "kv_a_layernorm"),
+            ("kv_b_proj.weight", "kv_b_proj.weight"), ("o_proj.weight", "o_proj.weight")]
+    for hk, ok in att:
+      mapping[hf + "self_attn." + hk] = ours + "self_attn." + ok
+    mapping[hf + "input_layernorm.weight"] = ours + "input_layernorm"
+    mapping[hf + "post_attention_layernorm.weight"] = ours + "post_attention_layernorm"
+    dense = lid < cfg.first_k_dense_replace or cfg.n_experts == 0
+    if dense:
+      for p in ("gate_proj", "up_proj", "down_proj"):
+        mapping[hf + f"mlp.{p}.weight"] = ours + f"mlp.{p}.weight"
+    else:
+      mapping[hf + "mlp.gate.weight"] = ours + "mlp.gate_weight"
+      mapping[hf + "mlp.gate.e_score_correction_bias"] = ours + "mlp.e_score_correction_bias"
+      for e in range(cfg.n_experts):
+        for p in ("gate_proj", "up_proj", "down_proj"):
+          mapping[hf + f"mlp.experts.{e}.{p}.weight"] = ours + f"mlp.experts.{e}.{p}.weight"
+      for p in ("gate_proj", "up_proj", "down_proj"):
+        mapping[hf + f"mlp.shared_experts.{p}.weight"] = ours + f"mlp.shared_experts.{p}.weight"
+  return mapping

@@ -59,6 +59,25 @@ def _qwen3_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=151936, tie=False, max
   }
 
 
+def _deepseek_v3_cfg():
+  # DeepSeek-V3 / R1 (671B): MLA + 256-expert MoE; 8x MI355X (2.3 TB HBM)
+  # holds it at bf16 via the ring. YaRN long-context scaling not implemented
+  # (default rope window).
+  return {
+    "model_type": "deepseek_v3", "hidden_size": 7168, "num_hidden_layers": 61,
+    "num_attention_heads": 128, "num_key_value_heads": 128,
+    "intermediate_size": 18432, "moe_intermediate_size": 2048,
+    "n_routed_experts": 256, "num_experts_per_tok": 8, "n_shared_experts": 1,
+    "n_group": 8, "topk_group": 4, "routed_scaling_factor": 2.5,
+    "norm_topk_prob": True, "first_k_dense_replace": 3,
+    "q_lora_rank": 1536, "kv_lora_rank": 512, "qk_rope_head_dim": 64,
+    "qk_nope_head_dim": 128, "v_head_dim": 128, "vocab_size": 129280,
+    "rope_theta": 10000.0, "rms_norm_eps": 1e-6,
+    "max_position_embeddings": 4096, "torch_dtype": "bfloat16",
+    "bos_token_id": 0, "eos_token_id": 1,
+  }
+
+
 def _gemma2_cfg(dim, n_layers, n_heads, n_kv, inter, head_dim, qpas, window=4096):
   return {
     "model_type": "gemma2", "hidden_size": dim, "num_hidden_layers": n_layers,
@@ -145,6 +164,9 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
   # gemma2 (separate decoder, models/gemma2.py)
   "gemma2-9b": _gemma2_cfg(3584, 42, 16, 8, 14336, 256, 256),
   "gemma2-27b": _gemma2_cfg(4608, 46, 32, 16, 36864, 128, 144),
+  # deepseek v3/r1 (MLA decoder, models/deepseek_v3.py)
+  "deepseek-v3": _deepseek_v3_cfg(),
+  "deepseek-r1": _deepseek_v3_cfg(),
   # phi-4-mini (llama-like enough for the generic decoder)
   "phi-4-mini": _llama_cfg(3072, 32, 24, 8, 8192, vocab=200064, theta=10000.0, max_pos=131072, tie=True),
   # tiny test model
@@ -187,19 +209,18 @@ model_cards: Dict[str, dict] = {
   "qwen-2.5-math-72b": {"layers": 80, "repo": {"TorchEngine": "Qwen/Qwen2.5-Math-72B-Instruct", "HIPEngine": "Qwen/Qwen2.5-Math-72B-Instruct"}},
   "deepseek-r1-distill-qwen-14b": {"layers": 48, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-14B", "HIPEngine": "deepseek-ai/DeepSeek-R1-Distill-Qwen-14B"}},
   "nemotron-70b": {"layers": 80, "repo": {"TorchEngine": "nvidia/Llama-3.1-Nemotron-70B-Instruct-HF", "HIPEngine": "nvidia/Llama-3.1-Nemotron-70B-Instruct-HF"}},
-  # Listed for registry parity with the reference but with NO supported engine
-  # here: deepseek v3/r1 (MLA + shared experts) and llava (vision tower) are
-  # not implemented by this decoder — the reference lists them too, but its
-  # torchtune GQA assembly cannot run them either.
-  # get_supported_models() filters them out.
+  # Listed for registry parity with the reference but with NO supported
+  # engine here: llava needs a vision tower + image path (the reference
+  # lists it too; its torchtune GQA assembly cannot run it either —
+  # SURVEY.md appendix). get_supported_models() filters it out.
   "qwen-3-0.6b": {"layers": 28, "repo": {"TorchEngine": "Qwen/Qwen3-0.6B", "HIPEngine": "Qwen/Qwen3-0.6B"}},
   "qwen-3-8b": {"layers": 36, "repo": {"TorchEngine": "Qwen/Qwen3-8B", "HIPEngine": "Qwen/Qwen3-8B"}},
   "qwen-3-32b": {"layers": 64, "repo": {"TorchEngine": "Qwen/Qwen3-32B", "HIPEngine": "Qwen/Qwen3-32B"}},
   "qwen-3-30b-a3b": {"layers": 48, "repo": {"TorchEngine": "Qwen/Qwen3-30B-A3B", "HIPEngine": "Qwen/Qwen3-30B-A3B"}},
   "gemma2-9b": {"layers": 42, "repo": {"TorchEngine": "google/gemma-2-9b-it", "HIPEngine": "google/gemma-2-9b-it"}},
   "gemma2-27b": {"layers": 46, "repo": {"TorchEngine": "google/gemma-2-27b-it", "HIPEngine": "google/gemma-2-27b-it"}},
-  "deepseek-r1": {"layers": 61, "repo": {}},
-  "deepseek-v3": {"layers": 61, "repo": {}},
+  "deepseek-r1": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1", "HIPEngine": "deepseek-ai/DeepSeek-R1"}},
+  "deepseek-v3": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-V3", "HIPEngine": "deepseek-ai/DeepSeek-V3"}},
   "llava-1.5-7b-hf": {"layers": 32, "repo": {}},
   "dummy": {"layers": 4, "repo": {"TorchEngine": "dummy", "HIPEngine": "dummy", "DummyEngine": "dummy"}},
 }
@@ -243,8 +264,8 @@ pretty_names = {
   "qwen-3-30b-a3b": "Qwen 3 30B A3B (MoE)",
   "gemma2-9b": "Gemma2 9B",
   "gemma2-27b": "Gemma2 27B",
-  "deepseek-r1": "DeepSeek R1 (unsupported arch)",
-  "deepseek-v3": "DeepSeek V3 (unsupported arch)",
+  "deepseek-r1": "DeepSeek R1",
+  "deepseek-v3": "DeepSeek V3",
   "llava-1.5-7b-hf": "LLaVa 1.5 7B (unsupported arch)",
   "phi-4-mini": "Phi-4 Mini",
   "dummy": "Dummy (test)",

@@ -20,9 +20,13 @@ from xotorch_amd.models.llama import ShardedModel, hf_key_map as _hf_key_map_lla
 
 def hf_key_map(shard, cfg):
   """Checkpoint key map, routed by architecture."""
-  if getattr(cfg, "model_type", "llama") == "gemma2":
+  mtype = getattr(cfg, "model_type", "llama")
+  if mtype == "gemma2":
     from xotorch_amd.models.gemma2 import hf_key_map_gemma2
     return hf_key_map_gemma2(shard, cfg)
+  if mtype in ("deepseek_v3", "deepseek_v2"):
+    from xotorch_amd.models.deepseek_v3 import hf_key_map_deepseek
+    return hf_key_map_deepseek(shard, cfg)
   return _hf_key_map_llama(shard, cfg)
 from xotorch_amd.shard import Shard
 
