@@ -221,6 +221,8 @@ model_cards: Dict[str, dict] = {
   "gemma2-27b": {"layers": 46, "repo": {"TorchEngine": "google/gemma-2-27b-it", "HIPEngine": "google/gemma-2-27b-it"}},
   "deepseek-r1": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1", "HIPEngine": "deepseek-ai/DeepSeek-R1"}},
   "deepseek-v3": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-V3", "HIPEngine": "deepseek-ai/DeepSeek-V3"}},
+  # llava needs PIL for the image path (not present in this environment);
+  # the reference's vision handling is also vestigial
   "llava-1.5-7b-hf": {"layers": 32, "repo": {}},
   "dummy": {"layers": 4, "repo": {"TorchEngine": "dummy", "HIPEngine": "dummy", "DummyEngine": "dummy"}},
 }
