@@ -69,10 +69,17 @@ def main():
     pass
   device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
   dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+  cfg_override = None
   if device == "cpu" and args.model == "llama-3-70b":
-    # CPU smoke path (no GPU in dev container): tiny model, same code path
+    # CPU smoke path (no GPU in dev container): tiny model, same code path;
+    # give it >= world layers so every rank gets a shard
+    from xotorch_amd.models.registry import builtin_config
     args.model = "dummy"
     args.prompt_len = min(args.prompt_len, 128)  # dummy max_seq_len is 256
+    world_hint = int(os.getenv("WORLD_SIZE", "1"))
+    if world_hint > 4:
+      cfg_override = dict(builtin_config("dummy"))
+      cfg_override["num_hidden_layers"] = world_hint
 
   from xotorch_amd.parallel.ring import RingPipeline
 
@@ -80,7 +87,7 @@ def main():
   ring = RingPipeline(
     model_id=args.model, rank=rank, world=world, device=device, dtype=dtype,
     mb_batch=args.mb_batch, prompt_len=args.prompt_len, max_gen=max_gen,
-    use_graphs=not args.no_graphs,
+    use_graphs=not args.no_graphs, cfg_override=cfg_override,
   )
   global_batch = args.mb_batch * ring.M
 

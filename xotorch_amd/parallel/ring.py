@@ -76,6 +76,11 @@ class RingPipeline:
     if self.total_len > self.cfg.max_seq_len:
       raise ValueError("prompt+gen exceeds model max_seq_len")
     shards = equal_ring_shards(model_id, self.cfg.n_layers, world)
+    if len(shards) < world:
+      raise ValueError(
+        f"ring world {world} exceeds the model's shardable layers "
+        f"({self.cfg.n_layers}): {len(shards)} non-empty shards"
+      )
     self.shard: Shard = shards[rank]
     self.is_first = self.shard.is_first_layer
     self.is_last = self.shard.is_last_layer
