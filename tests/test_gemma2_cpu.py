@@ -113,3 +113,42 @@ def test_gemma2_through_engine_and_node(monkeypatch):
     finally:
       await node.stop()
   asyncio.run(go())
+
+
+def test_gemma2_two_node_tcp_ring(monkeypatch, tmp_path):
+  """Gemma2 over the ring: two nodes split the layers over the TCP wire."""
+  import asyncio
+  from xotorch_amd.models import registry
+  from tests.test_node_cpu import _make_tcp_pair, run
+
+  tiny = dict(TINY)
+  tiny["model_type"] = "gemma2"
+  tiny["vocab_size"] = 256
+  monkeypatch.setitem(registry.BUILTIN_CONFIGS, "gemma2-tiny2", tiny)
+  monkeypatch.setitem(registry.model_cards, "gemma2-tiny2",
+                      {"layers": 4, "repo": {"TorchEngine": "dummy"}})
+
+  async def go():
+    import torch
+    from xotorch_amd.engine.torch_engine import TorchEngine
+    from xotorch_amd.models.registry import build_base_shard
+    nodes = await _make_tcp_pair(tmp_path, lambda: TorchEngine(device="cpu", dtype=torch.float32))
+    a, b = nodes
+    shard = build_base_shard("gemma2-tiny2", "TorchEngine")
+    done = asyncio.Event()
+    got = []
+    def on_token(rid, toks, fin):
+      got[:] = toks
+      if fin:
+        done.set()
+    a.on_token.register("g2-t").on_next(on_token)
+    await a.process_prompt(shard, "softcapped ring", "g2-tcp")
+    await asyncio.wait_for(done.wait(), 120)
+    assert len(got) >= 1
+    from xotorch_amd.models.gemma2 import Gemma2Model
+    assert isinstance(a.inference_engine.model, Gemma2Model)
+    assert isinstance(b.inference_engine.model, Gemma2Model)
+    for n in nodes:
+      await n.stop()
+    return True
+  assert run(go())
