@@ -95,8 +95,9 @@ def main():
   stats = ring.prefill()
   ttfts = torch.tensor(stats.ttft_ms, dtype=torch.float64, device="cpu")
   if world > 1:
-    # TTFT lives on the last stage; ship it to rank 0
-    if device == "cuda":
+    # TTFT lives on the last stage; ship it to rank 0 (gloo groups move it
+    # on host, nccl on device)
+    if device == "cuda" and dist.get_backend() != "gloo":
       ttfts = ttfts.to(device)
     dist.broadcast(ttfts, world - 1)
     ttfts = ttfts.cpu()

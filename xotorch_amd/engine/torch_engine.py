@@ -189,6 +189,9 @@ class TorchEngine(InferenceEngine):
   def _infer_tensor_sync(self, request_id: str, input_data: np.ndarray, inference_state: Optional[dict]):
     meta = ShardInferenceState.from_dict(inference_state)
     x = torch.from_numpy(np.ascontiguousarray(input_data))
+    if (inference_state or {}).get("wire_dtype") == "bfloat16" and x.dtype == torch.int16:
+      # bf16 hidden state shipped as raw int16 bytes (numpy has no bf16)
+      x = x.view(torch.bfloat16)
     is_tokens = x.dtype in (torch.int32, torch.int64)
     B = x.shape[0]
     S = x.shape[1]
@@ -222,11 +225,19 @@ class TorchEngine(InferenceEngine):
     new_pos = start_pos + S
     sess.state.curr_pos = new_pos
     sess.state.total_len = total_len
-    out_np = out.float().cpu().numpy()
     # merge pass-through metadata (max_tokens, traceparent, ...) so request
     # attributes survive the hop — engines own only their position keys
     out_state = dict(inference_state or {})
     out_state.update({"curr_pos": new_pos, "total_len": total_len, "batch": B})
+    if not self.shard.is_last_layer and out.dtype == torch.bfloat16:
+      # mid-ring hidden states travel as bf16 bytes (half the wire bytes of
+      # the reference's fp32-numpy hop); last-stage logits stay fp32 for the
+      # local sampling site
+      out_np = out.contiguous().cpu().view(torch.int16).numpy()
+      out_state["wire_dtype"] = "bfloat16"
+    else:
+      out_np = out.float().cpu().numpy()
+      out_state.pop("wire_dtype", None)
     return out_np, out_state
 
   # ---------- training (capability the reference declared but never built) ----------
@@ -300,6 +311,15 @@ class TorchEngine(InferenceEngine):
       self._lora_applied = True
       self._opt = None  # rebuild over the adapter params
 
+  def _ce_loss(self, out, tgt, lengths):
+    lens = torch.from_numpy(np.ascontiguousarray(lengths)).to(self.device)
+    S = tgt.shape[1]
+    mask = torch.arange(S, device=self.device)[None, :] < lens[:, None]
+    lg = out.float().reshape(-1, out.shape[-1])
+    lt = tgt.reshape(-1).long()
+    ce = torch.nn.functional.cross_entropy(lg, lt, reduction="none").reshape(tgt.shape)
+    return (ce * mask).sum() / mask.sum().clamp(min=1)
+
   def _train_sync(self, request_id, inputs, targets, lengths, loss, opt, do_step):
     self._maybe_apply_lora()
     model = self.model
@@ -307,6 +327,17 @@ class TorchEngine(InferenceEngine):
     model.train(do_step)
     x = torch.from_numpy(np.ascontiguousarray(inputs)).to(self.device)
     is_tokens = x.dtype in (torch.int32, torch.int64)
+    if not do_step and loss != "back_gradient":
+      # evaluation: loss only, under no_grad — a backward here would leave
+      # accumulated grads contaminating the next optimizer step
+      if not model.shard.is_last_layer:
+        raise ValueError("non-last shard evaluates by forwarding activations, not locally")
+      with torch.no_grad():
+        out = self._forward_nocache(x)
+        tgt = torch.from_numpy(np.ascontiguousarray(targets)).to(self.device)
+        loss_val = float(self._ce_loss(out, tgt, lengths))
+      model.train(was_training)
+      return loss_val, None
     if not is_tokens:
       x = x.to(self.dtype).requires_grad_(True)
     with torch.enable_grad():
@@ -319,13 +350,7 @@ class TorchEngine(InferenceEngine):
       else:
         tgt = torch.from_numpy(np.ascontiguousarray(targets)).to(self.device)
         if model.shard.is_last_layer:
-          lens = torch.from_numpy(np.ascontiguousarray(lengths)).to(self.device)
-          S = tgt.shape[1]
-          mask = torch.arange(S, device=self.device)[None, :] < lens[:, None]
-          lg = out.float().reshape(-1, out.shape[-1])
-          lt = tgt.reshape(-1).long()
-          ce = torch.nn.functional.cross_entropy(lg, lt, reduction="none").reshape(tgt.shape)
-          loss_t = (ce * mask).sum() / mask.sum().clamp(min=1)
+          loss_t = self._ce_loss(out, tgt, lengths)
           loss_t.backward()
           loss_val = float(loss_t.detach())
         else:

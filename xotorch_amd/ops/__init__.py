@@ -72,8 +72,9 @@ def rmsnorm_residual(
 
 
 def rope_apply(q, k, cos, sin, positions):
-  if _use_hip(q, k) and q.dtype == torch.bfloat16:
-    return _hip.rope_apply(q, k, cos, sin, positions)
+  """Standalone RoPE (training / oracle paths). On the GPU inference hot
+  path RoPE is fused into rope_qkv_append — there is deliberately no
+  separate HIP kernel for this op."""
   return torch_ref.rope_apply(q, k, cos, sin, positions)
 
 

@@ -83,7 +83,8 @@ class Server:
         result = msg.get("result")
         if result is None and msg.get("tensor_result") is not None:
           result = wire.unpack_tensor(msg["tensor_result"])
-        self.node.on_token.trigger_all(msg["request_id"], result, msg["is_finished"])
+        # finished results release this stage's KV session too (handle_result)
+        await self.node.handle_result(msg["request_id"], result, msg["is_finished"])
         return {"ok": True}
       if t == "status":
         self.node.on_opaque_status.trigger_all(msg.get("request_id", ""), msg["status"])

@@ -44,6 +44,11 @@ async def read_frame(reader: asyncio.StreamReader) -> Dict[str, Any]:
 
 def write_frame(writer: asyncio.StreamWriter, msg: Dict[str, Any]) -> None:
   payload = msgpack.packb(msg, use_bin_type=True)
+  if len(payload) > MAX_FRAME:
+    # fail on the SEND side with a clear error; otherwise the receiver
+    # rejects the oversized frame and kills the connection with no clue
+    # at the sender (e.g. a >256 MB activation hop at large batch)
+    raise ValueError(f"frame too large to send: {len(payload)} > {MAX_FRAME} bytes")
   writer.write(struct.pack("!I", len(payload)) + payload)
 
 

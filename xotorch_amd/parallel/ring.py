@@ -29,6 +29,7 @@ from xotorch_amd.models.config import ModelConfig, config_from_hf
 from xotorch_amd.models.llama import ShardedModel
 from xotorch_amd.models.registry import builtin_config
 from xotorch_amd.models.weights import fast_random_init_gpu, random_init
+from xotorch_amd.parallel.comm import RingComm
 from xotorch_amd.parallel.partitioning import Partition, map_partitions_to_shards
 from xotorch_amd.shard import Shard
 
@@ -86,6 +87,7 @@ class RingPipeline:
     self.is_last = self.shard.is_last_layer
     self.next_rank = (rank + 1) % world
     self.prev_rank = (rank - 1) % world
+    self.comm = RingComm(device)
 
     # --- model ---
     # construct at the target dtype so to_empty materializes bf16 directly
@@ -246,7 +248,7 @@ class RingPipeline:
           h = self._prefill_forward(mb, tokens, pos)
         else:
           hbuf = torch.empty(B, S, self.cfg.dim, dtype=self.dtype, device=self.device)
-          dist.recv(hbuf, self.prev_rank)
+          self.comm.recv(hbuf, self.prev_rank)
           h = self._prefill_forward(mb, hbuf, pos)
         if self.is_last:
           tok = h.argmax(dim=-1, keepdim=True)  # h is [B, V] logits
@@ -257,13 +259,13 @@ class RingPipeline:
             self.tok_buf[mb].copy_(tok)
           else:
             self._tok_out[mb].copy_(tok)
-            self._tok_send_req[mb] = dist.isend(self._tok_out[mb], self.next_rank)
+            self._tok_send_req[mb] = self.comm.isend(self._tok_out[mb], self.next_rank)
           if self.capture_tokens:
             self.generated[mb].append(tok.clone())
         else:
-          dist.send(h.contiguous(), self.next_rank)
+          self.comm.send(h, self.next_rank)
         if self.is_first and self.world > 1:
-          self._tok_recv_req[mb] = dist.irecv(self.tok_buf[mb], self.prev_rank)
+          self._tok_recv_req[mb] = self.comm.irecv(self.tok_buf[mb], self.prev_rank)
         # advance to first decode position
         self.positions[mb].fill_(S)
         self.seq_lens[mb].fill_(S + 1)
@@ -283,18 +285,18 @@ class RingPipeline:
         if self.is_first:
           self._tok_recv_req[mb].wait()
         else:
-          dist.recv(self.hid_buf[mb], self.prev_rank)
+          self.comm.recv(self.hid_buf[mb], self.prev_rank)
         if self.is_last and self._tok_send_req[mb] is not None:
           self._tok_send_req[mb].wait()
         self._run_mb(mb)
         if self.is_last:
-          self._tok_send_req[mb] = dist.isend(self._tok_out[mb], self.next_rank)
+          self._tok_send_req[mb] = self.comm.isend(self._tok_out[mb], self.next_rank)
           if self.capture_tokens:
             self.generated[mb].append(self._tok_out[mb].clone())
         else:
-          dist.send(self.hid_out[mb], self.next_rank)
+          self.comm.send(self.hid_out[mb], self.next_rank)
         if self.is_first:
-          self._tok_recv_req[mb] = dist.irecv(self.tok_buf[mb], self.prev_rank)
+          self._tok_recv_req[mb] = self.comm.irecv(self.tok_buf[mb], self.prev_rank)
 
   def _run_mb(self, mb: int):
     if self._graphs[mb] is not None:

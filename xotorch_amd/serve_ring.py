@@ -73,11 +73,17 @@ class RingWorker:
     dev = self.device
     hdr = torch.zeros(2, dtype=torch.int64, device=dev if self.world > 1 else "cpu")
     if self.rank == 0:
+      if tokens.shape[1] >= MAX_SEQ:
+        # reject BEFORE the ring is engaged (followers never see the request)
+        raise ValueError(f"prompt length {tokens.shape[1]} exceeds context {MAX_SEQ}")
       hdr[0] = tokens.shape[1]
       hdr[1] = max_new
     if self.world > 1:
       dist.broadcast(hdr, 0)
     S, max_new = int(hdr[0]), int(hdr[1])
+    # clamp so KV writes can never pass cache capacity (the HIP append
+    # kernel trusts pos < T)
+    max_new = min(max_new, MAX_SEQ - S)
     if self.world > 1:
       tok_bcast = torch.zeros(1, S, dtype=torch.int64, device=dev)
       if self.rank == 0:
