@@ -24,9 +24,10 @@ def _build_pair(seed=0):
   cfg = config_from_hf(raw, "gemma2-tiny")
   shard = Shard("gemma2-tiny", 0, cfg.n_layers - 1, cfg.n_layers)
   ours = Gemma2Model(cfg, shard).float()
+  from xotorch_amd.models.weights import remap_hf_state
   sd = ref.state_dict()
   mapping = hf_key_map_gemma2(shard, cfg)
-  ours.load_state_dict({v: sd[k] for k, v in mapping.items()}, strict=False)
+  ours.load_state_dict(remap_hf_state(sd, mapping), strict=False)
   ours.eval()
   return ref, ours, cfg, shard
 
@@ -66,9 +67,10 @@ def test_gemma2_split_equals_full():
     s0, s1 = Shard("gemma2-tiny", 0, 1, 4), Shard("gemma2-tiny", 2, 3, 4)
     sd = ref.state_dict()
     m0 = Gemma2Model(cfg, s0).float()
-    m0.load_state_dict({v: sd[k] for k, v in hf_key_map_gemma2(s0, cfg).items()}, strict=False)
+    from xotorch_amd.models.weights import remap_hf_state
+    m0.load_state_dict(remap_hf_state(sd, hf_key_map_gemma2(s0, cfg)), strict=False)
     m1 = Gemma2Model(cfg, s1).float()
-    m1.load_state_dict({v: sd[k] for k, v in hf_key_map_gemma2(s1, cfg).items()}, strict=False)
+    m1.load_state_dict(remap_hf_state(sd, hf_key_map_gemma2(s1, cfg)), strict=False)
     m0.eval(); m1.eval()
     c0, c1 = _caches(cfg, 2, B, S + 2), _caches(cfg, 2, B, S + 2)
     h = m0(x, caches=c0.caches, positions=torch.arange(S), start_pos=0)

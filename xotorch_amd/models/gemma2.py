@@ -9,11 +9,15 @@ final-logit soft-capping. The reference lists gemma2 model cards but its
 torchtune GQA assembly implements none of this
 (/root/reference/xotorch/inference/torch/models/general_mha.py:23-254 — no
 softcap/sliding-window/post-norms), so those cards cannot produce correct
-outputs there; here the family runs correctly on a plain-torch compute path
-(CPU or GPU) validated against transformers' Gemma2ForCausalLM. The
-hand-written CDNA4 kernels are NOT used for gemma2 (soft-capping is not in
-the MFMA attention kernels yet — docs/ROADMAP.md); this is a
-capability-completeness path, not a speed path.
+outputs there.
+
+On MI355X the family runs the SAME CDNA4 kernel path as llama: fused qkv
+XotLinear GEMMs (packed skinny kernels), the fused RoPE+KV-append kernel
+into the MFMA-packed cache, MFMA flash attention with softcap/window flags
+(hip_ops.hip attn_*_mfma), gemma RMSNorm via the HIP rmsnorm kernel's
+w_bias=1 mode, and a packed GeGLU kernel. head_dim must be 128 for the
+MFMA attention (gemma2-27b); other dims fall back to the eager oracle path
+(also the CPU path, validated against transformers' Gemma2ForCausalLM).
 
 Shard semantics match ShardedModel: first shard owns the (scaled) embedding,
 last owns the final norm + tied head; caches are the engine's LayerKV pairs.
@@ -26,19 +30,28 @@ from typing import List, Optional
 import torch
 import torch.nn as nn
 
+from xotorch_amd import ops
 from xotorch_amd.models.config import ModelConfig
+from xotorch_amd.models.llama import XotLinear
 from xotorch_amd.ops.torch_ref import rope_cos_sin
 from xotorch_amd.shard import Shard
 
 
 def _rms(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
-  xf = x.float()
-  out = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
-  return (out * (1.0 + w.float())).to(x.dtype)
+  # gemma convention: fp32 norm scaled by (1 + w); the HIP rmsnorm kernel
+  # runs this as its w_bias=1 mode on GPU
+  return ops.rmsnorm(x.contiguous(), w, eps, w_bias=1.0)
 
 
 def _softcap(x: torch.Tensor, cap: float) -> torch.Tensor:
   return torch.tanh(x / cap) * cap if cap else x
+
+
+def _use_mfma(x: torch.Tensor, hd: int, kv) -> bool:
+  return (
+    x.is_cuda and x.dtype == torch.bfloat16 and hd == 128
+    and len(kv) > 2 and kv[2] is not None and ops.hip_available()
+  )
 
 
 class Gemma2Attention(nn.Module):
@@ -46,32 +59,48 @@ class Gemma2Attention(nn.Module):
     super().__init__()
     self.cfg = cfg
     H, KVH, hd, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim, cfg.dim
-    self.q_proj = nn.Linear(D, H * hd, bias=False)
-    self.k_proj = nn.Linear(D, KVH * hd, bias=False)
-    self.v_proj = nn.Linear(D, KVH * hd, bias=False)
-    self.o_proj = nn.Linear(H * hd, D, bias=False)
+    # fused qkv: one [D, (H+2KVH)*hd] GEMM feeding the RoPE+append kernel
+    self.qkv_proj = XotLinear(D, (H + 2 * KVH) * hd, bias=False)
+    self.o_proj = XotLinear(H * hd, D, bias=False)
     self.scale = (cfg.query_pre_attn_scalar or cfg.head_dim) ** -0.5
     # even layers use the sliding window, odd layers are global (transformers
     # Gemma2DecoderLayer: is_sliding = not bool(layer_idx % 2))
     self.window = cfg.sliding_window if (layer_idx % 2 == 0) else 0
 
-  def forward(self, x, cos, sin, positions, kv, start_pos: int):
+  def forward(self, x, cos, sin, positions, kv, start_pos: int,
+              is_decode: bool = False, seq_lens=None):
     cfg = self.cfg
     B, S, _ = x.shape
     H, KVH, hd = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
-    from xotorch_amd.ops import torch_ref as tr
-    q = self.q_proj(x).view(B, S, H, hd)
-    k = self.k_proj(x).view(B, S, KVH, hd)
-    v = self.v_proj(x).view(B, S, KVH, hd)
-    q, k = tr.rope_apply(q, k, cos, sin, positions)
+    cap = cfg.attn_logit_softcapping or 0.0
+    qkv = self.qkv_proj(x)  # [B, S, (H+2KVH)*hd]
     k_cache, v_cache = kv[0], kv[1]
+    if _use_mfma(qkv, hd, kv):
+      kp, vp = kv[2], kv[3]
+      ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, KVH, hd, kp, vp)
+      q = qkv[:, :, : H * hd].view(B, S, H, hd)
+      if is_decode:
+        sl = seq_lens if seq_lens is not None else start_pos + 1
+        out = ops.attn_decode(q, k_cache, v_cache, sl, kp, vp,
+                              scale=self.scale, softcap=cap, window=self.window)
+      else:
+        out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S, kp, vp,
+                               scale=self.scale, softcap=cap, window=self.window)
+      return self.o_proj(out.reshape(B, S, H * hd))
+    # eager oracle path (CPU / hd != 128): split the fused projection
+    from xotorch_amd.ops import torch_ref as tr
+    q, k, v = torch.split(qkv, [H * hd, KVH * hd, KVH * hd], dim=-1)
+    q = q.view(B, S, H, hd)
+    k = k.view(B, S, KVH, hd)
+    v = v.view(B, S, KVH, hd)
+    q, k = tr.rope_apply(q, k, cos, sin, positions)
     tr.kv_append(k_cache, v_cache, k, v, start_pos)
     total = start_pos + S
     rep = H // KVH
     kk = k_cache[:, :, :total].repeat_interleave(rep, dim=1)  # [B, H, T, hd]
     vv = v_cache[:, :, :total].repeat_interleave(rep, dim=1)
     scores = torch.einsum("bshd,bhtd->bhst", q.float(), kk.float()) * self.scale
-    scores = _softcap(scores, cfg.attn_logit_softcapping)
+    scores = _softcap(scores, cap)
     qpos = torch.arange(start_pos, total, device=x.device)[:, None]
     kpos = torch.arange(0, total, device=x.device)[None, :]
     mask = kpos <= qpos
@@ -87,14 +116,20 @@ class Gemma2MLP(nn.Module):
   def __init__(self, cfg: ModelConfig):
     super().__init__()
     I, D = cfg.intermediate_dim, cfg.dim
-    self.gate_proj = nn.Linear(D, I, bias=False)
-    self.up_proj = nn.Linear(D, I, bias=False)
-    self.down_proj = nn.Linear(I, D, bias=False)
+    # fused [gate | up] GEMM + packed GeGLU kernel on GPU
+    self.gate_up_proj = XotLinear(D, 2 * I, bias=False)
+    self.down_proj = XotLinear(I, D, bias=False)
+    self.intermediate = I
 
   def forward(self, x):
-    return self.down_proj(
-      nn.functional.gelu(self.gate_proj(x), approximate="tanh") * self.up_proj(x)
-    )
+    gu = self.gate_up_proj(x)
+    if gu.is_cuda and gu.dtype == torch.bfloat16 and ops.hip_available():
+      h = ops.geglu_packed(gu)
+    else:
+      I = self.intermediate
+      h = (nn.functional.gelu(gu[..., :I].float(), approximate="tanh")
+           * gu[..., I:].float()).to(gu.dtype)
+    return self.down_proj(h)
 
 
 class Gemma2Layer(nn.Module):
@@ -109,10 +144,10 @@ class Gemma2Layer(nn.Module):
     self.pre_feedforward_layernorm = nn.Parameter(torch.zeros(D))
     self.post_feedforward_layernorm = nn.Parameter(torch.zeros(D))
 
-  def forward(self, h, cos, sin, positions, kv, start_pos):
+  def forward(self, h, cos, sin, positions, kv, start_pos, is_decode=False, seq_lens=None):
     res = h
     hs = _rms(h, self.input_layernorm, self.eps)
-    hs = self.self_attn(hs, cos, sin, positions, kv, start_pos)
+    hs = self.self_attn(hs, cos, sin, positions, kv, start_pos, is_decode, seq_lens)
     h = res + _rms(hs, self.post_attention_layernorm, self.eps)
     res = h
     hs = _rms(h, self.pre_feedforward_layernorm, self.eps)
@@ -151,8 +186,11 @@ class Gemma2Model(nn.Module):
   def local_layer_ids(self) -> List[int]:
     return list(range(self.shard.start_layer, self.shard.end_layer + 1))
 
-  def pack_decode_weights(self, reserve_bytes: int = 0) -> int:
-    return 0  # gemma2 runs the plain-torch path (no MFMA prepack yet)
+  def pack_decode_weights(self, reserve_bytes: int = 8 << 30) -> int:
+    """Same greedy prepack policy as ShardedModel (descending measured-win
+    order, free-memory guarded)."""
+    from xotorch_amd.models.llama import ShardedModel
+    return ShardedModel.pack_decode_weights(self, reserve_bytes)
 
   def head_weight(self):
     if hasattr(self, "embed_tokens"):
@@ -169,7 +207,8 @@ class Gemma2Model(nn.Module):
     if positions.dim() == 0:
       positions = positions.reshape(1)
     for idx, lid in enumerate(self.local_layer_ids):
-      h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, caches[idx], start_pos)
+      h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, caches[idx],
+                                start_pos, is_decode, seq_lens)
     if not self.shard.is_last_layer:
       return h
     if last_only and h.shape[1] > 1:
@@ -183,7 +222,9 @@ class Gemma2Model(nn.Module):
 
 
 def hf_key_map_gemma2(shard: Shard, cfg: ModelConfig):
-  """HF Gemma2ForCausalLM checkpoint keys → Gemma2Model state-dict keys."""
+  """HF Gemma2ForCausalLM checkpoint keys → Gemma2Model state-dict keys.
+  q/k/v concatenate into the fused qkv_proj, gate/up into gate_up_proj
+  (the loader concatenates `key#N` parts along dim 0 in N order)."""
   mapping = {}
   if shard.is_first_layer:
     mapping["model.embed_tokens.weight"] = "embed_tokens.weight"
@@ -194,10 +235,12 @@ def hf_key_map_gemma2(shard: Shard, cfg: ModelConfig):
   for lid in range(shard.start_layer, shard.end_layer + 1):
     hf = f"model.layers.{lid}."
     ours = f"layers.{lid}."
-    for p in ("q_proj", "k_proj", "v_proj", "o_proj"):
-      mapping[hf + f"self_attn.{p}.weight"] = ours + f"self_attn.{p}.weight"
-    for p in ("gate_proj", "up_proj", "down_proj"):
-      mapping[hf + f"mlp.{p}.weight"] = ours + f"mlp.{p}.weight"
+    for i, p in enumerate(("q_proj", "k_proj", "v_proj")):
+      mapping[hf + f"self_attn.{p}.weight"] = ours + f"self_attn.qkv_proj.weight#{i}"
+    mapping[hf + "self_attn.o_proj.weight"] = ours + "self_attn.o_proj.weight"
+    for i, p in enumerate(("gate_proj", "up_proj")):
+      mapping[hf + f"mlp.{p}.weight"] = ours + f"mlp.gate_up_proj.weight#{i}"
+    mapping[hf + "mlp.down_proj.weight"] = ours + "mlp.down_proj.weight"
     for p in ("input_layernorm", "post_attention_layernorm",
               "pre_feedforward_layernorm", "post_feedforward_layernorm"):
       mapping[hf + f"{p}.weight"] = ours + p

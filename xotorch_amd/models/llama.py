@@ -108,19 +108,27 @@ class XotLinear(nn.Linear):
           use = _PACKED_WINS.get(key)
           if use is None:
             if torch.cuda.is_current_stream_capturing():
-              use = True  # no timing under capture; normal warmup decides first
+              use = "packed"  # no timing under capture; normal warmup decides first
             else:
               wp, w, b = self.weight_packed, self.weight, self.bias
               t_packed = _time_us(lambda: hip.skinny_gemm_packed(x, wp, N, b))
+              t_xreg = _time_us(lambda: hip.skinny_gemm_packed_xreg(x, wp, N, b))
               t_blaslt = _time_us(lambda: torch.nn.functional.linear(x, w, b))
-              # require a clear (>5%) win: real packed wins measure 20%+ and
-              # borderline shapes would otherwise flip run-to-run with DVFS
-              use = t_packed < t_blaslt * 0.95
+              # require a clear (>5%) win over hipBLASLt: real wins measure
+              # 20%+ and borderline shapes would flip run-to-run with DVFS
+              use = "blaslt"
+              t_best = t_blaslt * 0.95
+              if t_packed < t_best:
+                use, t_best = "packed", t_packed
+              if t_xreg < t_best:
+                use = "xreg"
               _PACKED_WINS[key] = use
               if os.getenv("XOT_DEBUG", "0") != "0":
                 print(f"[xot] gemm auto-pick N={N} K={K} M={M}: packed {t_packed:.1f} us "
-                      f"vs blaslt {t_blaslt:.1f} us -> {'packed' if use else 'blaslt'}", flush=True)
-          if use:
+                      f"/ xreg {t_xreg:.1f} us vs blaslt {t_blaslt:.1f} us -> {use}", flush=True)
+          if use == "xreg":
+            return hip.skinny_gemm_packed_xreg(x, self.weight_packed, N, self.bias)
+          if use == "packed":
             return hip.skinny_gemm_packed(x, self.weight_packed, N, self.bias)
     return ops.linear(x, self.weight, self.bias)
 

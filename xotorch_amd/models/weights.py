@@ -91,6 +91,25 @@ def load_shard_weights(model: ShardedModel, model_dir: Path, device: str = "cpu"
   return len(state)
 
 
+def remap_hf_state(sd: Dict[str, torch.Tensor], mapping: Dict[str, str]) -> Dict[str, torch.Tensor]:
+  """Apply an hf_key_map to an in-memory HF state dict: rename keys and
+  concatenate `key#N` fused parts along dim 0 in N order (same semantics as
+  load_shard_weights' safetensors path; used by tests and converters)."""
+  state: Dict[str, torch.Tensor] = {}
+  packed: Dict[str, Dict[int, torch.Tensor]] = {}
+  for hf_key, our in mapping.items():
+    if hf_key not in sd:
+      continue
+    if "#" in our:
+      base, part = our.split("#")
+      packed.setdefault(base, {})[int(part)] = sd[hf_key]
+    else:
+      state[our] = sd[hf_key]
+  for base, parts in packed.items():
+    state[base] = torch.cat([parts[i] for i in sorted(parts)], dim=0)
+  return state
+
+
 @torch.no_grad()
 def random_init(model: ShardedModel, seed: int = 1234, std: float = 0.02) -> None:
   """Deterministic random init (synthetic-weights benches and tests).

@@ -57,18 +57,20 @@ def _use_hip(*tensors: torch.Tensor) -> bool:
   )
 
 
-def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float, w_bias: float = 0.0) -> torch.Tensor:
+  """w_bias=1.0 gives the gemma convention (scale by 1 + w, fp32 math)."""
   if _use_hip(x, weight) and x.dtype == torch.bfloat16:
-    return _hip.rmsnorm(x, weight, eps)
-  return torch_ref.rmsnorm(x, weight, eps)
+    return _hip.rmsnorm(x, weight, eps, w_bias)
+  return torch_ref.rmsnorm(x, weight, eps, w_bias)
 
 
 def rmsnorm_residual(
-  x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+  x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float,
+  w_bias: float = 0.0
 ) -> Tuple[torch.Tensor, torch.Tensor]:
   if _use_hip(x, residual, weight) and x.dtype == torch.bfloat16:
-    return _hip.rmsnorm_residual(x, residual, weight, eps)
-  return torch_ref.rmsnorm_residual(x, residual, weight, eps)
+    return _hip.rmsnorm_residual(x, residual, weight, eps, w_bias)
+  return torch_ref.rmsnorm_residual(x, residual, weight, eps, w_bias)
 
 
 def rope_apply(q, k, cos, sin, positions):
@@ -94,12 +96,14 @@ def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_
                             head_dim, q_norm, k_norm, norm_eps)
 
 
-def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int, kp=None, vp=None):
+def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int, kp=None, vp=None,
+                 scale=None, softcap: float = 0.0, window: int = 0):
   if q.is_cuda and q.dtype == torch.bfloat16 and kp is not None and q.shape[3] == 128 \
      and os.getenv("XOT_MFMA_ATTN", "1") == "1" and _use_hip(q):
-    # causal flash-forward on matrix cores, streaming the MFMA-packed cache
-    return _hip.attn_prefill_mfma(q, kp, vp, start_pos)
-  if q.is_cuda:
+    # causal flash-forward on matrix cores, streaming the MFMA-packed cache;
+    # softcap/window run gemma2 semantics inside the same kernel
+    return _hip.attn_prefill_mfma(q, kp, vp, start_pos, scale or 0.0, softcap, window)
+  if q.is_cuda and not softcap and not window:
     # fallback: sdpa (rocm flash/mem-efficient backends) in model dtype with
     # native GQA — hd != 128 or unpacked-cache path
     import torch.nn.functional as F
@@ -109,28 +113,32 @@ def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int, kp=None, vp=No
     k = k_cache[:, :, :total]
     v = v_cache[:, :, :total]
     if start_pos == 0:
-      out = F.scaled_dot_product_attention(qh, k, v, is_causal=True, enable_gqa=True)
+      out = F.scaled_dot_product_attention(qh, k, v, is_causal=True, enable_gqa=True, scale=scale)
     else:
       mask = torch.ones(S, total, dtype=torch.bool, device=q.device).tril(diagonal=start_pos)
-      out = F.scaled_dot_product_attention(qh, k, v, attn_mask=mask, enable_gqa=True)
+      out = F.scaled_dot_product_attention(qh, k, v, attn_mask=mask, enable_gqa=True, scale=scale)
     return out.transpose(1, 2).contiguous()
-  return torch_ref.attn_prefill(q, k_cache, v_cache, start_pos, s_len)
+  return torch_ref.attn_prefill(q, k_cache, v_cache, start_pos, s_len, scale, softcap, window)
 
 
-def attn_decode(q, k_cache, v_cache, seq_len, kp=None, vp=None):
+def attn_decode(q, k_cache, v_cache, seq_len, kp=None, vp=None,
+                scale=None, softcap: float = 0.0, window: int = 0):
   """seq_len: int or int32 device tensor [B] (per-sequence lengths).
   With MFMA-packed cache copies (kp, vp) the flash-decoding kernel scores on
-  matrix cores (v_mfma_f32_16x16x32_bf16, coalesced 1 KB cache streams)."""
+  matrix cores (v_mfma_f32_16x16x32_bf16, coalesced 1 KB cache streams).
+  softcap/window select gemma2 semantics (MFMA or torch_ref path only)."""
   if _use_hip(q) and q.dtype == torch.bfloat16:
     if not isinstance(seq_len, torch.Tensor):
       seq_len = torch.full((q.shape[0],), int(seq_len), dtype=torch.int32, device=q.device)
     rep = q.shape[2] // k_cache.shape[1]
     if kp is not None and rep <= 16 and os.getenv("XOT_MFMA_ATTN", "1") == "1":
-      return _hip.attn_decode_mfma(q, kp, vp, seq_len, k_cache.shape[2])
-    return _hip.attn_decode(q, k_cache, v_cache, seq_len)
+      return _hip.attn_decode_mfma(q, kp, vp, seq_len, k_cache.shape[2],
+                                   scale or 0.0, softcap, window)
+    if not softcap and not window and scale is None:
+      return _hip.attn_decode(q, k_cache, v_cache, seq_len)
   if isinstance(seq_len, torch.Tensor):
     seq_len = int(seq_len.max().item())
-  return torch_ref.attn_decode(q, k_cache, v_cache, seq_len)
+  return torch_ref.attn_decode(q, k_cache, v_cache, seq_len, scale, softcap, window)
 
 
 def swiglu(gate, up):
@@ -144,6 +152,13 @@ def swiglu_packed(gu):
   if _use_hip(gu) and gu.dtype == torch.bfloat16:
     return _hip.swiglu_packed(gu)
   return torch_ref.swiglu_packed(gu)
+
+
+def geglu_packed(gu):
+  """gelu_tanh(gate)*up on the packed [.., 2I] fused gate_up output (gemma2)."""
+  if _use_hip(gu) and gu.dtype == torch.bfloat16:
+    return _hip.geglu_packed(gu)
+  return torch_ref.geglu_packed(gu)
 
 
 _SKINNY = os.getenv("XOT_SKINNY", "1") == "1"

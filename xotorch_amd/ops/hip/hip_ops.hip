@@ -65,7 +65,7 @@ template <bool RESIDUAL>
 __global__ __launch_bounds__(256) void rmsnorm_kernel(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ res,
     const unsigned short* __restrict__ w, unsigned short* __restrict__ out,
-    unsigned short* __restrict__ res_out, int D, float eps) {
+    unsigned short* __restrict__ res_out, int D, float eps, float w_bias) {
   const int row = blockIdx.x;
   const size_t off = (size_t)row * D;
   float vals[64];
@@ -111,7 +111,7 @@ __global__ __launch_bounds__(256) void rmsnorm_kernel(
       ushort8 wv = *(const ushort8*)(w + i);
       ushort8 o;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) o[j] = f2b(vals[c * 8 + j] * inv * b2f(wv[j]));
+      for (int j = 0; j < 8; ++j) o[j] = f2b(vals[c * 8 + j] * inv * (b2f(wv[j]) + w_bias));
       *(ushort8*)(out + off + i) = o;
     }
   }
@@ -381,11 +381,16 @@ __global__ __launch_bounds__(256) void attn_decode_partial(
 // the VALU split-KV kernel above stays as the hd=64 / unpacked-cache path.
 // ---------------------------------------------------------------------------
 
+// softcap > 0 applies gemma2's attention-logit soft-capping
+// (s = cap * tanh(s / cap), after scale, before masking); window > 0 is the
+// sliding-window mask (query at sl-1 sees positions [sl - window, sl)).
+// Both are wave-uniform runtime flags: the llama path (0, 0) takes no tanh.
 __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, const int* __restrict__ seq_lens,
     float* __restrict__ ws_o, float* __restrict__ ws_ml,
-    int B, int H, int KVH, int T32, int nsplit, float scale, long long q_stride) {
+    int B, int H, int KVH, int T32, int nsplit, float scale, long long q_stride,
+    float softcap, int window) {
   const int wv = threadIdx.x >> 6;
   const int wid = blockIdx.x * 4 + wv;
   __shared__ unsigned short plds_all[4][16 * 48];  // 96 B row stride: bank-conflict-free b128 reads
@@ -397,8 +402,9 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   const int NQ = H / KVH;
   const int sl = seq_lens[b];
   const int chunk = ((T32 / nsplit + 31) >> 5) << 5;
-  const int c0 = split * chunk;
-  const int c1 = min(c0 + chunk, sl);
+  const int win_lo = (window > 0) ? max(0, sl - window) : 0;  // first visible pos
+  const int c0 = max(split * chunk, (win_lo >> 5) << 5);
+  const int c1 = min(split * chunk + chunk, sl);
   unsigned short* plds = plds_all[wv];
 
   // Q fragments: A[i = lane&15 (query head, clamped), k = (lane>>4)*8 + j]
@@ -441,11 +447,12 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int pos = t + h * 16 + col;
-      const bool ok = pos < c1;
+      const bool ok = pos < c1 && pos >= win_lo;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float s = ok ? sc[h][r] * scale : -INFINITY;
-        sc[h][r] = s;
+        float s = sc[h][r] * scale;
+        if (softcap > 0.f) s = softcap * tanhf(s * (1.f / softcap));
+        sc[h][r] = ok ? s : -INFINITY;
       }
     }
 #pragma unroll
@@ -527,7 +534,8 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, unsigned short* __restrict__ out,
     int B, int S, int H, int KVH, int T32, int start_pos,
-    long long q_bstride, long long q_sstride, float scale) {
+    long long q_bstride, long long q_sstride, float scale,
+    float softcap, int window) {
   // block -> (b, h, q-block)
   const int qblocks = (S + 127) >> 7;
   const int h = blockIdx.x % H;
@@ -601,10 +609,13 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
   };
 
   const int ntiles = (kv_end + 31) >> 5;
-  stage(0, 0);
+  // sliding window: the block's lowest q row sees nothing before
+  // start_pos + qb*128 - window + 1 — skip whole kv tiles below it
+  const int tp0 = (window > 0) ? max(0, (start_pos + qb * 128 - window + 1) >> 5) : 0;
+  stage(tp0, tp0 & 1);
   __syncthreads();
 
-  for (int tp = 0; tp < ntiles; ++tp) {
+  for (int tp = tp0; tp < ntiles; ++tp) {
     const int buf = tp & 1;
     if (tp + 1 < ntiles) stage(tp + 1, buf ^ 1);  // plain loads overlap compute
     const unsigned short* dk = kv_lds[buf];
@@ -631,8 +642,11 @@ __global__ __launch_bounds__(256) void attn_prefill_mfma_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = r0 + t * 16 + (lane >> 4) * 4 + r;
-          const bool ok = (pos <= start_pos + row) && (pos < kv_end);
-          sc[hh][r] = ok ? sc[hh][r] * scale : -INFINITY;
+          bool ok = (pos <= start_pos + row) && (pos < kv_end);
+          if (window > 0) ok = ok && (pos > start_pos + row - window);
+          float s = sc[hh][r] * scale;
+          if (softcap > 0.f) s = softcap * tanhf(s * (1.f / softcap));
+          sc[hh][r] = ok ? s : -INFINITY;
         }
       }
 #pragma unroll
@@ -770,6 +784,32 @@ __global__ __launch_bounds__(256) void swiglu_packed_kernel(
     for (int j = 0; j < 8; ++j) {
       const float gf = b2f(gv[j]);
       const float s = gf * __builtin_amdgcn_rcpf(1.f + __expf(-gf));
+      o[j] = f2b(s * b2f(uv[j]));
+    }
+    *(ushort8*)(out + i * 8) = o;
+  }
+}
+
+// packed GeGLU (gemma2): gelu_tanh(gate) * up on the fused [gate | up] GEMM
+// output — same stream shape as swiglu_packed, tanh-approximate gelu in fp32.
+__global__ __launch_bounds__(256) void geglu_packed_kernel(
+    const unsigned short* __restrict__ gu, unsigned short* __restrict__ out,
+    long long rows, int I) {
+  const long long n8 = rows * (I >> 3);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const int i8 = I >> 3;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n8; i += stride) {
+    const long long row = i / i8;
+    const long long col8 = i % i8;
+    const unsigned short* base = gu + row * (2LL * I) + col8 * 8;
+    ushort8 gv = *(const ushort8*)base;
+    ushort8 uv = *(const ushort8*)(base + I);
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = b2f(gv[j]);
+      const float g3 = 0.7978845608028654f * (gf + 0.044715f * gf * gf * gf);
+      const float s = 0.5f * gf * (1.f + tanhf(g3));
       o[j] = f2b(s * b2f(uv[j]));
     }
     *(ushort8*)(out + i * 8) = o;
@@ -1049,6 +1089,125 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v3: packed-A + register-resident-X variant for the M=128/256 decode shapes.
+// The LDS-staged v2 wins at M<=64, but at MT>=4 its X staging doubles
+// (16 KB/step), the barrier pairs serialize the A stream, and the staging
+// registers push the allocation to 156+ VGPRs (3 waves/SIMD).  Here X is
+// never staged: each lane loads its B fragments straight from global
+// (16 B per lane, same pattern as v1's B path).  X is tiny (M x K bf16,
+// 2-4 MB) and every n-tile re-reads it, so after the first touch the
+// fragments come from L2/L3, not HBM -- the gather that made v1's A path
+// 2.0 TB/s is harmless on the B side.  No LDS, no barriers; A keeps v2's
+// coalesced non-temporal 1 KB wave streams with depth-2 step prefetch.
+// B chunks are loaded just-in-time per 16-k chunk (bb[MT] live registers):
+// at MT=4 the allocation stays ~128 VGPRs = 4 waves/SIMD, and the vmcnt
+// wait on one chunk's L2 hit is covered by the other waves' MFMA clusters.
+// ---------------------------------------------------------------------------
+template <int MT, bool SPLIT>
+__global__ __launch_bounds__(256) void skinny_gemm_packed_xreg_kernel(
+    const unsigned short* __restrict__ Wp, const unsigned short* __restrict__ X,
+    unsigned short* __restrict__ Y, float* __restrict__ P,
+    const unsigned short* __restrict__ bias,
+    int N, long long K, int kc, int nsplit) {
+  const int ntiles = N >> 7;
+  const int tile = blockIdx.x % ntiles;
+  const int split = blockIdx.x / ntiles;
+  const int e = blockIdx.y;  // grouped mode (gridDim.y == 1 for plain GEMM)
+  Wp += (size_t)e * (size_t)N * (size_t)K;
+  X += (size_t)e * (size_t)(MT * 32) * (size_t)K;
+  const long long k0 = (long long)split * kc;
+  const long long k1 = min(k0 + (long long)kc, K);
+  if (k0 >= k1) return;
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int n32 = tile * 4 + wv;
+
+  const unsigned short* wp = Wp + ((size_t)n32 * (K >> 4) + (k0 >> 4)) * 512 + (size_t)lane * 8;
+  // B fragment base per m-tile: row (t*32 + lane&31), k-half (lane>>5)*8
+  const unsigned short* xt[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+    xt[t] = X + (size_t)(t * 32 + (lane & 31)) * K + k0 + ((lane >> 5) * 8);
+
+  floatx16 acc[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t) acc[t] = (floatx16)(0.f);
+
+  const int nsteps = (int)((k1 - k0) >> 6);  // 64-k steps, 4 k16-chunks each
+  bf16x8 a_buf[2][4];
+  const unsigned short* wp1 = (nsteps > 1) ? wp + 4 * 512 : wp;
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    a_buf[0][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
+    a_buf[1][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp1 + u * 512));
+  }
+  wp += 2 * 4 * 512;
+
+#define XRG_STEP(BUF)                                                                         \
+  {                                                                                           \
+    const long long kk = (long long)s * 64;                                                   \
+    _Pragma("unroll")                                                                         \
+    for (int u = 0; u < 4; ++u) {                                                             \
+      bf16x8 bb[MT];                                                                          \
+      _Pragma("unroll")                                                                       \
+      for (int t = 0; t < MT; ++t)                                                            \
+        bb[t] = *reinterpret_cast<const bf16x8*>(xt[t] + kk + u * 16);                        \
+      _Pragma("unroll")                                                                       \
+      for (int t = 0; t < MT; ++t)                                                            \
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_buf[BUF][u], bb[t], acc[t], 0, 0, 0); \
+    }                                                                                         \
+    if (s + 2 < nsteps) {                                                                     \
+      _Pragma("unroll")                                                                       \
+      for (int u = 0; u < 4; ++u)                                                             \
+        a_buf[BUF][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512)); \
+      wp += 4 * 512;                                                                          \
+    }                                                                                         \
+  }
+
+  int s = 0;
+  while (s + 2 <= nsteps) {
+    XRG_STEP(0);
+    ++s;
+    XRG_STEP(1);
+    ++s;
+  }
+  if (s < nsteps) XRG_STEP(0);
+#undef XRG_STEP
+
+  const int m_local = lane & 31;
+  const int rows_total = gridDim.y * MT * 32;
+  const int nbase = tile * 128 + wv * 32 + 4 * (lane >> 5);
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+    const int m = e * MT * 32 + t * 32 + m_local;
+    if (SPLIT) {
+      float* prow = P + ((size_t)split * rows_total + m) * N;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        floatx4 v4;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) v4[j] = acc[t][g * 4 + j];
+        *reinterpret_cast<floatx4*>(prow + nbase + g * 8) = v4;
+      }
+    } else {
+      unsigned short* yrow = Y + (size_t)m * N;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        unsigned short o[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float v = acc[t][g * 4 + j];
+          if (bias) v += b2f(bias[nbase + g * 8 + j]);
+          o[j] = f2b(v);
+        }
+        *reinterpret_cast<unsigned long long*>(yrow + nbase + g * 8) =
+            *reinterpret_cast<unsigned long long*>(o);
+      }
+    }
+  }
+}
+
 // fp8 (OCP e4m3) W8A8 variant: same structure as the bf16 packed kernel but
 // operands are e4m3 with per-output-channel weight scales s_w[n] and
 // per-token activation scales s_x[m] (dynamic, computed by quant_fp8_rows).
@@ -1273,7 +1432,7 @@ static int skinny_target_blocks() {
   return t;
 }
 
-torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps, double w_bias) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
   const int D = x.size(-1);
   CHK(D % 8 == 0 && D <= 16384);
@@ -1282,11 +1441,12 @@ torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
   hipLaunchKernelGGL((rmsnorm_kernel<false>), dim3(rows), dim3(256), 0, cur_stream(),
                      (const unsigned short*)x.data_ptr(), nullptr,
                      (const unsigned short*)w.data_ptr(), (unsigned short*)out.data_ptr(),
-                     nullptr, D, (float)eps);
+                     nullptr, D, (float)eps, (float)w_bias);
   return out;
 }
 
-std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res, torch::Tensor w, double eps) {
+std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res, torch::Tensor w, double eps,
+                                            double w_bias) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous() && res.is_contiguous());
   const int D = x.size(-1);
   CHK(D % 8 == 0 && D <= 16384);
@@ -1296,7 +1456,7 @@ std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res, 
   hipLaunchKernelGGL((rmsnorm_kernel<true>), dim3(rows), dim3(256), 0, cur_stream(),
                      (const unsigned short*)x.data_ptr(), (const unsigned short*)res.data_ptr(),
                      (const unsigned short*)w.data_ptr(), (unsigned short*)out.data_ptr(),
-                     (unsigned short*)res_out.data_ptr(), D, (float)eps);
+                     (unsigned short*)res_out.data_ptr(), D, (float)eps, (float)w_bias);
   return {out, res_out};
 }
 
@@ -1405,7 +1565,8 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, t
 
 // q: [B, 1, H, 128] (strided batch OK); kp/vp: the packed cache copies.
 torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor vp,
-                               torch::Tensor seq_lens, int64_t t_capacity) {
+                               torch::Tensor seq_lens, int64_t t_capacity,
+                               double scale_in, double softcap, int64_t window) {
   CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   CHK(q.stride(3) == 1 && q.stride(2) == q.size(3));
   CHK(seq_lens.dtype() == torch::kInt32 && seq_lens.is_cuda());
@@ -1424,14 +1585,14 @@ torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor 
   auto ws_ml = torch::empty({(long)B * H * nsplit * 2}, opts);
   auto out = torch::empty({q.size(0), q.size(1), q.size(2), q.size(3)},
                           torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
-  const float scale = 1.0f / sqrtf((float)hd);
+  const float scale = (scale_in > 0.0) ? (float)scale_in : 1.0f / sqrtf((float)hd);
   auto stream = cur_stream();
   const int waves = B * KVH * nsplit;
   hipLaunchKernelGGL(attn_decode_mfma_kernel, dim3((waves + 3) / 4), dim3(256), 0, stream,
                      (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
                      (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
                      ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,
-                     scale, q_stride);
+                     scale, q_stride, (float)softcap, (int)window);
   hipLaunchKernelGGL((attn_decode_merge<128>), dim3(B * H), dim3(128), 0, stream,
                      ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
                      (unsigned short*)out.data_ptr(), nsplit);
@@ -1441,7 +1602,8 @@ torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor 
 // q: [B, S, H, 128] (strided over b/s OK, head rows contiguous); kp/vp: the
 // packed cache copies already holding positions [0, start_pos + S).
 torch::Tensor attn_prefill_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor vp,
-                                int64_t start_pos) {
+                                int64_t start_pos, double scale_in, double softcap,
+                                int64_t window) {
   CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   CHK(q.stride(3) == 1 && q.stride(2) == q.size(3));
   CHK(kp.is_contiguous() && vp.is_contiguous());
@@ -1452,12 +1614,13 @@ torch::Tensor attn_prefill_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor
   CHK(start_pos + S <= T32);
   auto out = torch::empty({B, S, H, hd},
                           torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
-  const float scale = 1.0f / sqrtf((float)hd);
+  const float scale = (scale_in > 0.0) ? (float)scale_in : 1.0f / sqrtf((float)hd);
   const int qblocks = (S + 127) / 128;
   hipLaunchKernelGGL(attn_prefill_mfma_kernel, dim3(B * qblocks * H), dim3(256), 0, cur_stream(),
                      (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
                      (const unsigned short*)vp.data_ptr(), (unsigned short*)out.data_ptr(),
-                     B, S, H, KVH, T32, (int)start_pos, q.stride(0), q.stride(1), scale);
+                     B, S, H, KVH, T32, (int)start_pos, q.stride(0), q.stride(1), scale,
+                     (float)softcap, (int)window);
   return out;
 }
 
@@ -1495,6 +1658,22 @@ torch::Tensor swiglu_packed(torch::Tensor gu) {
   const long long n8 = rows * (I / 8);
   const int blocks = (int)std::min<long long>(2048, (n8 + 255) / 256);
   hipLaunchKernelGGL(swiglu_packed_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)gu.data_ptr(), (unsigned short*)out.data_ptr(), rows, I);
+  return out;
+}
+
+torch::Tensor geglu_packed(torch::Tensor gu) {
+  CHK(gu.is_cuda() && gu.dtype() == torch::kBFloat16 && gu.is_contiguous());
+  const int twoI = gu.size(-1);
+  CHK(twoI % 16 == 0);
+  const int I = twoI / 2;
+  const long long rows = gu.numel() / twoI;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = torch::empty(sizes, gu.options());
+  const long long n8 = rows * (I / 8);
+  const int blocks = (int)std::min<long long>(2048, (n8 + 255) / 256);
+  hipLaunchKernelGGL(geglu_packed_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
                      (const unsigned short*)gu.data_ptr(), (unsigned short*)out.data_ptr(), rows, I);
   return out;
 }
@@ -1620,6 +1799,59 @@ torch::Tensor skinny_gemm_packed(torch::Tensor x, torch::Tensor wp, int64_t N,
                        P.data_ptr<float>(), (unsigned short*)y.data_ptr(), bptr, MN, N, nsplit);
   }
 #undef SGP_CASE
+  return y;
+}
+
+// v3 (register-resident X) launcher — same contract as skinny_gemm_packed.
+torch::Tensor skinny_gemm_packed_xreg(torch::Tensor x, torch::Tensor wp, int64_t N,
+                                      c10::optional<torch::Tensor> bias) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  CHK(wp.is_cuda() && wp.dtype() == torch::kBFloat16 && wp.is_contiguous());
+  const long long K = wp.numel() / N;
+  const long long M = x.numel() / K;
+  CHK(M >= 32 && M <= 256 && M % 32 == 0);
+  CHK(N % 128 == 0 && K % 64 == 0);
+  const unsigned short* bptr = nullptr;
+  if (bias.has_value()) {
+    CHK(bias->is_contiguous() && bias->dtype() == torch::kBFloat16 && bias->numel() == N);
+    bptr = (const unsigned short*)bias->data_ptr();
+  }
+  auto sizes = x.sizes().vec();
+  sizes.back() = (long)N;
+  auto y = torch::empty(sizes, x.options());
+  const int ntiles = (int)(N / 128);
+  int nsplit = 1;
+  while (ntiles * nsplit < skinny_target_blocks() && (K / (nsplit * 2)) >= 1024 && nsplit < 16) nsplit *= 2;
+  int kc = (int)((K / nsplit + 63) / 64 * 64);
+  while ((long long)kc * (nsplit - 1) >= K) nsplit--;
+  auto stream = cur_stream();
+  const int MT = (int)(M / 32);
+  const dim3 grid(ntiles * nsplit), block(256);
+#define SGX_CASE(MTV) \
+  case MTV: \
+    if (nsplit == 1) { \
+      hipLaunchKernelGGL((skinny_gemm_packed_xreg_kernel<MTV, false>), grid, block, 0, stream, \
+                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                         (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit); \
+    } else { \
+      hipLaunchKernelGGL((skinny_gemm_packed_xreg_kernel<MTV, true>), grid, block, 0, stream, \
+                         (const unsigned short*)wp.data_ptr(), (const unsigned short*)x.data_ptr(), \
+                         nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit); \
+    } \
+    break;
+  if (nsplit == 1) {
+    torch::Tensor P;
+    switch (MT) { SGX_CASE(1) SGX_CASE(2) SGX_CASE(3) SGX_CASE(4) SGX_CASE(5) SGX_CASE(6) SGX_CASE(7) SGX_CASE(8) }
+  } else {
+    auto P = torch::empty({nsplit, M, (long long)N},
+                          torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+    switch (MT) { SGX_CASE(1) SGX_CASE(2) SGX_CASE(3) SGX_CASE(4) SGX_CASE(5) SGX_CASE(6) SGX_CASE(7) SGX_CASE(8) }
+    const long long MN = M * (long long)N;
+    const int blocks = (int)std::min<long long>(2048, (MN / 4 + 255) / 256);
+    hipLaunchKernelGGL(skinny_combine_kernel, dim3(blocks), dim3(256), 0, stream,
+                       P.data_ptr<float>(), (unsigned short*)y.data_ptr(), bptr, MN, N, nsplit);
+  }
+#undef SGX_CASE
   return y;
 }
 
@@ -1763,6 +1995,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("n"), py::arg("bias") = py::none());
   m.def("skinny_gemm", &skinny_gemm, "decode GEMM y = x @ w^T (+bias), bf16 MFMA weight-streaming",
         py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+  m.def("skinny_gemm_packed_xreg", &skinny_gemm_packed_xreg,
+        "decode GEMM on prepacked weights, register-resident X (M=128/256 shapes)",
+        py::arg("x"), py::arg("wp"), py::arg("N"), py::arg("bias") = py::none());
   m.def("skinny_gemm_packed", &skinny_gemm_packed,
         "decode GEMM on prepacked weights (MFMA fragment order)",
         py::arg("x"), py::arg("wp"), py::arg("n"), py::arg("bias") = py::none());
@@ -1770,8 +2005,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MoE grouped decode GEMM on stacked prepacked expert weights",
         py::arg("x"), py::arg("wp"), py::arg("n_experts"), py::arg("n"));
   m.def("swiglu_packed", &swiglu_packed, "SwiGLU on the packed [gate|up] GEMM output");
-  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
-  m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual add + RMSNorm");
+  m.def("geglu_packed", &geglu_packed, "GeGLU (tanh gelu) on the packed [gate|up] GEMM output");
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4); w_bias=1 gives gemma-style (1+w)",
+        py::arg("x"), py::arg("w"), py::arg("eps"), py::arg("w_bias") = 0.0);
+  m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual add + RMSNorm",
+        py::arg("x"), py::arg("res"), py::arg("w"), py::arg("eps"), py::arg("w_bias") = 0.0);
   m.def("rope_qkv_append", &rope_qkv_append,
         "fused (qk-RMSNorm +) RoPE + KV-cache append on packed qkv",
         py::arg("qkv"), py::arg("cos"), py::arg("sin"), py::arg("positions"), py::arg("kc"),
@@ -1781,9 +2019,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("norm_eps") = 1e-6);
   m.def("attn_decode", &attn_decode, "GQA decode attention (flash-decoding split-KV)");
   m.def("attn_decode_mfma", &attn_decode_mfma,
-        "GQA decode attention on matrix cores (packed cache, hd=128)");
+        "GQA decode attention on matrix cores (packed cache, hd=128); "
+        "softcap/window: gemma2 logit soft-capping and sliding window",
+        py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("seq_lens"), py::arg("t_capacity"),
+        py::arg("scale") = 0.0, py::arg("softcap") = 0.0, py::arg("window") = 0);
   m.def("attn_prefill_mfma", &attn_prefill_mfma,
-        "causal GQA prefill flash attention on matrix cores (packed cache, hd=128)");
+        "causal GQA prefill flash attention on matrix cores (packed cache, hd=128)",
+        py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("start_pos"),
+        py::arg("scale") = 0.0, py::arg("softcap") = 0.0, py::arg("window") = 0);
   m.def("mfma16_probe", &mfma16_probe, "v_mfma_f32_16x16x32_bf16 layout probe (tests)");
   m.def("swiglu", &swiglu, "SwiGLU activation");
 }

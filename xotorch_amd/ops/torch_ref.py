@@ -16,20 +16,22 @@ import torch
 import torch.nn.functional as F
 
 
-def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
-  """RMSNorm in fp32 math, cast back to input dtype."""
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float, w_bias: float = 0.0) -> torch.Tensor:
+  """RMSNorm in fp32 math, cast back to input dtype. w_bias=1 gives the
+  gemma convention (scale by 1 + w)."""
   xf = x.float()
   norm = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
-  return (norm * weight.float()).to(x.dtype)
+  return (norm * (weight.float() + w_bias)).to(x.dtype)
 
 
 def rmsnorm_residual(
-  x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+  x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float,
+  w_bias: float = 0.0
 ) -> Tuple[torch.Tensor, torch.Tensor]:
   """Fused residual-add + RMSNorm: returns (norm(x+residual), x+residual)."""
   s = (x.float() + residual.float())
   norm = s * torch.rsqrt(s.pow(2).mean(-1, keepdim=True) + eps)
-  return (norm * weight.float()).to(x.dtype), s.to(x.dtype)
+  return (norm * (weight.float() + w_bias)).to(x.dtype), s.to(x.dtype)
 
 
 def rope_cos_sin(
@@ -115,6 +117,9 @@ def attn_prefill(
   v_cache: torch.Tensor,
   start_pos: int,
   s_len: int,
+  scale: Optional[float] = None,
+  softcap: float = 0.0,
+  window: int = 0,
 ) -> torch.Tensor:
   """Causal GQA attention for a prefill chunk.
 
@@ -130,9 +135,19 @@ def attn_prefill(
   vals = v_cache[:, :, :total].repeat_interleave(H // KVH, dim=1)
   # mask: query i (absolute pos start_pos+i) sees keys 0..start_pos+i
   mask = torch.ones(S, total, dtype=torch.bool, device=q.device).tril(diagonal=start_pos)
-  out = F.scaled_dot_product_attention(
-    qh.float(), keys.float(), vals.float(), attn_mask=mask
-  )
+  if window and window > 0:
+    mask &= ~torch.ones(S, total, dtype=torch.bool, device=q.device).tril(diagonal=start_pos - window)
+  if scale is None:
+    scale = hd ** -0.5
+  if softcap and softcap > 0.0:
+    scores = torch.einsum("bhsd,bhtd->bhst", qh.float(), keys.float()) * scale
+    scores = torch.tanh(scores / softcap) * softcap
+    scores = scores.masked_fill(~mask[None, None], float("-inf"))
+    out = torch.softmax(scores, dim=-1) @ vals.float()
+  else:
+    out = F.scaled_dot_product_attention(
+      qh.float(), keys.float(), vals.float(), attn_mask=mask, scale=scale
+    )
   return out.transpose(1, 2).to(q.dtype)
 
 
@@ -141,6 +156,9 @@ def attn_decode(
   k_cache: torch.Tensor,
   v_cache: torch.Tensor,
   seq_len: int,
+  scale: Optional[float] = None,
+  softcap: float = 0.0,
+  window: int = 0,
 ) -> torch.Tensor:
   """Single-position GQA attention against the KV cache.
 
@@ -149,10 +167,18 @@ def attn_decode(
   """
   B, _, H, hd = q.shape
   KVH = k_cache.shape[1]
+  lo = max(0, seq_len - window) if window and window > 0 else 0
   qh = q.transpose(1, 2).float()  # [B,H,1,hd]
-  keys = k_cache[:, :, :seq_len].repeat_interleave(H // KVH, dim=1).float()
-  vals = v_cache[:, :, :seq_len].repeat_interleave(H // KVH, dim=1).float()
-  out = F.scaled_dot_product_attention(qh, keys, vals)
+  keys = k_cache[:, :, lo:seq_len].repeat_interleave(H // KVH, dim=1).float()
+  vals = v_cache[:, :, lo:seq_len].repeat_interleave(H // KVH, dim=1).float()
+  if scale is None:
+    scale = hd ** -0.5
+  if softcap and softcap > 0.0:
+    scores = (qh @ keys.transpose(-1, -2)) * scale
+    scores = torch.tanh(scores / softcap) * softcap
+    out = torch.softmax(scores, dim=-1) @ vals
+  else:
+    out = F.scaled_dot_product_attention(qh, keys, vals, scale=scale)
   return out.transpose(1, 2).to(q.dtype)
 
 
@@ -198,6 +224,13 @@ def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
 def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
   """silu(gate) * up, fp32 internally."""
   return (F.silu(gate.float()) * up.float()).to(gate.dtype)
+
+
+def geglu_packed(gu: torch.Tensor) -> torch.Tensor:
+  """gelu_tanh(gate) * up on the packed [.., 2I] fused gate_up output (gemma2)."""
+  I = gu.shape[-1] // 2
+  g, u = gu[..., :I].float(), gu[..., I:].float()
+  return (F.gelu(g, approximate="tanh") * u).to(gu.dtype)
 
 
 def softmax_sample(
