@@ -54,7 +54,7 @@ def main():
   assert hip is not None, "HIP extension missing"
   dev = "cuda"
   torch.manual_seed(7)
-  print(f"{'shape':<10} {'M':>4} {'blaslt':>9} {'packed':>9} {'xreg':>9}  best (TB/s of weight stream)")
+  print(f"{'shape':<10} {'M':>4} {'blaslt':>9} {'packed':>9} {'pk-bk64':>9} {'xreg':>9}  best (TB/s of weight stream)")
   for name, N, K in SHAPES:
     w = (torch.randn(N, K, device=dev) / K**0.5).to(torch.bfloat16)
     wp = ops.pack_decode_weight(w)
@@ -73,11 +73,14 @@ def main():
         continue
       t_bl = time_us(lambda: torch.nn.functional.linear(x, w))
       t_pk = time_us(lambda: hip.skinny_gemm_packed(x, wp, N, None))
+      os.environ["XOT_SKINNY_BK64"] = "1"
+      t_p64 = time_us(lambda: hip.skinny_gemm_packed(x, wp, N, None))
+      del os.environ["XOT_SKINNY_BK64"]
       t_xr = time_us(lambda: hip.skinny_gemm_packed_xreg(x, wp, N, None))
-      best = min(t_bl, t_pk, t_xr)
+      best = min(t_bl, t_pk, t_p64, t_xr)
       tbps = wbytes / best / 1e6
-      win = {t_bl: "blaslt", t_pk: "packed", t_xr: "xreg"}[best]
-      print(f"{name:<10} {M:>4} {t_bl:>9.1f} {t_pk:>9.1f} {t_xr:>9.1f}  {win} {tbps:.2f} TB/s")
+      win = {t_bl: "blaslt", t_pk: "packed", t_p64: "pk-bk64", t_xr: "xreg"}[best]
+      print(f"{name:<10} {M:>4} {t_bl:>9.1f} {t_pk:>9.1f} {t_p64:>9.1f} {t_xr:>9.1f}  {win} {tbps:.2f} TB/s")
     del w, wp
     torch.cuda.empty_cache()
 

@@ -985,24 +985,21 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   const int nsteps = (int)((k1 - k0) >> (BK == 64 ? 6 : 7));
   ushort8 xv[PPT];
   bf16x8 a_buf[2][BKC];
-  // per-thread staging addresses, strength-reduced (recomputing row/q per
-  // step costs ~10% on short-kc shapes)
-  const unsigned short* xpp[PPT];
-  int xso[PPT];
-#pragma unroll
-  for (int i = 0; i < PPT; ++i) {
-    const int p = tid + i * 256;
-    const int row = p / (BK / 8), q = p % (BK / 8);
-    xpp[i] = X + k0 + (size_t)row * K + q * 8;
-    xso[i] = row * XROW + q * 8;
-  }
+  // per-thread staging base + uniform per-piece offsets: piece p = tid+i*256
+  // has row = tid/(BK/8) + i*ROWS_PER_I and column-piece q = tid%(BK/8), so
+  // one VGPR pointer plus SGPR/immediate strides replaces the old per-piece
+  // pointer arrays (saves ~10 VGPRs -> a wave/SIMD at MT=4)
+  constexpr int ROWS_PER_I = 2048 / BK;  // rows advanced per i (256 threads)
+  const unsigned short* xpb = X + k0 + (size_t)(tid / (BK / 8)) * K + (tid % (BK / 8)) * 8;
+  const int xsb = (tid / (BK / 8)) * XROW + (tid % (BK / 8)) * 8;
+  const long long xstride_i = (long long)ROWS_PER_I * K;
 
   // prologue: stage step 0, preload A(0) and A(1).  The W stream is read
   // exactly once per launch -> non-temporal (L1-bypass) loads; depth-2
   // prefetch keeps 2*BK*32n*2B per wave in flight across staging barriers.
   const unsigned short* wp1 = (nsteps > 1) ? wp + BKC * 512 : wp;  // clamp: no OOB at nsteps==1
 #pragma unroll
-  for (int i = 0; i < PPT; ++i) xv[i] = *(const ushort8*)(xpp[i]);
+  for (int i = 0; i < PPT; ++i) xv[i] = *(const ushort8*)(xpb + i * xstride_i);
 #pragma unroll
   for (int u = 0; u < BKC; ++u) {
     a_buf[0][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
@@ -1010,7 +1007,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   }
   wp += 2 * BKC * 512;
 #pragma unroll
-  for (int i = 0; i < PPT; ++i) *(ushort8*)(xs + xso[i]) = xv[i];
+  for (int i = 0; i < PPT; ++i) *(ushort8*)(xs + xsb + i * ROWS_PER_I * XROW) = xv[i];
   __syncthreads();
 
   // NOTE: the A double-buffer index must be a compile-time constant — a
@@ -1021,7 +1018,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     if (!last) {                                                                               \
       _Pragma("unroll")                                                                        \
       for (int i = 0; i < PPT; ++i)                                                            \
-        xv[i] = *(const ushort8*)(xpp[i] + (size_t)(s + 1) * BK);                              \
+        xv[i] = *(const ushort8*)(xpb + i * xstride_i + (size_t)(s + 1) * BK);                 \
     }                                                                                          \
     _Pragma("unroll")                                                                          \
     for (int u = 0; u < BKC; ++u) {                                                            \
@@ -1041,7 +1038,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     if (!last) {                                                                               \
       __syncthreads();                                                                         \
       _Pragma("unroll")                                                                        \
-      for (int i = 0; i < PPT; ++i) *(ushort8*)(xs + xso[i]) = xv[i];                          \
+      for (int i = 0; i < PPT; ++i) *(ushort8*)(xs + xsb + i * ROWS_PER_I * XROW) = xv[i];     \
       __syncthreads();                                                                         \
     }                                                                                          \
   }
