@@ -126,3 +126,45 @@ def test_max_tokens_enforced():
       await client.close()
       await node.stop()
   run(go())
+
+
+def test_token_encode_route():
+  async def go():
+    node, client = await make_client()
+    r = await client.post("/v1/chat/token/encode", json={
+      "model": "dummy", "messages": [{"role": "user", "content": "hello world"}]})
+    data = await r.json()
+    assert r.status == 200 and data["num_tokens"] > 0 and isinstance(data["tokens"], list)
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())
+
+
+def test_cors_headers_on_every_route():
+  async def go():
+    node, client = await make_client()
+    for path in ("/healthcheck", "/v1/models", "/v1/topology"):
+      r = await client.get(path)
+      assert r.headers.get("Access-Control-Allow-Origin") == "*", path
+    r = await client.options("/v1/chat/completions")
+    assert r.status == 204
+    assert "POST" in r.headers.get("Access-Control-Allow-Methods", "")
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())
+
+
+def test_post_download_route():
+  async def go():
+    node, client = await make_client()
+    # no downloader wired -> 503 with a clear message; unknown model -> 400
+    r = await client.post("/download", json={"model": "nope"})
+    assert r.status == 400
+    r = await client.post("/download", json={"model": "llama-3-8b"})
+    assert r.status == 503
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())

@@ -100,12 +100,15 @@ class ChatGPTAPI:
     self.app.router.add_post("/quit", self.handle_quit)
     self.app.router.add_delete("/models/{model_name}", self.handle_delete_model)
     self.app.router.add_get("/v1/download/progress", self.handle_get_download_progress)
+    self.app.router.add_post("/v1/chat/token/encode", self.handle_post_chat_token_encode)
+    self.app.router.add_post("/download", self.handle_post_download)
     static_dir = Path(__file__).parent / "tinychat"
     if static_dir.exists():
       self.app.router.add_get("/", self.handle_root)
       self.app.router.add_static("/static/", static_dir, name="static")
     self.app.middlewares.append(self._timeout_middleware)
     self.app.middlewares.append(self._log_middleware)
+    self.app.middlewares.append(self._cors_middleware)
     # node token plumbing
     if node is not None:
       node.on_token.register("chatgpt-api-token-handler").on_next(self._on_token)
@@ -118,6 +121,24 @@ class ChatGPTAPI:
     if DEBUG >= 2:
       print(f"[api] {request.method} {request.path}")
     return await handler(request)
+
+  @web.middleware
+  async def _cors_middleware(self, request, handler):
+    # permissive CORS on every route (reference wraps each route in
+    # aiohttp_cors, chatgpt_api.py:208-223; tinychat runs cross-origin)
+    if request.method == "OPTIONS":
+      resp = web.Response(status=204)
+    else:
+      try:
+        resp = await handler(request)
+      except web.HTTPException as e:
+        resp = e
+    resp.headers["Access-Control-Allow-Origin"] = "*"
+    resp.headers["Access-Control-Allow-Methods"] = "GET, POST, DELETE, OPTIONS"
+    resp.headers["Access-Control-Allow-Headers"] = "Content-Type, Authorization"
+    if isinstance(resp, web.HTTPException):
+      raise resp
+    return resp
 
   @web.middleware
   async def _timeout_middleware(self, request, handler):
@@ -181,6 +202,36 @@ class ChatGPTAPI:
     if self.shard_downloader is not None and hasattr(self.shard_downloader, "progress"):
       return web.json_response(self.shard_downloader.progress())
     return web.json_response({})
+
+  async def handle_post_chat_token_encode(self, request):
+    """Tokenize a chat-templated conversation without running it
+    (reference chatgpt_api.py:210-211)."""
+    data = await request.json()
+    model_id = data.get("model") or self.default_model
+    messages = [Message(m.get("role", "user"), m.get("content", "")) for m in data.get("messages", [])]
+    tokenizer = getattr(self.node.inference_engine, "tokenizer", None)
+    if tokenizer is None:
+      repo = get_repo(model_id, self.inference_engine_classname)
+      try:
+        tokenizer = await resolve_tokenizer(repo)
+      except Exception:
+        from xotorch_amd.engine.tokenizers import DummyTokenizer
+        tokenizer = DummyTokenizer()
+    prompt = build_prompt(tokenizer, messages, data.get("tools"))
+    tokens = tokenizer.encode(prompt)
+    return web.json_response({"length": len(prompt), "num_tokens": len(tokens), "tokens": list(map(int, tokens))})
+
+  async def handle_post_download(self, request):
+    """Kick off a model download (reference chatgpt_api.py:221)."""
+    data = await request.json()
+    model_id = data.get("model")
+    if not model_id or model_id not in model_cards:
+      return web.json_response({"detail": f"unknown model {model_id}"}, status=400)
+    shard = build_base_shard(model_id, self.inference_engine_classname)
+    if self.shard_downloader is None or shard is None:
+      return web.json_response({"detail": "no downloader available"}, status=503)
+    asyncio.create_task(self.shard_downloader.ensure_shard(shard, self.inference_engine_classname))
+    return web.json_response({"status": "started", "model": model_id})
 
   async def handle_post_chat_completions(self, request):
     data = await request.json()

@@ -27,7 +27,8 @@ def build_parser() -> argparse.ArgumentParser:
   p.add_argument("command", nargs="?", choices=["run", "eval", "train", "serve"], help="one-shot verb (default: daemon)")
   p.add_argument("model_name", nargs="?", help="model id (see --list-models)")
   p.add_argument("--list-models", action="store_true")
-  p.add_argument("--prompt", type=str, default="Who are you?")
+  p.add_argument("--prompt", type=str, default=None,
+                 help="run: one-shot prompt; omit for the interactive chat TUI")
   p.add_argument("--max-generate-tokens", type=int, default=1024)
   p.add_argument("--default-temp", type=float, default=0.0)
   p.add_argument("--inference-engine", type=str, default=None, choices=[None, "torch", "hip", "dummy"])
@@ -95,6 +96,14 @@ async def run_model_cli(args):
   shard = build_base_shard(args.model_name, engine_classname)
   if shard is None:
     raise SystemExit(f"unknown model {args.model_name}")
+  if args.prompt is None:
+    # interactive chat REPL (reference main.py:381 -> viz/chat_tui.py:11)
+    from xotorch_amd.viz.chat_tui import run_chat_tui
+    try:
+      await run_chat_tui(node, args.model_name, engine_classname)
+    finally:
+      await node.stop()
+    return
   from xotorch_amd.engine.tokenizers import resolve_tokenizer
   from xotorch_amd.models.registry import get_repo
   tokenizer = await resolve_tokenizer(get_repo(args.model_name, engine_classname))

@@ -136,8 +136,6 @@ def attn_decode(q, k_cache, v_cache, seq_len, kp=None, vp=None,
                                    scale or 0.0, softcap, window)
     if not softcap and not window and scale is None:
       return _hip.attn_decode(q, k_cache, v_cache, seq_len)
-  if isinstance(seq_len, torch.Tensor):
-    seq_len = int(seq_len.max().item())
   return torch_ref.attn_decode(q, k_cache, v_cache, seq_len, scale, softcap, window)
 
 

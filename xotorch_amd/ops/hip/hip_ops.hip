@@ -141,7 +141,7 @@ __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
     int B, int S, int H, int KVH, int hd, int T,
     unsigned short* __restrict__ kpc, unsigned short* __restrict__ vpc, int T32,
     const unsigned short* __restrict__ qn, const unsigned short* __restrict__ kn,
-    float norm_eps) {
+    float norm_eps, int npos) {
   const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   const int slots = H + 2 * KVH;
@@ -150,7 +150,9 @@ __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
   const int bs = wid / slots;
   const int s = bs % S;
   const int b = bs / S;
-  const int pos = positions[s];
+  // positions: [S] (shared across batch) or [B*S] (per-row — continuous
+  // batching where every slot decodes at its own position)
+  const int pos = positions[(npos == B * S && npos != S) ? bs : s];
   const int hd2 = hd >> 1;
   unsigned short* row = qkv + ((size_t)bs * slots + slot) * hd;
   if (slot < H) {
@@ -1471,6 +1473,9 @@ void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
   const int T = kc.size(2);
   CHK(qkv.size(2) == (H + 2 * KVH) * hd);
   CHK(hd <= 128 && hd % 2 == 0);
+  const int npos = (int)positions.numel();
+  CHK(npos == S || npos == B * S);
+  CHK(positions.is_contiguous());
   unsigned short* kpc = nullptr;
   unsigned short* vpc = nullptr;
   int T32 = 0;
@@ -1496,7 +1501,7 @@ void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
                      (unsigned short*)qkv.data_ptr(), cos.data_ptr<float>(), sin.data_ptr<float>(),
                      positions.data_ptr<int>(), (unsigned short*)kc.data_ptr(),
                      (unsigned short*)vc.data_ptr(), B, S, H, KVH, hd, T, kpc, vpc, T32,
-                     qnp, knp, (float)norm_eps);
+                     qnp, knp, (float)norm_eps, npos);
 }
 
 template <int HD>

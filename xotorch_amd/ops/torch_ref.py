@@ -155,7 +155,7 @@ def attn_decode(
   q: torch.Tensor,
   k_cache: torch.Tensor,
   v_cache: torch.Tensor,
-  seq_len: int,
+  seq_len,
   scale: Optional[float] = None,
   softcap: float = 0.0,
   window: int = 0,
@@ -163,8 +163,21 @@ def attn_decode(
   """Single-position GQA attention against the KV cache.
 
   q: [B, 1, H, hd]; caches [B, KVH, T, hd] valid through seq_len.
+  seq_len: int, or an int tensor [B] of per-row lengths (ragged slots).
   Returns [B, 1, H, hd].
   """
+  if isinstance(seq_len, torch.Tensor) and seq_len.numel() > 1:
+    lens = [int(v) for v in seq_len.reshape(-1)]
+    if len(set(lens)) > 1:
+      outs = [
+        attn_decode(q[b:b + 1], k_cache[b:b + 1], v_cache[b:b + 1], lens[b],
+                    scale, softcap, window)
+        for b in range(q.shape[0])
+      ]
+      return torch.cat(outs, dim=0)
+    seq_len = lens[0]
+  elif isinstance(seq_len, torch.Tensor):
+    seq_len = int(seq_len.reshape(-1)[0])
   B, _, H, hd = q.shape
   KVH = k_cache.shape[1]
   lo = max(0, seq_len - window) if window and window > 0 else 0
@@ -209,6 +222,16 @@ def rope_qkv_append(
     qq = rmsnorm(q.float(), q_norm.float(), norm_eps).to(q.dtype)
   if k_norm is not None:
     kk = rmsnorm(k.float(), k_norm.float(), norm_eps).to(k.dtype)
+  if positions.numel() == B * S and positions.numel() != S:
+    # per-row positions (continuous batching): rope + append per batch row
+    pos_bs = positions.reshape(B, S)
+    q_r, k_r = rope_apply(qq, kk, cos, sin, pos_bs)
+    q.copy_(q_r)
+    for b in range(B):
+      p0 = int(pos_bs[b, 0])
+      k_cache[b:b + 1, :, p0:p0 + S] = k_r[b:b + 1].transpose(1, 2).to(k_cache.dtype)
+      v_cache[b:b + 1, :, p0:p0 + S] = v[b:b + 1].transpose(1, 2).to(v_cache.dtype)
+    return
   q_r, k_r = rope_apply(qq, kk, cos, sin, positions)
   q.copy_(q_r)
   start_pos = int(positions.reshape(-1)[0])

@@ -1,46 +1,93 @@
 """Single-node multi-GPU ring serving worker (launched via torchrun by
 `xot serve --gpus N`).
 
-Rank 0 hosts the ChatGPT-compatible HTTP API; every rank holds one layer
-shard. Per request: rank 0 tokenizes and broadcasts {S, max_new} + prompt
-tokens; the ring prefized+decodes with bf16 hidden send/recv over RCCL/xGMI;
-the last stage's sampled token is broadcast to all ranks each step (stage 0
-feeds it back in, rank 0 streams it to the client). This is the product
-serving path the reference implements with per-hop gRPC+JSON
- — here the activation hop is one xGMI link.
+The product serving path on MI355X: every rank holds one contiguous layer
+shard (ring partitioning), activations hop rank→rank as resident bf16 over
+RCCL/xGMI, and requests are served with SLOT-BASED CONTINUOUS BATCHING:
+
+- the KV cache is allocated once with batch dim = SLOTS; each request is
+  admitted into a free slot and decoded alongside the others (the reference
+  serves one request at a time over per-request caches,
+  /root/reference/xotorch/orchestration/node.py:145);
+- one decode tick advances EVERY active slot by one token in a single
+  hipGraph replay (per-slot positions / seq_lens are static device tensors
+  the graph reads, so admission/release never re-captures);
+- prompt prefill is CHUNKED through the ring (stage 0 forwards sequence
+  chunks so later stages overlap instead of idling through the whole
+  prompt — request-level TTFT, VERDICT round-1 #8);
+- rank 0 hosts the FULL ChatGPT-compatible API (the same ChatGPTAPI as the
+  elastic Node path — models/topology/tinychat/encode routes included) via
+  a thin Node adapter.
+
+Control plane per tick: rank 0 broadcasts a small header (op, slot, len,
+milli-temperature); ADMIT is followed by the prompt tokens; the last stage
+samples and broadcasts the per-slot tokens; rank 0 streams them to clients.
+All collectives go through RingComm, so N ranks verify on fewer GPUs (gloo
+staging) while the real deployment runs pure RCCL.
 """
 from __future__ import annotations
 
 import argparse
 import asyncio
-import json
+import os
+import queue
 import threading
 import time
 import uuid
-from queue import Queue
-from typing import List, Optional
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
 
 import torch
 import torch.distributed as dist
 
 from xotorch_amd.engine.kvcache import ShardKVCache
 from xotorch_amd.models.config import config_from_hf
-from xotorch_amd.models.llama import ShardedModel
 from xotorch_amd.models.registry import builtin_config, get_repo
 from xotorch_amd.models.weights import fast_random_init_gpu, load_shard_weights, random_init
-from xotorch_amd.parallel.comm import init_distributed
+from xotorch_amd.parallel.comm import RingComm, init_distributed
 from xotorch_amd.parallel.ring import equal_ring_shards
 
-MAX_SEQ = 4096
+OP_IDLE, OP_DECODE, OP_ADMIT, OP_RELEASE, OP_SHUTDOWN = 0, 1, 2, 3, 4
+
+MAX_SEQ = int(os.getenv("XOT_SERVE_MAX_SEQ", "4096"))
+SLOTS = int(os.getenv("XOT_SERVE_SLOTS", "8"))
+PREFILL_CHUNK = int(os.getenv("XOT_SERVE_PREFILL_CHUNK", "512"))
 
 
-class RingWorker:
+@dataclass
+class SlotState:
+  request_id: Optional[str] = None
+  prompt_len: int = 0
+  generated: int = 0
+  max_new: int = 0
+  temp: float = 0.0
+  done: bool = False  # finished but not yet released (release is its own tick)
+
+
+@dataclass
+class AdmitMsg:
+  request_id: str
+  tokens: torch.Tensor  # [1, S] int64 cpu
+  max_new: int
+  temp: float
+  t_submit: float = field(default_factory=time.perf_counter)
+
+
+class RingSlotWorker:
+  """One rank of the serving ring. Rank 0 drives the control loop."""
+
   def __init__(self, model_id: str, rank: int, world: int, device="cuda",
-               dtype=torch.bfloat16, model_dir: Optional[str] = None):
+               dtype=torch.bfloat16, model_dir: Optional[str] = None,
+               slots: int = SLOTS, max_seq: int = MAX_SEQ, use_graphs: bool = True):
+    self.model_id = model_id
     self.rank, self.world = rank, world
     self.device, self.dtype = device, dtype
+    self.slots, self.max_seq = slots, max_seq
+    self.use_graphs = use_graphs and device == "cuda"
     self.cfg = config_from_hf(builtin_config(model_id) or {}, model_id)
+    self.max_seq = min(self.max_seq, self.cfg.max_seq_len)
     self.shard = equal_ring_shards(model_id, self.cfg.n_layers, world)[rank]
+    from xotorch_amd.models.llama import ShardedModel
     prev_dtype = torch.get_default_dtype()
     torch.set_default_dtype(dtype)
     try:
@@ -58,74 +105,322 @@ class RingWorker:
       random_init(model)
     model.reset_rope()
     model.eval()
+    self.model = model
+    heads, k_dim, v_dim = self.cfg.kv_cache_dims()
+    self.cache = ShardKVCache(self.shard.get_layer_count(), slots, heads,
+                              self.max_seq, k_dim, dtype, device, v_dim=v_dim)
     if device == "cuda":
       torch.cuda.empty_cache()
-      model.pack_decode_weights(reserve_bytes=24 << 30)
-    self.model = model
-    self.cache = ShardKVCache(self.shard.get_layer_count(), 1, self.cfg.n_kv_heads,
-                              MAX_SEQ, self.cfg.head_dim, dtype, device)
+      model.pack_decode_weights(reserve_bytes=8 << 30)
+    self.comm = RingComm(device)
     self.next_rank = (rank + 1) % world
     self.prev_rank = (rank - 1) % world
+    self.is_first = self.shard.is_first_layer
+    self.is_last = self.shard.is_last_layer
 
-  def serve_request(self, tokens: Optional[torch.Tensor], max_new: int, temp: float,
-                    on_token=None) -> List[int]:
-    """Run one request through the ring. tokens given on rank 0 only."""
-    dev = self.device
-    hdr = torch.zeros(2, dtype=torch.int64, device=dev if self.world > 1 else "cpu")
-    if self.rank == 0:
-      if tokens.shape[1] >= MAX_SEQ:
-        # reject BEFORE the ring is engaged (followers never see the request)
-        raise ValueError(f"prompt length {tokens.shape[1]} exceeds context {MAX_SEQ}")
-      hdr[0] = tokens.shape[1]
-      hdr[1] = max_new
-    if self.world > 1:
-      dist.broadcast(hdr, 0)
-    S, max_new = int(hdr[0]), int(hdr[1])
-    # clamp so KV writes can never pass cache capacity (the HIP append
-    # kernel trusts pos < T)
-    max_new = min(max_new, MAX_SEQ - S)
-    if self.world > 1:
-      tok_bcast = torch.zeros(1, S, dtype=torch.int64, device=dev)
-      if self.rank == 0:
-        tok_bcast.copy_(tokens.to(dev))
-      dist.broadcast(tok_bcast, 0)
-      tokens = tok_bcast
+    D = self.cfg.dim
+    dev = device
+    # static per-slot state (graph-visible)
+    self.positions = torch.zeros(slots, dtype=torch.int32, device=dev)
+    self.seq_lens = torch.ones(slots, dtype=torch.int32, device=dev)
+    self.tok_buf = torch.zeros(slots, 1, dtype=torch.int64, device=dev)
+    self.hid_buf = torch.zeros(slots, 1, D, dtype=dtype, device=dev)
+    self.logits_buf: Optional[torch.Tensor] = None
+    self.hid_out: Optional[torch.Tensor] = None
+    # control/broadcast buffers
+    self.hdr = torch.zeros(4, dtype=torch.int64, device=dev)
+    self.tok_bcast = torch.zeros(slots, dtype=torch.int64, device=dev)
+    self.active = [False] * slots
+    self.slot_state: List[SlotState] = [SlotState() for _ in range(slots)]
+    self._graph: Optional[torch.cuda.CUDAGraph] = None
+    self.eos_token_id = self.cfg.eos_token_id
+
+  # ---------------- decode tick ----------------
+
+  def _decode_forward(self):
+    x = self.tok_buf if self.is_first else self.hid_buf
+    out = self.model(x, caches=self.cache.caches, positions=self.positions,
+                     start_pos=-1, is_decode=True, seq_lens=self.seq_lens)
+    if self.is_last:
+      if self.logits_buf is None:
+        self.logits_buf = torch.empty_like(out)
+      self.logits_buf.copy_(out)
     else:
-      tokens = tokens.to(dev)
-    total = min(MAX_SEQ, S + max_new)
-    out_tokens: List[int] = []
-    cur_tok = torch.zeros(1, 1, dtype=torch.int64, device=dev)
+      self.hid_out = out
+    self.positions.add_(1)
+    self.seq_lens.add_(1)
+
+  def _build_graph(self):
+    if not self.use_graphs:
+      return
+    self.positions.fill_(1)
+    self.seq_lens.fill_(2)
+    self.tok_buf.random_(0, self.cfg.vocab_size)
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+      with torch.inference_mode():
+        for _ in range(2):
+          self._decode_forward()
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
     with torch.inference_mode():
-      for step in range(max_new):
-        pos0 = S + step - 1 if step > 0 else 0
-        s_cur = 1 if step > 0 else S
-        positions = torch.arange(pos0, pos0 + s_cur, dtype=torch.int32, device=dev)
-        seq_lens = torch.full((1,), pos0 + s_cur, dtype=torch.int32, device=dev)
-        x = tokens if step == 0 else cur_tok
-        if self.shard.is_first_layer:
-          h = self.model(x, caches=self.cache.caches, positions=positions,
-                         start_pos=pos0, is_decode=(step > 0), seq_lens=seq_lens)
+      with torch.cuda.graph(g):
+        self._decode_forward()
+    self._graph = g
+    self._reset_inactive()
+
+  def _reset_inactive(self):
+    """Pin inactive slots at position 0 so graph-side position advance can
+    never run a dead slot's KV writes past capacity."""
+    idle = [i for i in range(self.slots) if not self.active[i]]
+    if idle:
+      idx = torch.tensor(idle, dtype=torch.int64, device=self.positions.device)
+      self.positions[idx] = 0
+      self.seq_lens[idx] = 1
+
+  def decode_tick(self):
+    with torch.inference_mode():
+      if not self.is_first and self.world > 1:
+        self.comm.recv(self.hid_buf, self.prev_rank)
+      if self._graph is not None:
+        self._graph.replay()
+      else:
+        self._decode_forward()
+      if not self.is_last and self.world > 1:
+        self.comm.send(self.hid_out, self.next_rank)
+      if self.is_last:
+        self._sample_all()
+      if self.world > 1:
+        self.comm.broadcast(self.tok_bcast, self.world - 1)
+      if self.is_first:
+        self.tok_buf.copy_(self.tok_bcast.view(self.slots, 1))
+      self._reset_inactive()
+
+  def _sample_all(self):
+    from xotorch_amd import ops
+    logits = self.logits_buf  # [slots, V]
+    greedy = logits.argmax(dim=-1)
+    toks = greedy
+    temps = [self.slot_state[i].temp for i in range(self.slots)]
+    if any(t > 0 for t in temps):
+      toks = greedy.clone()
+      for i, t in enumerate(temps):
+        if t > 0 and self.active[i]:
+          toks[i] = ops.softmax_sample(logits[i:i + 1], temperature=t, top_k=35)[0]
+    self.tok_bcast.copy_(toks)
+
+  # ---------------- prefill (chunked through the ring) ----------------
+
+  def _slot_caches(self, slot: int):
+    out = []
+    for layer in self.cache.caches:
+      sl = tuple(t[slot:slot + 1] if t is not None else None for t in layer)
+      out.append(type(layer)(*sl) if hasattr(layer, "_fields") else sl)
+    return out
+
+  def prefill_slot(self, slot: int, tokens: Optional[torch.Tensor], plen: int) -> Optional[int]:
+    """Run the prompt through the ring in chunks; returns first token on the
+    last stage (None elsewhere)."""
+    dev = self.device
+    caches = self._slot_caches(slot)
+    first_tok = None
+    with torch.inference_mode():
+      for c0 in range(0, plen, PREFILL_CHUNK):
+        c1 = min(c0 + PREFILL_CHUNK, plen)
+        pos = torch.arange(c0, c1, dtype=torch.int32, device=dev)
+        last_chunk = c1 == plen
+        if self.is_first:
+          x = tokens[:, c0:c1].to(dev)
+          h = self.model(x, caches=caches, positions=pos, start_pos=c0)
         else:
-          hbuf = torch.empty(1, s_cur, self.cfg.dim, dtype=self.dtype, device=dev)
-          dist.recv(hbuf, self.prev_rank)
-          h = self.model(hbuf, caches=self.cache.caches, positions=positions,
-                         start_pos=pos0, is_decode=(step > 0), seq_lens=seq_lens)
-        if self.shard.is_last_layer:
-          from xotorch_amd import ops
-          tok = ops.softmax_sample(h, temperature=temp, top_k=35 if temp > 0 else 0).view(1, 1)
-          cur_tok.copy_(tok)
-        elif self.world > 1:
-          dist.send(h.contiguous(), self.next_rank)
+          hbuf = torch.empty(1, c1 - c0, self.cfg.dim, dtype=self.dtype, device=dev)
+          self.comm.recv(hbuf, self.prev_rank)
+          h = self.model(hbuf, caches=caches, positions=pos, start_pos=c0)
+        if not self.is_last:
+          if self.world > 1:
+            self.comm.send(h.contiguous(), self.next_rank)
+        elif last_chunk:
+          st = self.slot_state[slot]
+          if st.temp > 0:
+            from xotorch_amd import ops
+            first_tok = int(ops.softmax_sample(h[:, -1] if h.dim() == 3 else h,
+                                               temperature=st.temp, top_k=35)[0])
+          else:
+            logits = h[:, -1] if h.dim() == 3 else h
+            first_tok = int(logits.argmax(dim=-1)[0])
+      # position state after prefill
+      self.positions[slot] = plen
+      self.seq_lens[slot] = plen + 1
+    return first_tok
+
+  # ---------------- control loop ----------------
+
+  def _bcast_hdr(self, op=OP_IDLE, slot=0, plen=0, extra=0):
+    if self.rank == 0:
+      self.hdr[0], self.hdr[1], self.hdr[2], self.hdr[3] = op, slot, plen, extra
+    if self.world > 1:
+      self.comm.broadcast(self.hdr, 0)
+    return [int(v) for v in self.hdr.cpu()]
+
+  def serve_forever(self, admit_q: Optional["queue.Queue"] = None, emit=None,
+                    idle_sleep: float = 0.002, max_ticks: Optional[int] = None):
+    """Run the ring loop. On rank 0, `admit_q` provides AdmitMsg and `emit`
+    receives (request_id, token, is_finished, meta). Followers pass None."""
+    pending_release: List[int] = []
+    inflight: Dict[int, AdmitMsg] = {}
+    ticks = 0
+    while max_ticks is None or ticks < max_ticks:
+      ticks += 1
+      op, slot, plen, extra = OP_IDLE, 0, 0, 0
+      if self.rank == 0:
+        if pending_release:
+          op, slot = OP_RELEASE, pending_release.pop(0)
+        else:
+          msg = None
+          free = next((i for i in range(self.slots) if not self.active[i]), None)
+          if admit_q is not None and free is not None:
+            try:
+              msg = admit_q.get_nowait()
+            except queue.Empty:
+              msg = None
+          if msg is not None and msg.tokens is None:  # shutdown sentinel
+            op = OP_SHUTDOWN
+          elif msg is not None:
+            op, slot, plen = OP_ADMIT, free, msg.tokens.shape[1]
+            extra = int(msg.temp * 1000)
+            inflight[free] = msg
+          elif any(self.active):
+            op = OP_DECODE
+          else:
+            op = OP_IDLE
+      op, slot, plen, extra = self._bcast_hdr(op, slot, plen, extra)
+
+      if op == OP_SHUTDOWN:
+        break
+      if op == OP_IDLE:
+        if self.rank == 0:
+          time.sleep(idle_sleep)
+        continue
+      if op == OP_RELEASE:
+        self.active[slot] = False
+        self.slot_state[slot] = SlotState()
+        self._reset_inactive()
+        continue
+      if op == OP_ADMIT:
+        if self.rank == 0:
+          msg = inflight[slot]
+          toks = msg.tokens.to(self.device)
+        else:
+          toks = torch.zeros(1, plen, dtype=torch.int64, device=self.device)
         if self.world > 1:
-          dist.broadcast(cur_tok, self.world - 1)
-        t = int(cur_tok[0, 0])
-        out_tokens.append(t)
-        if on_token is not None:
-          on_token(t)
-        eos = self.cfg.eos_token_id
-        if eos is not None and t == eos:
-          break
-    return out_tokens
+          self.comm.broadcast(toks, 0)
+        st = SlotState(prompt_len=plen, temp=extra / 1000.0)
+        self.slot_state[slot] = st
+        first = self.prefill_slot(slot, toks, plen)
+        # ship the first token to rank 0 (it finishes the admit bookkeeping)
+        self.tok_bcast.zero_()
+        if self.is_last and first is not None:
+          self.tok_bcast[slot] = first
+        if self.world > 1:
+          self.comm.broadcast(self.tok_bcast, self.world - 1)
+        if self.is_first:
+          self.tok_buf[slot, 0] = self.tok_bcast[slot]
+        self.active[slot] = True
+        if self.rank == 0:
+          msg = inflight.pop(slot)
+          st.request_id = msg.request_id
+          st.max_new = min(msg.max_new, self.max_seq - plen - 1)
+          st.generated = 1
+          tok = int(self.tok_bcast[slot])
+          fin = (self.eos_token_id is not None and tok == self.eos_token_id) or st.generated >= st.max_new
+          if emit:
+            emit(msg.request_id, tok, fin, {"ttft_s": time.perf_counter() - msg.t_submit})
+          if fin:
+            st.done = True
+            pending_release.append(slot)
+        continue
+      # OP_DECODE
+      self.decode_tick()
+      if self.rank == 0:
+        toks = self.tok_bcast.cpu()
+        for i in range(self.slots):
+          st = self.slot_state[i]
+          if not self.active[i] or st.request_id is None or st.done:
+            continue
+          tok = int(toks[i])
+          st.generated += 1
+          fin = (self.eos_token_id is not None and tok == self.eos_token_id) or st.generated >= st.max_new
+          if emit:
+            emit(st.request_id, tok, fin, {})
+          if fin:
+            st.done = True
+            pending_release.append(i)
+    return ticks
+
+
+# ---------------- rank 0: asyncio API adapter ----------------
+
+class _CallbackSystem:
+  """Duck-typed AsyncCallbackSystem surface the API uses."""
+
+  def __init__(self):
+    from xotorch_amd.helpers import AsyncCallbackSystem
+    self._cbs = AsyncCallbackSystem()
+
+  def register(self, name):
+    return self._cbs.register(name)
+
+  def trigger_all(self, *a):
+    self._cbs.trigger_all(*a)
+
+
+class RingAPINode:
+  """Node-shaped adapter over the ring worker: gives ChatGPTAPI the surface
+  it needs (process_prompt / on_token / inference_engine / topology) while
+  the actual decode runs in the ring thread."""
+
+  def __init__(self, worker: RingSlotWorker, tokenizer, loop):
+    self.worker = worker
+    self.tokenizer = tokenizer
+    self.loop = loop
+    self.on_token = _CallbackSystem()
+    self.admit_q: queue.Queue = queue.Queue()
+    self.request_meta: Dict[str, dict] = {}
+    # the API looks at node.inference_engine.{tokenizer, shard}
+    self.inference_engine = type("E", (), {})()
+    self.inference_engine.tokenizer = tokenizer
+    self.inference_engine.shard = worker.shard
+    self.server = None
+
+  @property
+  def current_topology(self):
+    from xotorch_amd.parallel.topology import Topology, DeviceCapabilities, DeviceFlops
+    topo = Topology()
+    for r in range(self.worker.world):
+      topo.update_node(f"rank{r}", DeviceCapabilities(
+        model="MI355X ring stage", chip="gfx950", memory=288 * 1024,
+        flops=DeviceFlops(fp32=0, fp16=2500.0, int8=0)))
+      topo.add_edge(f"rank{r}", f"rank{(r + 1) % self.worker.world}", "xgmi")
+    topo.active_node_id = "rank0"
+    return topo
+
+  def emit(self, request_id: str, token: int, finished: bool, meta: dict):
+    if meta.get("ttft_s") is not None:
+      self.request_meta[request_id] = meta
+    self.loop.call_soon_threadsafe(self.on_token.trigger_all, request_id, [token], finished)
+
+  async def process_prompt(self, shard, prompt: str, request_id=None, inference_state=None):
+    request_id = request_id or str(uuid.uuid4())
+    state = inference_state or {}
+    ids = self.tokenizer.encode(prompt)
+    if len(ids) >= self.worker.max_seq:
+      raise ValueError(f"prompt length {len(ids)} exceeds context {self.worker.max_seq}")
+    tokens = torch.tensor([ids], dtype=torch.int64)
+    max_new = int(state.get("max_tokens") or 256)
+    temp = float(state.get("temperature") or 0.0)
+    self.admit_q.put(AdmitMsg(request_id, tokens, max_new, temp))
+    return request_id
 
 
 def main():
@@ -133,90 +428,41 @@ def main():
   p.add_argument("--model", type=str, default="llama-3-8b")
   p.add_argument("--port", type=int, default=52415)
   p.add_argument("--model-dir", type=str, default=None)
+  p.add_argument("--slots", type=int, default=SLOTS)
+  p.add_argument("--no-graphs", action="store_true")
   args = p.parse_args()
   rank, world = init_distributed()
   device = "cuda" if torch.cuda.is_available() else "cpu"
   dtype = torch.bfloat16 if device == "cuda" else torch.float32
-  worker = RingWorker(args.model, rank, world, device, dtype, args.model_dir)
+  worker = RingSlotWorker(args.model, rank, world, device, dtype, args.model_dir,
+                          slots=args.slots, use_graphs=not args.no_graphs)
+  if device == "cuda":
+    worker._build_graph()
 
   if rank != 0:
-    # follower loop: serve forever (header broadcast wakes us per request)
-    while True:
-      worker.serve_request(None, 0, 0.0)
+    worker.serve_forever()
     return
 
-  # rank 0: HTTP API + ring driver
-  from aiohttp import web
-  from xotorch_amd.engine.tokenizers import resolve_tokenizer, DummyTokenizer
+  from xotorch_amd.api.chatgpt import ChatGPTAPI
+  from xotorch_amd.engine.tokenizers import DummyTokenizer, resolve_tokenizer
 
   loop = asyncio.new_event_loop()
   asyncio.set_event_loop(loop)
   try:
-    tokenizer = loop.run_until_complete(resolve_tokenizer(args.model_dir or get_repo(args.model, "HIPEngine")))
+    tokenizer = loop.run_until_complete(
+      resolve_tokenizer(args.model_dir or get_repo(args.model, "HIPEngine")))
   except Exception:
     tokenizer = DummyTokenizer()
-  ring_lock = threading.Lock()
-
-  async def completions(request):
-    data = await request.json()
-    messages = data.get("messages", [])
-    stream = data.get("stream", False)
-    temp = float(data.get("temperature", 0.0))
-    max_new = int(data.get("max_tokens") or 256)
-    try:
-      prompt = tokenizer.apply_chat_template(conversation=messages, tokenize=False, add_generation_prompt=True)
-    except Exception:
-      prompt = "\n".join(m.get("content", "") for m in messages)
-    ids = tokenizer.encode(prompt)
-    tokens = torch.tensor([ids], dtype=torch.int64)
-    rid = str(uuid.uuid4())
-    q: Queue = Queue()
-
-    def run_ring():
-      with ring_lock:
-        worker.serve_request(tokens, max_new, temp, on_token=lambda t: q.put(t))
-      q.put(None)
-
-    t = threading.Thread(target=run_ring, daemon=True)
-    t.start()
-    if stream:
-      resp = web.StreamResponse(headers={"Content-Type": "text/event-stream"})
-      await resp.prepare(request)
-      while True:
-        tok = await asyncio.get_running_loop().run_in_executor(None, q.get)
-        if tok is None:
-          break
-        chunk = {"id": rid, "object": "chat.completion.chunk", "model": args.model,
-                 "choices": [{"index": 0, "delta": {"content": tokenizer.decode([tok])}, "finish_reason": None}]}
-        await resp.write(f"data: {json.dumps(chunk)}\n\n".encode())
-      await resp.write(b"data: [DONE]\n\n")
-      await resp.write_eof()
-      return resp
-    toks = []
-    while True:
-      tok = await asyncio.get_running_loop().run_in_executor(None, q.get)
-      if tok is None:
-        break
-      toks.append(tok)
-    return web.json_response({
-      "id": rid, "object": "chat.completion", "model": args.model,
-      "choices": [{"index": 0, "message": {"role": "assistant", "content": tokenizer.decode(toks)},
-                   "finish_reason": "stop"}],
-      "usage": {"completion_tokens": len(toks)},
-    })
-
-  async def health(request):
-    return web.json_response({"status": "ok", "world": world, "model": args.model})
-
-  app = web.Application()
-  app.router.add_post("/v1/chat/completions", completions)
-  app.router.add_get("/healthcheck", health)
+  node = RingAPINode(worker, tokenizer, loop)
+  ring_thread = threading.Thread(
+    target=worker.serve_forever, args=(node.admit_q, node.emit), daemon=True)
+  ring_thread.start()
+  api = ChatGPTAPI(node, "TorchEngine", default_model=args.model)
 
   async def start():
-    runner = web.AppRunner(app)
-    await runner.setup()
-    await web.TCPSite(runner, "0.0.0.0", args.port).start()
-    print(f"ring server ({world} GPUs) on :{args.port}")
+    await api.run("0.0.0.0", args.port)
+    print(f"ring server ({world} ranks, {args.slots} slots, graphs="
+          f"{worker._graph is not None}) on :{args.port}", flush=True)
     await asyncio.Event().wait()
 
   loop.run_until_complete(start())
