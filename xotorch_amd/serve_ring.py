@@ -31,11 +31,16 @@ import argparse
 import asyncio
 import os
 import queue
+import sys
 import threading
 import time
 import uuid
 from dataclasses import dataclass, field
+from pathlib import Path
 from typing import Dict, List, Optional
+
+if __package__ in (None, ""):  # torchrun launches this file by path
+  sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 import torch.distributed as dist
