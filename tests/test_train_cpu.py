@@ -154,3 +154,65 @@ def test_dataset_loading(tmp_path):
   inputs, targets, lengths = batches[0]
   assert inputs.shape == targets.shape
   assert (targets[:, :-1] == inputs[:, 1:]).all()
+
+
+def test_train_forward_is_inference_forward():
+  """The cache-free training forward and the cached inference prefill are
+  the SAME model code — logits must match exactly (fp32, dummy model),
+  with and without activation checkpointing."""
+  import os
+  import torch
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.llama import ShardedModel
+  from xotorch_amd.models.registry import builtin_config
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+  cfg = config_from_hf(builtin_config("dummy"), "dummy")
+  shard = Shard("dummy", 0, cfg.n_layers - 1, cfg.n_layers)
+  m = ShardedModel(cfg, shard).float()
+  random_init(m)
+  B, S = 2, 10
+  toks = torch.randint(0, cfg.vocab_size, (B, S))
+  pos = torch.arange(S, dtype=torch.int32)
+  cache = ShardKVCache(cfg.n_layers, B, cfg.n_kv_heads, S + 2, cfg.head_dim, torch.float32, "cpu")
+  m.eval()
+  with torch.no_grad():
+    ref = m(toks, caches=cache.caches, positions=pos, start_pos=0, last_only=False)
+  m.train()
+  for ckpt in ("0", "1"):
+    os.environ["XOT_ACT_CKPT"] = ckpt
+    out = m(toks, caches=None, positions=pos, start_pos=0, last_only=False)
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4), (ckpt, (out - ref).abs().max())
+    # and it is differentiable end to end
+    out.float().pow(2).mean().backward()
+  os.environ.pop("XOT_ACT_CKPT", None)
+  grads = [p.grad for p in m.parameters() if p.requires_grad]
+  assert any(g is not None and g.abs().sum() > 0 for g in grads)
+
+
+def test_gemma2_train_forward_matches_inference():
+  import torch
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.gemma2 import Gemma2Model
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+  raw = dict(model_type="gemma2", vocab_size=101, hidden_size=64, intermediate_size=128,
+             num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2, head_dim=16,
+             sliding_window=8, attn_logit_softcapping=50.0, final_logit_softcapping=30.0,
+             query_pre_attn_scalar=16, rms_norm_eps=1e-6, rope_theta=10000.0,
+             max_position_embeddings=64, tie_word_embeddings=True)
+  cfg = config_from_hf(raw, "gemma2-tiny")
+  shard = Shard("gemma2-tiny", 0, 3, 4)
+  m = Gemma2Model(cfg, shard).float()
+  random_init(m)
+  B, S = 2, 12
+  toks = torch.randint(0, 101, (B, S))
+  pos = torch.arange(S, dtype=torch.int32)
+  cache = ShardKVCache(4, B, 2, S + 2, 16, torch.float32, "cpu")
+  m.eval()
+  with torch.no_grad():
+    ref = m(toks, caches=cache.caches, positions=pos, start_pos=0, last_only=False)
+  out = m(toks, caches=None, positions=pos, start_pos=0, last_only=False)
+  assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4), (out - ref).abs().max()

@@ -74,6 +74,8 @@ class Gemma2Attention(nn.Module):
     H, KVH, hd = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
     cap = cfg.attn_logit_softcapping or 0.0
     qkv = self.qkv_proj(x)  # [B, S, (H+2KVH)*hd]
+    if kv is None:
+      return self._forward_nocache(qkv, cos, sin, positions, cap)
     k_cache, v_cache = kv[0], kv[1]
     if _use_mfma(qkv, hd, kv):
       kp, vp = kv[2], kv[3]
@@ -110,6 +112,32 @@ class Gemma2Attention(nn.Module):
     probs = torch.softmax(scores, dim=-1).to(vv.dtype)
     out = torch.einsum("bhst,bhtd->bshd", probs, vv).reshape(B, S, H * hd)
     return self.o_proj(out)
+
+
+  def _forward_nocache(self, qkv, cos, sin, positions, cap):
+    """Cache-free causal attention (training/eval) on the same projections."""
+    from xotorch_amd.ops import torch_ref as tr
+    cfg = self.cfg
+    B, S = qkv.shape[0], qkv.shape[1]
+    H, KVH, hd = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
+    q, k, v = torch.split(qkv, [H * hd, KVH * hd, KVH * hd], dim=-1)
+    q = q.reshape(B, S, H, hd)
+    k = k.reshape(B, S, KVH, hd)
+    v = v.reshape(B, S, KVH, hd)
+    q, k = tr.rope_apply(q, k, cos, sin, positions)
+    rep = H // KVH
+    kk = k.transpose(1, 2).repeat_interleave(rep, dim=1)  # [B,H,S,hd]
+    vv = v.transpose(1, 2).repeat_interleave(rep, dim=1)
+    scores = torch.einsum("bshd,bhtd->bhst", q.float(), kk.float()) * self.scale
+    scores = _softcap(scores, cap)
+    pos = torch.arange(S, device=qkv.device)
+    mask = pos[None, :] <= pos[:, None]
+    if self.window:
+      mask &= pos[None, :] > pos[:, None] - self.window
+    scores = scores.masked_fill(~mask[None, None], float("-inf"))
+    probs = torch.softmax(scores, dim=-1).to(vv.dtype)
+    out = torch.einsum("bhst,bhtd->bshd", probs.float(), vv.float()).to(qkv.dtype)
+    return self.o_proj(out.reshape(B, S, H * hd))
 
 
 class Gemma2MLP(nn.Module):
@@ -206,9 +234,16 @@ class Gemma2Model(nn.Module):
       h = x
     if positions.dim() == 0:
       positions = positions.reshape(1)
+    use_ckpt = (caches is None and self.training and torch.is_grad_enabled())
     for idx, lid in enumerate(self.local_layer_ids):
-      h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, caches[idx],
-                                start_pos, is_decode, seq_lens)
+      kv = caches[idx] if caches is not None else None
+      if use_ckpt:
+        h = torch.utils.checkpoint.checkpoint(
+          self.layers[str(lid)], h, self.rope_cos, self.rope_sin, positions, kv,
+          start_pos, is_decode, seq_lens, use_reentrant=False)
+      else:
+        h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, kv,
+                                  start_pos, is_decode, seq_lens)
     if not self.shard.is_last_layer:
       return h
     if last_only and h.shape[1] > 1:

@@ -158,7 +158,37 @@ class Attention(nn.Module):
       self.q_norm = nn.Parameter(torch.ones(hd))
       self.k_norm = nn.Parameter(torch.ones(hd))
 
+  def _forward_nocache(self, x, cos, sin, positions):
+    """Cache-free causal attention (training/eval): the SAME modules and
+    weights as the inference path, autograd-safe ops (fp32 sdpa). This is
+    the one and only training forward — the engine calls the model, so the
+    two paths cannot drift (round-1 VERDICT weak #7)."""
+    import torch.nn.functional as F
+    B, S, _ = x.shape
+    cfg = self.cfg
+    H, KVH, hd = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
+    qkv = self.qkv_proj(x)
+    q, k, v = torch.split(qkv, [H * hd, KVH * hd, KVH * hd], dim=-1)
+    q = q.reshape(B, S, H, hd)
+    k = k.reshape(B, S, KVH, hd)
+    v = v.reshape(B, S, KVH, hd)
+    if cfg.qk_norm:
+      from xotorch_amd.ops import torch_ref as tr
+      q = tr.rmsnorm(q.float(), self.q_norm.float(), cfg.norm_eps).to(q.dtype)
+      k = tr.rmsnorm(k.float(), self.k_norm.float(), cfg.norm_eps).to(k.dtype)
+    q, k = ops.rope_apply(q, k, cos, sin, positions)
+    rep = H // KVH
+    out = F.scaled_dot_product_attention(
+      q.transpose(1, 2).float(),
+      k.transpose(1, 2).repeat_interleave(rep, dim=1).float(),
+      v.transpose(1, 2).repeat_interleave(rep, dim=1).float(),
+      is_causal=True,
+    ).to(x.dtype)
+    return self.o_proj(out.transpose(1, 2).reshape(B, S, H * hd))
+
   def forward(self, x, cos, sin, positions, kv, start_pos: int, is_decode: bool, seq_lens=None):
+    if kv is None:
+      return self._forward_nocache(x, cos, sin, positions)
     B, S, _ = x.shape
     cfg = self.cfg
     H, hd = cfg.n_heads, cfg.head_dim
@@ -463,8 +493,20 @@ class ShardedModel(nn.Module):
     else:
       h = x
     cos, sin = self.rope_cos, self.rope_sin
+    # caches=None selects the cache-free training forward; activation
+    # checkpointing (on by default when training under grad) trades the
+    # per-layer activation residency for a recompute in backward
+    use_ckpt = (caches is None and self.training and torch.is_grad_enabled()
+                and os.getenv("XOT_ACT_CKPT", "1") == "1")
     for idx, lid in enumerate(self.local_layer_ids):
-      h = self.layers[str(lid)](h, cos, sin, positions, caches[idx], start_pos, is_decode, seq_lens)
+      layer = self.layers[str(lid)]
+      kv = caches[idx] if caches is not None else None
+      if use_ckpt:
+        h = torch.utils.checkpoint.checkpoint(
+          layer, h, cos, sin, positions, kv, start_pos, is_decode, seq_lens,
+          use_reentrant=False)
+      else:
+        h = layer(h, cos, sin, positions, kv, start_pos, is_decode, seq_lens)
     if not self.shard.is_last_layer:
       return h
     if last_only and h.shape[1] > 1:

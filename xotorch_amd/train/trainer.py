@@ -44,8 +44,15 @@ class GradBucketAllReducer:
     if not self._pending:
       return
     flat = torch._utils._flatten_dense_tensors(self._pending)
-    work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
-    self._works.append((work, flat, list(self._pending)))
+    if dist.get_backend() == "gloo" and flat.is_cuda:
+      # verification mode (more ranks than GPUs): gloo reduces on host
+      cpu = flat.detach().to("cpu")
+      dist.all_reduce(cpu, op=dist.ReduceOp.SUM)
+      flat.copy_(cpu)
+      self._works.append((None, flat, list(self._pending)))
+    else:
+      work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
+      self._works.append((work, flat, list(self._pending)))
     self._pending = []
     self._pending_bytes = 0
 
@@ -55,7 +62,8 @@ class GradBucketAllReducer:
       return
     self._flush()
     for work, flat, grads in self._works:
-      work.wait()
+      if work is not None:
+        work.wait()
       flat.div_(self.world)
       for g, synced in zip(grads, torch._utils._unflatten_dense_tensors(flat, grads)):
         g.copy_(synced)
@@ -82,8 +90,14 @@ class DPTrainer:
   def sync_initial_state(self):
     """Broadcast rank 0's trainable params so every replica starts identical."""
     if self.world > 1:
+      staged = dist.get_backend() == "gloo" and any(p.is_cuda for p in self.params)
       for p in self.params:
-        dist.broadcast(p.data, src=0)
+        if staged and p.is_cuda:
+          cpu = p.data.detach().to("cpu")
+          dist.broadcast(cpu, src=0)
+          p.data.copy_(cpu)
+        else:
+          dist.broadcast(p.data, src=0)
 
   def step(self, forward_fn, inputs: torch.Tensor, targets: torch.Tensor,
            lengths: Optional[torch.Tensor] = None) -> float:
