@@ -260,8 +260,9 @@ class TorchEngine(InferenceEngine):
     """Training/eval forward THROUGH THE MODEL's own cache-free path
     (caches=None): one forward implementation for inference and training —
     drift between them is impossible by construction (round-1 VERDICT
-    weak #7). Activation checkpointing is on inside the model when
-    training under grad (XOT_ACT_CKPT=0 disables)."""
+    weak #7). Activation checkpointing inside the model is opt-in
+    (XOT_ACT_CKPT=1): measured 380 vs 283 ms/step on 8B LoRA B=8 — worth it
+    only when activations would not fit."""
     S = x.shape[1]
     positions = torch.arange(0, S, dtype=torch.int32, device=self.device)
     return self.model(x, caches=None, positions=positions, start_pos=0, last_only=False)

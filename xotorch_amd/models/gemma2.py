@@ -25,6 +25,7 @@ last owns the final norm + tied head; caches are the engine's LayerKV pairs.
 from __future__ import annotations
 
 import math
+import os
 from typing import List, Optional
 
 import torch
@@ -234,7 +235,8 @@ class Gemma2Model(nn.Module):
       h = x
     if positions.dim() == 0:
       positions = positions.reshape(1)
-    use_ckpt = (caches is None and self.training and torch.is_grad_enabled())
+    use_ckpt = (caches is None and self.training and torch.is_grad_enabled()
+                and os.getenv("XOT_ACT_CKPT", "0") == "1")
     for idx, lid in enumerate(self.local_layer_ids):
       kv = caches[idx] if caches is not None else None
       if use_ckpt:

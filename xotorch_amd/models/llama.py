@@ -497,7 +497,7 @@ class ShardedModel(nn.Module):
     # checkpointing (on by default when training under grad) trades the
     # per-layer activation residency for a recompute in backward
     use_ckpt = (caches is None and self.training and torch.is_grad_enabled()
-                and os.getenv("XOT_ACT_CKPT", "1") == "1")
+                and os.getenv("XOT_ACT_CKPT", "0") == "1")
     for idx, lid in enumerate(self.local_layer_ids):
       layer = self.layers[str(lid)]
       kv = caches[idx] if caches is not None else None
