@@ -587,3 +587,104 @@ def test_gemma2_gpu_hip_path_matches_eager():
             start_pos=S, is_decode=True, seq_lens=sl)
     lr2 = mc(nxt.cpu(), caches=cache_ref.caches, positions=torch.tensor([S]), start_pos=S, is_decode=True)
     assert (lg2.float().cpu().argmax(-1) == lr2.argmax(-1)).all()
+
+
+# ---------------- MLA (DeepSeek) absorbed-latent decode ----------------
+
+def _mla_pack_ref(lat, rot, T32):
+  """Python reference packer for the MLA fragment layouts."""
+  B, T = lat.shape[0], lat.shape[1]
+  full = torch.cat([lat, rot], dim=-1)  # [B, T, 576]
+  kp = torch.zeros(B, T32 // 16, 18, 64, 8, dtype=lat.dtype, device=lat.device)
+  vp = torch.zeros(B, 32, T32 // 32, 64, 8, dtype=lat.dtype, device=lat.device)
+  for pos in range(T):
+    for d in range(576):
+      kp[:, pos >> 4, d >> 5, ((d & 31) >> 3) * 16 + (pos & 15), d & 7] = full[:, pos, d]
+      if d < 512:
+        vp[:, d >> 4, pos >> 5, ((pos & 31) >> 3) * 16 + (d & 15), pos & 7] = full[:, pos, d]
+  return kp, vp
+
+
+def test_mla_append_layout(hip):
+  from xotorch_amd.ops import _hip_ops
+  B, T = 2, 48
+  t32 = 64
+  lat = bt(B, T, 512, seed=101)
+  rot = bt(B, T, 64, seed=102)
+  kp = torch.zeros(B, t32 // 16, 18, 64, 8, dtype=torch.bfloat16, device="cuda")
+  vp = torch.zeros(B, 32, t32 // 32, 64, 8, dtype=torch.bfloat16, device="cuda")
+  pos = torch.arange(T, dtype=torch.int32, device="cuda")
+  _hip_ops.mla_append(lat, rot, pos, kp, vp)
+  kp_ref, vp_ref = _mla_pack_ref(lat, rot, t32)
+  assert torch.equal(kp, kp_ref)
+  assert torch.equal(vp, vp_ref)
+
+
+@pytest.mark.parametrize("B,H,T,sl", [(2, 8, 64, 50), (1, 16, 128, 128), (3, 4, 96, 33)])
+def test_attn_decode_mla_vs_ref(hip, B, H, T, sl):
+  from xotorch_amd.ops import _hip_ops
+  torch.manual_seed(0)
+  lat = bt(B, T, 512, seed=111, scale=0.5)
+  rot = bt(B, T, 64, seed=112, scale=0.5)
+  q = bt(B, H, 576, seed=113, scale=0.2)
+  t32 = (T + 31) // 32 * 32
+  kp = torch.zeros(B, t32 // 16, 18, 64, 8, dtype=torch.bfloat16, device="cuda")
+  vp = torch.zeros(B, 32, t32 // 32, 64, 8, dtype=torch.bfloat16, device="cuda")
+  _hip_ops.mla_append(lat, rot, torch.arange(T, dtype=torch.int32, device="cuda"), kp, vp)
+  sl_t = torch.full((B,), sl, dtype=torch.int32, device="cuda")
+  scale = 576 ** -0.5
+  out = _hip_ops.attn_decode_mla(q, kp, vp, sl_t, scale).float()
+  # reference: MQA over [lat | rot], PV over lat
+  full = torch.cat([lat, rot], dim=-1).float()[:, :sl]       # [B, sl, 576]
+  scores = torch.einsum("bhd,btd->bht", q.float(), full) * scale
+  p = torch.softmax(scores, dim=-1)
+  ref = torch.einsum("bht,btl->bhl", p, lat.float()[:, :sl])
+  assert torch.allclose(out, ref, atol=3e-2, rtol=3e-2), (out - ref).abs().max()
+
+
+def test_deepseek_gpu_decode_matches_eager():
+  """Tiny MLA model (real 512/64 latent dims): GPU HIP decode path vs the
+  CPU fp32 eager oracle — prefill + cached decode steps."""
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.deepseek_v3 import DeepseekV3Model
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+  raw = dict(model_type="deepseek_v3", vocab_size=512, hidden_size=256, intermediate_size=512,
+             num_hidden_layers=2, num_attention_heads=8, num_key_value_heads=8,
+             kv_lora_rank=512, qk_rope_head_dim=64, qk_nope_head_dim=128, v_head_dim=128,
+             q_lora_rank=0, first_k_dense_replace=2, n_routed_experts=0,
+             rms_norm_eps=1e-6, rope_theta=10000.0, max_position_embeddings=128)
+  cfg = config_from_hf(raw, "ds-tiny")
+  shard = Shard("ds-tiny", 0, 1, 2)
+  torch.manual_seed(3)
+  m = DeepseekV3Model(cfg, shard).to("cuda").to(torch.bfloat16)
+  random_init(m)
+  m.reset_rope()
+  m.eval()
+  mc = DeepseekV3Model(cfg, shard).float()
+  mc.load_state_dict({k: v.float().cpu() for k, v in m.state_dict().items()}, strict=False)
+  mc.reset_rope()
+  mc.eval()
+  B, S = 2, 33
+  toks = torch.randint(0, 512, (B, S), device="cuda")
+  heads, kd, vd = cfg.kv_cache_dims()
+  cg = ShardKVCache(2, B, heads, S + 8, kd, torch.bfloat16, "cuda", v_dim=vd)
+  cc = ShardKVCache(2, B, heads, S + 8, kd, torch.float32, "cpu", v_dim=vd)
+  assert cg.caches[0].kp is not None, "MLA packed cache copies missing"
+  with torch.inference_mode():
+    pos = torch.arange(S, dtype=torch.int32, device="cuda")
+    lg = m(toks, caches=cg.caches, positions=pos, start_pos=0)
+    lr = mc(toks.cpu(), caches=cc.caches, positions=torch.arange(S), start_pos=0)
+    assert (lg.float().cpu().argmax(-1) == lr.argmax(-1)).all()
+    nxt = lg.argmax(-1, keepdim=True)
+    for step in range(3):
+      p = S + step
+      lg = m(nxt, caches=cg.caches, positions=torch.tensor([p], dtype=torch.int32, device="cuda"),
+             start_pos=p, is_decode=True)
+      lr = mc(nxt.cpu(), caches=cc.caches, positions=torch.tensor([p]), start_pos=p, is_decode=True)
+      agree = (lg.float().cpu().argmax(-1) == lr.argmax(-1)).float().mean()
+      assert agree >= 0.5, (step, agree)
+      assert torch.allclose(lg.float().cpu(), lr.float(), atol=0.5, rtol=0.1), \
+        (step, (lg.float().cpu() - lr.float()).abs().max())
+      nxt = lg.argmax(-1, keepdim=True)

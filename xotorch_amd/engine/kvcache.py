@@ -51,6 +51,14 @@ class ShardKVCache:
     self.caches: List[LayerKV] = []
     vd = v_dim if v_dim is not None else head_dim
     packed = _want_packed(device, head_dim, dtype) and vd == head_dim
+    # MLA latent cache (k = 512-dim latent, v = 64-dim roped shared key,
+    # one kv "head"): fragment-packed copies for the absorbed-MQA MFMA
+    # decode kernel (18 qk chunks / 32 pv groups — hip_ops.hip MLA section)
+    mla_packed = (
+      str(device).startswith("cuda") and dtype == torch.bfloat16
+      and n_kv_heads == 1 and head_dim == 512 and vd == 64
+      and os.getenv("XOT_MFMA_ATTN", "1") == "1"
+    )
     t32 = (capacity + 31) // 32 * 32
     for _ in range(n_layers):
       k = torch.zeros(batch, n_kv_heads, capacity, head_dim, dtype=dtype, device=device)
@@ -58,6 +66,10 @@ class ShardKVCache:
       if packed:
         kp = torch.zeros(batch, n_kv_heads, t32 // 16, 4, 64, 8, dtype=dtype, device=device)
         vp = torch.zeros(batch, n_kv_heads, 8, t32 // 32, 64, 8, dtype=dtype, device=device)
+        self.caches.append(LayerKV(k, v, kp, vp))
+      elif mla_packed:
+        kp = torch.zeros(batch, t32 // 16, 18, 64, 8, dtype=dtype, device=device)
+        vp = torch.zeros(batch, 32, t32 // 32, 64, 8, dtype=dtype, device=device)
         self.caches.append(LayerKV(k, v, kp, vp))
       else:
         self.caches.append(LayerKV(k, v))

@@ -94,6 +94,18 @@ class MLAttention(nn.Module):
     lat_c, rot_c = kv[0], kv[1]
     lat_c[:, 0, start_pos: start_pos + S] = kv_nope
     rot_c[:, 0, start_pos: start_pos + S] = k_rot[:, :, 0, :]
+    kp = kv[2] if len(kv) > 2 else None
+    if kp is not None and x.is_cuda and x.dtype == torch.bfloat16:
+      # keep the fragment-packed copies appended (prefill AND decode) so
+      # the MFMA decode kernel can stream them
+      from xotorch_amd.ops import _load_hip
+      hip = _load_hip()
+      if hip is not None:
+        hip.mla_append(kv_nope.to(x.dtype).contiguous(),
+                       k_rot[:, :, 0, :].to(x.dtype).contiguous(),
+                       positions.to(torch.int32).contiguous(), kp, kv[3])
+        if S == 1 and start_pos > 0 and not torch.is_grad_enabled():
+          return self._decode_mfma(x, q_pass, q_rot, kv, start_pos, hip)
     total = start_pos + S
     lat = lat_c[:, 0, :total]                                   # [B, T, kv_lora]
     krot = rot_c[:, 0, :total]                                  # [B, T, rope_d]
@@ -111,6 +123,25 @@ class MLAttention(nn.Module):
     probs = torch.softmax(scores, dim=-1)
     out = torch.einsum("bhst,bthd->bshd", probs, v.float()).to(x.dtype)
     return self.o_proj(out.reshape(B, S, H * vd))
+
+
+  def _decode_mfma(self, x, q_pass, q_rot, kv, start_pos: int, hip):
+    """Absorbed-latent MFMA decode: kv_b is folded into q and out, so
+    attention runs as MQA over the packed 1152 B/token latent stream
+    (hip_ops.hip attn_decode_mla)."""
+    cfg = self.cfg
+    B, S, H = q_pass.shape[0], q_pass.shape[1], cfg.n_heads
+    nope, vd, lat = cfg.qk_nope_head_dim, cfg.v_head_dim, cfg.kv_lora_rank
+    if not hasattr(self, "_w_k"):
+      W = self.kv_b_proj.weight.view(H, nope + vd, lat)
+      self._w_k = W[:, :nope, :].contiguous()   # [H, nope, lat]
+      self._w_v = W[:, nope:, :].contiguous()   # [H, vd, lat]
+    q_lat = torch.einsum("bshn,hnl->bshl", q_pass.to(x.dtype), self._w_k)
+    qfull = torch.cat([q_lat, q_rot.to(x.dtype)], dim=-1).reshape(B, H, lat + cfg.qk_rope_head_dim)
+    sl = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
+    out_lat = hip.attn_decode_mla(qfull.contiguous(), kv[2], kv[3], sl, self.scale)
+    out = torch.einsum("bhl,hdl->bhd", out_lat, self._w_v).to(x.dtype)
+    return self.o_proj(out.reshape(B, 1, H * vd))
 
 
 class DsMLP(nn.Module):
@@ -216,7 +247,8 @@ class DeepseekV3Model(nn.Module):
     return list(range(self.shard.start_layer, self.shard.end_layer + 1))
 
   def pack_decode_weights(self, reserve_bytes: int = 0) -> int:
-    return 0  # plain-torch compute path (no MFMA prepack yet)
+    return 0  # GEMMs stay on hipBLASLt; MLA decode attention runs the
+    # absorbed-latent MFMA kernel (attn_decode_mla) via the packed cache
 
   def head_weight(self):
     return self.lm_head.weight
@@ -224,6 +256,8 @@ class DeepseekV3Model(nn.Module):
   def forward(self, x, caches, positions, start_pos: int, is_decode: bool = False,
               seq_lens=None, last_only: bool = True):
     cfg = self.cfg
+    if caches is None:
+      raise NotImplementedError("MLA training forward is not implemented (inference only)")
     h = self.embed_tokens(x) if x.dtype in (torch.int32, torch.int64) else x
     if positions.dim() == 0:
       positions = positions.reshape(1)

@@ -747,6 +747,197 @@ __global__ __launch_bounds__(128) void attn_decode_merge(
 }
 
 // ---------------------------------------------------------------------------
+// MLA (DeepSeek V2/V3) decode attention on matrix cores — ABSORBED form.
+//
+// The kv_b projection is folded into the query and output (DeepSeek's own
+// inference trick): q_lat[h] = q_nope[h] @ W_k[h] lives in latent space, so
+// decode attention is MQA over the cached per-token latent(512)+rope(64)
+// stream — 1152 B/token TOTAL regardless of head count (the reference's
+// torchtune engine cannot run MLA at all; a GQA cache for V3 would be
+// n_heads*576 per token).  Per 32-position tile and 16-head group:
+//   scores[16h][16p]: 18 x v_mfma_f32_16x16x32_bf16 per half-tile
+//     (A = q fragments re-gathered from L1, B = 1 KB coalesced packed-cache
+//      streams — same fragment geometry as the GQA kernel)
+//   online softmax via 4 x shfl over 16-lane column groups (shared code shape)
+//   out_lat[16h][512] += 32 x mfma (P from LDS image, V = packed latent)
+// One wave per (b, head-group, split); fp32 (m, l, o[512]) partials merged
+// by attn_decode_merge512.  Latent+rope cache copies are kept in the SAME
+// fragment-packed layouts as the GQA kp/vp (18 qk-chunks / 32 pv-groups),
+// appended by mla_append_kernel.
+// ---------------------------------------------------------------------------
+
+#define MLA_LAT 512
+#define MLA_ROPE 64
+#define MLA_DQK 576   // latent + rope, the absorbed q/k width
+#define MLA_CHQK 18   // 32-dim MFMA chunks across DQK
+#define MLA_GPV 32    // 16-dim output groups across LAT
+
+__global__ __launch_bounds__(256) void mla_append_kernel(
+    const unsigned short* __restrict__ lat, const unsigned short* __restrict__ rot,
+    const int* __restrict__ positions, unsigned short* __restrict__ kp,
+    unsigned short* __restrict__ vp, int B, int S, int T32, int npos) {
+  // wave per (b, s); lane covers dims d = lane*9 .. lane*9+8 (576 = 64*9)
+  const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (wid >= B * S) return;
+  const int lane = threadIdx.x & 63;
+  const int s = wid % S;
+  const int b = wid / S;
+  const int pos = positions[(npos == B * S && npos != S) ? wid : s];
+  const unsigned short* lrow = lat + ((size_t)b * S + s) * MLA_LAT;
+  const unsigned short* rrow = rot + ((size_t)b * S + s) * MLA_ROPE;
+  const size_t kbase = (size_t)b * (T32 >> 4) * MLA_CHQK * 512 + (size_t)(pos >> 4) * MLA_CHQK * 512;
+  const size_t vbase = (size_t)b * MLA_GPV * (T32 >> 5) * 512;
+#pragma unroll
+  for (int i = 0; i < 9; ++i) {
+    const int d = lane * 9 + i;
+    const unsigned short v = (d < MLA_LAT) ? lrow[d] : rrow[d - MLA_LAT];
+    // K-fragment image (16 pos x 32 dims per chunk)
+    kp[kbase + (size_t)(d >> 5) * 512 + (((d & 31) >> 3) * 16 + (pos & 15)) * 8 + (d & 7)] = v;
+    if (d < MLA_LAT) {
+      // V-fragment image (32 pos x 16 dims per group)
+      vp[vbase + ((size_t)(d >> 4) * (T32 >> 5) + (pos >> 5)) * 512
+         + (((pos & 31) >> 3) * 16 + (d & 15)) * 8 + (pos & 7)] = v;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void attn_decode_mla_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
+    const unsigned short* __restrict__ vp, const int* __restrict__ seq_lens,
+    float* __restrict__ ws_o, float* __restrict__ ws_ml,
+    int B, int H, int T32, int nsplit, float scale) {
+  const int wv = threadIdx.x >> 6;
+  const int wid = blockIdx.x * 4 + wv;
+  const int hgroups = (H + 15) >> 4;
+  __shared__ unsigned short plds_all[4][16 * 48];
+  if (wid >= B * hgroups * nsplit) return;
+  const int split = wid % nsplit;
+  const int hg = (wid / nsplit) % hgroups;
+  const int b = wid / (nsplit * hgroups);
+  const int lane = threadIdx.x & 63;
+  const int sl = seq_lens[b];
+  const int chunk = ((T32 / nsplit + 31) >> 5) << 5;
+  const int c0 = split * chunk;
+  const int c1 = min(c0 + chunk, sl);
+  unsigned short* plds = plds_all[wv];
+
+  const int nq = min(16, H - hg * 16);
+  const int qi = min(lane & 15, nq - 1);
+  const unsigned short* qrow = q + ((size_t)b * H + hg * 16 + qi) * MLA_DQK + (lane >> 4) * 8;
+
+  float m[4], lsum[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m[r] = -INFINITY; lsum[r] = 0.f; }
+  floatx4 acco[MLA_GPV];
+#pragma unroll
+  for (int g = 0; g < MLA_GPV; ++g) acco[g] = (floatx4)(0.f);
+
+  const size_t kbase = (size_t)b * (T32 >> 4) * MLA_CHQK * 512;
+  const size_t vbase = (size_t)b * MLA_GPV * (T32 >> 5) * 512;
+  const int col = lane & 15;
+
+  for (int t = c0; t < c1; t += 32) {
+    floatx4 sc[2];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) sc[h] = (floatx4)(0.f);
+    // scores over the 18 qk chunks; q fragments re-gathered per chunk
+    // (L1-resident after the first tile), k streamed packed
+    for (int c = 0; c < MLA_CHQK; ++c) {
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(qrow + c * 32);
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const unsigned short* kt = kp + kbase
+            + ((size_t)((t >> 4) + h) * MLA_CHQK + c) * 512 + (size_t)lane * 8;
+        const bf16x8 kb = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt));
+        sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, kb, sc[h], 0, 0, 0);
+      }
+    }
+    float rmax[4];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int pos = t + h * 16 + col;
+      const bool ok = pos < c1;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) sc[h][r] = ok ? sc[h][r] * scale : -INFINITY;
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) rmax[r] = fmaxf(sc[0][r], sc[1][r]);
+#pragma unroll
+    for (int mm = 1; mm < 16; mm <<= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) rmax[r] = fmaxf(rmax[r], __shfl_xor(rmax[r], mm));
+    float alpha[4], rsum[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float mn = fmaxf(m[r], rmax[r]);
+      alpha[r] = (lsum[r] > 0.f) ? __expf(m[r] - mn) : 0.f;
+      m[r] = mn;
+      const float p0 = __expf(sc[0][r] - mn);
+      const float p1 = __expf(sc[1][r] - mn);
+      sc[0][r] = p0;
+      sc[1][r] = p1;
+      rsum[r] = p0 + p1;
+    }
+#pragma unroll
+    for (int mm = 1; mm < 16; mm <<= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) rsum[r] += __shfl_xor(rsum[r], mm);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) lsum[r] = lsum[r] * alpha[r] + rsum[r];
+#pragma unroll
+    for (int g = 0; g < MLA_GPV; ++g)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acco[g][r] *= alpha[r];
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        plds[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = f2b(sc[h][r]);
+    const bf16x8 pf = *reinterpret_cast<const bf16x8*>(plds + (lane & 15) * 48 + (lane >> 4) * 8);
+    const unsigned short* vt = vp + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
+    for (int g = 0; g < MLA_GPV; ++g) {
+      const bf16x8 vb = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16x8*>(vt + (size_t)g * (T32 >> 5) * 512));
+      acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vb, acco[g], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow_i = (lane >> 4) * 4 + r;
+    if (qrow_i < nq) {
+      const size_t pidx = ((size_t)(b * H + hg * 16 + qrow_i) * nsplit + split);
+      for (int g = 0; g < MLA_GPV; ++g) ws_o[pidx * MLA_LAT + g * 16 + col] = acco[g][r];
+      if (col == 0) {
+        ws_ml[pidx * 2 + 0] = m[r];
+        ws_ml[pidx * 2 + 1] = lsum[r];
+      }
+    }
+  }
+}
+
+// wide merge for the MLA 512-dim latent output: 256 threads, 2 dims each
+__global__ __launch_bounds__(256) void attn_decode_merge512(
+    const float* __restrict__ ws_o, const float* __restrict__ ws_ml,
+    unsigned short* __restrict__ out, int nsplit) {
+  const int bh = blockIdx.x;
+  float M = -INFINITY;
+  for (int i = 0; i < nsplit; ++i) M = fmaxf(M, ws_ml[((size_t)bh * nsplit + i) * 2 + 0]);
+#pragma unroll
+  for (int e = 0; e < 2; ++e) {
+    const int d = threadIdx.x + e * 256;
+    float L = 0.f, O = 0.f;
+    for (int i = 0; i < nsplit; ++i) {
+      const float lg = ws_ml[((size_t)bh * nsplit + i) * 2 + 1];
+      const float alpha = (lg > 0.f) ? __expf(ws_ml[((size_t)bh * nsplit + i) * 2 + 0] - M) : 0.f;
+      L += alpha * lg;
+      O += alpha * ws_o[((size_t)bh * nsplit + i) * MLA_LAT + d];
+    }
+    out[(size_t)bh * MLA_LAT + d] = f2b(L > 0.f ? O / L : 0.f);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // SwiGLU: silu(gate) * up, elementwise, vectorized
 // ---------------------------------------------------------------------------
 
@@ -1626,6 +1817,57 @@ torch::Tensor attn_prefill_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor
   return out;
 }
 
+// lat: [B, S, 512] bf16 (post-RMSNorm latent), rot: [B, S, 64] bf16 (roped
+// shared key); kp: [B, T32/16, 18, 64, 8], vp: [B, 32, T32/32, 64, 8].
+void mla_append(torch::Tensor lat, torch::Tensor rot, torch::Tensor positions,
+                torch::Tensor kp, torch::Tensor vp) {
+  CHK(lat.is_cuda() && lat.dtype() == torch::kBFloat16 && lat.is_contiguous());
+  CHK(rot.is_cuda() && rot.dtype() == torch::kBFloat16 && rot.is_contiguous());
+  CHK(positions.dtype() == torch::kInt32 && positions.is_cuda() && positions.is_contiguous());
+  CHK(kp.is_contiguous() && vp.is_contiguous());
+  const int B = lat.size(0), S = lat.size(1);
+  CHK(lat.size(2) == MLA_LAT && rot.size(2) == MLA_ROPE);
+  const int T32 = (int)kp.size(1) * 16;
+  const int npos = (int)positions.numel();
+  CHK(npos == S || npos == B * S);
+  const int waves = B * S;
+  hipLaunchKernelGGL(mla_append_kernel, dim3((waves + 3) / 4), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)lat.data_ptr(), (const unsigned short*)rot.data_ptr(),
+                     positions.data_ptr<int>(), (unsigned short*)kp.data_ptr(),
+                     (unsigned short*)vp.data_ptr(), B, S, T32, npos);
+}
+
+// q: [B, H, 576] bf16 (absorbed latent+rope query); returns out_lat
+// [B, H, 512] bf16 (latent-space attention output, pre kv_b-v expansion).
+torch::Tensor attn_decode_mla(torch::Tensor q, torch::Tensor kp, torch::Tensor vp,
+                              torch::Tensor seq_lens, double scale_in) {
+  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  CHK(seq_lens.dtype() == torch::kInt32 && seq_lens.is_cuda());
+  CHK(kp.is_contiguous() && vp.is_contiguous());
+  const int B = q.size(0), H = q.size(1);
+  CHK(q.size(2) == MLA_DQK);
+  const int T32 = (int)kp.size(1) * 16;
+  const int hgroups = (H + 15) / 16;
+  int nsplit = 1;
+  while (B * hgroups * nsplit * 2 < 4096 && (T32 / (nsplit * 2)) >= 32 && nsplit < 32) nsplit *= 2;
+  auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
+  auto ws_o = torch::empty({(long)B * H * nsplit * MLA_LAT}, opts);
+  auto ws_ml = torch::empty({(long)B * H * nsplit * 2}, opts);
+  auto out = torch::empty({(long)B, (long)H, (long)MLA_LAT},
+                          torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
+  const float scale = (scale_in > 0.0) ? (float)scale_in : 1.0f / sqrtf((float)MLA_DQK);
+  auto stream = cur_stream();
+  const int waves = B * hgroups * nsplit;
+  hipLaunchKernelGGL(attn_decode_mla_kernel, dim3((waves + 3) / 4), dim3(256), 0, stream,
+                     (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
+                     (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
+                     ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, T32, nsplit, scale);
+  hipLaunchKernelGGL(attn_decode_merge512, dim3(B * H), dim3(256), 0, stream,
+                     ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
+                     (unsigned short*)out.data_ptr(), nsplit);
+  return out;
+}
+
 torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
   CHK(a.is_cuda() && a.dtype() == torch::kBFloat16 && a.is_contiguous() && a.numel() == 16 * 32);
   CHK(b.is_cuda() && b.dtype() == torch::kBFloat16 && b.is_contiguous() && b.numel() == 32 * 16);
@@ -2029,6 +2271,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "causal GQA prefill flash attention on matrix cores (packed cache, hd=128)",
         py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("start_pos"),
         py::arg("scale") = 0.0, py::arg("softcap") = 0.0, py::arg("window") = 0);
+  m.def("mla_append", &mla_append,
+        "append MLA latent+rope token stream into the fragment-packed cache");
+  m.def("attn_decode_mla", &attn_decode_mla,
+        "MLA decode attention (absorbed latent MQA) on matrix cores",
+        py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("seq_lens"),
+        py::arg("scale") = 0.0);
   m.def("mfma16_probe", &mfma16_probe, "v_mfma_f32_16x16x32_bf16 layout probe (tests)");
   m.def("swiglu", &swiglu, "SwiGLU activation");
 }
