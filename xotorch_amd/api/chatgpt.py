@@ -175,9 +175,21 @@ class ChatGPTAPI:
 
   async def handle_get_initial_models(self, request):
     out = {}
+    local_repos = set()
+    try:
+      from xotorch_amd.download.downloader import models_dir
+      md = models_dir()
+      if md.exists():
+        local_repos = {p.name for p in md.iterdir() if p.is_dir()}
+    except Exception:
+      pass
     for mid in get_supported_models():
+      repo = get_repo(mid, self.inference_engine_classname) or ""
+      folder = repo.replace("/", "--")
       out[mid] = {
-        "name": pretty_name(mid), "downloaded": None, "download_percentage": None,
+        "name": pretty_name(mid),
+        "downloaded": (folder in local_repos) if local_repos else None,
+        "download_percentage": None,
         "total_size": None, "total_downloaded": None, "loading": False,
       }
     return web.json_response(out)

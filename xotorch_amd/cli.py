@@ -45,6 +45,8 @@ def build_parser() -> argparse.ArgumentParser:
   p.add_argument("--chatgpt-api-response-timeout", type=int, default=120)
   p.add_argument("--disable-tui", action="store_true")
   p.add_argument("--gpus", type=int, default=1, help="serve: GPUs for the single-node RCCL ring")
+  p.add_argument("--slots", type=int, default=8, help="serve: concurrent KV slots (continuous batching)")
+  p.add_argument("--no-graphs", action="store_true", help="serve: disable hipGraph decode capture")
   # train/eval
   p.add_argument("--data", type=str, default=None)
   p.add_argument("--epochs", type=int, default=1)
@@ -213,7 +215,10 @@ def serve_ring(args):
          f"--nproc-per-node={args.gpus}", "--master-addr", "127.0.0.1",
          "--master-port", str(find_available_port("127.0.0.1")), str(script),
          "--model", args.model_name or "llama-3-8b",
-         "--port", str(args.chatgpt_api_port)]
+         "--port", str(args.chatgpt_api_port),
+         "--slots", str(args.slots)]
+  if args.no_graphs:
+    cmd.append("--no-graphs")
   raise SystemExit(subprocess.run(cmd).returncode)
 
 
