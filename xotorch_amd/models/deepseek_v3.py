@@ -69,7 +69,8 @@ class MLAttention(nn.Module):
     self.o_proj = nn.Linear(H * cfg.v_head_dim, D, bias=cfg.attn_bias)
     self.scale = self.qk_head_dim ** -0.5  # default rope (no yarn mscale)
 
-  def forward(self, x, cos, sin, positions, kv, start_pos: int):
+  def forward(self, x, cos, sin, positions, kv, start_pos: int,
+              is_decode: bool = False, seq_lens=None):
     cfg = self.cfg
     B, S, _ = x.shape
     H = cfg.n_heads
@@ -89,23 +90,36 @@ class MLAttention(nn.Module):
     q_rot = _rope(q_rot, cs, sn, cfg.rope_interleave)
     k_rot = _rope(k_rot.view(B, S, 1, rope_d), cs, sn, cfg.rope_interleave)
 
+    kp = kv[2] if len(kv) > 2 else None
+    hip = None
+    if kp is not None and x.is_cuda and x.dtype == torch.bfloat16:
+      from xotorch_amd.ops import _load_hip
+      hip = _load_hip()
+    decode = is_decode or (S == 1 and (start_pos > 0 or start_pos < 0))
+    if hip is not None and decode and not torch.is_grad_enabled():
+      # MFMA decode fast path: packed-cache append with DEVICE positions
+      # (graph-capturable — no host sync), absorbed-latent attention.
+      # The plain cache is not touched: decode never reads it, and prefill
+      # rebuilt it before this point.
+      hip.mla_append(kv_nope.to(x.dtype).contiguous(),
+                     k_rot[:, :, 0, :].to(x.dtype).contiguous(),
+                     positions.to(torch.int32).contiguous(), kp, kv[3])
+      if seq_lens is None:
+        seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
+      return self._decode_mfma(x, q_pass, q_rot, kv, seq_lens, hip)
+    if start_pos < 0:  # ring decode contract: derive from positions (host sync, eager only)
+      start_pos = int(positions.reshape(-1)[0])
     # latent cache: k tensor <- kv_nope [B,1,T,kv_lora], v tensor <- roped
     # shared key [B,1,T,rope_d]
     lat_c, rot_c = kv[0], kv[1]
     lat_c[:, 0, start_pos: start_pos + S] = kv_nope
     rot_c[:, 0, start_pos: start_pos + S] = k_rot[:, :, 0, :]
-    kp = kv[2] if len(kv) > 2 else None
-    if kp is not None and x.is_cuda and x.dtype == torch.bfloat16:
-      # keep the fragment-packed copies appended (prefill AND decode) so
-      # the MFMA decode kernel can stream them
-      from xotorch_amd.ops import _load_hip
-      hip = _load_hip()
-      if hip is not None:
-        hip.mla_append(kv_nope.to(x.dtype).contiguous(),
-                       k_rot[:, :, 0, :].to(x.dtype).contiguous(),
-                       positions.to(torch.int32).contiguous(), kp, kv[3])
-        if S == 1 and start_pos > 0 and not torch.is_grad_enabled():
-          return self._decode_mfma(x, q_pass, q_rot, kv, start_pos, hip)
+    if hip is not None:
+      # keep the fragment-packed copies appended at prefill so later decode
+      # steps can stream them
+      hip.mla_append(kv_nope.to(x.dtype).contiguous(),
+                     k_rot[:, :, 0, :].to(x.dtype).contiguous(),
+                     positions.to(torch.int32).contiguous(), kp, kv[3])
     total = start_pos + S
     lat = lat_c[:, 0, :total]                                   # [B, T, kv_lora]
     krot = rot_c[:, 0, :total]                                  # [B, T, rope_d]
@@ -125,7 +139,7 @@ class MLAttention(nn.Module):
     return self.o_proj(out.reshape(B, S, H * vd))
 
 
-  def _decode_mfma(self, x, q_pass, q_rot, kv, start_pos: int, hip):
+  def _decode_mfma(self, x, q_pass, q_rot, kv, seq_lens, hip):
     """Absorbed-latent MFMA decode: kv_b is folded into q and out, so
     attention runs as MQA over the packed 1152 B/token latent stream
     (hip_ops.hip attn_decode_mla)."""
@@ -138,8 +152,7 @@ class MLAttention(nn.Module):
       self._w_v = W[:, nope:, :].contiguous()   # [H, vd, lat]
     q_lat = torch.einsum("bshn,hnl->bshl", q_pass.to(x.dtype), self._w_k)
     qfull = torch.cat([q_lat, q_rot.to(x.dtype)], dim=-1).reshape(B, H, lat + cfg.qk_rope_head_dim)
-    sl = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
-    out_lat = hip.attn_decode_mla(qfull.contiguous(), kv[2], kv[3], sl, self.scale)
+    out_lat = hip.attn_decode_mla(qfull.contiguous(), kv[2], kv[3], seq_lens, self.scale)
     out = torch.einsum("bhl,hdl->bhd", out_lat, self._w_v).to(x.dtype)
     return self.o_proj(out.reshape(B, 1, H * vd))
 
@@ -211,8 +224,9 @@ class DsLayer(nn.Module):
     self.input_layernorm = nn.Parameter(torch.ones(cfg.dim))
     self.post_attention_layernorm = nn.Parameter(torch.ones(cfg.dim))
 
-  def forward(self, h, cos, sin, positions, kv, start_pos):
-    h = h + self.self_attn(_rms(h, self.input_layernorm, self.eps), cos, sin, positions, kv, start_pos)
+  def forward(self, h, cos, sin, positions, kv, start_pos, is_decode=False, seq_lens=None):
+    h = h + self.self_attn(_rms(h, self.input_layernorm, self.eps), cos, sin, positions, kv,
+                           start_pos, is_decode, seq_lens)
     return h + self.mlp(_rms(h, self.post_attention_layernorm, self.eps))
 
 
@@ -262,7 +276,8 @@ class DeepseekV3Model(nn.Module):
     if positions.dim() == 0:
       positions = positions.reshape(1)
     for idx, lid in enumerate(self.local_layer_ids):
-      h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, caches[idx], start_pos)
+      h = self.layers[str(lid)](h, self.rope_cos, self.rope_sin, positions, caches[idx],
+                                start_pos, is_decode, seq_lens)
     if not self.shard.is_last_layer:
       return h
     if last_only and h.shape[1] > 1:

@@ -91,6 +91,8 @@ class Gemma2Attention(nn.Module):
                                scale=self.scale, softcap=cap, window=self.window)
       return self.o_proj(out.reshape(B, S, H * hd))
     # eager oracle path (CPU / hd != 128): split the fused projection
+    if start_pos < 0:  # ring decode contract: derive from positions
+      start_pos = int(positions.reshape(-1)[0])
     from xotorch_amd.ops import torch_ref as tr
     q, k, v = torch.split(qkv, [H * hd, KVH * hd, KVH * hd], dim=-1)
     q = q.view(B, S, H, hd)

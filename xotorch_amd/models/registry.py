@@ -167,6 +167,21 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
   # deepseek v3/r1 (MLA decoder, models/deepseek_v3.py)
   "deepseek-v3": _deepseek_v3_cfg(),
   "deepseek-r1": _deepseek_v3_cfg(),
+  # DeepSeek-V2-Lite (15.7B): same MLA geometry (512/64 latent) at a size a
+  # single MI355X holds — the measurable MLA config
+  "deepseek-v2-lite": {
+    "model_type": "deepseek_v2", "hidden_size": 2048, "num_hidden_layers": 27,
+    "num_attention_heads": 16, "num_key_value_heads": 16,
+    "intermediate_size": 10944, "moe_intermediate_size": 1408,
+    "n_routed_experts": 64, "num_experts_per_tok": 6, "n_shared_experts": 2,
+    "n_group": 1, "topk_group": 1, "routed_scaling_factor": 1.0,
+    "norm_topk_prob": False, "first_k_dense_replace": 1,
+    "q_lora_rank": 0, "kv_lora_rank": 512, "qk_rope_head_dim": 64,
+    "qk_nope_head_dim": 128, "v_head_dim": 128, "vocab_size": 102400,
+    "rope_theta": 10000.0, "rms_norm_eps": 1e-6,
+    "max_position_embeddings": 4096, "torch_dtype": "bfloat16",
+    "bos_token_id": 100000, "eos_token_id": 100001,
+  },
   # phi-4-mini (llama-like enough for the generic decoder)
   "phi-4-mini": _llama_cfg(3072, 32, 24, 8, 8192, vocab=200064, theta=10000.0, max_pos=131072, tie=True),
   # tiny test model
@@ -221,6 +236,7 @@ model_cards: Dict[str, dict] = {
   "gemma2-27b": {"layers": 46, "repo": {"TorchEngine": "google/gemma-2-27b-it", "HIPEngine": "google/gemma-2-27b-it"}},
   "deepseek-r1": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1", "HIPEngine": "deepseek-ai/DeepSeek-R1"}},
   "deepseek-v3": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-V3", "HIPEngine": "deepseek-ai/DeepSeek-V3"}},
+  "deepseek-v2-lite": {"layers": 27, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-V2-Lite-Chat", "HIPEngine": "deepseek-ai/DeepSeek-V2-Lite-Chat"}},
   # llava needs PIL for the image path (not present in this environment);
   # the reference's vision handling is also vestigial
   "llava-1.5-7b-hf": {"layers": 32, "repo": {}},
@@ -267,6 +283,7 @@ pretty_names = {
   "gemma2-9b": "Gemma2 9B",
   "gemma2-27b": "Gemma2 27B",
   "deepseek-r1": "DeepSeek R1",
+  "deepseek-v2-lite": "DeepSeek V2 Lite (16B MLA)",
   "deepseek-v3": "DeepSeek V3",
   "llava-1.5-7b-hf": "LLaVa 1.5 7B (unsupported arch)",
   "phi-4-mini": "Phi-4 Mini",

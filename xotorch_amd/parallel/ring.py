@@ -25,8 +25,8 @@ import torch
 import torch.distributed as dist
 
 from xotorch_amd.engine.kvcache import ShardKVCache
+from xotorch_amd.models import model_class_for
 from xotorch_amd.models.config import ModelConfig, config_from_hf
-from xotorch_amd.models.llama import ShardedModel
 from xotorch_amd.models.registry import builtin_config
 from xotorch_amd.models.weights import fast_random_init_gpu, random_init
 from xotorch_amd.parallel.comm import RingComm
@@ -93,11 +93,12 @@ class RingPipeline:
     # construct at the target dtype so to_empty materializes bf16 directly
     # (fp32-then-cast would peak at 2x the weight bytes and leave the
     # allocator's reserved pool at that level)
+    model_cls = model_class_for(self.cfg)
     prev_dtype = torch.get_default_dtype()
     torch.set_default_dtype(dtype)
     try:
       with torch.device("meta"):
-        model = ShardedModel(self.cfg, self.shard)
+        model = model_cls(self.cfg, self.shard)
     finally:
       torch.set_default_dtype(prev_dtype)
     model = model.to_empty(device=device)
@@ -125,9 +126,10 @@ class RingPipeline:
 
     # --- per-micro-batch state ---
     B = mb_batch
+    kv_heads, k_dim, v_dim = self.cfg.kv_cache_dims()
     self.caches = [
-      ShardKVCache(self.shard.get_layer_count(), B, self.cfg.n_kv_heads, self.total_len,
-                   self.cfg.head_dim, dtype, device)
+      ShardKVCache(self.shard.get_layer_count(), B, kv_heads, self.total_len,
+                   k_dim, dtype, device, v_dim=v_dim)
       for _ in range(self.M)
     ]
     self.positions = [torch.zeros(1, dtype=torch.int32, device=device) for _ in range(self.M)]

@@ -92,12 +92,13 @@ class RingSlotWorker:
     self.cfg = config_from_hf(builtin_config(model_id) or {}, model_id)
     self.max_seq = min(self.max_seq, self.cfg.max_seq_len)
     self.shard = equal_ring_shards(model_id, self.cfg.n_layers, world)[rank]
-    from xotorch_amd.models.llama import ShardedModel
+    from xotorch_amd.models import model_class_for
+    model_cls = model_class_for(self.cfg)
     prev_dtype = torch.get_default_dtype()
     torch.set_default_dtype(dtype)
     try:
       with torch.device("meta"):
-        model = ShardedModel(self.cfg, self.shard)
+        model = model_cls(self.cfg, self.shard)
     finally:
       torch.set_default_dtype(prev_dtype)
     model = model.to_empty(device=device).to(dtype)
