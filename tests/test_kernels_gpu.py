@@ -688,3 +688,33 @@ def test_deepseek_gpu_decode_matches_eager():
       assert torch.allclose(lg.float().cpu(), lr.float(), atol=0.5, rtol=0.1), \
         (step, (lg.float().cpu() - lr.float()).abs().max())
       nxt = lg.argmax(-1, keepdim=True)
+
+
+def test_dsmoe_grouped_matches_eager_loop():
+  """DeepSeek MoE static grouped decode path vs the dynamic eager loop."""
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.deepseek_v3 import DsMoE
+  raw = dict(model_type="deepseek_v2", vocab_size=64, hidden_size=256, intermediate_size=512,
+             moe_intermediate_size=256, n_routed_experts=8, num_experts_per_tok=2,
+             n_shared_experts=1, n_group=1, topk_group=1, routed_scaling_factor=1.0,
+             norm_topk_prob=False, first_k_dense_replace=0,
+             num_attention_heads=4, num_key_value_heads=4, kv_lora_rank=512,
+             qk_rope_head_dim=64, qk_nope_head_dim=64, v_head_dim=64,
+             rms_norm_eps=1e-6, max_position_embeddings=64)
+  cfg = config_from_hf(raw, "dsmoe-tiny")
+  torch.manual_seed(11)
+  moe = DsMoE(cfg).to("cuda").to(torch.bfloat16)
+  with torch.no_grad():
+    moe.gate_weight.normal_(0, 0.5)
+    for e in moe.experts:
+      for p in e.parameters():
+        p.normal_(0, 0.05)
+    for p in moe.shared_experts.parameters():
+      p.normal_(0, 0.05)
+  x = (torch.randn(1, 48, 256, device="cuda") * 0.5).to(torch.bfloat16)
+  with torch.inference_mode():
+    ref = moe(x).float()          # eager dynamic loop (wp not packed yet)
+    moe.pack_grouped()
+    assert moe.wp_gate_up is not None
+    out = moe(x).float()          # static grouped path
+  assert torch.allclose(out, ref, atol=5e-2, rtol=5e-2), (out - ref).abs().max()

@@ -200,9 +200,75 @@ class DsMoE(nn.Module):
       w = w / (w.sum(dim=-1, keepdim=True) + 1e-20)
     return idx, w * cfg.routed_scaling_factor
 
+  wp_gate_up = None
+  wp_down = None
+
+  def pack_grouped(self):
+    """Stack per-expert [gate|up] and down prepacks for the single-launch
+    grouped MFMA decode GEMM (same machinery as llama MoEMLP)."""
+    if self.wp_gate_up is not None:
+      return
+    first = self.experts[0].gate_proj.weight
+    if not (first.is_cuda and first.dtype == torch.bfloat16):
+      return
+    from xotorch_amd import ops
+    gu, dn = [], []
+    for e in self.experts:
+      w = torch.cat([e.gate_proj.weight.detach(), e.up_proj.weight.detach()], dim=0)
+      gu.append(ops.pack_decode_weight(w))
+      dn.append(ops.pack_decode_weight(e.down_proj.weight.detach()))
+    self.wp_gate_up = torch.stack(gu).contiguous()
+    self.wp_down = torch.stack(dn).contiguous()
+
+  def _forward_decode(self, flat):
+    """Static-shape routed path (graph-capturable): sort token-expert pairs,
+    pad to a fixed per-expert capacity, one grouped MFMA launch per GEMM."""
+    from xotorch_amd import ops
+    from xotorch_amd.ops import _load_hip
+    hip = _load_hip()
+    cfg = self.cfg
+    T, D = flat.shape
+    E, k = cfg.n_experts, cfg.n_experts_per_tok
+    dev = flat.device
+    idx, w = self.route(flat)                                 # [T, k]
+    A = T * k
+    expert_of = idx.reshape(A)
+    token_of = torch.arange(T, device=dev).repeat_interleave(k)
+    w_of = w.reshape(A)
+    order = torch.argsort(expert_of)
+    sorted_token = token_of[order]
+    sorted_w = w_of[order]
+    counts = (expert_of.unsqueeze(0) == torch.arange(E, device=dev).unsqueeze(1)).sum(1)
+    offsets = torch.cumsum(counts, 0) - counts
+    C = max(32, -(-T // 32) * 32)
+    c_idx = torch.arange(C, device=dev)
+    pos = offsets.unsqueeze(1) + c_idx.unsqueeze(0)
+    valid = c_idx.unsqueeze(0) < counts.unsqueeze(1)
+    pos_c = pos.clamp(max=A - 1)
+    gather_tok = sorted_token[pos_c.reshape(-1)]
+    scale = torch.where(valid, sorted_w[pos_c], torch.zeros((), dtype=sorted_w.dtype, device=dev))
+    xg = flat[gather_tok].to(flat.dtype)                      # [E*C, D]
+    I = cfg.moe_intermediate_dim
+    gu = hip.skinny_gemm_grouped(xg.view(E, C, D).contiguous(), self.wp_gate_up, E, 2 * I)
+    h = ops.swiglu_packed(gu.view(E * C, 2 * I))
+    y = hip.skinny_gemm_grouped(h.view(E, C, I), self.wp_down, E, D).view(E * C, D)
+    out = torch.zeros(T, D, dtype=torch.float32, device=dev)
+    out.index_add_(0, gather_tok, y.float() * scale.reshape(-1, 1).float())
+    return out
+
   def forward(self, x):
     B, S, D = x.shape
     flat = x.view(-1, D)
+    T = flat.shape[0]
+    if (self.wp_gate_up is not None and x.is_cuda and not torch.is_grad_enabled()
+        and 1 <= T <= 256):
+      out = self._forward_decode(flat).to(x.dtype) + self.shared_experts(flat)
+      return out.view(B, S, D)
+    if torch.cuda.is_available() and x.is_cuda and torch.cuda.is_current_stream_capturing():
+      raise RuntimeError(
+        "DsMoE dynamic routing is not graph-capturable: call "
+        "model.pack_decode_weights() first (grouped static path), or bench "
+        "with --no-graphs")
     idx, w = self.route(flat)
     out = torch.zeros_like(flat, dtype=torch.float32)
     for e in range(self.cfg.n_experts):
@@ -261,8 +327,17 @@ class DeepseekV3Model(nn.Module):
     return list(range(self.shard.start_layer, self.shard.end_layer + 1))
 
   def pack_decode_weights(self, reserve_bytes: int = 0) -> int:
-    return 0  # GEMMs stay on hipBLASLt; MLA decode attention runs the
-    # absorbed-latent MFMA kernel (attn_decode_mla) via the packed cache
+    """Dense GEMMs stay on hipBLASLt; MLA decode attention runs the
+    absorbed-latent MFMA kernel via the packed cache; MoE experts get the
+    stacked grouped prepack (single-launch decode GEMMs, graph-capturable
+    static routing)."""
+    packed = 0
+    for mod in self.modules():
+      if isinstance(mod, DsMoE):
+        mod.pack_grouped()
+        if mod.wp_gate_up is not None:
+          packed += mod.wp_gate_up.numel() * 2 + mod.wp_down.numel() * 2
+    return packed
 
   def head_weight(self):
     return self.lm_head.weight
