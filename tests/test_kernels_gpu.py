@@ -718,3 +718,32 @@ def test_dsmoe_grouped_matches_eager_loop():
     assert moe.wp_gate_up is not None
     out = moe(x).float()          # static grouped path
   assert torch.allclose(out, ref, atol=5e-2, rtol=5e-2), (out - ref).abs().max()
+
+
+def test_moe_build_combine_kernels():
+  """moe_build/moe_combine vs the torch glue semantics."""
+  from xotorch_amd.ops import _hip_ops
+  torch.manual_seed(4)
+  T, k, E = 48, 4, 16
+  C = 64
+  idx = torch.randint(0, E, (T, k), dtype=torch.int32, device="cuda")
+  w = torch.rand(T, k, dtype=torch.float32, device="cuda")
+  gather_tok, inv_pos = _hip_ops.moe_build(idx, E, C)
+  # every (t, j) pair landed in its expert's slot range and maps back
+  for t in range(T):
+    for j in range(k):
+      pos = int(inv_pos[t, j])
+      e = int(idx[t, j])
+      assert e * C <= pos < (e + 1) * C
+      assert int(gather_tok[pos]) == t
+  # per-expert slot ranks are unique
+  assert len(set(int(v) for v in inv_pos.reshape(-1))) == T * k
+  # combine = weighted sum over the pair rows
+  D = 128
+  y = (torch.randn(E * C, D, device="cuda") * 0.5).to(torch.bfloat16)
+  out = _hip_ops.moe_combine(y, inv_pos, w).float()
+  ref = torch.zeros(T, D, device="cuda")
+  for t in range(T):
+    for j in range(k):
+      ref[t] += w[t, j] * y[int(inv_pos[t, j])].float()
+  assert torch.allclose(out, ref, atol=2e-2, rtol=2e-2), (out - ref).abs().max()

@@ -231,30 +231,17 @@ class DsMoE(nn.Module):
     E, k = cfg.n_experts, cfg.n_experts_per_tok
     dev = flat.device
     idx, w = self.route(flat)                                 # [T, k]
-    A = T * k
-    expert_of = idx.reshape(A)
-    token_of = torch.arange(T, device=dev).repeat_interleave(k)
-    w_of = w.reshape(A)
-    order = torch.argsort(expert_of)
-    sorted_token = token_of[order]
-    sorted_w = w_of[order]
-    counts = (expert_of.unsqueeze(0) == torch.arange(E, device=dev).unsqueeze(1)).sum(1)
-    offsets = torch.cumsum(counts, 0) - counts
     C = max(32, -(-T // 32) * 32)
-    c_idx = torch.arange(C, device=dev)
-    pos = offsets.unsqueeze(1) + c_idx.unsqueeze(0)
-    valid = c_idx.unsqueeze(0) < counts.unsqueeze(1)
-    pos_c = pos.clamp(max=A - 1)
-    gather_tok = sorted_token[pos_c.reshape(-1)]
-    scale = torch.where(valid, sorted_w[pos_c], torch.zeros((), dtype=sorted_w.dtype, device=dev))
-    xg = flat[gather_tok].to(flat.dtype)                      # [E*C, D]
+    # counting-sort routing + deterministic combine as single HIP launches
+    # (the torch glue — argsort/cumsum/index_add — was the measured top
+    # decode cost, profiles/r02_new_decoders_profile.md)
+    gather_tok, inv_pos = hip.moe_build(idx.to(torch.int32).contiguous(), E, C)
+    xg = flat[gather_tok.long()].to(flat.dtype)               # [E*C, D]
     I = cfg.moe_intermediate_dim
     gu = hip.skinny_gemm_grouped(xg.view(E, C, D).contiguous(), self.wp_gate_up, E, 2 * I)
     h = ops.swiglu_packed(gu.view(E * C, 2 * I))
     y = hip.skinny_gemm_grouped(h.view(E, C, I), self.wp_down, E, D).view(E * C, D)
-    out = torch.zeros(T, D, dtype=torch.float32, device=dev)
-    out.index_add_(0, gather_tok, y.float() * scale.reshape(-1, 1).float())
-    return out
+    return hip.moe_combine(y, inv_pos, w.float().contiguous())
 
   def forward(self, x):
     B, S, D = x.shape
@@ -262,8 +249,8 @@ class DsMoE(nn.Module):
     T = flat.shape[0]
     if (self.wp_gate_up is not None and x.is_cuda and not torch.is_grad_enabled()
         and 1 <= T <= 256):
-      out = self._forward_decode(flat).to(x.dtype) + self.shared_experts(flat)
-      return out.view(B, S, D)
+      out = self._forward_decode(flat) + self.shared_experts(flat)
+      return out.to(x.dtype).view(B, S, D)
     if torch.cuda.is_available() and x.is_cuda and torch.cuda.is_current_stream_capturing():
       raise RuntimeError(
         "DsMoE dynamic routing is not graph-capturable: call "

@@ -303,22 +303,31 @@ class MoEMLP(nn.Module):
     E, k = self.n_experts, self.top_k
     dev = flat.device
     weights, selected = self._route(flat)                     # [T, k]
-    A = T * k
-    expert_of = selected.reshape(A)                           # [A]
-    token_of = torch.arange(T, device=dev).repeat_interleave(k)
-    w_of = weights.reshape(A)
-    order = torch.argsort(expert_of)                          # static shape [A]
-    sorted_token = token_of[order]
-    sorted_w = w_of[order]
-    counts = (expert_of.unsqueeze(0) == torch.arange(E, device=dev).unsqueeze(1)).sum(1)  # [E]
-    offsets = torch.cumsum(counts, 0) - counts                # exclusive prefix
+    from xotorch_amd.ops import _load_hip
+    hip0 = _load_hip()
     C = max(32, -(-T // 32) * 32)                             # capacity (lossless: count_e <= T)
-    c_idx = torch.arange(C, device=dev)
-    pos = offsets.unsqueeze(1) + c_idx.unsqueeze(0)           # [E, C]
-    valid = c_idx.unsqueeze(0) < counts.unsqueeze(1)
-    pos_c = pos.clamp(max=A - 1)
-    gather_tok = sorted_token[pos_c.reshape(-1)]              # [E*C]
-    scale = torch.where(valid, sorted_w[pos_c], torch.zeros((), dtype=sorted_w.dtype, device=dev))
+    if hip0 is not None and flat.is_cuda:
+      # single-launch counting-sort + deterministic combine (replaces the
+      # argsort/cumsum/index_add torch glue — measured top MoE decode cost)
+      gather_tok, inv_pos = hip0.moe_build(selected.to(torch.int32).contiguous(), E, C)
+      gather_tok = gather_tok.long()
+      scale = None
+    else:
+      A = T * k
+      expert_of = selected.reshape(A)                         # [A]
+      token_of = torch.arange(T, device=dev).repeat_interleave(k)
+      w_of = weights.reshape(A)
+      order = torch.argsort(expert_of)                        # static shape [A]
+      sorted_token = token_of[order]
+      sorted_w = w_of[order]
+      counts = (expert_of.unsqueeze(0) == torch.arange(E, device=dev).unsqueeze(1)).sum(1)
+      offsets = torch.cumsum(counts, 0) - counts              # exclusive prefix
+      c_idx = torch.arange(C, device=dev)
+      pos = offsets.unsqueeze(1) + c_idx.unsqueeze(0)         # [E, C]
+      valid = c_idx.unsqueeze(0) < counts.unsqueeze(1)
+      pos_c = pos.clamp(max=A - 1)
+      gather_tok = sorted_token[pos_c.reshape(-1)]            # [E*C]
+      scale = torch.where(valid, sorted_w[pos_c], torch.zeros((), dtype=sorted_w.dtype, device=dev))
     xg = flat[gather_tok]                                     # [E*C, D]
     if self.wp_gate_up_fp8 is not None and not torch.is_grad_enabled():
       from xotorch_amd.ops import _load_hip
@@ -343,6 +352,9 @@ class MoEMLP(nn.Module):
       for e in range(E):
         outs.append(self.experts[e](xe[e]))
       y = torch.cat(outs, dim=0)
+    if scale is None:
+      return hip0.moe_combine(y.view(E * C, D).to(torch.bfloat16).contiguous(), inv_pos,
+                              weights.float().contiguous()).float()
     out = torch.zeros(T, D, dtype=torch.float32, device=dev)
     out.index_add_(0, gather_tok, y.float() * scale.reshape(-1, 1))
     return out

@@ -747,6 +747,79 @@ __global__ __launch_bounds__(128) void attn_decode_merge(
 }
 
 // ---------------------------------------------------------------------------
+// MoE decode-routing glue: the static-capacity routed path needs
+// (token-expert pairs -> per-expert-sorted gather list) and the weighted
+// combine. In torch this is ~8 launches per MoE layer per step (argsort,
+// repeat_interleave, cumsum, eq+sum, clamp, where, index_add) — measured as
+// the TOP decode cost on DeepSeek-V2-Lite (profiles/r02_new_decoders_...).
+// Here: ONE single-workgroup counting-sort kernel (A = T*k pairs <= 2048,
+// E <= 256 fits LDS) + ONE deterministic gather-combine kernel (no atomics:
+// out[t] = sum_j w[t,j] * y[pos(t,j)]).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void moe_build_kernel(
+    const int* __restrict__ idx, int* __restrict__ gather_tok,
+    int* __restrict__ inv_pos, int T, int k, int E, int C) {
+  __shared__ int counts[256];
+  __shared__ int offsets[256];
+  const int tid = threadIdx.x;
+  const int A = T * k;
+  for (int e = tid; e < E; e += 256) counts[e] = 0;
+  __syncthreads();
+  for (int a = tid; a < A; a += 256) atomicAdd(&counts[idx[a]], 1);
+  __syncthreads();
+  if (tid == 0) {  // E <= 256: serial prefix is ~E adds, negligible here
+    int run = 0;
+    for (int e = 0; e < E; ++e) {
+      offsets[e] = run;
+      run += counts[e];
+    }
+  }
+  __syncthreads();
+  for (int e = tid; e < E; e += 256) counts[e] = 0;  // reuse as cursors
+  // default: invalid capacity slots gather token 0 (scale is 0 via inv_pos)
+  for (int p = tid; p < E * C; p += 256) gather_tok[p] = 0;
+  __syncthreads();
+  for (int a = tid; a < A; a += 256) {
+    const int e = idx[a];
+    const int r = atomicAdd(&counts[e], 1);           // rank within expert
+    const int global_rank = offsets[e] + r;           // dense order
+    // dense position -> padded [E, C] position
+    const int pos = e * C + r;                        // r < C always (count_e <= T <= C)
+    gather_tok[pos] = a / k;
+    inv_pos[a] = pos;
+    (void)global_rank;
+  }
+}
+
+// out[t, :] = sum_j w[t, j] * y[inv_pos[t*k + j], :]   (fp32 accumulate)
+__global__ __launch_bounds__(256) void moe_combine_kernel(
+    const unsigned short* __restrict__ y, const int* __restrict__ inv_pos,
+    const float* __restrict__ w, unsigned short* __restrict__ out,
+    int T, int k, int D) {
+  const long long n8 = (long long)T * (D >> 3);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const int d8 = D >> 3;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n8; i += stride) {
+    const int t = (int)(i / d8);
+    const int c8 = (int)(i % d8) * 8;
+    float acc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+    for (int j = 0; j < k; ++j) {
+      const float wj = w[t * k + j];
+      const ushort8 v = *(const ushort8*)(y + (size_t)inv_pos[t * k + j] * D + c8);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) acc[q] += wj * b2f(v[q]);
+    }
+    ushort8 o;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) o[q] = f2b(acc[q]);
+    *(ushort8*)(out + (size_t)t * D + c8) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // MLA (DeepSeek V2/V3) decode attention on matrix cores — ABSORBED form.
 //
 // The kv_b projection is folded into the query and output (DeepSeek's own
@@ -1817,6 +1890,40 @@ torch::Tensor attn_prefill_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor
   return out;
 }
 
+// idx: [T, k] int32 expert choices -> (gather_tok int32 [E*C], inv_pos
+// int32 [T, k]). One workgroup (counting sort in LDS).
+std::vector<torch::Tensor> moe_build(torch::Tensor idx, int64_t E, int64_t C) {
+  CHK(idx.is_cuda() && idx.dtype() == torch::kInt32 && idx.is_contiguous() && idx.dim() == 2);
+  const int T = idx.size(0), k = idx.size(1);
+  CHK(E <= 256 && T * k <= 8192 && T <= C);
+  auto opts = torch::TensorOptions().dtype(torch::kInt32).device(idx.device());
+  auto gather_tok = torch::empty({E * C}, opts);
+  auto inv_pos = torch::empty({T, k}, opts);
+  hipLaunchKernelGGL(moe_build_kernel, dim3(1), dim3(256), 0, cur_stream(),
+                     idx.data_ptr<int>(), gather_tok.data_ptr<int>(),
+                     inv_pos.data_ptr<int>(), T, k, (int)E, (int)C);
+  return {gather_tok, inv_pos};
+}
+
+// y: [E*C, D] bf16 expert outputs, inv_pos [T,k] int32, w [T,k] fp32 ->
+// out [T, D] bf16 (deterministic weighted combine, no atomics).
+torch::Tensor moe_combine(torch::Tensor y, torch::Tensor inv_pos, torch::Tensor w) {
+  CHK(y.is_cuda() && y.dtype() == torch::kBFloat16 && y.is_contiguous());
+  CHK(inv_pos.dtype() == torch::kInt32 && inv_pos.is_contiguous());
+  CHK(w.dtype() == torch::kFloat32 && w.is_contiguous());
+  const int T = inv_pos.size(0), k = inv_pos.size(1);
+  const int D = y.size(-1);
+  CHK(D % 8 == 0);
+  auto out = torch::empty({(long)T, (long)D},
+                          torch::TensorOptions().dtype(torch::kBFloat16).device(y.device()));
+  const long long n8 = (long long)T * (D / 8);
+  const int blocks = (int)std::min<long long>(1024, (n8 + 255) / 256);
+  hipLaunchKernelGGL(moe_combine_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)y.data_ptr(), inv_pos.data_ptr<int>(),
+                     w.data_ptr<float>(), (unsigned short*)out.data_ptr(), T, k, D);
+  return out;
+}
+
 // lat: [B, S, 512] bf16 (post-RMSNorm latent), rot: [B, S, 64] bf16 (roped
 // shared key); kp: [B, T32/16, 18, 64, 8], vp: [B, 32, T32/32, 64, 8].
 void mla_append(torch::Tensor lat, torch::Tensor rot, torch::Tensor positions,
@@ -2271,6 +2378,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "causal GQA prefill flash attention on matrix cores (packed cache, hd=128)",
         py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("start_pos"),
         py::arg("scale") = 0.0, py::arg("softcap") = 0.0, py::arg("window") = 0);
+  m.def("moe_build", &moe_build,
+        "MoE decode routing: counting-sort token-expert pairs to padded per-expert slots");
+  m.def("moe_combine", &moe_combine,
+        "MoE decode combine: out[t] = sum_j w[t,j] * y[pos(t,j)] (deterministic)");
   m.def("mla_append", &mla_append,
         "append MLA latent+rope token stream into the fragment-packed cache");
   m.def("attn_decode_mla", &attn_decode_mla,
