@@ -1955,8 +1955,12 @@ torch::Tensor attn_decode_mla(torch::Tensor q, torch::Tensor kp, torch::Tensor v
   CHK(q.size(2) == MLA_DQK);
   const int T32 = (int)kp.size(1) * 16;
   const int hgroups = (H + 15) / 16;
+  // fill the chip: B*hgroups is small for lite MLA configs (64 x 1), so
+  // split the kv range aggressively — one 32-pos tile per split is fine
+  // (empty splits merge as -inf partials). 128 WGs on 256 CUs measured
+  // 1.1 TB/s; 256+ WGs is the floor for the packed stream.
   int nsplit = 1;
-  while (B * hgroups * nsplit * 2 < 4096 && (T32 / (nsplit * 2)) >= 32 && nsplit < 32) nsplit *= 2;
+  while (B * hgroups * nsplit * 2 < 4096 && (T32 / (nsplit * 2)) >= 16 && nsplit < 64) nsplit *= 2;
   auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
   auto ws_o = torch::empty({(long)B * H * nsplit * MLA_LAT}, opts);
   auto ws_ml = torch::empty({(long)B * H * nsplit * 2}, opts);

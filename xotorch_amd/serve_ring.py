@@ -67,6 +67,8 @@ class SlotState:
   max_new: int = 0
   temp: float = 0.0
   done: bool = False  # finished but not yet released (release is its own tick)
+  ttft_s: float = 0.0
+  t_admit: float = 0.0
 
 
 @dataclass
@@ -260,6 +262,12 @@ class RingSlotWorker:
       self.seq_lens[slot] = plen + 1
     return first_tok
 
+  def _log_done(self, st: SlotState):
+    dt = max(1e-9, time.perf_counter() - st.t_admit)
+    print(f"[serve] {str(st.request_id)[:8]} done: prompt {st.prompt_len}, "
+          f"{st.generated} tokens, TTFT {st.ttft_s * 1000:.0f} ms, "
+          f"{st.generated / dt:.1f} tok/s", flush=True)
+
   # ---------------- control loop ----------------
 
   def _bcast_hdr(self, op=OP_IDLE, slot=0, plen=0, extra=0):
@@ -338,12 +346,15 @@ class RingSlotWorker:
           st.request_id = msg.request_id
           st.max_new = min(msg.max_new, self.max_seq - plen - 1)
           st.generated = 1
+          st.ttft_s = time.perf_counter() - msg.t_submit
+          st.t_admit = msg.t_submit
           tok = int(self.tok_bcast[slot])
           fin = (self.eos_token_id is not None and tok == self.eos_token_id) or st.generated >= st.max_new
           if emit:
-            emit(msg.request_id, tok, fin, {"ttft_s": time.perf_counter() - msg.t_submit})
+            emit(msg.request_id, tok, fin, {"ttft_s": st.ttft_s})
           if fin:
             st.done = True
+            self._log_done(st)
             pending_release.append(slot)
         continue
       # OP_DECODE
@@ -361,6 +372,7 @@ class RingSlotWorker:
             emit(st.request_id, tok, fin, {})
           if fin:
             st.done = True
+            self._log_done(st)
             pending_release.append(i)
     return ticks
 
