@@ -72,3 +72,21 @@ def test_concurrent_slots_isolated(workers):
   other = [int(v) for v in rng.integers(0, 32000, 15)]
   both, _ = _serve(g, [("same", ids, 6), ("noise", other, 6)])
   assert both["same"] == solo["solo"]
+
+
+def test_spec_decode_invariant_gpu():
+  """Speculative output == target-only greedy on silicon (bf16, 1b+1b pair
+  with different seeds: mismatched draft, exact-output invariant)."""
+  from xotorch_amd.engine.spec import SpeculativeDecoder
+  sd = SpeculativeDecoder.from_model_ids("llama-3.2-1b", "llama-3.2-1b",
+                                         device="cuda", gamma=3, seed=11)
+  # different draft weights
+  from xotorch_amd.models.weights import fast_random_init_gpu
+  fast_random_init_gpu(sd.draft, seed=77)
+  sd.draft.reset_rope()
+  prompt = torch.randint(0, 32000, (1, 20), device="cuda")
+  toks, stats = sd.generate(prompt, max_new=24)
+  sd.reset()
+  ref = sd.generate_plain(prompt, max_new=24)
+  assert toks == ref, (toks, ref)
+  assert stats.rounds > 0
