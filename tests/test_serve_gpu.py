@@ -75,14 +75,18 @@ def test_concurrent_slots_isolated(workers):
 
 
 def test_spec_decode_invariant_gpu():
-  """Speculative output == target-only greedy on silicon (bf16, 1b+1b pair
-  with different seeds: mismatched draft, exact-output invariant)."""
+  """Speculative output == target-only greedy on silicon (fp32: the chunked
+  verify pass and single-token steps hit different GEMM shapes, so bf16
+  logits differ in the last bits and random-init near-tie argmaxes can
+  flip — exactness is a fp32 property; at bf16 spec decode carries the
+  same caveat as every production implementation)."""
   from xotorch_amd.engine.spec import SpeculativeDecoder
   sd = SpeculativeDecoder.from_model_ids("llama-3.2-1b", "llama-3.2-1b",
-                                         device="cuda", gamma=3, seed=11)
+                                         device="cuda", dtype=torch.float32,
+                                         gamma=3, seed=11)
   # different draft weights
-  from xotorch_amd.models.weights import fast_random_init_gpu
-  fast_random_init_gpu(sd.draft, seed=77)
+  from xotorch_amd.models.weights import random_init
+  random_init(sd.draft, seed=77)
   sd.draft.reset_rope()
   prompt = torch.randint(0, 32000, (1, 20), device="cuda")
   toks, stats = sd.generate(prompt, max_new=24)
