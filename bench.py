@@ -153,6 +153,7 @@ def main():
         "micro_batches": ring.M,
         "mb_batch": args.mb_batch,
         "p50_ttft_ms": round(p50_ttft, 1),
+        "ttft_semantics": "whole-batch pipelined prefill (request-level TTFT: serve path)",
         "sampling": "greedy",
         "hip_graphs": not args.no_graphs and device == "cuda",
       },

@@ -150,3 +150,17 @@ def test_write_frame_rejects_oversize(monkeypatch):
 
   with pytest.raises(ValueError, match="frame too large"):
     wire.write_frame(W(), {"data": b"x" * 2048})
+
+
+def test_top_p_sampling():
+  """top_p narrows the candidate set (the reference parses but ignores it)."""
+  from xotorch_amd.ops import torch_ref as tr
+  g = torch.Generator().manual_seed(0)
+  logits = torch.tensor([[10.0, 9.0, 0.0, -5.0, -5.0]]).repeat(256, 1)
+  toks = tr.softmax_sample(logits, temperature=1.0, top_p=0.5, generator=g)
+  assert set(toks.tolist()) == {0}, "p=0.5 keeps only the top token here"
+  toks = tr.softmax_sample(logits, temperature=1.0, top_p=0.95, generator=g)
+  assert set(toks.tolist()) <= {0, 1}
+  # top_p=0 disables (all tokens samplable at high temp)
+  toks = tr.softmax_sample(logits * 0.01, temperature=5.0, top_p=0.0, generator=g)
+  assert len(set(toks.tolist())) >= 3

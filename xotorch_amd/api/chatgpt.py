@@ -281,7 +281,13 @@ class ChatGPTAPI:
         state["max_tokens"] = int(chat_request.max_tokens)
       if chat_request.temperature:
         state["temperature"] = float(chat_request.temperature)
-      await self.node.process_prompt(shard, prompt, request_id, state or None)
+      if data.get("top_p") is not None:
+        state["top_p"] = float(data["top_p"])
+      try:
+        await self.node.process_prompt(shard, prompt, request_id, state or None)
+      except ValueError as e:
+        # e.g. prompt longer than the serving context — a client error
+        return web.json_response({"detail": str(e)}, status=400)
       if stream:
         return await self._stream_response(request, request_id, model_id, tokenizer)
       return await self._full_response(request_id, model_id, tokenizer)

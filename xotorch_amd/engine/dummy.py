@@ -31,7 +31,7 @@ class DummyEngine(InferenceEngine):
     await self.ensure_shard(shard)
     return self.tokenizer.decode(list(np.asarray(tokens).reshape(-1)))
 
-  async def sample(self, x: np.ndarray, temp: float = 0.0, top_k: int = 0) -> np.ndarray:
+  async def sample(self, x: np.ndarray, temp: float = 0.0, top_k: int = 0, top_p: float = 0.0) -> np.ndarray:
     return np.asarray(x).reshape(np.asarray(x).shape[0], -1)[:, -1].astype(np.int64) % 256
 
   async def infer_tensor(

@@ -26,7 +26,7 @@ class InferenceEngine(ABC):
     ...
 
   @abstractmethod
-  async def sample(self, x: np.ndarray, temp: float = 0.0, top_k: int = 0) -> np.ndarray:
+  async def sample(self, x: np.ndarray, temp: float = 0.0, top_k: int = 0, top_p: float = 0.0) -> np.ndarray:
     ...
 
   @abstractmethod

@@ -170,11 +170,12 @@ class TorchEngine(InferenceEngine):
     await self.ensure_shard(shard)
     return self.tokenizer.decode(list(np.asarray(tokens).reshape(-1)))
 
-  async def sample(self, x: np.ndarray, temp: float = TEMP, top_k: int = TOP_K) -> np.ndarray:
+  async def sample(self, x: np.ndarray, temp: float = TEMP, top_k: int = TOP_K,
+                   top_p: float = 0.0) -> np.ndarray:
     logits = torch.from_numpy(np.asarray(x))
     if logits.dim() == 3:
       logits = logits[:, -1, :]
-    tok = ops.softmax_sample(logits, temperature=temp, top_k=top_k)
+    tok = ops.softmax_sample(logits, temperature=temp, top_k=top_k, top_p=top_p)
     return tok.cpu().numpy()
 
   # ---------- inference ----------

@@ -199,6 +199,7 @@ def linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] =
   return torch.nn.functional.linear(x, weight, bias)
 
 
-def softmax_sample(logits, temperature: float = 0.0, top_k: int = 0, generator=None):
+def softmax_sample(logits, temperature: float = 0.0, top_k: int = 0, generator=None,
+                   top_p: float = 0.0):
   # torch ops here are graph-capturable (argmax / exponential+argmax)
-  return torch_ref.softmax_sample(logits, temperature, top_k, generator)
+  return torch_ref.softmax_sample(logits, temperature, top_k, generator, top_p)

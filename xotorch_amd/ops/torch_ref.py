@@ -261,8 +261,10 @@ def softmax_sample(
   temperature: float = 0.0,
   top_k: int = 0,
   generator: Optional[torch.Generator] = None,
+  top_p: float = 0.0,
 ) -> torch.Tensor:
-  """Temperature + top-k sampling via the exponential trick; temp 0 → argmax.
+  """Temperature + top-k + nucleus (top-p) sampling via the exponential
+  trick; temp 0 → argmax. (The reference parses but ignores top_p.)
 
   logits: [B, V] → token ids [B].
   """
@@ -272,6 +274,13 @@ def softmax_sample(
   if top_k and top_k > 0 and top_k < logits.shape[-1]:
     kth = torch.topk(logits, top_k, dim=-1).values[..., -1, None]
     logits = torch.where(logits < kth, torch.full_like(logits, float("-inf")), logits)
+  if top_p and 0.0 < top_p < 1.0:
+    sorted_logits, sorted_idx = torch.sort(logits, descending=True, dim=-1)
+    cum = torch.softmax(sorted_logits, dim=-1).cumsum(dim=-1)
+    # keep tokens until cumulative prob exceeds top_p (always keep the first)
+    cut = cum - torch.softmax(sorted_logits, dim=-1) >= top_p
+    sorted_logits = sorted_logits.masked_fill(cut, float("-inf"))
+    logits = torch.full_like(logits, float("-inf")).scatter(-1, sorted_idx, sorted_logits)
   probs = torch.softmax(logits, dim=-1)
   q = torch.empty_like(probs).exponential_(1, generator=generator)
   return (probs / q).argmax(dim=-1)

@@ -242,7 +242,8 @@ class Node:
     if shard.is_last_layer:
       # sample a token from the logits
       temp = (inference_state or {}).get("temperature", self.default_sample_temperature)
-      token = (await self.inference_engine.sample(result, temp=float(temp))).reshape(-1)
+      top_p = float((inference_state or {}).get("top_p", 0.0) or 0.0)
+      token = (await self.inference_engine.sample(result, temp=float(temp), top_p=top_p)).reshape(-1)
       tok = int(token[0])
       buffered, _ = self.buffered_token_output.setdefault(request_id, ([], False))
       buffered.append(tok)
