@@ -110,3 +110,22 @@ def test_four_stage_ring_matches_single():
     four = json.loads(Path(d, "tokens_w4.json").read_text())
     # world=1 runs M=1 micro-batch; world=4 runs M=4 — compare the shared mb 0
     assert ref[0] == four[0], "4-stage ring tokens diverged from single-process decode"
+
+
+@pytest.mark.timeout(240)
+def test_seq_chunked_prefill_matches(monkeypatch):
+  """Sequence-chunked multi-rank prefill produces identical tokens."""
+  from xotorch_amd.helpers import find_available_port
+  with tempfile.TemporaryDirectory() as d:
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    run_ring(1, d, 0)
+    os.environ["XOT_RING_PREFILL_SEQ_CHUNK"] = "5"  # 12-token prompt -> 3 chunks
+    try:
+      port = find_available_port("127.0.0.1")
+      mp.spawn(_worker, args=(2, d, port), nprocs=2, join=True)
+    finally:
+      os.environ.pop("XOT_RING_PREFILL_SEQ_CHUNK", None)
+    ref = json.loads(Path(d, "tokens_w1.json").read_text())
+    two = json.loads(Path(d, "tokens_w2.json").read_text())
+    assert ref == two, "seq-chunked prefill diverged"

@@ -198,14 +198,15 @@ class RingPipeline:
 
   # ---------- prefill ----------
 
-  def _prefill_forward(self, mb: int, x: torch.Tensor, pos: torch.Tensor) -> torch.Tensor:
+  def _prefill_forward(self, mb: int, x: torch.Tensor, pos: torch.Tensor,
+                       start_pos: int = 0) -> torch.Tensor:
     """Prefill one micro-batch through this shard in batch chunks: peak
     activation memory (the fused gate_up output is B*S*2I bf16 — 15 GB at
     B=256, S=512 on 70B) stays bounded at the chunk size regardless of B."""
     B = x.shape[0]
     bc = min(B, int(os.getenv("XOT_PREFILL_CHUNK", "128")))
     if bc >= B:
-      return self.model(x, caches=self.caches[mb].caches, positions=pos, start_pos=0)
+      return self.model(x, caches=self.caches[mb].caches, positions=pos, start_pos=start_pos)
     outs = []
     for c0 in range(0, B, bc):
       c1 = min(c0 + bc, B)
@@ -215,7 +216,7 @@ class RingPipeline:
           sliced.append((layer[0][c0:c1], layer[1][c0:c1], layer[2][c0:c1], layer[3][c0:c1]))
         else:
           sliced.append((layer[0][c0:c1], layer[1][c0:c1]))
-      outs.append(self.model(x[c0:c1], caches=sliced, positions=pos, start_pos=0))
+      outs.append(self.model(x[c0:c1], caches=sliced, positions=pos, start_pos=start_pos))
     return torch.cat(outs, dim=0)
 
   def prefill(self, prompts: Optional[List[torch.Tensor]] = None) -> RingStats:
