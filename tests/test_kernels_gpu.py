@@ -747,3 +747,17 @@ def test_moe_build_combine_kernels():
     for j in range(k):
       ref[t] += w[t, j] * y[int(inv_pos[t, j])].float()
   assert torch.allclose(out, ref, atol=2e-2, rtol=2e-2), (out - ref).abs().max()
+
+
+@pytest.mark.parametrize("M,I,N", [(64, 512, 256), (128, 1024, 384), (96, 768, 128)])
+def test_skinny_gemm_packed_swiglu(hip, M, I, N):
+  """Fused silu(gate)*up + down-GEMM vs the two-step reference."""
+  from xotorch_amd import ops
+  from xotorch_amd.ops import _hip_ops, torch_ref
+  gu = bt(M, 2 * I, seed=121, scale=0.5)
+  w = bt(N, I, seed=122, scale=1.0 / I ** 0.5)
+  wp = ops.pack_decode_weight(w)
+  y = _hip_ops.skinny_gemm_packed_swiglu(gu, wp, N, None).float()
+  h = torch_ref.swiglu_packed(gu.float())
+  ref = h.float() @ w.float().T
+  assert torch.allclose(y, ref, atol=5e-2, rtol=5e-2), (y - ref).abs().max()

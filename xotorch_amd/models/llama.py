@@ -221,7 +221,36 @@ class MLP(nn.Module):
     self.down_proj = XotLinear(I, cfg.dim, bias=False)
 
   def forward(self, x):
-    return self.down_proj(ops.swiglu_packed(self.gate_up_proj(x)))
+    gu = self.gate_up_proj(x)
+    dp = self.down_proj
+    if (dp.weight_packed is not None and gu.is_cuda and gu.dtype == torch.bfloat16
+        and not torch.is_grad_enabled() and gu.is_contiguous()
+        and os.getenv("XOT_FUSE_SWIGLU", "1") == "1"):
+      I = self.intermediate
+      M = gu.numel() // (2 * I)
+      if 32 <= M <= 256 and M % 32 == 0:
+        from xotorch_amd.ops import _load_hip
+        hip = _load_hip()
+        if hip is not None:
+          N = dp.weight.shape[0]
+          key = ("fuse_swiglu", N, I, M)
+          use = _PACKED_WINS.get(key)
+          if use is None:
+            if torch.cuda.is_current_stream_capturing():
+              use = "fused"
+            else:
+              wp = dp.weight_packed
+              t_fused = _time_us(lambda: hip.skinny_gemm_packed_swiglu(gu, wp, N, dp.bias))
+              t_split = _time_us(lambda: dp(ops.swiglu_packed(gu)))
+              use = "fused" if t_fused < t_split * 0.98 else "split"
+              _PACKED_WINS[key] = use
+              if os.getenv("XOT_DEBUG", "0") != "0":
+                print(f"[xot] swiglu-fuse auto-pick N={N} I={I} M={M}: fused {t_fused:.1f} us "
+                      f"vs split {t_split:.1f} us -> {use}", flush=True)
+          if use == "fused":
+            y = hip.skinny_gemm_packed_swiglu(gu, dp.weight_packed, N, dp.bias)
+            return y.view(list(x.shape[:-1]) + [N])
+    return self.down_proj(ops.swiglu_packed(gu))
 
 
 class MoEMLP(nn.Module):

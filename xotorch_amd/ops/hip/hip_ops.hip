@@ -1213,12 +1213,17 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 // Mixtral-class decode each expert e (blockIdx.y) computes its own
 // [C, N] = [C, K] @ Wp_e^T over a fixed per-expert token capacity C — one
 // launch covers every expert, weights stream from the stacked prepack.
-template <int MT, bool SPLIT, int BK>
+template <int MT, bool SPLIT, int BK, bool SWIGLU = false>
 __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     const unsigned short* __restrict__ Wp, const unsigned short* __restrict__ X,
     unsigned short* __restrict__ Y, float* __restrict__ P,
     const unsigned short* __restrict__ bias,
-    int N, long long K, int kc, int nsplit) {
+    int N, long long K, int kc, int nsplit, long long ldx = 0) {
+  // SWIGLU: X is the fused [rows, 2K] gate|up GEMM output; the staging
+  // computes silu(gate)*up on the fly (row stride ldx = 2K, up at +K) —
+  // the intermediate activation tensor never exists (saves its write +
+  // read, ~29 MB/layer at 70B B=128, and one launch per layer).
+  if (!SWIGLU) ldx = K;
   constexpr int BKC = BK / 16;   // MFMA k-chunks (A fragments) per K-step
   constexpr int XROW = BK + 8;   // padded LDS row stride (elems): conflict-free b128
   const int ntiles = N >> 7;
@@ -1256,16 +1261,26 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
   // one VGPR pointer plus SGPR/immediate strides replaces the old per-piece
   // pointer arrays (saves ~10 VGPRs -> a wave/SIMD at MT=4)
   constexpr int ROWS_PER_I = 2048 / BK;  // rows advanced per i (256 threads)
-  const unsigned short* xpb = X + k0 + (size_t)(tid / (BK / 8)) * K + (tid % (BK / 8)) * 8;
+  const unsigned short* xpb = X + k0 + (size_t)(tid / (BK / 8)) * ldx + (tid % (BK / 8)) * 8;
   const int xsb = (tid / (BK / 8)) * XROW + (tid % (BK / 8)) * 8;
-  const long long xstride_i = (long long)ROWS_PER_I * K;
+  const long long xstride_i = (long long)ROWS_PER_I * ldx;
 
   // prologue: stage step 0, preload A(0) and A(1).  The W stream is read
   // exactly once per launch -> non-temporal (L1-bypass) loads; depth-2
   // prefetch keeps 2*BK*32n*2B per wave in flight across staging barriers.
   const unsigned short* wp1 = (nsteps > 1) ? wp + BKC * 512 : wp;  // clamp: no OOB at nsteps==1
 #pragma unroll
-  for (int i = 0; i < PPT; ++i) xv[i] = *(const ushort8*)(xpb + i * xstride_i);
+  for (int i = 0; i < PPT; ++i) {
+    xv[i] = *(const ushort8*)(xpb + i * xstride_i);
+    if (SWIGLU) {
+      const ushort8 up = *(const ushort8*)(xpb + i * xstride_i + K);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float g = b2f(xv[i][j]);
+        xv[i][j] = f2b(g * __builtin_amdgcn_rcpf(1.f + __expf(-g)) * b2f(up[j]));
+      }
+    }
+  }
 #pragma unroll
   for (int u = 0; u < BKC; ++u) {
     a_buf[0][u] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(wp + u * 512));
@@ -1283,8 +1298,17 @@ __global__ __launch_bounds__(256) void skinny_gemm_packed_kernel(
     const bool last = (s == nsteps - 1);                                                       \
     if (!last) {                                                                               \
       _Pragma("unroll")                                                                        \
-      for (int i = 0; i < PPT; ++i)                                                            \
+      for (int i = 0; i < PPT; ++i) {                                                          \
         xv[i] = *(const ushort8*)(xpb + i * xstride_i + (size_t)(s + 1) * BK);                 \
+        if (SWIGLU) {                                                                          \
+          const ushort8 up_ = *(const ushort8*)(xpb + i * xstride_i + (size_t)(s + 1) * BK + K); \
+          _Pragma("unroll")                                                                    \
+          for (int j_ = 0; j_ < 8; ++j_) {                                                     \
+            const float g_ = b2f(xv[i][j_]);                                                   \
+            xv[i][j_] = f2b(g_ * __builtin_amdgcn_rcpf(1.f + __expf(-g_)) * b2f(up_[j_]));     \
+          }                                                                                    \
+        }                                                                                      \
+      }                                                                                        \
     }                                                                                          \
     _Pragma("unroll")                                                                          \
     for (int u = 0; u < BKC; ++u) {                                                            \
@@ -2210,6 +2234,76 @@ torch::Tensor skinny_gemm_packed_xreg(torch::Tensor x, torch::Tensor wp, int64_t
   return y;
 }
 
+// Fused SwiGLU + down-projection decode GEMM: gu is the [M, 2K] fused
+// gate|up GEMM output, wp the prepacked down_proj ([N/32,K/16,64,8]);
+// computes y = (silu(gate) * up) @ W_down^T with the activation applied in
+// the X-staging — the intermediate [M, K] tensor never materializes.
+torch::Tensor skinny_gemm_packed_swiglu(torch::Tensor gu, torch::Tensor wp, int64_t N,
+                                        c10::optional<torch::Tensor> bias) {
+  CHK(gu.is_cuda() && gu.dtype() == torch::kBFloat16 && gu.is_contiguous());
+  CHK(wp.is_cuda() && wp.dtype() == torch::kBFloat16 && wp.is_contiguous());
+  const long long K = wp.numel() / N;
+  const long long M = gu.numel() / (2 * K);
+  CHK(gu.size(-1) == 2 * K);
+  CHK(M >= 32 && M <= 256 && M % 32 == 0);
+  CHK(N % 128 == 0 && K % 64 == 0);
+  const unsigned short* bptr = nullptr;
+  if (bias.has_value()) {
+    CHK(bias->is_contiguous() && bias->dtype() == torch::kBFloat16 && bias->numel() == N);
+    bptr = (const unsigned short*)bias->data_ptr();
+  }
+  auto sizes = gu.sizes().vec();
+  sizes.back() = (long)N;
+  auto y = torch::empty(sizes, gu.options());
+  const int ntiles = (int)(N / 128);
+  int nsplit = 1;
+  while (ntiles * nsplit < skinny_target_blocks() && (K / (nsplit * 2)) >= 1024 && nsplit < 16) nsplit *= 2;
+  int kc = (int)((K / nsplit + 63) / 64 * 64);
+  while ((long long)kc * (nsplit - 1) >= K) nsplit--;
+  auto stream = cur_stream();
+  const int MT = (int)(M / 32);
+  const bool bk128 = (K % 128 == 0) && (kc % 128 == 0) && kc >= 2048
+                     && getenv("XOT_SKINNY_BK64") == nullptr;
+  const dim3 grid(ntiles * nsplit), block(256);
+  const long long ldx = 2 * K;
+#define SGS_CASE(MTV) \
+  case MTV: \
+    if (nsplit == 1) { \
+      if (bk128) \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false, 128, true>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)gu.data_ptr(), \
+                           (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit, ldx); \
+      else \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, false, 64, true>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)gu.data_ptr(), \
+                           (unsigned short*)y.data_ptr(), nullptr, bptr, (int)N, K, kc, nsplit, ldx); \
+    } else { \
+      if (bk128) \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true, 128, true>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)gu.data_ptr(), \
+                           nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit, ldx); \
+      else \
+        hipLaunchKernelGGL((skinny_gemm_packed_kernel<MTV, true, 64, true>), grid, block, 0, stream, \
+                           (const unsigned short*)wp.data_ptr(), (const unsigned short*)gu.data_ptr(), \
+                           nullptr, P.data_ptr<float>(), nullptr, (int)N, K, kc, nsplit, ldx); \
+    } \
+    break;
+  if (nsplit == 1) {
+    torch::Tensor P;
+    switch (MT) { SGS_CASE(1) SGS_CASE(2) SGS_CASE(3) SGS_CASE(4) SGS_CASE(5) SGS_CASE(6) SGS_CASE(7) SGS_CASE(8) }
+  } else {
+    auto P = torch::empty({nsplit, M, (long long)N},
+                          torch::TensorOptions().dtype(torch::kFloat32).device(gu.device()));
+    switch (MT) { SGS_CASE(1) SGS_CASE(2) SGS_CASE(3) SGS_CASE(4) SGS_CASE(5) SGS_CASE(6) SGS_CASE(7) SGS_CASE(8) }
+    const long long MN = M * (long long)N;
+    const int blocks = (int)std::min<long long>(2048, (MN / 4 + 255) / 256);
+    hipLaunchKernelGGL(skinny_combine_kernel, dim3(blocks), dim3(256), 0, stream,
+                       P.data_ptr<float>(), (unsigned short*)y.data_ptr(), bptr, MN, N, nsplit);
+  }
+#undef SGS_CASE
+  return y;
+}
+
 // MoE grouped decode GEMM: x [E, C, K] (C = per-expert token capacity,
 // 32..256, %32), wp: stacked prepacks [E, N/32, K/16, 64, 8] -> y [E, C, N].
 // One launch for all experts (blockIdx.y = expert).
@@ -2353,6 +2447,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm_packed_xreg", &skinny_gemm_packed_xreg,
         "decode GEMM on prepacked weights, register-resident X (M=128/256 shapes)",
         py::arg("x"), py::arg("wp"), py::arg("N"), py::arg("bias") = py::none());
+  m.def("skinny_gemm_packed_swiglu", &skinny_gemm_packed_swiglu,
+        "fused silu(gate)*up + down-proj decode GEMM on the prepacked weight",
+        py::arg("gu"), py::arg("wp"), py::arg("n"), py::arg("bias") = py::none());
   m.def("skinny_gemm_packed", &skinny_gemm_packed,
         "decode GEMM on prepacked weights (MFMA fragment order)",
         py::arg("x"), py::arg("wp"), py::arg("n"), py::arg("bias") = py::none());
