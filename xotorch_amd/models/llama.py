@@ -223,7 +223,7 @@ class MLP(nn.Module):
   def forward(self, x):
     gu = self.gate_up_proj(x)
     dp = self.down_proj
-    if (dp.weight_packed is not None and gu.is_cuda and gu.dtype == torch.bfloat16
+    if (getattr(dp, "weight_packed", None) is not None and gu.is_cuda and gu.dtype == torch.bfloat16
         and not torch.is_grad_enabled() and gu.is_contiguous()
         and os.getenv("XOT_FUSE_SWIGLU", "1") == "1"):
       I = self.intermediate
