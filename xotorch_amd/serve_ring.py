@@ -431,7 +431,7 @@ class RingAPINode:
   async def process_prompt(self, shard, prompt: str, request_id=None, inference_state=None):
     request_id = request_id or str(uuid.uuid4())
     state = inference_state or {}
-    ids = self.tokenizer.encode(prompt)
+    ids = self.tokenizer.encode(prompt) or [self.tokenizer.eos_token_id or 0]
     if len(ids) >= self.worker.max_seq:
       raise ValueError(f"prompt length {len(ids)} exceeds context {self.worker.max_seq}")
     tokens = torch.tensor([ids], dtype=torch.int64)
