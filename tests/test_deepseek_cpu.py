@@ -205,3 +205,48 @@ def test_deepseek_two_node_tcp_ring(monkeypatch, tmp_path):
       await n.stop()
     return True
   assert run(go())
+
+
+def test_mla_per_row_positions():
+  """Continuous-batching contract: decode with ragged per-row positions
+  matches per-row sequential decode (CPU fp32, tiny MLA)."""
+  import copy
+  import torch
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.deepseek_v3 import DeepseekV3Model
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+  raw = dict(model_type="deepseek_v2", vocab_size=128, hidden_size=64, intermediate_size=96,
+             num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=4,
+             kv_lora_rank=16, qk_rope_head_dim=8, qk_nope_head_dim=16, v_head_dim=16,
+             q_lora_rank=0, first_k_dense_replace=2, n_routed_experts=0,
+             rms_norm_eps=1e-6, max_position_embeddings=64)
+  cfg = config_from_hf(raw, "mla-rows")
+  shard = Shard("mla-rows", 0, 1, 2)
+  torch.manual_seed(0)
+  m = DeepseekV3Model(cfg, shard).float()
+  random_init(m, std=0.15)  # big enough that wrong-context bugs are visible
+  m.eval()
+  B = 3
+  lens = [5, 9, 7]
+  h, kd, vd = cfg.kv_cache_dims()
+  cache = ShardKVCache(2, B, h, 32, kd, torch.float32, "cpu", v_dim=vd)
+  toks = torch.randint(0, 128, (B, 12))
+  with torch.inference_mode():
+    for b, L in enumerate(lens):
+      sliced = [tuple(t[b:b + 1] if t is not None else None for t in layer) for layer in cache.caches]
+      m(toks[b:b + 1, :L], caches=sliced, positions=torch.arange(L), start_pos=0)
+    nxt = toks[:, :1]
+    # references FIRST, on cache clones (decode writes the cache)
+    refs = []
+    for b, L in enumerate(lens):
+      cc = [tuple(t[b:b + 1].clone() if t is not None else None for t in layer) for layer in cache.caches]
+      refs.append(m(nxt[b:b + 1], caches=cc, positions=torch.tensor([L], dtype=torch.int32),
+                    start_pos=L, is_decode=True))
+    pos = torch.tensor(lens, dtype=torch.int32)
+    out = m(nxt, caches=cache.caches, positions=pos, start_pos=-1, is_decode=True,
+            seq_lens=torch.tensor([l + 1 for l in lens], dtype=torch.int32))
+    for b in range(B):
+      assert torch.allclose(out[b:b + 1], refs[b], atol=1e-5), \
+        (b, float((out[b:b + 1] - refs[b]).abs().max()))

@@ -154,3 +154,40 @@ def test_gemma2_two_node_tcp_ring(monkeypatch, tmp_path):
       await n.stop()
     return True
   assert run(go())
+
+
+def test_gemma2_per_row_positions():
+  """Ragged per-slot decode equals per-row sequential decode (eager path)."""
+  import torch
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.gemma2 import Gemma2Model
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+  raw = dict(TINY)
+  raw["model_type"] = "gemma2"
+  cfg = config_from_hf(raw, "gemma2-tiny")
+  shard = Shard("gemma2-tiny", 0, cfg.n_layers - 1, cfg.n_layers)
+  torch.manual_seed(2)
+  m = Gemma2Model(cfg, shard).float()
+  random_init(m, std=0.15)
+  m.eval()
+  B, lens = 3, [4, 8, 6]
+  cache = ShardKVCache(cfg.n_layers, B, cfg.n_kv_heads, 32, cfg.head_dim, torch.float32, "cpu")
+  toks = torch.randint(0, cfg.vocab_size, (B, 10))
+  with torch.inference_mode():
+    for b, L in enumerate(lens):
+      sliced = [tuple(t[b:b + 1] if t is not None else None for t in layer) for layer in cache.caches]
+      m(toks[b:b + 1, :L], caches=sliced, positions=torch.arange(L), start_pos=0)
+    nxt = toks[:, :1]
+    refs = []
+    for b, L in enumerate(lens):
+      cc = [tuple(t[b:b + 1].clone() if t is not None else None for t in layer) for layer in cache.caches]
+      refs.append(m(nxt[b:b + 1], caches=cc, positions=torch.tensor([L], dtype=torch.int32),
+                    start_pos=L, is_decode=True))
+    out = m(nxt, caches=cache.caches, positions=torch.tensor(lens, dtype=torch.int32),
+            start_pos=-1, is_decode=True,
+            seq_lens=torch.tensor([l + 1 for l in lens], dtype=torch.int32))
+    for b in range(B):
+      assert torch.allclose(out[b:b + 1], refs[b], atol=1e-5), \
+        (b, float((out[b:b + 1] - refs[b]).abs().max()))

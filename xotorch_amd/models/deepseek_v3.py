@@ -39,9 +39,14 @@ def _rms(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
 
 
 def _rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, interleave: bool) -> torch.Tensor:
-  """x: [B, S, H, D]; cos/sin: [S, D/2] (per-pair angles)."""
-  c = cos[None, :, None, :].float()
-  s = sin[None, :, None, :].float()
+  """x: [B, S, H, D]; cos/sin: [S, D/2] shared across batch, or [B, S, D/2]
+  per-row (continuous-batching slots decode at different positions)."""
+  if cos.dim() == 3:
+    c = cos[:, :, None, :].float()
+    s = sin[:, :, None, :].float()
+  else:
+    c = cos[None, :, None, :].float()
+    s = sin[None, :, None, :].float()
   xf = x.float()
   if interleave:
     x1, x2 = xf[..., 0::2], xf[..., 1::2]
@@ -86,7 +91,12 @@ class MLAttention(nn.Module):
     kv_nope, k_rot = ckv[..., : cfg.kv_lora_rank], ckv[..., cfg.kv_lora_rank:]
     kv_nope = _rms(kv_nope, self.kv_a_layernorm, cfg.norm_eps)
 
-    cs, sn = cos[positions.reshape(-1)], sin[positions.reshape(-1)]
+    pidx = positions.reshape(-1).long()
+    cs, sn = cos[pidx], sin[pidx]
+    if pidx.numel() == B * S and pidx.numel() != S:
+      # per-row positions (continuous-batching slots)
+      cs = cs.view(B, S, -1)
+      sn = sn.view(B, S, -1)
     q_rot = _rope(q_rot, cs, sn, cfg.rope_interleave)
     k_rot = _rope(k_rot.view(B, S, 1, rope_d), cs, sn, cfg.rope_interleave)
 
@@ -107,8 +117,18 @@ class MLAttention(nn.Module):
       if seq_lens is None:
         seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
       return self._decode_mfma(x, q_pass, q_rot, kv, seq_lens, hip)
-    if start_pos < 0:  # ring decode contract: derive from positions (host sync, eager only)
-      start_pos = int(positions.reshape(-1)[0])
+    if start_pos < 0:  # ring/serve decode contract: derive from positions
+      prow = positions.reshape(-1)
+      if prow.numel() == B * S and S == 1 and B > 1 and int(prow.min()) != int(prow.max()):
+        # ragged per-row positions on the EAGER path (continuous-batching
+        # slots on CPU / no packed cache): correctness-first row loop
+        outs = []
+        for b in range(B):
+          outs.append(self.forward(x[b:b + 1], cos, sin, prow[b:b + 1],
+                                   tuple(t[b:b + 1] if t is not None else None for t in kv),
+                                   int(prow[b]), is_decode, None))
+        return torch.cat(outs, dim=0)
+      start_pos = int(prow[0])
     # latent cache: k tensor <- kv_nope [B,1,T,kv_lora], v tensor <- roped
     # shared key [B,1,T,rope_d]
     lat_c, rot_c = kv[0], kv[1]

@@ -91,8 +91,17 @@ class Gemma2Attention(nn.Module):
                                scale=self.scale, softcap=cap, window=self.window)
       return self.o_proj(out.reshape(B, S, H * hd))
     # eager oracle path (CPU / hd != 128): split the fused projection
-    if start_pos < 0:  # ring decode contract: derive from positions
-      start_pos = int(positions.reshape(-1)[0])
+    if start_pos < 0:  # ring/serve decode contract: derive from positions
+      prow = positions.reshape(-1)
+      if prow.numel() == B * S and S == 1 and B > 1 and int(prow.min()) != int(prow.max()):
+        # ragged per-row slot positions: correctness-first row loop
+        outs = []
+        for b in range(B):
+          outs.append(self.forward(x[b:b + 1], cos, sin, prow[b:b + 1],
+                                   tuple(t[b:b + 1] if t is not None else None for t in kv),
+                                   int(prow[b]), is_decode, None))
+        return torch.cat(outs, dim=0)
+      start_pos = int(prow[0])
     from xotorch_amd.ops import torch_ref as tr
     q, k, v = torch.split(qkv, [H * hd, KVH * hd, KVH * hd], dim=-1)
     q = q.view(B, S, H, hd)
