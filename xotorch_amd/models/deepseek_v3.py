@@ -28,11 +28,16 @@ import torch
 import torch.nn as nn
 
 from xotorch_amd.models.config import ModelConfig
+from xotorch_amd.models.llama import XotLinear
 from xotorch_amd.ops.torch_ref import rope_cos_sin
 from xotorch_amd.shard import Shard
 
 
 def _rms(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+  if x.is_cuda and x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16 \
+     and x.shape[-1] % 8 == 0 and x.shape[-1] <= 16384:
+    from xotorch_amd import ops
+    return ops.rmsnorm(x.contiguous(), w, eps)  # one HIP launch vs ~4 torch ops
   xf = x.float()
   out = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
   return (w.float() * out).to(x.dtype)
@@ -65,13 +70,13 @@ class MLAttention(nn.Module):
     if cfg.q_lora_rank:
       self.q_a_proj = nn.Linear(D, cfg.q_lora_rank, bias=cfg.attn_bias)
       self.q_a_layernorm = nn.Parameter(torch.ones(cfg.q_lora_rank))
-      self.q_b_proj = nn.Linear(cfg.q_lora_rank, H * self.qk_head_dim, bias=False)
+      self.q_b_proj = XotLinear(cfg.q_lora_rank, H * self.qk_head_dim, bias=False)
     else:
-      self.q_proj = nn.Linear(D, H * self.qk_head_dim, bias=False)
+      self.q_proj = XotLinear(D, H * self.qk_head_dim, bias=False)
     self.kv_a_proj_with_mqa = nn.Linear(D, cfg.kv_lora_rank + cfg.qk_rope_head_dim, bias=cfg.attn_bias)
     self.kv_a_layernorm = nn.Parameter(torch.ones(cfg.kv_lora_rank))
     self.kv_b_proj = nn.Linear(cfg.kv_lora_rank, H * (cfg.qk_nope_head_dim + cfg.v_head_dim), bias=False)
-    self.o_proj = nn.Linear(H * cfg.v_head_dim, D, bias=cfg.attn_bias)
+    self.o_proj = XotLinear(H * cfg.v_head_dim, D, bias=cfg.attn_bias)
     self.scale = self.qk_head_dim ** -0.5  # default rope (no yarn mscale)
 
   def forward(self, x, cos, sin, positions, kv, start_pos: int,
@@ -181,9 +186,9 @@ class DsMLP(nn.Module):
   def __init__(self, cfg: ModelConfig, intermediate: int):
     super().__init__()
     D = cfg.dim
-    self.gate_proj = nn.Linear(D, intermediate, bias=False)
-    self.up_proj = nn.Linear(D, intermediate, bias=False)
-    self.down_proj = nn.Linear(intermediate, D, bias=False)
+    self.gate_proj = XotLinear(D, intermediate, bias=False)
+    self.up_proj = XotLinear(D, intermediate, bias=False)
+    self.down_proj = XotLinear(intermediate, D, bias=False)
 
   def forward(self, x):
     return self.down_proj(nn.functional.silu(self.gate_proj(x)) * self.up_proj(x))
@@ -334,11 +339,17 @@ class DeepseekV3Model(nn.Module):
     return list(range(self.shard.start_layer, self.shard.end_layer + 1))
 
   def pack_decode_weights(self, reserve_bytes: int = 0) -> int:
-    """Dense GEMMs stay on hipBLASLt; MLA decode attention runs the
-    absorbed-latent MFMA kernel via the packed cache; MoE experts get the
-    stacked grouped prepack (single-launch decode GEMMs, graph-capturable
-    static routing)."""
+    """MLA projections + dense/shared MLPs get the packed skinny-GEMM
+    prepack (auto-picked per shape vs hipBLASLt); MoE experts get the
+    stacked grouped prepack; routed per-expert linears are NOT individually
+    packed (the grouped copy already covers them). MLA decode attention
+    runs the absorbed-latent MFMA kernel via the packed cache."""
     packed = 0
+    for name, mod in self.named_modules():
+      if isinstance(mod, XotLinear) and mod.packable() and ".experts." not in name:
+        mod.pack_decode()
+        if mod.weight_packed is not None:
+          packed += mod.weight_packed.numel() * 2
     for mod in self.modules():
       if isinstance(mod, DsMoE):
         mod.pack_grouped()
