@@ -110,3 +110,12 @@ def test_8b_two_ranks_one_gpu_fits():
     for _ in range(1)
   ) * 2  # both ranks share the device; worst-rank x2 is an upper bound
   assert total < HBM_BYTES * 0.7
+
+
+def test_405b_ring_fits_hbm():
+  """llama-3.1-405b across 8 ranks: the 2.3 TB ring holds it at bf16
+  (B=64 per micro-batch; B=128 exceeds the comfortable ceiling — use
+  --mb-batch 64)."""
+  need = ring_rank_bytes("llama-3.1-405b", 8, mb_batch=64, prompt_len=512,
+                         max_gen=256, include_packed=False)
+  assert need < HBM_BYTES * 0.85, f"{need/2**30:.1f} GiB"

@@ -133,3 +133,31 @@ def test_two_rank_ring_matches_single(tmp_path):
   got = json.loads((tmp_path / "tokens.json").read_text())
   for rid, ids, max_new in reqs:
     assert got[rid] == oracle_tokens(ids, max_new), rid
+
+
+def test_cancel_releases_slot():
+  """Cancelling a long request frees its slot for the next one."""
+  import time as _t
+  w = RingSlotWorker(TINY_ID, 0, 1, device="cpu", dtype=torch.float32,
+                     slots=1, max_seq=2048, use_graphs=False)
+  q = queue.Queue()
+  got = {}
+  done = threading.Event()
+
+  def emit(rid, tok, fin, meta):
+    got.setdefault(rid, []).append((tok, fin, meta))
+    if fin and rid == "short":
+      done.set()
+
+  ids = list(np.random.default_rng(2).integers(0, 200, 6))
+  q.put(AdmitMsg("long", torch.tensor([ids], dtype=torch.int64), 2000, 0.0))
+  t = threading.Thread(target=w.serve_forever, args=(q, emit), daemon=True)
+  t.start()
+  _t.sleep(0.15)         # long request occupies the only slot
+  w.cancel("long")
+  q.put(AdmitMsg("short", torch.tensor([ids], dtype=torch.int64), 3, 0.0))
+  assert done.wait(60), "slot was not freed by cancellation"
+  q.put(AdmitMsg("stop", None, 0, 0.0))
+  t.join(timeout=30)
+  assert len(got["long"]) < 240  # cancelled well before the max_seq-clamped max_new
+  assert got["long"][-1][2].get("cancelled") or got["long"][-1][1]

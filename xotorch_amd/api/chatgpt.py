@@ -302,13 +302,21 @@ class ChatGPTAPI:
     await response.prepare(request)
     all_tokens: List[int] = []
     finished = False
-    while not finished:
-      tokens, finished = await asyncio.wait_for(self.token_queues[request_id].get(), self.response_timeout)
-      all_tokens.extend(tokens)
-      decoded = tokenizer.decode(tokens) if tokens else ""
-      chunk = generate_completion(request_id, tokens, decoded, model_id,
-                                  "stop" if finished else None, True, "chat.completion.chunk")
-      await response.write(f"data: {json.dumps(chunk)}\n\n".encode())
+    try:
+      while not finished:
+        tokens, finished = await asyncio.wait_for(self.token_queues[request_id].get(), self.response_timeout)
+        all_tokens.extend(tokens)
+        decoded = tokenizer.decode(tokens) if tokens else ""
+        chunk = generate_completion(request_id, tokens, decoded, model_id,
+                                    "stop" if finished else None, True, "chat.completion.chunk")
+        await response.write(f"data: {json.dumps(chunk)}\n\n".encode())
+    except (asyncio.CancelledError, ConnectionResetError):
+      # client went away mid-stream: free the slot/session instead of
+      # generating to max_tokens for nobody
+      cancel = getattr(self.node, "cancel_request", None)
+      if cancel is not None:
+        cancel(request_id)
+      raise
     await response.write(b"data: [DONE]\n\n")
     await response.write_eof()
     return response

@@ -277,10 +277,18 @@ class RingSlotWorker:
       self.comm.broadcast(self.hdr, 0)
     return [int(v) for v in self.hdr.cpu()]
 
+  def cancel(self, request_id: str):
+    """Mark a request for release (client went away). Thread-safe; consumed
+    by the rank-0 control loop on its next tick."""
+    self._cancelled.add(request_id)
+
+  _cancelled: set = None  # set in serve_forever (rank 0)
+
   def serve_forever(self, admit_q: Optional["queue.Queue"] = None, emit=None,
                     idle_sleep: float = 0.002, max_ticks: Optional[int] = None):
     """Run the ring loop. On rank 0, `admit_q` provides AdmitMsg and `emit`
     receives (request_id, token, is_finished, meta). Followers pass None."""
+    self._cancelled = set()
     pending_release: List[int] = []
     inflight: Dict[int, AdmitMsg] = {}
     ticks = 0
@@ -288,6 +296,15 @@ class RingSlotWorker:
       ticks += 1
       op, slot, plen, extra = OP_IDLE, 0, 0, 0
       if self.rank == 0:
+        if self._cancelled:
+          for i in range(self.slots):
+            st = self.slot_state[i]
+            if self.active[i] and st.request_id in self._cancelled and not st.done:
+              st.done = True
+              if emit:
+                emit(st.request_id, 0, True, {"cancelled": True})
+              pending_release.append(i)
+          self._cancelled.clear()
         if pending_release:
           op, slot = OP_RELEASE, pending_release.pop(0)
         else:
@@ -427,6 +444,9 @@ class RingAPINode:
     if meta.get("ttft_s") is not None:
       self.request_meta[request_id] = meta
     self.loop.call_soon_threadsafe(self.on_token.trigger_all, request_id, [token], finished)
+
+  def cancel_request(self, request_id: str):
+    self.worker.cancel(request_id)
 
   async def process_prompt(self, shard, prompt: str, request_id=None, inference_state=None):
     request_id = request_id or str(uuid.uuid4())
