@@ -761,3 +761,36 @@ def test_skinny_gemm_packed_swiglu(hip, M, I, N):
   h = torch_ref.swiglu_packed(gu.float())
   ref = h.float() @ w.float().T
   assert torch.allclose(y, ref, atol=5e-2, rtol=5e-2), (y - ref).abs().max()
+
+
+def test_mla_prep_append_matches_torch():
+  """Fused rms+rope+dual-cache-append vs the torch reference path."""
+  from xotorch_amd.ops import _hip_ops
+  from xotorch_amd.models.deepseek_v3 import _rms, _rope
+  from xotorch_amd.ops.torch_ref import rope_cos_sin
+  torch.manual_seed(8)
+  B, S, T = 2, 20, 32
+  t32 = 32
+  ckv = bt(B, S, 576, seed=131, scale=0.5)
+  w = bt(512, seed=132, scale=0.3) + 1.0
+  cos, sin = rope_cos_sin(64, 64, 10000.0, device="cuda")
+  lat_c = torch.zeros(B, 1, T, 512, dtype=torch.bfloat16, device="cuda")
+  rot_c = torch.zeros(B, 1, T, 64, dtype=torch.bfloat16, device="cuda")
+  kp = torch.zeros(B, t32 // 16, 18, 64, 8, dtype=torch.bfloat16, device="cuda")
+  vp = torch.zeros(B, 32, t32 // 32, 64, 8, dtype=torch.bfloat16, device="cuda")
+  pos = torch.arange(S, dtype=torch.int32, device="cuda")
+  for interleave in (False, True):
+    _hip_ops.mla_prep_append(ckv, w.contiguous(), cos, sin, pos, lat_c, rot_c, kp, vp,
+                             1e-6, interleave)
+    lat_ref = _rms(ckv[..., :512], w, 1e-6)
+    rot_ref = _rope(ckv[..., 512:].view(B, S, 1, 64), cos[pos.long()], sin[pos.long()],
+                    interleave)[:, :, 0, :]
+    assert torch.allclose(lat_c[:, 0, :S].float(), lat_ref.float(), atol=2e-2), \
+      (lat_c[:, 0, :S].float() - lat_ref.float()).abs().max()
+    assert torch.allclose(rot_c[:, 0, :S].float(), rot_ref.float(), atol=2e-2), \
+      (rot_c[:, 0, :S].float() - rot_ref.float()).abs().max()
+    # packed copies match the reference packer of (lat, rot)
+    kp_ref, vp_ref = _mla_pack_ref(lat_c[:, 0, :S], rot_c[:, 0, :S], t32)
+    assert torch.equal(kp, kp_ref)
+    assert torch.equal(vp, vp_ref)
+    kp.zero_(); vp.zero_(); lat_c.zero_(); rot_c.zero_()

@@ -93,8 +93,6 @@ class MLAttention(nn.Module):
     q_pass, q_rot = q[..., :nope], q[..., nope:]
 
     ckv = self.kv_a_proj_with_mqa(x)
-    kv_nope, k_rot = ckv[..., : cfg.kv_lora_rank], ckv[..., cfg.kv_lora_rank:]
-    kv_nope = _rms(kv_nope, self.kv_a_layernorm, cfg.norm_eps)
 
     pidx = positions.reshape(-1).long()
     cs, sn = cos[pidx], sin[pidx]
@@ -103,7 +101,6 @@ class MLAttention(nn.Module):
       cs = cs.view(B, S, -1)
       sn = sn.view(B, S, -1)
     q_rot = _rope(q_rot, cs, sn, cfg.rope_interleave)
-    k_rot = _rope(k_rot.view(B, S, 1, rope_d), cs, sn, cfg.rope_interleave)
 
     kp = kv[2] if len(kv) > 2 else None
     hip = None
@@ -111,40 +108,40 @@ class MLAttention(nn.Module):
       from xotorch_amd.ops import _load_hip
       hip = _load_hip()
     decode = is_decode or (S == 1 and (start_pos > 0 or start_pos < 0))
-    if hip is not None and decode and not torch.is_grad_enabled():
-      # MFMA decode fast path: packed-cache append with DEVICE positions
-      # (graph-capturable — no host sync), absorbed-latent attention.
-      # The plain cache is not touched: decode never reads it, and prefill
-      # rebuilt it before this point.
-      hip.mla_append(kv_nope.to(x.dtype).contiguous(),
-                     k_rot[:, :, 0, :].to(x.dtype).contiguous(),
-                     positions.to(torch.int32).contiguous(), kp, kv[3])
-      if seq_lens is None:
-        seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
-      return self._decode_mfma(x, q_pass, q_rot, kv, seq_lens, hip)
-    if start_pos < 0:  # ring/serve decode contract: derive from positions
-      prow = positions.reshape(-1)
-      if prow.numel() == B * S and S == 1 and B > 1 and int(prow.min()) != int(prow.max()):
-        # ragged per-row positions on the EAGER path (continuous-batching
-        # slots on CPU / no packed cache): correctness-first row loop
-        outs = []
-        for b in range(B):
-          outs.append(self.forward(x[b:b + 1], cos, sin, prow[b:b + 1],
-                                   tuple(t[b:b + 1] if t is not None else None for t in kv),
-                                   int(prow[b]), is_decode, None))
-        return torch.cat(outs, dim=0)
-      start_pos = int(prow[0])
-    # latent cache: k tensor <- kv_nope [B,1,T,kv_lora], v tensor <- roped
-    # shared key [B,1,T,rope_d]
-    lat_c, rot_c = kv[0], kv[1]
-    lat_c[:, 0, start_pos: start_pos + S] = kv_nope
-    rot_c[:, 0, start_pos: start_pos + S] = k_rot[:, :, 0, :]
     if hip is not None:
-      # keep the fragment-packed copies appended at prefill so later decode
-      # steps can stream them
-      hip.mla_append(kv_nope.to(x.dtype).contiguous(),
-                     k_rot[:, :, 0, :].to(x.dtype).contiguous(),
-                     positions.to(torch.int32).contiguous(), kp, kv[3])
+      # fused KV-side prep: latent RMSNorm + shared-key rope + plain AND
+      # fragment-packed cache writes in ONE launch (device positions —
+      # graph-capturable; replaces ~10 torch launches per layer)
+      hip.mla_prep_append(ckv.contiguous(), self.kv_a_layernorm, cos, sin,
+                          positions.to(torch.int32).contiguous(), kv[0], kv[1],
+                          kp, kv[3], cfg.norm_eps, cfg.rope_interleave)
+      if decode and not torch.is_grad_enabled():
+        if seq_lens is None:
+          seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
+        return self._decode_mfma(x, q_pass, q_rot, kv, seq_lens, hip)
+      if start_pos < 0:
+        start_pos = int(positions.reshape(-1)[0])
+    else:
+      kv_nope, k_rot = ckv[..., : cfg.kv_lora_rank], ckv[..., cfg.kv_lora_rank:]
+      kv_nope = _rms(kv_nope, self.kv_a_layernorm, cfg.norm_eps)
+      k_rot = _rope(k_rot.view(B, S, 1, rope_d), cs, sn, cfg.rope_interleave)
+      if start_pos < 0:  # ring/serve decode contract: derive from positions
+        prow = positions.reshape(-1)
+        if prow.numel() == B * S and S == 1 and B > 1 and int(prow.min()) != int(prow.max()):
+          # ragged per-row positions on the EAGER path (continuous-batching
+          # slots on CPU / no packed cache): correctness-first row loop
+          outs = []
+          for b in range(B):
+            outs.append(self.forward(x[b:b + 1], cos, sin, prow[b:b + 1],
+                                     tuple(t[b:b + 1] if t is not None else None for t in kv),
+                                     int(prow[b]), is_decode, None))
+          return torch.cat(outs, dim=0)
+        start_pos = int(prow[0])
+      # latent cache: k tensor <- kv_nope [B,1,T,kv_lora], v tensor <- roped
+      # shared key [B,1,T,rope_d]
+      lat_c, rot_c = kv[0], kv[1]
+      lat_c[:, 0, start_pos: start_pos + S] = kv_nope
+      rot_c[:, 0, start_pos: start_pos + S] = k_rot[:, :, 0, :]
     total = start_pos + S
     lat = lat_c[:, 0, :total]                                   # [B, T, kv_lora]
     krot = rot_c[:, 0, :total]                                  # [B, T, rope_d]

@@ -845,6 +845,74 @@ __global__ __launch_bounds__(256) void moe_combine_kernel(
 #define MLA_CHQK 18   // 32-dim MFMA chunks across DQK
 #define MLA_GPV 32    // 16-dim output groups across LAT
 
+// Fused MLA KV-side prep: raw kv_a output [B,S,576] -> RMSNorm the 512-dim
+// latent (fp32 math, weight w), rope the 64-dim shared key at `positions`
+// (interleaved or half convention), and write BOTH cache layouts (plain
+// [B,1,T,512]/[B,1,T,64] for the eager prefill reader, fragment-packed
+// kp/vp for the MFMA decode kernel). Replaces ~10 torch launches per layer
+// (norm ops, rope ops, two .contiguous() copies, slice writes) with one.
+// One wave per (b, s): 512 = 64 lanes x 8 for the rms reduce.
+__global__ __launch_bounds__(256) void mla_prep_append_kernel(
+    const unsigned short* __restrict__ ckv, const unsigned short* __restrict__ w,
+    const float* __restrict__ cosb, const float* __restrict__ sinb,
+    const int* __restrict__ positions,
+    unsigned short* __restrict__ lat_c, unsigned short* __restrict__ rot_c,
+    unsigned short* __restrict__ kp, unsigned short* __restrict__ vp,
+    int B, int S, int T, int T32, float eps, int interleave, int npos) {
+  const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (wid >= B * S) return;
+  const int lane = threadIdx.x & 63;
+  const int s = wid % S;
+  const int b = wid / S;
+  const int pos = positions[(npos == B * S && npos != S) ? wid : s];
+  const unsigned short* row = ckv + (size_t)wid * MLA_DQK;
+  // ---- rms over the 512 latent dims ----
+  float v[8];
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    v[j] = b2f(row[lane * 8 + j]);
+    acc += v[j] * v[j];
+  }
+#pragma unroll
+  for (int m = 32; m > 0; m >>= 1) acc += __shfl_xor(acc, m);
+  const float inv = rsqrtf(acc / (float)MLA_LAT + eps);
+  unsigned short lat[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) lat[j] = f2b(v[j] * inv * b2f(w[lane * 8 + j]));
+  // plain latent cache + packed copies
+  unsigned short* lrow = lat_c + ((size_t)b * T + pos) * MLA_LAT;
+  const size_t kbase = ((size_t)b * (T32 >> 4) + (pos >> 4)) * MLA_CHQK * 512;
+  const size_t vbase = (size_t)b * MLA_GPV * (T32 >> 5) * 512;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int d = lane * 8 + j;
+    lrow[d] = lat[j];
+    kp[kbase + (size_t)(d >> 5) * 512 + (((d & 31) >> 3) * 16 + (pos & 15)) * 8 + (d & 7)] = lat[j];
+    vp[vbase + ((size_t)(d >> 4) * (T32 >> 5) + (pos >> 5)) * 512
+       + (((pos & 31) >> 3) * 16 + (d & 15)) * 8 + (pos & 7)] = lat[j];
+  }
+  // ---- rope the 64-dim shared key (lanes 0..31 handle one pair each) ----
+  unsigned short* rrow = rot_c + ((size_t)b * T + pos) * MLA_ROPE;
+  if (lane < 32) {
+    const float c = cosb[(size_t)pos * 32 + lane];
+    const float sn = sinb[(size_t)pos * 32 + lane];
+    int i1, i2;
+    if (interleave) { i1 = lane * 2; i2 = lane * 2 + 1; }
+    else            { i1 = lane;     i2 = lane + 32;    }
+    const float x1 = b2f(row[MLA_LAT + i1]);
+    const float x2 = b2f(row[MLA_LAT + i2]);
+    // output is cat([x1c - x2s, x2c + x1s]) in HALF layout (matches _rope)
+    const unsigned short r1 = f2b(x1 * c - x2 * sn);
+    const unsigned short r2 = f2b(x2 * c + x1 * sn);
+    rrow[lane] = r1;
+    rrow[lane + 32] = r2;
+    const int d1 = MLA_LAT + lane, d2 = MLA_LAT + lane + 32;
+    kp[kbase + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = r1;
+    kp[kbase + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = r2;
+  }
+}
+
 __global__ __launch_bounds__(256) void mla_append_kernel(
     const unsigned short* __restrict__ lat, const unsigned short* __restrict__ rot,
     const int* __restrict__ positions, unsigned short* __restrict__ kp,
@@ -1948,6 +2016,31 @@ torch::Tensor moe_combine(torch::Tensor y, torch::Tensor inv_pos, torch::Tensor 
   return out;
 }
 
+// ckv: [B, S, 576] raw kv_a output; w: [512] bf16 norm weight; cos/sin
+// fp32 [maxT, 32]; lat_c/rot_c: the plain caches viewed [B, T, 512]/[B, T, 64].
+void mla_prep_append(torch::Tensor ckv, torch::Tensor w, torch::Tensor cos,
+                     torch::Tensor sin, torch::Tensor positions,
+                     torch::Tensor lat_c, torch::Tensor rot_c,
+                     torch::Tensor kp, torch::Tensor vp, double eps, bool interleave) {
+  CHK(ckv.is_cuda() && ckv.dtype() == torch::kBFloat16 && ckv.is_contiguous());
+  CHK(w.dtype() == torch::kBFloat16 && w.is_contiguous() && w.numel() == MLA_LAT);
+  CHK(positions.dtype() == torch::kInt32 && positions.is_contiguous());
+  CHK(lat_c.is_contiguous() && rot_c.is_contiguous() && kp.is_contiguous() && vp.is_contiguous());
+  const int B = ckv.size(0), S = ckv.size(1);
+  CHK(ckv.size(2) == MLA_DQK);
+  const int T = lat_c.numel() / (B * MLA_LAT);
+  const int T32 = (int)kp.size(1) * 16;
+  const int npos = (int)positions.numel();
+  CHK(npos == S || npos == B * S);
+  const int waves = B * S;
+  hipLaunchKernelGGL(mla_prep_append_kernel, dim3((waves + 3) / 4), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)ckv.data_ptr(), (const unsigned short*)w.data_ptr(),
+                     cos.data_ptr<float>(), sin.data_ptr<float>(), positions.data_ptr<int>(),
+                     (unsigned short*)lat_c.data_ptr(), (unsigned short*)rot_c.data_ptr(),
+                     (unsigned short*)kp.data_ptr(), (unsigned short*)vp.data_ptr(),
+                     B, S, T, T32, (float)eps, interleave ? 1 : 0, npos);
+}
+
 // lat: [B, S, 512] bf16 (post-RMSNorm latent), rot: [B, S, 64] bf16 (roped
 // shared key); kp: [B, T32/16, 18, 64, 8], vp: [B, 32, T32/32, 64, 8].
 void mla_append(torch::Tensor lat, torch::Tensor rot, torch::Tensor positions,
@@ -2483,6 +2576,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MoE decode routing: counting-sort token-expert pairs to padded per-expert slots");
   m.def("moe_combine", &moe_combine,
         "MoE decode combine: out[t] = sum_j w[t,j] * y[pos(t,j)] (deterministic)");
+  m.def("mla_prep_append", &mla_prep_append,
+        "fused MLA kv prep: latent RMSNorm + shared-key rope + both cache layouts");
   m.def("mla_append", &mla_append,
         "append MLA latent+rope token stream into the fragment-packed cache");
   m.def("attn_decode_mla", &attn_decode_mla,
