@@ -142,6 +142,7 @@ class MLAttention(nn.Module):
       lat_c, rot_c = kv[0], kv[1]
       lat_c[:, 0, start_pos: start_pos + S] = kv_nope
       rot_c[:, 0, start_pos: start_pos + S] = k_rot[:, :, 0, :]
+    lat_c, rot_c = kv[0], kv[1]
     total = start_pos + S
     lat = lat_c[:, 0, :total]                                   # [B, T, kv_lora]
     krot = rot_c[:, 0, :total]                                  # [B, T, rope_d]
