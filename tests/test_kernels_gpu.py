@@ -794,3 +794,46 @@ def test_mla_prep_append_matches_torch():
     assert torch.equal(kp, kp_ref)
     assert torch.equal(vp, vp_ref)
     kp.zero_(); vp.zero_(); lat_c.zero_(); rot_c.zero_()
+
+
+def test_moe_route_softmax_mode(hip):
+  from xotorch_amd.ops import _hip_ops
+  torch.manual_seed(5)
+  T, E, k = 48, 64, 2
+  logits = torch.randn(T, E, device="cuda", dtype=torch.float32)
+  idx, w = _hip_ops.moe_route(logits, None, k, 0)
+  ref_w, ref_idx = torch.topk(torch.softmax(logits, dim=-1), k, dim=-1)
+  ref_w = ref_w / ref_w.sum(dim=-1, keepdim=True)
+  assert torch.equal(idx.long(), ref_idx)
+  assert torch.allclose(w, ref_w, atol=1e-5), (w - ref_w).abs().max()
+
+
+@pytest.mark.parametrize("E,ng,tg,k,norm", [(64, 1, 1, 6, False), (256, 8, 4, 8, True)])
+def test_moe_route_deepseek_mode(hip, E, ng, tg, k, norm):
+  from xotorch_amd.ops import _hip_ops
+  torch.manual_seed(6)
+  T = 40
+  logits = torch.randn(T, E, device="cuda", dtype=torch.float32)
+  bias = torch.randn(E, device="cuda", dtype=torch.float32) * 0.1
+  scale = 2.5
+  idx, w = _hip_ops.moe_route(logits, bias, k, 1, ng, tg, scale, norm)
+  # torch reference (DsMoE.route semantics)
+  scores = logits.sigmoid()
+  choice = scores + bias
+  group_scores = choice.view(T, ng, E // ng).topk(min(2, E // ng), dim=-1)[0].sum(-1)
+  gidx = torch.topk(group_scores, k=tg, dim=-1, sorted=False)[1]
+  gmask = torch.zeros_like(group_scores).scatter_(1, gidx, 1)
+  smask = gmask[:, :, None].expand(-1, ng, E // ng).reshape(T, E)
+  masked = choice.masked_fill(~smask.bool(), float("-inf"))
+  ref_idx = torch.topk(masked, k=k, dim=-1, sorted=True)[1]
+  ref_w = scores.gather(1, ref_idx)
+  if norm:
+    ref_w = ref_w / (ref_w.sum(dim=-1, keepdim=True) + 1e-20)
+  ref_w = ref_w * scale
+  # compare as (expert -> weight) sets per row (tie order may differ)
+  for t in range(T):
+    a = dict(zip(idx[t].tolist(), w[t].tolist()))
+    b = dict(zip(ref_idx[t].tolist(), ref_w[t].tolist()))
+    assert set(a) == set(b), (t, a, b)
+    for e in a:
+      assert abs(a[e] - b[e]) < 1e-4, (t, e, a[e], b[e])

@@ -253,7 +253,12 @@ class DsMoE(nn.Module):
     T, D = flat.shape
     E, k = cfg.n_experts, cfg.n_experts_per_tok
     dev = flat.device
-    idx, w = self.route(flat)                                 # [T, k]
+    cfg = self.cfg
+    logits = nn.functional.linear(flat.float(), self.gate_weight.float())
+    idx, w = hip.moe_route(logits.contiguous(), self.e_score_correction_bias.float().contiguous(),
+                           cfg.n_experts_per_tok, 1, max(1, cfg.n_group),
+                           max(1, cfg.topk_group), cfg.routed_scaling_factor,
+                           bool(cfg.norm_topk_prob))
     C = max(32, -(-T // 32) * 32)
     # counting-sort routing + deterministic combine as single HIP launches
     # (the torch glue — argsort/cumsum/index_add — was the measured top

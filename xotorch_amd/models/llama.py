@@ -331,10 +331,15 @@ class MoEMLP(nn.Module):
     T, D = flat.shape
     E, k = self.n_experts, self.top_k
     dev = flat.device
-    weights, selected = self._route(flat)                     # [T, k]
     from xotorch_amd.ops import _load_hip
     hip0 = _load_hip()
     C = max(32, -(-T // 32) * 32)                             # capacity (lossless: count_e <= T)
+    if hip0 is not None and flat.is_cuda:
+      # fused router (softmax-topk-renorm in one launch)
+      logits = self.gate(flat).float()
+      selected, weights = hip0.moe_route(logits.contiguous(), None, self.top_k, 0)
+    else:
+      weights, selected = self._route(flat)                   # [T, k]
     if hip0 is not None and flat.is_cuda:
       # single-launch counting-sort + deterministic combine (replaces the
       # argsort/cumsum/index_add torch glue — measured top MoE decode cost)

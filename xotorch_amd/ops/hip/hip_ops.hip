@@ -757,6 +757,129 @@ __global__ __launch_bounds__(128) void attn_decode_merge(
 // out[t] = sum_j w[t,j] * y[pos(t,j)]).
 // ---------------------------------------------------------------------------
 
+// Fused MoE router: one wave per token row over the gate logits [T, E].
+// mode 0: softmax -> top-k -> renormalize (Mixtral / qwen3-moe).
+// mode 1: sigmoid + e-score bias -> group-limited top-k (top-2-sum group
+//         scores, keep topk_group groups) -> gather sigmoid weights ->
+//         optional renorm -> routed scaling (DeepSeek V2/V3).
+// Replaces ~4-10 torch launches per MoE layer per step. E <= 256, k <= 8.
+__global__ __launch_bounds__(256) void moe_route_kernel(
+    const float* __restrict__ logits, const float* __restrict__ bias,
+    int* __restrict__ idx, float* __restrict__ w,
+    int T, int E, int k, int mode, int n_group, int topk_group,
+    float routed_scale, int norm_topk) {
+  const int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= T) return;
+  const int lane = threadIdx.x & 63;
+  // per-lane 4 contiguous elements (E <= 256)
+  float v[4], s[4], c[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int e = lane * 4 + j;
+    v[j] = (e < E) ? logits[(size_t)row * E + e] : -INFINITY;
+  }
+  if (mode == 0) {
+    // softmax over the row
+    float mx = fmaxf(fmaxf(v[0], v[1]), fmaxf(v[2], v[3]));
+#pragma unroll
+    for (int m = 32; m > 0; m >>= 1) mx = fmaxf(mx, __shfl_xor(mx, m));
+    float sum = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      s[j] = (v[j] == -INFINITY) ? 0.f : __expf(v[j] - mx);
+      sum += s[j];
+    }
+#pragma unroll
+    for (int m = 32; m > 0; m >>= 1) sum += __shfl_xor(sum, m);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      s[j] /= sum;
+      c[j] = (v[j] == -INFINITY) ? -INFINITY : s[j];
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      s[j] = 1.f / (1.f + __expf(-v[j]));
+      const int e = lane * 4 + j;
+      c[j] = (e < E) ? s[j] + (bias ? bias[e] : 0.f) : -INFINITY;
+    }
+    if (n_group > 1) {
+      // group score = sum of top-2 choice values in the group; keep the
+      // topk_group best groups, mask the rest
+      const int gsize = E / n_group;  // elements per group (>= 4 assumed)
+      float gs[8];
+      for (int g = 0; g < n_group; ++g) {
+        float m1 = -INFINITY, m2 = -INFINITY;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int e = lane * 4 + j;
+          const float cv = (e >= g * gsize && e < (g + 1) * gsize) ? c[j] : -INFINITY;
+          if (cv > m1) { m2 = m1; m1 = cv; }
+          else if (cv > m2) { m2 = cv; }
+        }
+        // wave-combine per-lane top2
+#pragma unroll
+        for (int m = 32; m > 0; m >>= 1) {
+          const float o1 = __shfl_xor(m1, m);
+          const float o2 = __shfl_xor(m2, m);
+          if (o1 > m1) { m2 = fmaxf(m1, o2); m1 = o1; }
+          else { m2 = fmaxf(m2, o1); }
+        }
+        gs[g] = m1 + (m2 == -INFINITY ? 0.f : m2);
+      }
+      // select topk_group groups (all lanes hold identical gs)
+      unsigned int keep = 0;
+      for (int t = 0; t < topk_group; ++t) {
+        int best = -1;
+        float bv = -INFINITY;
+        for (int g = 0; g < n_group; ++g)
+          if (!(keep >> g & 1) && gs[g] > bv) { bv = gs[g]; best = g; }
+        if (best >= 0) keep |= 1u << best;
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int e = lane * 4 + j;
+        if (e < E && !(keep >> (e / gsize) & 1)) c[j] = -INFINITY;
+      }
+    }
+  }
+  // iterative wave-wide top-k on c; weights come from s
+  float wsum = 0.f;
+  float wk[8];
+  int ik[8];
+  for (int t = 0; t < k; ++t) {
+    float mx = fmaxf(fmaxf(c[0], c[1]), fmaxf(c[2], c[3]));
+#pragma unroll
+    for (int m = 32; m > 0; m >>= 1) mx = fmaxf(mx, __shfl_xor(mx, m));
+    // lowest index attaining the max (torch.topk tie convention)
+    int my = INT_MAX;
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      if (c[j] == mx && lane * 4 + j < my) my = lane * 4 + j;
+#pragma unroll
+    for (int m = 32; m > 0; m >>= 1) my = min(my, __shfl_xor(my, m));
+    const int owner = my >> 2, slot = my & 3;
+    float sv;
+    switch (slot) {
+      case 0: sv = __shfl(s[0], owner); break;
+      case 1: sv = __shfl(s[1], owner); break;
+      case 2: sv = __shfl(s[2], owner); break;
+      default: sv = __shfl(s[3], owner); break;
+    }
+    ik[t] = my;
+    wk[t] = sv;
+    wsum += sv;
+    if (lane == owner) c[slot] = -INFINITY;
+  }
+  if (lane == 0) {
+    const float inv = (mode == 0 || norm_topk) ? 1.f / (wsum + 1e-20f) : 1.f;
+    for (int t = 0; t < k; ++t) {
+      idx[(size_t)row * k + t] = ik[t];
+      w[(size_t)row * k + t] = wk[t] * inv * routed_scale;
+    }
+  }
+}
+
 __global__ __launch_bounds__(256) void moe_build_kernel(
     const int* __restrict__ idx, int* __restrict__ gather_tok,
     int* __restrict__ inv_pos, int T, int k, int E, int C) {
@@ -1982,6 +2105,29 @@ torch::Tensor attn_prefill_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor
   return out;
 }
 
+// logits: [T, E] fp32 gate output -> (idx int32 [T,k], w fp32 [T,k]).
+std::vector<torch::Tensor> moe_route(torch::Tensor logits, c10::optional<torch::Tensor> bias,
+                                     int64_t k, int64_t mode, int64_t n_group,
+                                     int64_t topk_group, double routed_scale, bool norm_topk) {
+  CHK(logits.is_cuda() && logits.dtype() == torch::kFloat32 && logits.is_contiguous() && logits.dim() == 2);
+  const int T = logits.size(0), E = logits.size(1);
+  CHK(E <= 256 && k >= 1 && k <= 8 && n_group >= 1 && n_group <= 8);
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHK(bias->is_contiguous() && bias->dtype() == torch::kFloat32 && bias->numel() == E);
+    bptr = bias->data_ptr<float>();
+  }
+  auto iopts = torch::TensorOptions().dtype(torch::kInt32).device(logits.device());
+  auto fopts = torch::TensorOptions().dtype(torch::kFloat32).device(logits.device());
+  auto idx = torch::empty({(long)T, (long)k}, iopts);
+  auto w = torch::empty({(long)T, (long)k}, fopts);
+  hipLaunchKernelGGL(moe_route_kernel, dim3((T + 3) / 4), dim3(256), 0, cur_stream(),
+                     logits.data_ptr<float>(), bptr, idx.data_ptr<int>(), w.data_ptr<float>(),
+                     T, E, (int)k, (int)mode, (int)n_group, (int)topk_group,
+                     (float)routed_scale, norm_topk ? 1 : 0);
+  return {idx, w};
+}
+
 // idx: [T, k] int32 expert choices -> (gather_tok int32 [E*C], inv_pos
 // int32 [T, k]). One workgroup (counting sort in LDS).
 std::vector<torch::Tensor> moe_build(torch::Tensor idx, int64_t E, int64_t C) {
@@ -2572,6 +2718,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "causal GQA prefill flash attention on matrix cores (packed cache, hd=128)",
         py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("start_pos"),
         py::arg("scale") = 0.0, py::arg("softcap") = 0.0, py::arg("window") = 0);
+  m.def("moe_route", &moe_route,
+        "fused MoE router: softmax-topk (mode 0) / sigmoid+group-limited (mode 1)",
+        py::arg("logits"), py::arg("bias") = py::none(), py::arg("k") = 2,
+        py::arg("mode") = 0, py::arg("n_group") = 1, py::arg("topk_group") = 1,
+        py::arg("routed_scale") = 1.0, py::arg("norm_topk") = false);
   m.def("moe_build", &moe_build,
         "MoE decode routing: counting-sort token-expert pairs to padded per-expert slots");
   m.def("moe_combine", &moe_combine,
