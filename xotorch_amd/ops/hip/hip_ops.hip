@@ -387,6 +387,7 @@ __global__ __launch_bounds__(256) void attn_decode_partial(
 // (s = cap * tanh(s / cap), after scale, before masking); window > 0 is the
 // sliding-window mask (query at sl-1 sees positions [sl - window, sl)).
 // Both are wave-uniform runtime flags: the llama path (0, 0) takes no tanh.
+template <bool PREFETCH = false>
 __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, const int* __restrict__ seq_lens,
@@ -431,6 +432,20 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   const size_t vbase = (size_t)(b * KVH + kvh) * 8 * (T32 >> 5) * 512;
   const int col = lane & 15;
 
+  // PREFETCH: the next tile's 8 K fragments are loaded during the previous
+  // tile's PV phase (kbuf registers, +32 VGPR) so the score MFMAs never
+  // wait on HBM; measured A/B via XOT_ATTN_PREFETCH.
+  bf16x8 kbuf[8];
+  if (PREFETCH && c0 < c1) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const unsigned short* kt = kp + kbase + ((size_t)((c0 >> 4) + h)) * 2048 + (size_t)lane * 8;
+#pragma unroll
+      for (int c = 0; c < 4; ++c)
+        kbuf[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+    }
+  }
+
   for (int t = c0; t < c1; t += 32) {
     // ---- scores: two 16-position half-tiles ----
     floatx4 sc[2];
@@ -440,7 +455,8 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
       const unsigned short* kt = kp + kbase + ((size_t)((t >> 4) + h)) * 2048 + (size_t)lane * 8;
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
-        const bf16x8 kb = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+        const bf16x8 kb = PREFETCH ? kbuf[h * 4 + c]
+            : __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
         sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kb, sc[h], 0, 0, 0);
       }
     }
@@ -492,6 +508,16 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
       for (int r = 0; r < 4; ++r)
         plds[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = f2b(sc[h][r]);
     const bf16x8 pf = *reinterpret_cast<const bf16x8*>(plds + (lane & 15) * 48 + (lane >> 4) * 8);
+    if (PREFETCH && t + 32 < c1) {
+      // issue next tile's K stream now; it completes under the PV MFMAs
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const unsigned short* kt = kp + kbase + ((size_t)(((t + 32) >> 4) + h)) * 2048 + (size_t)lane * 8;
+#pragma unroll
+        for (int c = 0; c < 4; ++c)
+          kbuf[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+      }
+    }
     // ---- PV: out[16q][g*16..] += P x V ----
     const unsigned short* vt = vp + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
 #pragma unroll
@@ -2069,11 +2095,19 @@ torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor 
   const float scale = (scale_in > 0.0) ? (float)scale_in : 1.0f / sqrtf((float)hd);
   auto stream = cur_stream();
   const int waves = B * KVH * nsplit;
-  hipLaunchKernelGGL(attn_decode_mfma_kernel, dim3((waves + 3) / 4), dim3(256), 0, stream,
-                     (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
-                     (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
-                     ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,
-                     scale, q_stride, (float)softcap, (int)window);
+  const bool prefetch = getenv("XOT_ATTN_PREFETCH") != nullptr;
+  if (prefetch)
+    hipLaunchKernelGGL((attn_decode_mfma_kernel<true>), dim3((waves + 3) / 4), dim3(256), 0, stream,
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
+                       (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
+                       ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,
+                       scale, q_stride, (float)softcap, (int)window);
+  else
+    hipLaunchKernelGGL((attn_decode_mfma_kernel<false>), dim3((waves + 3) / 4), dim3(256), 0, stream,
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
+                       (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
+                       ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,
+                       scale, q_stride, (float)softcap, (int)window);
   hipLaunchKernelGGL((attn_decode_merge<128>), dim3(B * H), dim3(128), 0, stream,
                      ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
                      (unsigned short*)out.data_ptr(), nsplit);
