@@ -2095,7 +2095,9 @@ torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor 
   const float scale = (scale_in > 0.0) ? (float)scale_in : 1.0f / sqrtf((float)hd);
   auto stream = cur_stream();
   const int waves = B * KVH * nsplit;
-  const bool prefetch = getenv("XOT_ATTN_PREFETCH") != nullptr;
+  // default ON: measured -1.1 ms/step on 70B B=128 and -0.8 on 8B B=256
+  const char* pf_env = getenv("XOT_ATTN_PREFETCH");
+  const bool prefetch = (pf_env == nullptr) || (pf_env[0] != '0');
   if (prefetch)
     hipLaunchKernelGGL((attn_decode_mfma_kernel<true>), dim3((waves + 3) / 4), dim3(256), 0, stream,
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
