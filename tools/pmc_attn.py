@@ -1,6 +1,7 @@
 """PMC microbench target: the headline decode-attention shape (70B B=128)."""
 import sys
-sys.path.insert(0, ".")
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 import torch
 from xotorch_amd.ops import _hip_ops as hip
 

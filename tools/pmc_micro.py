@@ -1,5 +1,6 @@
 import sys
-sys.path.insert(0, ".")
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 import torch
 from xotorch_amd.ops import _hip_ops as hip
 from xotorch_amd.ops import pack_decode_weight
