@@ -451,6 +451,10 @@ class RingAPINode:
   async def process_prompt(self, shard, prompt: str, request_id=None, inference_state=None):
     request_id = request_id or str(uuid.uuid4())
     state = inference_state or {}
+    if shard is not None and shard.model_id != self.worker.model_id:
+      raise ValueError(
+        f"this ring serves {self.worker.model_id}; requested {shard.model_id} "
+        f"(restart `xot serve {shard.model_id}` to switch)")
     ids = self.tokenizer.encode(prompt) or [self.tokenizer.eos_token_id or 0]
     if len(ids) >= self.worker.max_seq:
       raise ValueError(f"prompt length {len(ids)} exceeds context {self.worker.max_seq}")
