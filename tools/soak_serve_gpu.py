@@ -17,18 +17,24 @@ def emit(rid, tok, fin, meta):
     remaining.discard(rid)
     if not remaining:
       done.set()
-N = 24
+import os
+N = int(os.getenv("SOAK_N", "24"))
 for i in range(N):
   rid = f"r{i}"
   remaining.add(rid)
   ids = [int(v) for v in rng.integers(0, 32000, int(rng.integers(4, 40)))]
   q.put(AdmitMsg(rid, torch.tensor([ids], dtype=torch.int64), int(rng.integers(3, 20)), 0.0))
+mem0 = torch.cuda.memory_allocated()
 t = threading.Thread(target=w.serve_forever, args=(q, emit), daemon=True)
 t0 = time.perf_counter()
 t.start()
-ok = done.wait(300)
+ok = done.wait(600)
 q.put(AdmitMsg("stop", None, 0, 0.0))
 t.join(timeout=30)
 assert ok, f"soak failed, remaining: {remaining}"
+mem1 = torch.cuda.memory_allocated()
+growth = (mem1 - mem0) / 2**20
+assert growth < 256, f"memory grew {growth:.0f} MiB over the soak (slot/KV leak?)"
 print(f"serve soak: {N} requests, {sum(len(v) for v in got.values())} tokens in "
-      f"{time.perf_counter()-t0:.1f}s, graphs={w._graph is not None}, ok")
+      f"{time.perf_counter()-t0:.1f}s, graphs={w._graph is not None}, "
+      f"mem growth {growth:.1f} MiB, ok")
