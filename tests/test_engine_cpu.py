@@ -128,7 +128,7 @@ def test_sessions_cleaned_after_finish_and_error():
     from xotorch_amd.engine.torch_engine import TorchEngine
     from xotorch_amd.models.registry import build_base_shard
     from xotorch_amd.orchestration.node import Node
-    eng = TorchEngine()
+    eng = TorchEngine(device="cpu", dtype=torch.float32)
     node = Node("sess-test", None, eng, None, max_generate_tokens=6)
     await node.start(0)
     try:
@@ -161,7 +161,7 @@ def test_engine_preserves_passthrough_state():
     from xotorch_amd.engine.torch_engine import TorchEngine
     from xotorch_amd.models.registry import build_full_shard
     import numpy as np
-    eng = TorchEngine()
+    eng = TorchEngine(device="cpu", dtype=torch.float32)
     shard = build_full_shard("dummy", "TorchEngine")
     toks = np.array([[1, 2, 3]], dtype=np.int64)
     state_in = {"max_tokens": 7, "traceparent": "00-abc-def-01"}

@@ -95,7 +95,7 @@ def test_gemma2_through_engine_and_node(monkeypatch):
     from xotorch_amd.engine.torch_engine import TorchEngine
     from xotorch_amd.models.registry import build_base_shard
     from xotorch_amd.orchestration.node import Node
-    eng = TorchEngine()
+    eng = TorchEngine(device="cpu", dtype=torch.float32)
     node = Node("g2-test", None, eng, None, max_generate_tokens=6)
     await node.start(0)
     try:

@@ -1,6 +1,7 @@
 """Node orchestration tests: single-node decode, two-node TCP ring (the
 reference's multi-node-without-cluster pattern — SURVEY.md §4), elastic
 topology, ring training over the wire."""
+import torch
 import asyncio
 import json
 
@@ -191,7 +192,7 @@ async def _ctx_exhaustion():
   from xotorch_amd.orchestration.node import Node
   from xotorch_amd.parallel.partitioning import RingMemoryWeightedPartitioningStrategy
 
-  eng = TorchEngine()
+  eng = TorchEngine(device="cpu", dtype=torch.float32)
   node = Node("ctx-test", None, eng, None, RingMemoryWeightedPartitioningStrategy(),
               max_generate_tokens=100000)
   await node.start(0)
@@ -227,7 +228,7 @@ async def _concurrent():
   from xotorch_amd.orchestration.node import Node
   from xotorch_amd.parallel.partitioning import RingMemoryWeightedPartitioningStrategy
 
-  eng = TorchEngine()
+  eng = TorchEngine(device="cpu", dtype=torch.float32)
   node = Node("conc-test", None, eng, None, RingMemoryWeightedPartitioningStrategy(),
               max_generate_tokens=24)
   await node.start(0)
@@ -266,7 +267,7 @@ def test_coordinate_save_roundtrip(tmp_path):
     from xotorch_amd.engine.torch_engine import TorchEngine
     from xotorch_amd.models.registry import build_base_shard
     from xotorch_amd.orchestration.node import Node
-    eng = TorchEngine()
+    eng = TorchEngine(device="cpu", dtype=torch.float32)
     node = Node("ckpt-test", None, eng, None)
     await node.start(0)
     try:
