@@ -837,3 +837,87 @@ def test_moe_route_deepseek_mode(hip, E, ng, tg, k, norm):
     assert set(a) == set(b), (t, a, b)
     for e in a:
       assert abs(a[e] - b[e]) < 1e-4, (t, e, a[e], b[e])
+
+
+# ---------------- fp8 KV cache (opt-in XOT_FP8_KV) ----------------
+
+def test_fp8_kv_append_and_decode(hip):
+  """fp8 packed-cache append (e4m3 + per-row scales) and the fp8 decode
+  attention vs the bf16 reference path."""
+  from xotorch_amd.ops import _hip_ops, torch_ref
+  torch.manual_seed(3)
+  B, S, H, KVH, hd, T = 2, 40, 8, 4, 128, 64
+  t32 = 64
+  qkv = bt(B, S, (H + 2 * KVH) * hd, seed=141, scale=0.5)
+  qkv8 = qkv.clone()
+  kc = torch.zeros(B, KVH, T, hd, dtype=torch.bfloat16, device="cuda")
+  vc = torch.zeros_like(kc)
+  kc8, vc8 = kc.clone(), vc.clone()
+  kp = torch.zeros(B, KVH, t32 // 16, 4, 64, 8, dtype=torch.bfloat16, device="cuda")
+  vp = torch.zeros(B, KVH, 8, t32 // 32, 64, 8, dtype=torch.bfloat16, device="cuda")
+  kp8 = torch.zeros(B, KVH, t32 // 16, 4, 64, 8, dtype=torch.uint8, device="cuda")
+  vp8 = torch.zeros(B, KVH, 8, t32 // 32, 64, 8, dtype=torch.uint8, device="cuda")
+  ksc = torch.ones(B, KVH, t32, dtype=torch.float32, device="cuda")
+  vsc = torch.ones_like(ksc)
+  from xotorch_amd.ops.torch_ref import rope_cos_sin
+  cos, sin = rope_cos_sin(hd, T, 10000.0, device="cuda")
+  pos = torch.arange(S, dtype=torch.int32, device="cuda")
+  _hip_ops.rope_qkv_append(qkv, cos, sin, pos, kc, vc, H, KVH, hd, kp, vp)
+  _hip_ops.rope_qkv_append(qkv8, cos, sin, pos, kc8, vc8, H, KVH, hd, kp8, vp8,
+                           k_scale=ksc, v_scale=vsc)
+  # plain caches identical; fp8 copies dequantize close to the bf16 rows
+  assert torch.equal(kc, kc8) and torch.equal(vc, vc8)
+  assert (ksc[:, :, :S] != 1.0).any() and (vsc[:, :, :S] != 1.0).any()
+  # decode: fp8 path vs bf16 path vs torch reference
+  q = qkv[:, -1:, : H * hd].view(B, 1, H, hd).contiguous()
+  sl = torch.full((B,), S, dtype=torch.int32, device="cuda")
+  out_bf = _hip_ops.attn_decode_mfma(q, kp, vp, sl, T).float()
+  out_f8 = _hip_ops.attn_decode_mfma(q, kp8, vp8, sl, T, k_scale=ksc, v_scale=vsc).float()
+  ref = torch_ref.attn_decode(q, kc, vc, S).float()
+  assert torch.allclose(out_bf, ref, atol=3e-2, rtol=3e-2)
+  # fp8 carries e4m3 quantization error (~6% per element): compare loosely
+  err = (out_f8 - ref).abs().max().item()
+  scale_ref = ref.abs().max().item()
+  assert err < 0.12 * max(scale_ref, 1.0) + 0.05, (err, scale_ref)
+
+
+@pytest.mark.gpu
+def test_fp8_kv_end_to_end_model(monkeypatch):
+  """XOT_FP8_KV=1 end to end on a tiny hd=128 llama: argmax agreement with
+  the bf16-cache run."""
+  monkeypatch.setenv("XOT_FP8_KV", "1")
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.llama import ShardedModel
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+  raw = dict(model_type="llama", vocab_size=512, hidden_size=256, intermediate_size=512,
+             num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+             head_dim=128, rms_norm_eps=1e-6, rope_theta=10000.0,
+             max_position_embeddings=128)
+  cfg = config_from_hf(raw, "fp8kv-tiny")
+  shard = Shard("fp8kv-tiny", 0, 1, 2)
+  torch.manual_seed(9)
+  m = ShardedModel(cfg, shard).to("cuda").to(torch.bfloat16)
+  random_init(m)
+  m.reset_rope()
+  m.eval()
+  B, S = 2, 24
+  toks = torch.randint(0, 512, (B, S), device="cuda")
+  c8 = ShardKVCache(2, B, 2, S + 8, 128, torch.bfloat16, "cuda")
+  assert c8.caches[0].kp.dtype == torch.uint8 and c8.caches[0].ksc is not None
+  monkeypatch.setenv("XOT_FP8_KV", "0")
+  cb = ShardKVCache(2, B, 2, S + 8, 128, torch.bfloat16, "cuda")
+  assert cb.caches[0].kp.dtype == torch.bfloat16
+  with torch.inference_mode():
+    pos = torch.arange(S, dtype=torch.int32, device="cuda")
+    l8 = m(toks, caches=c8.caches, positions=pos, start_pos=0)
+    lb = m(toks, caches=cb.caches, positions=pos, start_pos=0)
+    assert (l8.argmax(-1) == lb.argmax(-1)).all()  # prefill identical (plain cache)
+    nxt = lb.argmax(-1, keepdim=True)
+    sl = torch.full((B,), S + 1, dtype=torch.int32, device="cuda")
+    p1 = torch.tensor([S], dtype=torch.int32, device="cuda")
+    d8 = m(nxt, caches=c8.caches, positions=p1, start_pos=S, is_decode=True, seq_lens=sl)
+    db = m(nxt, caches=cb.caches, positions=p1, start_pos=S, is_decode=True, seq_lens=sl)
+    assert torch.allclose(d8.float(), db.float(), atol=0.5, rtol=0.1), \
+      (d8.float() - db.float()).abs().max()

@@ -24,8 +24,10 @@ import torch
 class LayerKV(NamedTuple):
   k: torch.Tensor
   v: torch.Tensor
-  kp: Optional[torch.Tensor] = None  # MFMA-packed K (cuda, hd=128 only)
+  kp: Optional[torch.Tensor] = None  # MFMA-packed K (cuda, hd=128; bf16 or e4m3 bytes)
   vp: Optional[torch.Tensor] = None  # MFMA-packed V
+  ksc: Optional[torch.Tensor] = None  # fp8 mode: per-row K scales [B, KVH, T32]
+  vsc: Optional[torch.Tensor] = None  # fp8 mode: per-row V scales
 
 
 def _want_packed(device: str, head_dim: int, dtype: torch.dtype) -> bool:
@@ -51,6 +53,7 @@ class ShardKVCache:
     self.caches: List[LayerKV] = []
     vd = v_dim if v_dim is not None else head_dim
     packed = _want_packed(device, head_dim, dtype) and vd == head_dim
+    fp8_kv = packed and os.getenv("XOT_FP8_KV", "0") == "1"
     # MLA latent cache (k = 512-dim latent, v = 64-dim roped shared key,
     # one kv "head"): fragment-packed copies for the absorbed-MQA MFMA
     # decode kernel (18 qk chunks / 32 pv groups — hip_ops.hip MLA section)
@@ -63,7 +66,15 @@ class ShardKVCache:
     for _ in range(n_layers):
       k = torch.zeros(batch, n_kv_heads, capacity, head_dim, dtype=dtype, device=device)
       v = torch.zeros(batch, n_kv_heads, capacity, vd, dtype=dtype, device=device)
-      if packed:
+      if fp8_kv:
+        # e4m3 packed copies (half the decode stream) + per-row scales;
+        # plain bf16 cache stays for prefill
+        kp = torch.zeros(batch, n_kv_heads, t32 // 16, 4, 64, 8, dtype=torch.uint8, device=device)
+        vp = torch.zeros(batch, n_kv_heads, 8, t32 // 32, 64, 8, dtype=torch.uint8, device=device)
+        ksc = torch.ones(batch, n_kv_heads, t32, dtype=torch.float32, device=device)
+        vsc = torch.ones(batch, n_kv_heads, t32, dtype=torch.float32, device=device)
+        self.caches.append(LayerKV(k, v, kp, vp, ksc, vsc))
+      elif packed:
         kp = torch.zeros(batch, n_kv_heads, t32 // 16, 4, 64, 8, dtype=dtype, device=device)
         vp = torch.zeros(batch, n_kv_heads, 8, t32 // 32, 64, 8, dtype=dtype, device=device)
         self.caches.append(LayerKV(k, v, kp, vp))

@@ -195,15 +195,17 @@ class Attention(nn.Module):
     k_cache, v_cache = kv[0], kv[1]
     kp = kv[2] if len(kv) > 2 else None  # MFMA-packed cache copies (GPU, hd=128)
     vp = kv[3] if len(kv) > 3 else None
+    ksc = kv[4] if len(kv) > 4 else None  # fp8-KV mode scales
+    vsc = kv[5] if len(kv) > 5 else None
     qkv = self.qkv_proj(x)  # [B, S, (H+2KVH)*hd]
     qn = self.q_norm if cfg.qk_norm else None
     kn = self.k_norm if cfg.qk_norm else None
     ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, cfg.n_kv_heads, hd, kp, vp,
-                        qn, kn, cfg.norm_eps)
+                        qn, kn, cfg.norm_eps, ksc, vsc)
     q = qkv[:, :, : H * hd].view(B, S, H, hd)  # strided view; kernels accept it
     if is_decode:
       sl = seq_lens if seq_lens is not None else start_pos + 1
-      out = ops.attn_decode(q, k_cache, v_cache, sl, kp, vp)
+      out = ops.attn_decode(q, k_cache, v_cache, sl, kp, vp, k_scale=ksc, v_scale=vsc)
     else:
       out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S, kp, vp)
     return self.o_proj(out.reshape(B, S, H * hd))

@@ -82,7 +82,7 @@ def rope_apply(q, k, cos, sin, positions):
 
 def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_kv_heads: int,
                     head_dim: int, kp=None, vp=None, q_norm=None, k_norm=None,
-                    norm_eps: float = 1e-6):
+                    norm_eps: float = 1e-6, k_scale=None, v_scale=None):
   """Fused on the packed qkv GEMM output [B,S,(H+2KVH)*hd]: (optionally
   per-head-RMSNorm q/k — qwen3), RoPE-rotate the q heads in place, rotate k
   heads into the cache, copy v heads into the cache. When the MFMA-packed
@@ -90,7 +90,7 @@ def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_
   """
   if _use_hip(qkv) and qkv.dtype == torch.bfloat16:
     _hip.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads, head_dim,
-                         kp, vp, q_norm, k_norm, norm_eps)
+                         kp, vp, q_norm, k_norm, norm_eps, k_scale, v_scale)
     return
   torch_ref.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads, n_kv_heads,
                             head_dim, q_norm, k_norm, norm_eps)
@@ -99,9 +99,11 @@ def rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, n_heads: int, n_
 def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int, kp=None, vp=None,
                  scale=None, softcap: float = 0.0, window: int = 0):
   if q.is_cuda and q.dtype == torch.bfloat16 and kp is not None and q.shape[3] == 128 \
+     and kp.dtype == torch.bfloat16 \
      and os.getenv("XOT_MFMA_ATTN", "1") == "1" and _use_hip(q):
     # causal flash-forward on matrix cores, streaming the MFMA-packed cache;
-    # softcap/window run gemma2 semantics inside the same kernel
+    # softcap/window run gemma2 semantics inside the same kernel (fp8-KV
+    # mode keeps prefill on the plain bf16 cache via the sdpa path below)
     return _hip.attn_prefill_mfma(q, kp, vp, start_pos, scale or 0.0, softcap, window)
   if q.is_cuda and not softcap and not window:
     # fallback: sdpa (rocm flash/mem-efficient backends) in model dtype with
@@ -122,7 +124,8 @@ def attn_prefill(q, k_cache, v_cache, start_pos: int, s_len: int, kp=None, vp=No
 
 
 def attn_decode(q, k_cache, v_cache, seq_len, kp=None, vp=None,
-                scale=None, softcap: float = 0.0, window: int = 0):
+                scale=None, softcap: float = 0.0, window: int = 0,
+                k_scale=None, v_scale=None):
   """seq_len: int or int32 device tensor [B] (per-sequence lengths).
   With MFMA-packed cache copies (kp, vp) the flash-decoding kernel scores on
   matrix cores (v_mfma_f32_16x16x32_bf16, coalesced 1 KB cache streams).
@@ -133,7 +136,7 @@ def attn_decode(q, k_cache, v_cache, seq_len, kp=None, vp=None,
     rep = q.shape[2] // k_cache.shape[1]
     if kp is not None and rep <= 16 and os.getenv("XOT_MFMA_ATTN", "1") == "1":
       return _hip.attn_decode_mfma(q, kp, vp, seq_len, k_cache.shape[2],
-                                   scale or 0.0, softcap, window)
+                                   scale or 0.0, softcap, window, k_scale, v_scale)
     if not softcap and not window and scale is None:
       return _hip.attn_decode(q, k_cache, v_cache, seq_len)
   return torch_ref.attn_decode(q, k_cache, v_cache, seq_len, scale, softcap, window)

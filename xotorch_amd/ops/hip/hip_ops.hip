@@ -134,6 +134,9 @@ __global__ __launch_bounds__(256) void rmsnorm_kernel(
 // Optional qwen3 per-head q/k RMSNorm (qn/kn weights [hd], fused before the
 // rotation): the slot's wave reduces sum-of-squares over the head row with
 // shfl_xor, then scales while rotating.
+// fp8 packed mode (kp8 non-null): the packed copies are OCP e4m3 bytes with
+// per-row scales sk/sv [B*KVH*T32] — half the decode stream and ~60% of the
+// dual-cache residency (plain cache stays bf16 for prefill). XOT_FP8_KV=1.
 __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
     unsigned short* __restrict__ qkv, const float* __restrict__ cosb,
     const float* __restrict__ sinb, const int* __restrict__ positions,
@@ -141,7 +144,9 @@ __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
     int B, int S, int H, int KVH, int hd, int T,
     unsigned short* __restrict__ kpc, unsigned short* __restrict__ vpc, int T32,
     const unsigned short* __restrict__ qn, const unsigned short* __restrict__ kn,
-    float norm_eps, int npos) {
+    float norm_eps, int npos,
+    unsigned char* __restrict__ kp8, unsigned char* __restrict__ vp8,
+    float* __restrict__ sk, float* __restrict__ sv) {
   const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   const int slots = H + 2 * KVH;
@@ -195,22 +200,39 @@ __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
         kx2 *= inv * b2f(kn[lane + hd2]);
       }
     }
+    float f1 = 0.f, f2v = 0.f;
+    {
+      const float c = (lane < hd2) ? cosb[(size_t)pos * hd2 + lane] : 0.f;
+      const float sn = (lane < hd2) ? sinb[(size_t)pos * hd2 + lane] : 0.f;
+      f1 = kx1 * c - kx2 * sn;
+      f2v = kx2 * c + kx1 * sn;
+    }
     if (lane < hd2) {
-      const float c = cosb[(size_t)pos * hd2 + lane];
-      const float sn = sinb[(size_t)pos * hd2 + lane];
-      const float x1 = kx1;
-      const float x2 = kx2;
-      const unsigned short r1 = f2b(x1 * c - x2 * sn);
-      const unsigned short r2 = f2b(x2 * c + x1 * sn);
-      dst[lane] = r1;
-      dst[lane + hd2] = r2;
+      dst[lane] = f2b(f1);
+      dst[lane + hd2] = f2b(f2v);
       if (kpc) {
         // packed K: element for hd-dim d at
         //   [(b*KVH+h)][pos>>4][d>>5][((d&31)>>3)*16 + (pos&15)][d&7]
         const size_t base = ((size_t)(b * KVH + h) * (T32 >> 4) + (pos >> 4)) * 2048;
         const int d1 = lane, d2 = lane + hd2;
-        kpc[base + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = r1;
-        kpc[base + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = r2;
+        kpc[base + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = f2b(f1);
+        kpc[base + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = f2b(f2v);
+      }
+    }
+    if (kp8) {
+      float mx = fmaxf(fabsf(f1), fabsf(f2v));
+#pragma unroll
+      for (int m = 32; m > 0; m >>= 1) mx = fmaxf(mx, __shfl_xor(mx, m));
+      const float scl = fmaxf(mx, 1e-12f) / 448.f;
+      const float inv = 1.f / scl;
+      if (lane == 0) sk[(size_t)(b * KVH + h) * T32 + pos] = scl;
+      if (lane < hd2) {
+        const size_t base = ((size_t)(b * KVH + h) * (T32 >> 4) + (pos >> 4)) * 2048;
+        const int d1 = lane, d2 = lane + hd2;
+        const unsigned short p1 = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(f1 * inv, 0.f, 0, false);
+        const unsigned short p2 = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(f2v * inv, 0.f, 0, false);
+        kp8[base + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = (unsigned char)(p1 & 0xff);
+        kp8[base + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = (unsigned char)(p2 & 0xff);
       }
     }
   } else {
@@ -227,6 +249,27 @@ __global__ __launch_bounds__(256) void rope_qkv_append_kernel(
       for (int e = 0; e < 2; ++e) {
         const int d = lane * 2 + e;
         vpc[((gbase + (d >> 4)) * (T32 >> 5) + tp) * 512 + (qt * 16 + (d & 15)) * 8 + j] = row[d];
+      }
+    }
+    if (vp8) {
+      float mx = 0.f;
+      if (lane * 2 < hd) {
+        mx = fmaxf(fabsf(b2f(row[lane * 2])), fabsf(b2f(row[lane * 2 + 1])));
+      }
+#pragma unroll
+      for (int m = 32; m > 0; m >>= 1) mx = fmaxf(mx, __shfl_xor(mx, m));
+      const float scl = fmaxf(mx, 1e-12f) / 448.f;
+      const float inv = 1.f / scl;
+      if (lane == 0) sv[(size_t)(b * KVH + h) * T32 + pos] = scl;
+      if (lane * 2 < hd) {
+        const int tp = pos >> 5, qt = (pos & 31) >> 3, j = pos & 7;
+        const size_t gbase = (size_t)(b * KVH + h) * 8;
+#pragma unroll
+        for (int e = 0; e < 2; ++e) {
+          const int d = lane * 2 + e;
+          const unsigned short pk = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(b2f(row[d]) * inv, 0.f, 0, false);
+          vp8[((gbase + (d >> 4)) * (T32 >> 5) + tp) * 512 + (qt * 16 + (d & 15)) * 8 + j] = (unsigned char)(pk & 0xff);
+        }
       }
     }
   }
@@ -387,13 +430,18 @@ __global__ __launch_bounds__(256) void attn_decode_partial(
 // (s = cap * tanh(s / cap), after scale, before masking); window > 0 is the
 // sliding-window mask (query at sl-1 sees positions [sl - window, sl)).
 // Both are wave-uniform runtime flags: the llama path (0, 0) takes no tanh.
-template <bool PREFETCH = false>
+// FP8: the packed cache copies are OCP e4m3 bytes with per-row scales
+// sk/sv (appended by rope_qkv_append's fp8 mode) — half the stream bytes.
+// q is quantized per head in-kernel; scores rescale by s_q[row]*s_k[col];
+// PV keeps a per-tile running scale R (s_v folded into the quantized P).
+template <bool PREFETCH = false, bool FP8 = false>
 __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, const int* __restrict__ seq_lens,
     float* __restrict__ ws_o, float* __restrict__ ws_ml,
     int B, int H, int KVH, int T32, int nsplit, float scale, long long q_stride,
-    float softcap, int window) {
+    float softcap, int window,
+    const float* __restrict__ sk = nullptr, const float* __restrict__ sv = nullptr) {
   const int wv = threadIdx.x >> 6;
   const int wid = blockIdx.x * 4 + wv;
   __shared__ unsigned short plds_all[4][16 * 48];  // 96 B row stride: bank-conflict-free b128 reads
@@ -420,6 +468,33 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     for (int c = 0; c < 4; ++c)
       qf[c] = *reinterpret_cast<const bf16x8*>(qrow + c * 32 + (lane >> 4) * 8);
   }
+  long qf8[4];
+  float srow[4];
+  if (FP8) {
+    float mx = 0.f;
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) mx = fmaxf(mx, fabsf((float)qf[c][j]));
+    mx = fmaxf(mx, __shfl_xor(mx, 16));
+    mx = fmaxf(mx, __shfl_xor(mx, 32));
+    const float sq = fmaxf(mx, 1e-12f) / 448.f;
+    const float inv = 1.f / sq;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      unsigned char o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j += 2) {
+        const unsigned short pk = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(
+            (float)qf[c][j] * inv, (float)qf[c][j + 1] * inv, 0, false);
+        o[j] = (unsigned char)(pk & 0xff);
+        o[j + 1] = (unsigned char)(pk >> 8);
+      }
+      qf8[c] = *reinterpret_cast<const long*>(o);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) srow[r] = __shfl(sq, (lane >> 4) * 4 + r);
+  }
 
   float m[4], lsum[4];
   floatx4 acco[8];
@@ -436,13 +511,25 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   // tile's PV phase (kbuf registers, +32 VGPR) so the score MFMAs never
   // wait on HBM; measured A/B via XOT_ATTN_PREFETCH.
   bf16x8 kbuf[8];
+  long kbuf8[8];
+  const unsigned char* kp8 = reinterpret_cast<const unsigned char*>(kp);
+  const unsigned char* vp8 = reinterpret_cast<const unsigned char*>(vp);
+  const size_t ssbase = (size_t)(b * KVH + kvh) * T32;
+  float Rscale = 1.f;
   if (PREFETCH && c0 < c1) {
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      const unsigned short* kt = kp + kbase + ((size_t)((c0 >> 4) + h)) * 2048 + (size_t)lane * 8;
+      if (FP8) {
+        const unsigned char* kt = kp8 + kbase + ((size_t)((c0 >> 4) + h)) * 2048 + (size_t)lane * 8;
 #pragma unroll
-      for (int c = 0; c < 4; ++c)
-        kbuf[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+        for (int c = 0; c < 4; ++c)
+          kbuf8[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const long*>(kt + c * 512));
+      } else {
+        const unsigned short* kt = kp + kbase + ((size_t)((c0 >> 4) + h)) * 2048 + (size_t)lane * 8;
+#pragma unroll
+        for (int c = 0; c < 4; ++c)
+          kbuf[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+      }
     }
   }
 
@@ -452,12 +539,22 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       sc[h] = (floatx4)(0.f);
-      const unsigned short* kt = kp + kbase + ((size_t)((t >> 4) + h)) * 2048 + (size_t)lane * 8;
+      if (FP8) {
+        const unsigned char* kt = kp8 + kbase + ((size_t)((t >> 4) + h)) * 2048 + (size_t)lane * 8;
 #pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        const bf16x8 kb = PREFETCH ? kbuf[h * 4 + c]
-            : __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
-        sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kb, sc[h], 0, 0, 0);
+        for (int c = 0; c < 4; ++c) {
+          const long kb = PREFETCH ? kbuf8[h * 4 + c]
+              : __builtin_nontemporal_load(reinterpret_cast<const long*>(kt + c * 512));
+          sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(qf8[c], kb, sc[h], 0, 0, 0);
+        }
+      } else {
+        const unsigned short* kt = kp + kbase + ((size_t)((t >> 4) + h)) * 2048 + (size_t)lane * 8;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const bf16x8 kb = PREFETCH ? kbuf[h * 4 + c]
+              : __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+          sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kb, sc[h], 0, 0, 0);
+        }
       }
     }
     // ---- online softmax (rows r are this lane's 4 query heads) ----
@@ -466,9 +563,11 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     for (int h = 0; h < 2; ++h) {
       const int pos = t + h * 16 + col;
       const bool ok = pos < c1 && pos >= win_lo;
+      const float skc = FP8 ? sk[ssbase + min(pos, T32 - 1)] : 1.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float s = sc[h][r] * scale;
+        if (FP8) s *= srow[r] * skc;
         if (softcap > 0.f) s = softcap * tanhf(s * (1.f / softcap));
         sc[h][r] = ok ? s : -INFINITY;
       }
@@ -497,34 +596,80 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
       for (int r = 0; r < 4; ++r) rsum[r] += __shfl_xor(rsum[r], mm);
 #pragma unroll
     for (int r = 0; r < 4; ++r) lsum[r] = lsum[r] * alpha[r] + rsum[r];
+    float s_pv = 1.f, rfac = 1.f;
+    if (FP8) {
+      // per-tile V scale ceiling; acco is kept in units of 1/R so quantized
+      // P' = p * s_v[pos] / s_pv enters at full e4m3 range
+      float mv = sv[ssbase + min(t + (lane & 31), T32 - 1)];
+#pragma unroll
+      for (int m_ = 32; m_ > 0; m_ >>= 1) mv = fmaxf(mv, __shfl_xor(mv, m_));
+      s_pv = fmaxf(mv, 1e-12f);
+      rfac = Rscale / s_pv;
+      Rscale = s_pv;
+    }
 #pragma unroll
     for (int g = 0; g < 8; ++g)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) acco[g][r] *= alpha[r];
-    // ---- P -> LDS (bf16, 96 B row stride) and back as A fragments ----
+      for (int r = 0; r < 4; ++r) acco[g][r] *= alpha[r] * (FP8 ? rfac : 1.f);
+    // ---- P -> LDS (bf16 / fp8 bytes, conflict-padded) and back as A fragments ----
+    long pf8 = 0;
+    bf16x8 pf;
+    if (FP8) {
+      unsigned char* pl8 = reinterpret_cast<unsigned char*>(plds);
 #pragma unroll
-    for (int h = 0; h < 2; ++h)
+      for (int h = 0; h < 2; ++h) {
+        const int pos = t + h * 16 + col;
+        const float svv = sv[ssbase + min(pos, T32 - 1)];
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        plds[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = f2b(sc[h][r]);
-    const bf16x8 pf = *reinterpret_cast<const bf16x8*>(plds + (lane & 15) * 48 + (lane >> 4) * 8);
+        for (int r = 0; r < 4; ++r) {
+          const unsigned short pk = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(
+              sc[h][r] * svv / s_pv, 0.f, 0, false);
+          pl8[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = (unsigned char)(pk & 0xff);
+        }
+      }
+      pf8 = *reinterpret_cast<const long*>(pl8 + (lane & 15) * 48 + (lane >> 4) * 8);
+    } else {
+#pragma unroll
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          plds[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = f2b(sc[h][r]);
+      pf = *reinterpret_cast<const bf16x8*>(plds + (lane & 15) * 48 + (lane >> 4) * 8);
+    }
     if (PREFETCH && t + 32 < c1) {
       // issue next tile's K stream now; it completes under the PV MFMAs
 #pragma unroll
       for (int h = 0; h < 2; ++h) {
-        const unsigned short* kt = kp + kbase + ((size_t)(((t + 32) >> 4) + h)) * 2048 + (size_t)lane * 8;
+        if (FP8) {
+          const unsigned char* kt = kp8 + kbase + ((size_t)(((t + 32) >> 4) + h)) * 2048 + (size_t)lane * 8;
 #pragma unroll
-        for (int c = 0; c < 4; ++c)
-          kbuf[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+          for (int c = 0; c < 4; ++c)
+            kbuf8[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const long*>(kt + c * 512));
+        } else {
+          const unsigned short* kt = kp + kbase + ((size_t)(((t + 32) >> 4) + h)) * 2048 + (size_t)lane * 8;
+#pragma unroll
+          for (int c = 0; c < 4; ++c)
+            kbuf[h * 4 + c] = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt + c * 512));
+        }
       }
     }
     // ---- PV: out[16q][g*16..] += P x V ----
-    const unsigned short* vt = vp + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
+    if (FP8) {
+      const unsigned char* vt = vp8 + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
 #pragma unroll
-    for (int g = 0; g < 8; ++g) {
-      const bf16x8 vb = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16x8*>(vt + (size_t)g * (T32 >> 5) * 512));
-      acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vb, acco[g], 0, 0, 0);
+      for (int g = 0; g < 8; ++g) {
+        const long vb = __builtin_nontemporal_load(
+            reinterpret_cast<const long*>(vt + (size_t)g * (T32 >> 5) * 512));
+        acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(pf8, vb, acco[g], 0, 0, 0);
+      }
+    } else {
+      const unsigned short* vt = vp + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
+#pragma unroll
+      for (int g = 0; g < 8; ++g) {
+        const bf16x8 vb = __builtin_nontemporal_load(
+            reinterpret_cast<const bf16x8*>(vt + (size_t)g * (T32 >> 5) * 512));
+        acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vb, acco[g], 0, 0, 0);
+      }
     }
   }
 
@@ -535,7 +680,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     if (qrow < NQ) {
       const size_t pidx = ((size_t)(b * H + qh_base + qrow) * nsplit + split);
 #pragma unroll
-      for (int g = 0; g < 8; ++g) ws_o[pidx * 128 + g * 16 + col] = acco[g][r];
+      for (int g = 0; g < 8; ++g) ws_o[pidx * 128 + g * 16 + col] = acco[g][r] * (FP8 ? Rscale : 1.f);
       if (col == 0) {
         ws_ml[pidx * 2 + 0] = m[r];
         ws_ml[pidx * 2 + 1] = lsum[r];
@@ -1969,7 +2114,8 @@ void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
                      int64_t n_heads, int64_t n_kv_heads, int64_t head_dim,
                      c10::optional<torch::Tensor> kp, c10::optional<torch::Tensor> vp,
                      c10::optional<torch::Tensor> q_norm, c10::optional<torch::Tensor> k_norm,
-                     double norm_eps) {
+                     double norm_eps,
+                     c10::optional<torch::Tensor> k_scale, c10::optional<torch::Tensor> v_scale) {
   CHK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
   CHK(kc.is_contiguous() && vc.is_contiguous());
   CHK(positions.dtype() == torch::kInt32 && positions.is_cuda());
@@ -1983,12 +2129,27 @@ void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
   CHK(positions.is_contiguous());
   unsigned short* kpc = nullptr;
   unsigned short* vpc = nullptr;
+  unsigned char* kp8 = nullptr;
+  unsigned char* vp8 = nullptr;
+  float* skp = nullptr;
+  float* svp = nullptr;
   int T32 = 0;
   if (kp.has_value() && vp.has_value()) {
     CHK(hd == 128 && kp->is_contiguous() && vp->is_contiguous());
-    kpc = (unsigned short*)kp->data_ptr();
-    vpc = (unsigned short*)vp->data_ptr();
     T32 = (int)kp->size(2) * 16;
+    if (kp->dtype() == torch::kUInt8) {
+      // fp8 packed mode: per-row e4m3 scales required
+      CHK(k_scale.has_value() && v_scale.has_value());
+      CHK(k_scale->dtype() == torch::kFloat32 && k_scale->is_contiguous());
+      CHK(v_scale->dtype() == torch::kFloat32 && v_scale->is_contiguous());
+      kp8 = (unsigned char*)kp->data_ptr();
+      vp8 = (unsigned char*)vp->data_ptr();
+      skp = k_scale->data_ptr<float>();
+      svp = v_scale->data_ptr<float>();
+    } else {
+      kpc = (unsigned short*)kp->data_ptr();
+      vpc = (unsigned short*)vp->data_ptr();
+    }
   }
   const unsigned short* qnp = nullptr;
   const unsigned short* knp = nullptr;
@@ -2006,7 +2167,7 @@ void rope_qkv_append(torch::Tensor qkv, torch::Tensor cos, torch::Tensor sin,
                      (unsigned short*)qkv.data_ptr(), cos.data_ptr<float>(), sin.data_ptr<float>(),
                      positions.data_ptr<int>(), (unsigned short*)kc.data_ptr(),
                      (unsigned short*)vc.data_ptr(), B, S, H, KVH, hd, T, kpc, vpc, T32,
-                     qnp, knp, (float)norm_eps, npos);
+                     qnp, knp, (float)norm_eps, npos, kp8, vp8, skp, svp);
 }
 
 template <int HD>
@@ -2073,11 +2234,21 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc, t
 // q: [B, 1, H, 128] (strided batch OK); kp/vp: the packed cache copies.
 torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor vp,
                                torch::Tensor seq_lens, int64_t t_capacity,
-                               double scale_in, double softcap, int64_t window) {
+                               double scale_in, double softcap, int64_t window,
+                               c10::optional<torch::Tensor> k_scale,
+                               c10::optional<torch::Tensor> v_scale) {
   CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   CHK(q.stride(3) == 1 && q.stride(2) == q.size(3));
   CHK(seq_lens.dtype() == torch::kInt32 && seq_lens.is_cuda());
   CHK(kp.is_contiguous() && vp.is_contiguous());
+  const bool fp8 = kp.dtype() == torch::kUInt8;
+  const float* skp = nullptr;
+  const float* svp = nullptr;
+  if (fp8) {
+    CHK(k_scale.has_value() && v_scale.has_value());
+    skp = k_scale->data_ptr<float>();
+    svp = v_scale->data_ptr<float>();
+  }
   const int B = q.size(0), H = q.size(2), hd = q.size(3);
   TORCH_CHECK(hd == 128, "attn_decode_mfma requires head_dim 128");
   const int KVH = kp.size(1);
@@ -2098,18 +2269,12 @@ torch::Tensor attn_decode_mfma(torch::Tensor q, torch::Tensor kp, torch::Tensor 
   // default ON: measured -1.1 ms/step on 70B B=128 and -0.8 on 8B B=256
   const char* pf_env = getenv("XOT_ATTN_PREFETCH");
   const bool prefetch = (pf_env == nullptr) || (pf_env[0] != '0');
-  if (prefetch)
-    hipLaunchKernelGGL((attn_decode_mfma_kernel<true>), dim3((waves + 3) / 4), dim3(256), 0, stream,
-                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
-                       (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
-                       ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,
-                       scale, q_stride, (float)softcap, (int)window);
-  else
-    hipLaunchKernelGGL((attn_decode_mfma_kernel<false>), dim3((waves + 3) / 4), dim3(256), 0, stream,
-                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
-                       (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
-                       ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,
-                       scale, q_stride, (float)softcap, (int)window);
+#define ADM_LAUNCH(PF, F8)   hipLaunchKernelGGL((attn_decode_mfma_kernel<PF, F8>), dim3((waves + 3) / 4), dim3(256), 0, stream,                      (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),                      (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),                      ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, KVH, T32, nsplit,                      scale, q_stride, (float)softcap, (int)window, skp, svp)
+  if (prefetch && fp8) ADM_LAUNCH(true, true);
+  else if (prefetch) ADM_LAUNCH(true, false);
+  else if (fp8) ADM_LAUNCH(false, true);
+  else ADM_LAUNCH(false, false);
+#undef ADM_LAUNCH
   hipLaunchKernelGGL((attn_decode_merge<128>), dim3(B * H), dim3(128), 0, stream,
                      ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
                      (unsigned short*)out.data_ptr(), nsplit);
@@ -2743,13 +2908,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("vc"), py::arg("n_heads"), py::arg("n_kv_heads"), py::arg("head_dim"),
         py::arg("kp") = py::none(), py::arg("vp") = py::none(),
         py::arg("q_norm") = py::none(), py::arg("k_norm") = py::none(),
-        py::arg("norm_eps") = 1e-6);
+        py::arg("norm_eps") = 1e-6,
+        py::arg("k_scale") = py::none(), py::arg("v_scale") = py::none());
   m.def("attn_decode", &attn_decode, "GQA decode attention (flash-decoding split-KV)");
   m.def("attn_decode_mfma", &attn_decode_mfma,
         "GQA decode attention on matrix cores (packed cache, hd=128); "
         "softcap/window: gemma2 logit soft-capping and sliding window",
         py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("seq_lens"), py::arg("t_capacity"),
-        py::arg("scale") = 0.0, py::arg("softcap") = 0.0, py::arg("window") = 0);
+        py::arg("scale") = 0.0, py::arg("softcap") = 0.0, py::arg("window") = 0,
+        py::arg("k_scale") = py::none(), py::arg("v_scale") = py::none());
   m.def("attn_prefill_mfma", &attn_prefill_mfma,
         "causal GQA prefill flash attention on matrix cores (packed cache, hd=128)",
         py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("start_pos"),
