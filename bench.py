@@ -143,7 +143,9 @@ def main():
       "higher_is_better": True,
       "scaling": "weak",
       "vs_baseline": None,
-      "dtype": (args.dtype + "+fp8-w8a8" if os.getenv("XOT_FP8_GEMM", "0") == "1" else args.dtype),
+      "dtype": (args.dtype
+                + ("+fp8-w8a8" if os.getenv("XOT_FP8_GEMM", "0") == "1" else "")
+                + ("+fp8-kv" if os.getenv("XOT_FP8_KV", "0") == "1" else "")),
       "data": "synthetic",
       "config": {
         "model": args.model,
