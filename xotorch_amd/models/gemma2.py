@@ -80,12 +80,16 @@ class Gemma2Attention(nn.Module):
     k_cache, v_cache = kv[0], kv[1]
     if _use_mfma(qkv, hd, kv):
       kp, vp = kv[2], kv[3]
-      ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, KVH, hd, kp, vp)
+      ksc = kv[4] if len(kv) > 4 else None
+      vsc = kv[5] if len(kv) > 5 else None
+      ops.rope_qkv_append(qkv, cos, sin, positions, k_cache, v_cache, H, KVH, hd, kp, vp,
+                          k_scale=ksc, v_scale=vsc)
       q = qkv[:, :, : H * hd].view(B, S, H, hd)
       if is_decode:
         sl = seq_lens if seq_lens is not None else start_pos + 1
         out = ops.attn_decode(q, k_cache, v_cache, sl, kp, vp,
-                              scale=self.scale, softcap=cap, window=self.window)
+                              scale=self.scale, softcap=cap, window=self.window,
+                              k_scale=ksc, v_scale=vsc)
       else:
         out = ops.attn_prefill(q, k_cache, v_cache, start_pos, S, kp, vp,
                                scale=self.scale, softcap=cap, window=self.window)
