@@ -921,3 +921,25 @@ def test_fp8_kv_end_to_end_model(monkeypatch):
     db = m(nxt, caches=cb.caches, positions=p1, start_pos=S, is_decode=True, seq_lens=sl)
     assert torch.allclose(d8.float(), db.float(), atol=0.5, rtol=0.1), \
       (d8.float() - db.float()).abs().max()
+
+
+def test_fp8_prefill_scaled_mm(monkeypatch):
+  """XOT_FP8_PREFILL: XotLinear prefill GEMMs via e4m3 scaled_mm within
+  quantization tolerance of the bf16 path."""
+  from xotorch_amd.models.llama import XotLinear
+  torch.manual_seed(12)
+  lin = XotLinear(512, 1024, bias=True).to("cuda").to(torch.bfloat16)
+  x = (torch.randn(2, 300, 512, device="cuda") * 0.3).to(torch.bfloat16)
+  with torch.inference_mode():
+    ref = lin(x).float()
+    monkeypatch.setenv("XOT_FP8_PREFILL", "1")
+    out = lin(x).float()
+  rel = (out - ref).abs().max().item() / max(ref.abs().max().item(), 1.0)
+  assert rel < 0.08, rel
+  # decode shapes (M <= 256) stay on the bf16 path
+  xd = (torch.randn(64, 512, device="cuda") * 0.3).to(torch.bfloat16)
+  with torch.inference_mode():
+    d1 = lin(xd)
+    monkeypatch.setenv("XOT_FP8_PREFILL", "0")
+    d0 = lin(xd)
+  assert torch.equal(d1, d0)
