@@ -94,15 +94,20 @@ class XotLinear(nn.Linear):
     fp8 weight copy is cached lazily."""
     N, K = self.weight.shape
     M = x.numel() // K
-    if self._w8 is None:
-      w = self.weight.detach().float()
-      sw = w.abs().amax(dim=1, keepdim=True).clamp(min=1e-12) / 448.0
-      self._w8 = (w / sw).clamp(-448, 448).to(torch.float8_e4m3fn)
-      self._w8_scale = sw
-    x2 = x.reshape(M, K).float()
-    sx = x2.abs().amax(dim=1, keepdim=True).clamp(min=1e-12) / 448.0
-    x8 = (x2 / sx).clamp(-448, 448).to(torch.float8_e4m3fn)
-    y = torch._scaled_mm(x8, self._w8.t(), scale_a=sx, scale_b=self._w8_scale.t(),
+    w8, sw = self._w8, self._w8_scale
+    if w8 is None:
+      w = self.weight.detach()
+      sw = (w.float().abs().amax(dim=1, keepdim=True).clamp(min=1e-12) / 448.0)
+      w8 = (w / sw.to(w.dtype)).clamp(-448, 448).to(torch.float8_e4m3fn)
+      free, _ = torch.cuda.mem_get_info()
+      if free > (12 << 30):  # persist the fp8 copy only with HBM headroom
+        self._w8, self._w8_scale = w8, sw
+    # activation quantization stays in bf16 (a fp32 image of a 65k x 8k
+    # prefill activation would be 2 GB per layer)
+    x2 = x.reshape(M, K)
+    sx = (x2.abs().amax(dim=1, keepdim=True).float().clamp(min=1e-12) / 448.0)
+    x8 = (x2 / sx.to(x2.dtype)).clamp(-448, 448).to(torch.float8_e4m3fn)
+    y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw.t(),
                          out_dtype=torch.bfloat16)
     if self.bias is not None:
       y = y + self.bias
@@ -114,7 +119,10 @@ class XotLinear(nn.Linear):
       N, K = self.weight.shape
       M = x.numel() // K
       if M > 256 and N % 16 == 0 and K % 16 == 0 and x.is_contiguous():
-        return self._fp8_prefill(x)
+        try:
+          return self._fp8_prefill(x)
+        except torch.cuda.OutOfMemoryError:
+          torch.cuda.empty_cache()  # fall back to the bf16 path
     wp8 = getattr(self, "weight_packed_fp8", None)
     if wp8 is not None and x.is_cuda and x.dtype == torch.bfloat16 and not torch.is_grad_enabled():
       N, K = self.weight.shape
