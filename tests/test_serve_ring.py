@@ -161,3 +161,19 @@ def test_cancel_releases_slot():
   t.join(timeout=30)
   assert len(got["long"]) < 240  # cancelled well before the max_seq-clamped max_new
   assert got["long"][-1][2].get("cancelled") or got["long"][-1][1]
+
+
+@pytest.mark.timeout(300)
+def test_three_rank_ring_matches_single(tmp_path):
+  """Odd-world serving ring (middle stage is neither first nor last)."""
+  import json
+  import torch.multiprocessing as mp
+  from xotorch_amd.helpers import find_available_port
+  rng = np.random.default_rng(17)
+  reqs = [("a", [int(v) for v in rng.integers(0, 200, 5)], 4),
+          ("b", [int(v) for v in rng.integers(0, 200, 9)], 4)]
+  port = find_available_port("127.0.0.1")
+  mp.spawn(_ring_worker, args=(3, port, str(tmp_path), reqs), nprocs=3, join=True)
+  got = json.loads((tmp_path / "tokens.json").read_text())
+  for rid, ids, max_new in reqs:
+    assert got[rid] == oracle_tokens(ids, max_new), rid

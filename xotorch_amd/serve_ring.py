@@ -289,6 +289,7 @@ class RingSlotWorker:
     """Run the ring loop. On rank 0, `admit_q` provides AdmitMsg and `emit`
     receives (request_id, token, is_finished, meta). Followers pass None."""
     self._cancelled = set()
+    idle_ticks = 0
     pending_release: List[int] = []
     inflight: Dict[int, AdmitMsg] = {}
     ticks = 0
@@ -327,11 +328,16 @@ class RingSlotWorker:
             op = OP_IDLE
       op, slot, plen, extra = self._bcast_hdr(op, slot, plen, extra)
 
+      if op != OP_IDLE:
+        idle_ticks = 0
       if op == OP_SHUTDOWN:
         break
       if op == OP_IDLE:
         if self.rank == 0:
-          time.sleep(idle_sleep)
+          # adaptive backoff: a long-idle ring stops burning control-plane
+          # broadcasts (each idle tick is an NCCL op on every rank)
+          idle_ticks += 1
+          time.sleep(min(idle_sleep * (1 + idle_ticks // 50), 0.05))
         continue
       if op == OP_RELEASE:
         self.active[slot] = False
