@@ -943,3 +943,26 @@ def test_fp8_prefill_scaled_mm(monkeypatch):
     monkeypatch.setenv("XOT_FP8_PREFILL", "0")
     d0 = lin(xd)
   assert torch.equal(d1, d0)
+
+
+def test_mla_q_prep(hip):
+  from xotorch_amd.ops import _hip_ops
+  from xotorch_amd.models.deepseek_v3 import _rope
+  from xotorch_amd.ops.torch_ref import rope_cos_sin
+  torch.manual_seed(13)
+  B, H, nope = 3, 8, 128
+  q = bt(B, 1, H, nope + 64, seed=151, scale=0.4)
+  q_lat = bt(B, 1, H, 512, seed=152, scale=0.4)
+  cos, sin = rope_cos_sin(64, 128, 10000.0, device="cuda")
+  for npos, pos in ((1, torch.tensor([37], dtype=torch.int32, device="cuda")),
+                    (B, torch.tensor([5, 90, 44], dtype=torch.int32, device="cuda"))):
+    for interleave in (False, True):
+      qf = _hip_ops.mla_q_prep(q, q_lat, cos, sin, pos, nope, interleave)
+      assert torch.equal(qf[:, :, :512], q_lat.view(B, H, 512))
+      cs = cos[pos.long()]
+      sn = sin[pos.long()]
+      if npos == B:
+        cs, sn = cs.view(B, 1, -1), sn.view(B, 1, -1)
+      ref = _rope(q[:, :, :, nope:], cs, sn, interleave)[:, 0]  # [B, H, 64]
+      assert torch.allclose(qf[:, :, 512:].float(), ref.float(), atol=2e-2), \
+        (npos, interleave, (qf[:, :, 512:].float() - ref.float()).abs().max())

@@ -94,14 +94,6 @@ class MLAttention(nn.Module):
 
     ckv = self.kv_a_proj_with_mqa(x)
 
-    pidx = positions.reshape(-1).long()
-    cs, sn = cos[pidx], sin[pidx]
-    if pidx.numel() == B * S and pidx.numel() != S:
-      # per-row positions (continuous-batching slots)
-      cs = cs.view(B, S, -1)
-      sn = sn.view(B, S, -1)
-    q_rot = _rope(q_rot, cs, sn, cfg.rope_interleave)
-
     kp = kv[2] if len(kv) > 2 else None
     hip = None
     if kp is not None and x.is_cuda and x.dtype == torch.bfloat16:
@@ -118,10 +110,17 @@ class MLAttention(nn.Module):
       if decode and not torch.is_grad_enabled():
         if seq_lens is None:
           seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
-        return self._decode_mfma(x, q_pass, q_rot, kv, seq_lens, hip)
+        return self._decode_mfma(x, q, q_pass, kv, seq_lens, hip, positions, cos, sin)
       if start_pos < 0:
         start_pos = int(positions.reshape(-1)[0])
+      q_rot = self._rope_q(q_rot, cos, sin, positions, B, S)
     else:
+      q_rot = self._rope_q(q_rot, cos, sin, positions, B, S)
+      pidx = positions.reshape(-1).long()
+      cs, sn = cos[pidx], sin[pidx]
+      if pidx.numel() == B * S and pidx.numel() != S:
+        cs = cs.view(B, S, -1)
+        sn = sn.view(B, S, -1)
       kv_nope, k_rot = ckv[..., : cfg.kv_lora_rank], ckv[..., cfg.kv_lora_rank:]
       kv_nope = _rms(kv_nope, self.kv_a_layernorm, cfg.norm_eps)
       k_rot = _rope(k_rot.view(B, S, 1, rope_d), cs, sn, cfg.rope_interleave)
@@ -162,10 +161,19 @@ class MLAttention(nn.Module):
     return self.o_proj(out.reshape(B, S, H * vd))
 
 
-  def _decode_mfma(self, x, q_pass, q_rot, kv, seq_lens, hip):
+  def _rope_q(self, q_rot, cos, sin, positions, B, S):
+    pidx = positions.reshape(-1).long()
+    cs, sn = cos[pidx], sin[pidx]
+    if pidx.numel() == B * S and pidx.numel() != S:
+      cs = cs.view(B, S, -1)
+      sn = sn.view(B, S, -1)
+    return _rope(q_rot, cs, sn, self.cfg.rope_interleave)
+
+  def _decode_mfma(self, x, q_raw, q_pass, kv, seq_lens, hip, positions, cos, sin):
     """Absorbed-latent MFMA decode: kv_b is folded into q and out, so
     attention runs as MQA over the packed 1152 B/token latent stream
-    (hip_ops.hip attn_decode_mla)."""
+    (hip_ops.hip attn_decode_mla); the q tail rope + assembly is one
+    kernel (mla_q_prep)."""
     cfg = self.cfg
     B, S, H = q_pass.shape[0], q_pass.shape[1], cfg.n_heads
     nope, vd, lat = cfg.qk_nope_head_dim, cfg.v_head_dim, cfg.kv_lora_rank
@@ -174,8 +182,10 @@ class MLAttention(nn.Module):
       self._w_k = W[:, :nope, :].contiguous()   # [H, nope, lat]
       self._w_v = W[:, nope:, :].contiguous()   # [H, vd, lat]
     q_lat = torch.einsum("bshn,hnl->bshl", q_pass.to(x.dtype), self._w_k)
-    qfull = torch.cat([q_lat, q_rot.to(x.dtype)], dim=-1).reshape(B, H, lat + cfg.qk_rope_head_dim)
-    out_lat = hip.attn_decode_mla(qfull.contiguous(), kv[2], kv[3], seq_lens, self.scale)
+    qfull = hip.mla_q_prep(q_raw.contiguous(), q_lat.contiguous(), cos, sin,
+                           positions.to(torch.int32).contiguous(),
+                           nope, cfg.rope_interleave)
+    out_lat = hip.attn_decode_mla(qfull, kv[2], kv[3], seq_lens, self.scale)
     out = torch.einsum("bhl,hdl->bhd", out_lat, self._w_v).to(x.dtype)
     return self.o_proj(out.reshape(B, 1, H * vd))
 

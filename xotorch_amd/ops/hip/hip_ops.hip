@@ -1207,6 +1207,41 @@ __global__ __launch_bounds__(256) void mla_prep_append_kernel(
   }
 }
 
+// q-side finisher: rope the 64-dim query tail and assemble the absorbed
+// query [B, H, 576] = [q_lat(512) | rope(q_rot)(64)] in one launch
+// (replaces ~6 torch ops per layer: rope mul/cat x2, cat, contiguous).
+// One wave per (b, h): q strided [B, S=1, H, 192], q_lat [B, 1, H, 512].
+__global__ __launch_bounds__(256) void mla_q_prep_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ q_lat,
+    const float* __restrict__ cosb, const float* __restrict__ sinb,
+    const int* __restrict__ positions, unsigned short* __restrict__ qfull,
+    int B, int H, long long q_bstride, int nope, int interleave, int npos) {
+  const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (wid >= B * H) return;
+  const int lane = threadIdx.x & 63;
+  const int h = wid % H;
+  const int b = wid / H;
+  const unsigned short* lsrc = q_lat + ((size_t)b * H + h) * MLA_LAT;
+  unsigned short* dst = qfull + ((size_t)b * H + h) * MLA_DQK;
+  // copy the 512 absorbed-latent dims (8 per lane)
+#pragma unroll
+  for (int j = 0; j < 8; ++j) dst[lane * 8 + j] = lsrc[lane * 8 + j];
+  // rope the 64-dim tail (lanes 0..31, one pair each)
+  if (lane < 32) {
+    const int pos = positions[(npos == B) ? b : 0];
+    const float c = cosb[(size_t)pos * 32 + lane];
+    const float sn = sinb[(size_t)pos * 32 + lane];
+    const unsigned short* qrot = q + (size_t)b * q_bstride + (size_t)h * (nope + MLA_ROPE) + nope;
+    int i1, i2;
+    if (interleave) { i1 = lane * 2; i2 = lane * 2 + 1; }
+    else            { i1 = lane;     i2 = lane + 32;    }
+    const float x1 = b2f(qrot[i1]);
+    const float x2 = b2f(qrot[i2]);
+    dst[MLA_LAT + lane] = f2b(x1 * c - x2 * sn);
+    dst[MLA_LAT + lane + 32] = f2b(x2 * c + x1 * sn);
+  }
+}
+
 __global__ __launch_bounds__(256) void mla_append_kernel(
     const unsigned short* __restrict__ lat, const unsigned short* __restrict__ rot,
     const int* __restrict__ positions, unsigned short* __restrict__ kp,
@@ -2363,6 +2398,31 @@ torch::Tensor moe_combine(torch::Tensor y, torch::Tensor inv_pos, torch::Tensor 
   return out;
 }
 
+// q: [B, 1, H, nope+64] bf16 contiguous (raw projections, nope part is the
+// un-roped passthrough); q_lat: [B, 1, H, 512] absorbed latent query;
+// returns qfull [B, H, 576] with the roped 64-dim tail appended.
+torch::Tensor mla_q_prep(torch::Tensor q, torch::Tensor q_lat, torch::Tensor cos,
+                         torch::Tensor sin, torch::Tensor positions,
+                         int64_t nope, bool interleave) {
+  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  CHK(q_lat.is_cuda() && q_lat.dtype() == torch::kBFloat16 && q_lat.is_contiguous());
+  CHK(positions.dtype() == torch::kInt32 && positions.is_contiguous());
+  const int B = q.size(0), H = q.size(2);
+  CHK(q.size(1) == 1 && q.size(3) == nope + MLA_ROPE);
+  CHK(q_lat.numel() == (long long)B * H * MLA_LAT);
+  const int npos = (int)positions.numel();
+  CHK(npos == 1 || npos == B);
+  auto qfull = torch::empty({(long)B, (long)H, (long)MLA_DQK},
+                            torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
+  const int waves = B * H;
+  hipLaunchKernelGGL(mla_q_prep_kernel, dim3((waves + 3) / 4), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(), (const unsigned short*)q_lat.data_ptr(),
+                     cos.data_ptr<float>(), sin.data_ptr<float>(), positions.data_ptr<int>(),
+                     (unsigned short*)qfull.data_ptr(), B, H, (long long)H * (nope + MLA_ROPE),
+                     (int)nope, interleave ? 1 : 0, npos);
+  return qfull;
+}
+
 // ckv: [B, S, 576] raw kv_a output; w: [512] bf16 norm weight; cos/sin
 // fp32 [maxT, 32]; lat_c/rot_c: the plain caches viewed [B, T, 512]/[B, T, 64].
 void mla_prep_append(torch::Tensor ckv, torch::Tensor w, torch::Tensor cos,
@@ -2930,6 +2990,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MoE decode routing: counting-sort token-expert pairs to padded per-expert slots");
   m.def("moe_combine", &moe_combine,
         "MoE decode combine: out[t] = sum_j w[t,j] * y[pos(t,j)] (deterministic)");
+  m.def("mla_q_prep", &mla_q_prep,
+        "MLA absorbed-query finisher: rope the 64-dim tail + assemble [B,H,576]");
   m.def("mla_prep_append", &mla_prep_append,
         "fused MLA kv prep: latent RMSNorm + shared-key rope + both cache layouts");
   m.def("mla_append", &mla_append,
