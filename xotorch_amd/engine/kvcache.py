@@ -78,6 +78,11 @@ class ShardKVCache:
         kp = torch.zeros(batch, n_kv_heads, t32 // 16, 4, 64, 8, dtype=dtype, device=device)
         vp = torch.zeros(batch, n_kv_heads, 8, t32 // 32, 64, 8, dtype=dtype, device=device)
         self.caches.append(LayerKV(k, v, kp, vp))
+      elif mla_packed and os.getenv("XOT_FP8_KV", "0") == "1":
+        kp = torch.zeros(batch, t32 // 16, 18, 64, 8, dtype=torch.uint8, device=device)
+        vp = torch.zeros(batch, 32, t32 // 32, 64, 8, dtype=torch.uint8, device=device)
+        ksc = torch.ones(batch, t32, dtype=torch.float32, device=device)
+        self.caches.append(LayerKV(k, v, kp, vp, ksc))
       elif mla_packed:
         kp = torch.zeros(batch, t32 // 16, 18, 64, 8, dtype=dtype, device=device)
         vp = torch.zeros(batch, 32, t32 // 32, 64, 8, dtype=dtype, device=device)

@@ -106,7 +106,8 @@ class MLAttention(nn.Module):
       # graph-capturable; replaces ~10 torch launches per layer)
       hip.mla_prep_append(ckv.contiguous(), self.kv_a_layernorm, cos, sin,
                           positions.to(torch.int32).contiguous(), kv[0], kv[1],
-                          kp, kv[3], cfg.norm_eps, cfg.rope_interleave)
+                          kp, kv[3], cfg.norm_eps, cfg.rope_interleave,
+                          kv[4] if len(kv) > 4 else None)
       if decode and not torch.is_grad_enabled():
         if seq_lens is None:
           seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=x.device)
@@ -182,10 +183,15 @@ class MLAttention(nn.Module):
       self._w_k = W[:, :nope, :].contiguous()   # [H, nope, lat]
       self._w_v = W[:, nope:, :].contiguous()   # [H, vd, lat]
     q_lat = torch.einsum("bshn,hnl->bshl", q_pass.to(x.dtype), self._w_k)
-    qfull = hip.mla_q_prep(q_raw.contiguous(), q_lat.contiguous(), cos, sin,
-                           positions.to(torch.int32).contiguous(),
-                           nope, cfg.rope_interleave)
-    out_lat = hip.attn_decode_mla(qfull, kv[2], kv[3], seq_lens, self.scale)
+    fp8 = kv[2].dtype == torch.uint8
+    prep = hip.mla_q_prep(q_raw.contiguous(), q_lat.contiguous(), cos, sin,
+                          positions.to(torch.int32).contiguous(),
+                          nope, cfg.rope_interleave, fp8)
+    if fp8:
+      out_lat = hip.attn_decode_mla(prep[0], kv[2], kv[3], seq_lens, self.scale,
+                                    prep[1], kv[4])
+    else:
+      out_lat = hip.attn_decode_mla(prep[0], kv[2], kv[3], seq_lens, self.scale)
     out = torch.einsum("bhl,hdl->bhd", out_lat, self._w_v).to(x.dtype)
     return self.o_proj(out.reshape(B, 1, H * vd))
 

@@ -1152,7 +1152,9 @@ __global__ __launch_bounds__(256) void mla_prep_append_kernel(
     const int* __restrict__ positions,
     unsigned short* __restrict__ lat_c, unsigned short* __restrict__ rot_c,
     unsigned short* __restrict__ kp, unsigned short* __restrict__ vp,
-    int B, int S, int T, int T32, float eps, int interleave, int npos) {
+    int B, int S, int T, int T32, float eps, int interleave, int npos,
+    unsigned char* __restrict__ kp8, unsigned char* __restrict__ vp8,
+    float* __restrict__ ks) {
   const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (wid >= B * S) return;
   const int lane = threadIdx.x & 63;
@@ -1171,23 +1173,11 @@ __global__ __launch_bounds__(256) void mla_prep_append_kernel(
 #pragma unroll
   for (int m = 32; m > 0; m >>= 1) acc += __shfl_xor(acc, m);
   const float inv = rsqrtf(acc / (float)MLA_LAT + eps);
-  unsigned short lat[8];
+  float lat[8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) lat[j] = f2b(v[j] * inv * b2f(w[lane * 8 + j]));
-  // plain latent cache + packed copies
-  unsigned short* lrow = lat_c + ((size_t)b * T + pos) * MLA_LAT;
-  const size_t kbase = ((size_t)b * (T32 >> 4) + (pos >> 4)) * MLA_CHQK * 512;
-  const size_t vbase = (size_t)b * MLA_GPV * (T32 >> 5) * 512;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const int d = lane * 8 + j;
-    lrow[d] = lat[j];
-    kp[kbase + (size_t)(d >> 5) * 512 + (((d & 31) >> 3) * 16 + (pos & 15)) * 8 + (d & 7)] = lat[j];
-    vp[vbase + ((size_t)(d >> 4) * (T32 >> 5) + (pos >> 5)) * 512
-       + (((pos & 31) >> 3) * 16 + (d & 15)) * 8 + (pos & 7)] = lat[j];
-  }
+  for (int j = 0; j < 8; ++j) lat[j] = v[j] * inv * b2f(w[lane * 8 + j]);
   // ---- rope the 64-dim shared key (lanes 0..31 handle one pair each) ----
-  unsigned short* rrow = rot_c + ((size_t)b * T + pos) * MLA_ROPE;
+  float r1 = 0.f, r2 = 0.f;
   if (lane < 32) {
     const float c = cosb[(size_t)pos * 32 + lane];
     const float sn = sinb[(size_t)pos * 32 + lane];
@@ -1197,13 +1187,58 @@ __global__ __launch_bounds__(256) void mla_prep_append_kernel(
     const float x1 = b2f(row[MLA_LAT + i1]);
     const float x2 = b2f(row[MLA_LAT + i2]);
     // output is cat([x1c - x2s, x2c + x1s]) in HALF layout (matches _rope)
-    const unsigned short r1 = f2b(x1 * c - x2 * sn);
-    const unsigned short r2 = f2b(x2 * c + x1 * sn);
-    rrow[lane] = r1;
-    rrow[lane + 32] = r2;
-    const int d1 = MLA_LAT + lane, d2 = MLA_LAT + lane + 32;
-    kp[kbase + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = r1;
-    kp[kbase + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = r2;
+    r1 = x1 * c - x2 * sn;
+    r2 = x2 * c + x1 * sn;
+  }
+  // plain caches (bf16, prefill reader) + bf16 packed copies
+  unsigned short* lrow = lat_c + ((size_t)b * T + pos) * MLA_LAT;
+  unsigned short* rrow = rot_c + ((size_t)b * T + pos) * MLA_ROPE;
+  const size_t kbase = ((size_t)b * (T32 >> 4) + (pos >> 4)) * MLA_CHQK * 512;
+  const size_t vbase = (size_t)b * MLA_GPV * (T32 >> 5) * 512;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int d = lane * 8 + j;
+    lrow[d] = f2b(lat[j]);
+    if (kp) {
+      kp[kbase + (size_t)(d >> 5) * 512 + (((d & 31) >> 3) * 16 + (pos & 15)) * 8 + (d & 7)] = f2b(lat[j]);
+      vp[vbase + ((size_t)(d >> 4) * (T32 >> 5) + (pos >> 5)) * 512
+         + (((pos & 31) >> 3) * 16 + (d & 15)) * 8 + (pos & 7)] = f2b(lat[j]);
+    }
+  }
+  if (lane < 32) {
+    rrow[lane] = f2b(r1);
+    rrow[lane + 32] = f2b(r2);
+    if (kp) {
+      const int d1 = MLA_LAT + lane, d2 = MLA_LAT + lane + 32;
+      kp[kbase + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = f2b(r1);
+      kp[kbase + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = f2b(r2);
+    }
+  }
+  if (kp8) {
+    // fp8 packed copies: one scale per token row over all 576 dims
+    float mx = fmaxf(fabsf(r1), fabsf(r2));
+#pragma unroll
+    for (int j = 0; j < 8; ++j) mx = fmaxf(mx, fabsf(lat[j]));
+#pragma unroll
+    for (int m = 32; m > 0; m >>= 1) mx = fmaxf(mx, __shfl_xor(mx, m));
+    const float scl = fmaxf(mx, 1e-12f) / 448.f;
+    const float qinv = 1.f / scl;
+    if (lane == 0) ks[(size_t)b * T32 + pos] = scl;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = lane * 8 + j;
+      const unsigned short pk = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(lat[j] * qinv, 0.f, 0, false);
+      kp8[kbase + (size_t)(d >> 5) * 512 + (((d & 31) >> 3) * 16 + (pos & 15)) * 8 + (d & 7)] = (unsigned char)(pk & 0xff);
+      vp8[vbase + ((size_t)(d >> 4) * (T32 >> 5) + (pos >> 5)) * 512
+          + (((pos & 31) >> 3) * 16 + (d & 15)) * 8 + (pos & 7)] = (unsigned char)(pk & 0xff);
+    }
+    if (lane < 32) {
+      const int d1 = MLA_LAT + lane, d2 = MLA_LAT + lane + 32;
+      const unsigned short p1 = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(r1 * qinv, 0.f, 0, false);
+      const unsigned short p2 = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(r2 * qinv, 0.f, 0, false);
+      kp8[kbase + (size_t)(d1 >> 5) * 512 + (((d1 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d1 & 7)] = (unsigned char)(p1 & 0xff);
+      kp8[kbase + (size_t)(d2 >> 5) * 512 + (((d2 & 31) >> 3) * 16 + (pos & 15)) * 8 + (d2 & 7)] = (unsigned char)(p2 & 0xff);
+    }
   }
 }
 
@@ -1215,18 +1250,18 @@ __global__ __launch_bounds__(256) void mla_q_prep_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ q_lat,
     const float* __restrict__ cosb, const float* __restrict__ sinb,
     const int* __restrict__ positions, unsigned short* __restrict__ qfull,
-    int B, int H, long long q_bstride, int nope, int interleave, int npos) {
+    int B, int H, long long q_bstride, int nope, int interleave, int npos,
+    unsigned char* __restrict__ q8, float* __restrict__ sq) {
   const int wid = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (wid >= B * H) return;
   const int lane = threadIdx.x & 63;
   const int h = wid % H;
   const int b = wid / H;
   const unsigned short* lsrc = q_lat + ((size_t)b * H + h) * MLA_LAT;
-  unsigned short* dst = qfull + ((size_t)b * H + h) * MLA_DQK;
-  // copy the 512 absorbed-latent dims (8 per lane)
+  float lv[8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) dst[lane * 8 + j] = lsrc[lane * 8 + j];
-  // rope the 64-dim tail (lanes 0..31, one pair each)
+  for (int j = 0; j < 8; ++j) lv[j] = b2f(lsrc[lane * 8 + j]);
+  float r1 = 0.f, r2 = 0.f;
   if (lane < 32) {
     const int pos = positions[(npos == B) ? b : 0];
     const float c = cosb[(size_t)pos * 32 + lane];
@@ -1237,8 +1272,39 @@ __global__ __launch_bounds__(256) void mla_q_prep_kernel(
     else            { i1 = lane;     i2 = lane + 32;    }
     const float x1 = b2f(qrot[i1]);
     const float x2 = b2f(qrot[i2]);
-    dst[MLA_LAT + lane] = f2b(x1 * c - x2 * sn);
-    dst[MLA_LAT + lane + 32] = f2b(x2 * c + x1 * sn);
+    r1 = x1 * c - x2 * sn;
+    r2 = x2 * c + x1 * sn;
+  }
+  if (q8 == nullptr) {
+    unsigned short* dst = qfull + ((size_t)b * H + h) * MLA_DQK;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dst[lane * 8 + j] = f2b(lv[j]);
+    if (lane < 32) {
+      dst[MLA_LAT + lane] = f2b(r1);
+      dst[MLA_LAT + lane + 32] = f2b(r2);
+    }
+    return;
+  }
+  // fp8 output: one e4m3 scale per (b, h) query row over all 576 dims
+  float mx = fmaxf(fabsf(r1), fabsf(r2));
+#pragma unroll
+  for (int j = 0; j < 8; ++j) mx = fmaxf(mx, fabsf(lv[j]));
+#pragma unroll
+  for (int m = 32; m > 0; m >>= 1) mx = fmaxf(mx, __shfl_xor(mx, m));
+  const float scl = fmaxf(mx, 1e-12f) / 448.f;
+  const float inv = 1.f / scl;
+  if (lane == 0) sq[(size_t)b * H + h] = scl;
+  unsigned char* dst8 = q8 + ((size_t)b * H + h) * MLA_DQK;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const unsigned short pk = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(lv[j] * inv, 0.f, 0, false);
+    dst8[lane * 8 + j] = (unsigned char)(pk & 0xff);
+  }
+  if (lane < 32) {
+    const unsigned short p1 = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(r1 * inv, 0.f, 0, false);
+    const unsigned short p2 = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(r2 * inv, 0.f, 0, false);
+    dst8[MLA_LAT + lane] = (unsigned char)(p1 & 0xff);
+    dst8[MLA_LAT + lane + 32] = (unsigned char)(p2 & 0xff);
   }
 }
 
@@ -1271,11 +1337,13 @@ __global__ __launch_bounds__(256) void mla_append_kernel(
   }
 }
 
+template <bool FP8 = false>
 __global__ __launch_bounds__(256) void attn_decode_mla_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ kp,
     const unsigned short* __restrict__ vp, const int* __restrict__ seq_lens,
     float* __restrict__ ws_o, float* __restrict__ ws_ml,
-    int B, int H, int T32, int nsplit, float scale) {
+    int B, int H, int T32, int nsplit, float scale,
+    const float* __restrict__ sq = nullptr, const float* __restrict__ ks = nullptr) {
   const int wv = threadIdx.x >> 6;
   const int wid = blockIdx.x * 4 + wv;
   const int hgroups = (H + 15) >> 4;
@@ -1294,6 +1362,16 @@ __global__ __launch_bounds__(256) void attn_decode_mla_kernel(
   const int nq = min(16, H - hg * 16);
   const int qi = min(lane & 15, nq - 1);
   const unsigned short* qrow = q + ((size_t)b * H + hg * 16 + qi) * MLA_DQK + (lane >> 4) * 8;
+  const unsigned char* qrow8 = reinterpret_cast<const unsigned char*>(q)
+      + ((size_t)b * H + hg * 16 + qi) * MLA_DQK + (lane >> 4) * 8;
+  const unsigned char* kp8 = reinterpret_cast<const unsigned char*>(kp);
+  const unsigned char* vp8 = reinterpret_cast<const unsigned char*>(vp);
+  float srow[4], Rscale = 1.f;
+  if (FP8) {
+    const float sqv = sq[(size_t)b * H + hg * 16 + qi];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) srow[r] = __shfl(sqv, (lane >> 4) * 4 + r);
+  }
 
   float m[4], lsum[4];
 #pragma unroll
@@ -1313,13 +1391,24 @@ __global__ __launch_bounds__(256) void attn_decode_mla_kernel(
     // scores over the 18 qk chunks; q fragments re-gathered per chunk
     // (L1-resident after the first tile), k streamed packed
     for (int c = 0; c < MLA_CHQK; ++c) {
-      const bf16x8 a = *reinterpret_cast<const bf16x8*>(qrow + c * 32);
+      if (FP8) {
+        const long a = *reinterpret_cast<const long*>(qrow8 + c * 32);
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        const unsigned short* kt = kp + kbase
-            + ((size_t)((t >> 4) + h) * MLA_CHQK + c) * 512 + (size_t)lane * 8;
-        const bf16x8 kb = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt));
-        sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, kb, sc[h], 0, 0, 0);
+        for (int h = 0; h < 2; ++h) {
+          const unsigned char* kt = kp8 + kbase
+              + ((size_t)((t >> 4) + h) * MLA_CHQK + c) * 512 + (size_t)lane * 8;
+          const long kb = __builtin_nontemporal_load(reinterpret_cast<const long*>(kt));
+          sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, kb, sc[h], 0, 0, 0);
+        }
+      } else {
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(qrow + c * 32);
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          const unsigned short* kt = kp + kbase
+              + ((size_t)((t >> 4) + h) * MLA_CHQK + c) * 512 + (size_t)lane * 8;
+          const bf16x8 kb = __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(kt));
+          sc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, kb, sc[h], 0, 0, 0);
+        }
       }
     }
     float rmax[4];
@@ -1327,8 +1416,13 @@ __global__ __launch_bounds__(256) void attn_decode_mla_kernel(
     for (int h = 0; h < 2; ++h) {
       const int pos = t + h * 16 + col;
       const bool ok = pos < c1;
+      const float skc = FP8 ? ks[(size_t)b * T32 + min(pos, T32 - 1)] : 1.f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) sc[h][r] = ok ? sc[h][r] * scale : -INFINITY;
+      for (int r = 0; r < 4; ++r) {
+        float sv_ = sc[h][r] * scale;
+        if (FP8) sv_ *= srow[r] * skc;
+        sc[h][r] = ok ? sv_ : -INFINITY;
+      }
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) rmax[r] = fmaxf(sc[0][r], sc[1][r]);
@@ -1354,21 +1448,54 @@ __global__ __launch_bounds__(256) void attn_decode_mla_kernel(
       for (int r = 0; r < 4; ++r) rsum[r] += __shfl_xor(rsum[r], mm);
 #pragma unroll
     for (int r = 0; r < 4; ++r) lsum[r] = lsum[r] * alpha[r] + rsum[r];
+    float s_pv = 1.f, rfac = 1.f;
+    if (FP8) {
+      float mv = ks[(size_t)b * T32 + min(t + (lane & 31), T32 - 1)];
+#pragma unroll
+      for (int m_ = 32; m_ > 0; m_ >>= 1) mv = fmaxf(mv, __shfl_xor(mv, m_));
+      s_pv = fmaxf(mv, 1e-12f);
+      rfac = Rscale / s_pv;
+      Rscale = s_pv;
+    }
 #pragma unroll
     for (int g = 0; g < MLA_GPV; ++g)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) acco[g][r] *= alpha[r];
+      for (int r = 0; r < 4; ++r) acco[g][r] *= alpha[r] * (FP8 ? rfac : 1.f);
+    long pf8 = 0;
+    bf16x8 pf;
+    if (FP8) {
+      unsigned char* pl8 = reinterpret_cast<unsigned char*>(plds);
 #pragma unroll
-    for (int h = 0; h < 2; ++h)
+      for (int h = 0; h < 2; ++h) {
+        const int pos = t + h * 16 + col;
+        const float svv = ks[(size_t)b * T32 + min(pos, T32 - 1)];
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        plds[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = f2b(sc[h][r]);
-    const bf16x8 pf = *reinterpret_cast<const bf16x8*>(plds + (lane & 15) * 48 + (lane >> 4) * 8);
-    const unsigned short* vt = vp + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
-    for (int g = 0; g < MLA_GPV; ++g) {
-      const bf16x8 vb = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16x8*>(vt + (size_t)g * (T32 >> 5) * 512));
-      acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vb, acco[g], 0, 0, 0);
+        for (int r = 0; r < 4; ++r) {
+          const unsigned short pk = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(
+              sc[h][r] * svv / s_pv, 0.f, 0, false);
+          pl8[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = (unsigned char)(pk & 0xff);
+        }
+      }
+      pf8 = *reinterpret_cast<const long*>(pl8 + (lane & 15) * 48 + (lane >> 4) * 8);
+      const unsigned char* vt8 = vp8 + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
+      for (int g = 0; g < MLA_GPV; ++g) {
+        const long vb = __builtin_nontemporal_load(
+            reinterpret_cast<const long*>(vt8 + (size_t)g * (T32 >> 5) * 512));
+        acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(pf8, vb, acco[g], 0, 0, 0);
+      }
+    } else {
+#pragma unroll
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          plds[((lane >> 4) * 4 + r) * 48 + h * 16 + col] = f2b(sc[h][r]);
+      pf = *reinterpret_cast<const bf16x8*>(plds + (lane & 15) * 48 + (lane >> 4) * 8);
+      const unsigned short* vt = vp + vbase + (size_t)(t >> 5) * 512 + (size_t)lane * 8;
+      for (int g = 0; g < MLA_GPV; ++g) {
+        const bf16x8 vb = __builtin_nontemporal_load(
+            reinterpret_cast<const bf16x8*>(vt + (size_t)g * (T32 >> 5) * 512));
+        acco[g] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vb, acco[g], 0, 0, 0);
+      }
     }
   }
 
@@ -1377,7 +1504,8 @@ __global__ __launch_bounds__(256) void attn_decode_mla_kernel(
     const int qrow_i = (lane >> 4) * 4 + r;
     if (qrow_i < nq) {
       const size_t pidx = ((size_t)(b * H + hg * 16 + qrow_i) * nsplit + split);
-      for (int g = 0; g < MLA_GPV; ++g) ws_o[pidx * MLA_LAT + g * 16 + col] = acco[g][r];
+      for (int g = 0; g < MLA_GPV; ++g)
+        ws_o[pidx * MLA_LAT + g * 16 + col] = acco[g][r] * (FP8 ? Rscale : 1.f);
       if (col == 0) {
         ws_ml[pidx * 2 + 0] = m[r];
         ws_ml[pidx * 2 + 1] = lsum[r];
@@ -2401,9 +2529,9 @@ torch::Tensor moe_combine(torch::Tensor y, torch::Tensor inv_pos, torch::Tensor 
 // q: [B, 1, H, nope+64] bf16 contiguous (raw projections, nope part is the
 // un-roped passthrough); q_lat: [B, 1, H, 512] absorbed latent query;
 // returns qfull [B, H, 576] with the roped 64-dim tail appended.
-torch::Tensor mla_q_prep(torch::Tensor q, torch::Tensor q_lat, torch::Tensor cos,
-                         torch::Tensor sin, torch::Tensor positions,
-                         int64_t nope, bool interleave) {
+std::vector<torch::Tensor> mla_q_prep(torch::Tensor q, torch::Tensor q_lat, torch::Tensor cos,
+                                      torch::Tensor sin, torch::Tensor positions,
+                                      int64_t nope, bool interleave, bool fp8) {
   CHK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
   CHK(q_lat.is_cuda() && q_lat.dtype() == torch::kBFloat16 && q_lat.is_contiguous());
   CHK(positions.dtype() == torch::kInt32 && positions.is_contiguous());
@@ -2412,15 +2540,28 @@ torch::Tensor mla_q_prep(torch::Tensor q, torch::Tensor q_lat, torch::Tensor cos
   CHK(q_lat.numel() == (long long)B * H * MLA_LAT);
   const int npos = (int)positions.numel();
   CHK(npos == 1 || npos == B);
+  const int waves = B * H;
+  if (fp8) {
+    auto q8 = torch::empty({(long)B, (long)H, (long)MLA_DQK},
+                           torch::TensorOptions().dtype(torch::kUInt8).device(q.device()));
+    auto sq = torch::empty({(long)B, (long)H},
+                           torch::TensorOptions().dtype(torch::kFloat32).device(q.device()));
+    hipLaunchKernelGGL(mla_q_prep_kernel, dim3((waves + 3) / 4), dim3(256), 0, cur_stream(),
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)q_lat.data_ptr(),
+                       cos.data_ptr<float>(), sin.data_ptr<float>(), positions.data_ptr<int>(),
+                       nullptr, B, H, (long long)H * (nope + MLA_ROPE),
+                       (int)nope, interleave ? 1 : 0, npos,
+                       (unsigned char*)q8.data_ptr(), sq.data_ptr<float>());
+    return {q8, sq};
+  }
   auto qfull = torch::empty({(long)B, (long)H, (long)MLA_DQK},
                             torch::TensorOptions().dtype(torch::kBFloat16).device(q.device()));
-  const int waves = B * H;
   hipLaunchKernelGGL(mla_q_prep_kernel, dim3((waves + 3) / 4), dim3(256), 0, cur_stream(),
                      (const unsigned short*)q.data_ptr(), (const unsigned short*)q_lat.data_ptr(),
                      cos.data_ptr<float>(), sin.data_ptr<float>(), positions.data_ptr<int>(),
                      (unsigned short*)qfull.data_ptr(), B, H, (long long)H * (nope + MLA_ROPE),
-                     (int)nope, interleave ? 1 : 0, npos);
-  return qfull;
+                     (int)nope, interleave ? 1 : 0, npos, nullptr, nullptr);
+  return {qfull};
 }
 
 // ckv: [B, S, 576] raw kv_a output; w: [512] bf16 norm weight; cos/sin
@@ -2428,7 +2569,8 @@ torch::Tensor mla_q_prep(torch::Tensor q, torch::Tensor q_lat, torch::Tensor cos
 void mla_prep_append(torch::Tensor ckv, torch::Tensor w, torch::Tensor cos,
                      torch::Tensor sin, torch::Tensor positions,
                      torch::Tensor lat_c, torch::Tensor rot_c,
-                     torch::Tensor kp, torch::Tensor vp, double eps, bool interleave) {
+                     torch::Tensor kp, torch::Tensor vp, double eps, bool interleave,
+                     c10::optional<torch::Tensor> k_scale) {
   CHK(ckv.is_cuda() && ckv.dtype() == torch::kBFloat16 && ckv.is_contiguous());
   CHK(w.dtype() == torch::kBFloat16 && w.is_contiguous() && w.numel() == MLA_LAT);
   CHK(positions.dtype() == torch::kInt32 && positions.is_contiguous());
@@ -2440,12 +2582,26 @@ void mla_prep_append(torch::Tensor ckv, torch::Tensor w, torch::Tensor cos,
   const int npos = (int)positions.numel();
   CHK(npos == S || npos == B * S);
   const int waves = B * S;
+  unsigned short* kpc = nullptr;
+  unsigned short* vpc = nullptr;
+  unsigned char* kp8 = nullptr;
+  unsigned char* vp8 = nullptr;
+  float* ksp = nullptr;
+  if (kp.dtype() == torch::kUInt8) {
+    CHK(k_scale.has_value() && k_scale->dtype() == torch::kFloat32 && k_scale->is_contiguous());
+    kp8 = (unsigned char*)kp.data_ptr();
+    vp8 = (unsigned char*)vp.data_ptr();
+    ksp = k_scale->data_ptr<float>();
+  } else {
+    kpc = (unsigned short*)kp.data_ptr();
+    vpc = (unsigned short*)vp.data_ptr();
+  }
   hipLaunchKernelGGL(mla_prep_append_kernel, dim3((waves + 3) / 4), dim3(256), 0, cur_stream(),
                      (const unsigned short*)ckv.data_ptr(), (const unsigned short*)w.data_ptr(),
                      cos.data_ptr<float>(), sin.data_ptr<float>(), positions.data_ptr<int>(),
                      (unsigned short*)lat_c.data_ptr(), (unsigned short*)rot_c.data_ptr(),
-                     (unsigned short*)kp.data_ptr(), (unsigned short*)vp.data_ptr(),
-                     B, S, T, T32, (float)eps, interleave ? 1 : 0, npos);
+                     kpc, vpc, B, S, T, T32, (float)eps, interleave ? 1 : 0, npos,
+                     kp8, vp8, ksp);
 }
 
 // lat: [B, S, 512] bf16 (post-RMSNorm latent), rot: [B, S, 64] bf16 (roped
@@ -2471,8 +2627,12 @@ void mla_append(torch::Tensor lat, torch::Tensor rot, torch::Tensor positions,
 // q: [B, H, 576] bf16 (absorbed latent+rope query); returns out_lat
 // [B, H, 512] bf16 (latent-space attention output, pre kv_b-v expansion).
 torch::Tensor attn_decode_mla(torch::Tensor q, torch::Tensor kp, torch::Tensor vp,
-                              torch::Tensor seq_lens, double scale_in) {
-  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+                              torch::Tensor seq_lens, double scale_in,
+                              c10::optional<torch::Tensor> q_scale,
+                              c10::optional<torch::Tensor> k_scale) {
+  const bool fp8 = kp.dtype() == torch::kUInt8;
+  CHK(q.is_cuda() && q.is_contiguous());
+  CHK(fp8 ? (q.dtype() == torch::kUInt8) : (q.dtype() == torch::kBFloat16));
   CHK(seq_lens.dtype() == torch::kInt32 && seq_lens.is_cuda());
   CHK(kp.is_contiguous() && vp.is_contiguous());
   const int B = q.size(0), H = q.size(1);
@@ -2493,10 +2653,24 @@ torch::Tensor attn_decode_mla(torch::Tensor q, torch::Tensor kp, torch::Tensor v
   const float scale = (scale_in > 0.0) ? (float)scale_in : 1.0f / sqrtf((float)MLA_DQK);
   auto stream = cur_stream();
   const int waves = B * hgroups * nsplit;
-  hipLaunchKernelGGL(attn_decode_mla_kernel, dim3((waves + 3) / 4), dim3(256), 0, stream,
-                     (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
-                     (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
-                     ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, T32, nsplit, scale);
+  const float* sqp = nullptr;
+  const float* ksp = nullptr;
+  if (fp8) {
+    CHK(q_scale.has_value() && k_scale.has_value());
+    sqp = q_scale->data_ptr<float>();
+    ksp = k_scale->data_ptr<float>();
+  }
+  if (fp8)
+    hipLaunchKernelGGL((attn_decode_mla_kernel<true>), dim3((waves + 3) / 4), dim3(256), 0, stream,
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
+                       (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
+                       ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, T32, nsplit, scale,
+                       sqp, ksp);
+  else
+    hipLaunchKernelGGL((attn_decode_mla_kernel<false>), dim3((waves + 3) / 4), dim3(256), 0, stream,
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kp.data_ptr(),
+                       (const unsigned short*)vp.data_ptr(), seq_lens.data_ptr<int>(),
+                       ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(), B, H, T32, nsplit, scale);
   hipLaunchKernelGGL(attn_decode_merge512, dim3(B * H), dim3(256), 0, stream,
                      ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
                      (unsigned short*)out.data_ptr(), nsplit);
@@ -2991,15 +3165,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_combine", &moe_combine,
         "MoE decode combine: out[t] = sum_j w[t,j] * y[pos(t,j)] (deterministic)");
   m.def("mla_q_prep", &mla_q_prep,
-        "MLA absorbed-query finisher: rope the 64-dim tail + assemble [B,H,576]");
+        "MLA absorbed-query finisher: rope the 64-dim tail + assemble [B,H,576]; "
+        "fp8=True returns (q8, per-head scales)",
+        py::arg("q"), py::arg("q_lat"), py::arg("cos"), py::arg("sin"), py::arg("positions"),
+        py::arg("nope"), py::arg("interleave") = false, py::arg("fp8") = false);
   m.def("mla_prep_append", &mla_prep_append,
-        "fused MLA kv prep: latent RMSNorm + shared-key rope + both cache layouts");
+        "fused MLA kv prep: latent RMSNorm + shared-key rope + both cache layouts",
+        py::arg("ckv"), py::arg("w"), py::arg("cos"), py::arg("sin"), py::arg("positions"),
+        py::arg("lat_c"), py::arg("rot_c"), py::arg("kp"), py::arg("vp"), py::arg("eps"),
+        py::arg("interleave") = false, py::arg("k_scale") = py::none());
   m.def("mla_append", &mla_append,
         "append MLA latent+rope token stream into the fragment-packed cache");
   m.def("attn_decode_mla", &attn_decode_mla,
-        "MLA decode attention (absorbed latent MQA) on matrix cores",
+        "MLA decode attention (absorbed latent MQA) on matrix cores; fp8 cache mode",
         py::arg("q"), py::arg("kp"), py::arg("vp"), py::arg("seq_lens"),
-        py::arg("scale") = 0.0);
+        py::arg("scale") = 0.0, py::arg("q_scale") = py::none(), py::arg("k_scale") = py::none());
   m.def("mfma16_probe", &mfma16_probe, "v_mfma_f32_16x16x32_bf16 layout probe (tests)");
   m.def("swiglu", &swiglu, "SwiGLU activation");
 }
