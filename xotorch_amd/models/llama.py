@@ -357,6 +357,12 @@ class MoEMLP(nn.Module):
     T = flat.shape[0]
     if x.is_cuda and not torch.is_grad_enabled() and T <= 256:
       return self._forward_decode(flat).view(B, S, D).to(x.dtype)
+    if x.is_cuda and not torch.is_grad_enabled() and self.wp_gate_up is not None:
+      # prefill: run the static grouped path over 256-token chunks — the
+      # per-expert where/index eager loop is launch-bound (measured: the
+      # top prefill cost on 128-expert qwen3-moe)
+      outs = [self._forward_decode(c) for c in flat.split(256)]
+      return torch.cat(outs, dim=0).view(B, S, D).to(x.dtype)
     weights, selected = self._route(flat)
     out = torch.zeros_like(flat, dtype=torch.float32)
     for e in range(self.n_experts):
