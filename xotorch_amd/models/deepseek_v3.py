@@ -295,11 +295,6 @@ class DsMoE(nn.Module):
         and 1 <= T <= 256):
       out = self._forward_decode(flat) + self.shared_experts(flat)
       return out.to(x.dtype).view(B, S, D)
-    if self.wp_gate_up is not None and x.is_cuda and not torch.is_grad_enabled():
-      # prefill via 256-token chunks of the static grouped path
-      outs = [self._forward_decode(c) for c in flat.split(256)]
-      out = torch.cat(outs, dim=0) + self.shared_experts(flat)
-      return out.to(x.dtype).view(B, S, D)
     if torch.cuda.is_available() and x.is_cuda and torch.cuda.is_current_stream_capturing():
       raise RuntimeError(
         "DsMoE dynamic routing is not graph-capturable: call "

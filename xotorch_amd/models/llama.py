@@ -357,12 +357,10 @@ class MoEMLP(nn.Module):
     T = flat.shape[0]
     if x.is_cuda and not torch.is_grad_enabled() and T <= 256:
       return self._forward_decode(flat).view(B, S, D).to(x.dtype)
-    if x.is_cuda and not torch.is_grad_enabled() and self.wp_gate_up is not None:
-      # prefill: run the static grouped path over 256-token chunks — the
-      # per-expert where/index eager loop is launch-bound (measured: the
-      # top prefill cost on 128-expert qwen3-moe)
-      outs = [self._forward_decode(c) for c in flat.split(256)]
-      return torch.cat(outs, dim=0).view(B, S, D).to(x.dtype)
+    # NOTE: chunking the static grouped path over 256-token slices for
+    # prefill was measured 1.5-2.5x WORSE TTFT (every chunk re-streams the
+    # full expert weights; the eager loop streams each expert once with a
+    # large-M GEMM) — keep the eager loop for T > 256.
     weights, selected = self._route(flat)
     out = torch.zeros_like(flat, dtype=torch.float32)
     for e in range(self.n_experts):
