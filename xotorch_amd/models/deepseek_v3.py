@@ -301,6 +301,28 @@ class DsMoE(nn.Module):
         "model.pack_decode_weights() first (grouped static path), or bench "
         "with --no-graphs")
     idx, w = self.route(flat)
+    if x.is_cuda and not torch.is_grad_enabled():
+      # prefill: sorted contiguous per-expert slices (one argsort + gather
+      # instead of E where/eq scans)
+      k = self.cfg.n_experts_per_tok
+      dev = flat.device
+      expert_of = idx.reshape(-1)
+      order = torch.argsort(expert_of)
+      tok_sorted = torch.arange(T, device=dev).repeat_interleave(k)[order]
+      w_sorted = w.reshape(-1)[order]
+      counts = torch.bincount(expert_of, minlength=self.cfg.n_experts).cpu().tolist()
+      xg = flat[tok_sorted]
+      ys = []
+      off = 0
+      for e, c in enumerate(counts):
+        if c:
+          ys.append(self.experts[e](xg[off:off + c]))
+        off += c
+      y = torch.cat(ys, dim=0).float() * w_sorted[:, None].float()
+      out = torch.zeros(T, D, dtype=torch.float32, device=dev)
+      out.index_add_(0, tok_sorted, y)
+      out = out.to(x.dtype) + self.shared_experts(flat)
+      return out.view(B, S, D)
     out = torch.zeros_like(flat, dtype=torch.float32)
     for e in range(self.cfg.n_experts):
       tok, kk = torch.where(idx == e)
