@@ -114,6 +114,12 @@ class MLAttention(nn.Module):
         return self._decode_mfma(x, q, q_pass, kv, seq_lens, hip, positions, cos, sin)
       if start_pos < 0:
         start_pos = int(positions.reshape(-1)[0])
+      if not torch.is_grad_enabled():
+        # GPU prefill: absorbed-MQA flash attention over the latent cache
+        # (the eager expanded path materializes [B,H,S,T] fp32 scores —
+        # 67 GB transient per layer at B=64, S=512)
+        q_rot = self._rope_q(q_rot, cos, sin, positions, B, S)
+        return self._prefill_absorbed(x, q_pass, q_rot, kv, start_pos, S)
       q_rot = self._rope_q(q_rot, cos, sin, positions, B, S)
     else:
       q_rot = self._rope_q(q_rot, cos, sin, positions, B, S)
@@ -169,6 +175,30 @@ class MLAttention(nn.Module):
       cs = cs.view(B, S, -1)
       sn = sn.view(B, S, -1)
     return _rope(q_rot, cs, sn, self.cfg.rope_interleave)
+
+  def _prefill_absorbed(self, x, q_pass, q_rot, kv, start_pos: int, S: int):
+    """Prefill with the SAME absorbed-latent trick as decode: MQA flash
+    attention (sdpa) over [latent | roped-key], value = latent; kv_b's v
+    half applied to the output. O(S) memory instead of O(S*T) fp32."""
+    import torch.nn.functional as F
+    cfg = self.cfg
+    B, H = x.shape[0], cfg.n_heads
+    nope, vd, lat = cfg.qk_nope_head_dim, cfg.v_head_dim, cfg.kv_lora_rank
+    if not hasattr(self, "_w_k"):
+      W = self.kv_b_proj.weight.view(H, nope + vd, lat)
+      self._w_k = W[:, :nope, :].contiguous()
+      self._w_v = W[:, nope:, :].contiguous()
+    total = start_pos + S
+    q_lat = torch.einsum("bshn,hnl->bshl", q_pass.to(x.dtype), self._w_k)
+    qfull = torch.cat([q_lat, q_rot.to(x.dtype)], dim=-1).permute(0, 2, 1, 3)  # [B,H,S,576]
+    kfull = torch.cat([kv[0][:, 0, :total], kv[1][:, 0, :total]], dim=-1)[:, None]  # [B,1,T,576]
+    vlat = kv[0][:, 0, :total][:, None]                                             # [B,1,T,512]
+    mask = torch.ones(S, total, dtype=torch.bool, device=x.device).tril(diagonal=start_pos)
+    out_lat = F.scaled_dot_product_attention(
+      qfull, kfull, vlat, attn_mask=mask, scale=self.scale, enable_gqa=True
+    )  # [B,H,S,512]
+    out = torch.einsum("bhsl,hdl->bshd", out_lat, self._w_v).to(x.dtype)
+    return self.o_proj(out.reshape(B, S, H * vd))
 
   def _decode_mfma(self, x, q_raw, q_pass, kv, seq_lens, hip, positions, cos, sin):
     """Absorbed-latent MFMA decode: kv_b is folded into q and out, so
