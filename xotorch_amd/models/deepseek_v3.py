@@ -24,8 +24,6 @@ from __future__ import annotations
 
 from typing import List
 
-import os
-
 import torch
 import torch.nn as nn
 
@@ -117,9 +115,9 @@ class MLAttention(nn.Module):
       if start_pos < 0:
         start_pos = int(positions.reshape(-1)[0])
       if not torch.is_grad_enabled():
-        # GPU prefill: absorbed-MQA flash attention over the latent cache
-        # (the eager expanded path materializes [B,H,S,T] fp32 scores —
-        # 67 GB transient per layer at B=64, S=512)
+        # GPU prefill: query-chunked bf16 attention (the eager path below
+        # materializes [B,H,S,T] fp32 scores — 67 GB transient per layer
+        # at B=64, S=512)
         q_rot = self._rope_q(q_rot, cos, sin, positions, B, S)
         return self._prefill_absorbed(x, q_pass, q_rot, kv, start_pos, S)
       q_rot = self._rope_q(q_rot, cos, sin, positions, B, S)
