@@ -177,3 +177,40 @@ def test_three_rank_ring_matches_single(tmp_path):
   got = json.loads((tmp_path / "tokens.json").read_text())
   for rid, ids, max_new in reqs:
     assert got[rid] == oracle_tokens(ids, max_new), rid
+
+def test_prefill_interleaves_with_decode(monkeypatch):
+  """A long prompt prefills in chunks WHILE an active slot keeps emitting:
+  tokens from the decoding request must appear between the second request's
+  admission and its first token, and both must still match their oracles."""
+  monkeypatch.setattr("xotorch_amd.serve_ring.PREFILL_CHUNK", 2)
+  w = RingSlotWorker(TINY_ID, 0, 1, device="cpu", dtype=torch.float32,
+                     slots=4, max_seq=128, use_graphs=False)
+  rng = np.random.default_rng(23)
+  ids_a = list(rng.integers(0, 200, 2))    # one-chunk prompt: activates first
+  ids_b = list(rng.integers(0, 200, 10))   # five chunks: prefills while a decodes
+  order = []
+  q = queue.Queue()
+  done = threading.Event()
+  remaining = {"a", "b"}
+
+  def emit(rid, tok, fin, meta):
+    order.append((rid, tok))
+    if fin:
+      remaining.discard(rid)
+      if not remaining:
+        done.set()
+
+  q.put(AdmitMsg("a", torch.tensor([ids_a], dtype=torch.int64), 12, 0.0))
+  q.put(AdmitMsg("b", torch.tensor([ids_b], dtype=torch.int64), 4, 0.0))
+  t = threading.Thread(target=w.serve_forever, args=(q, emit), daemon=True)
+  t.start()
+  assert done.wait(120)
+  q.put(AdmitMsg("stop", None, 0, 0.0))
+  t.join(timeout=30)
+  got_a = [tok for rid, tok in order if rid == "a"]
+  got_b = [tok for rid, tok in order if rid == "b"]
+  assert got_a == oracle_tokens(ids_a, 12)
+  assert got_b == oracle_tokens(ids_b, 4)
+  first_b = next(i for i, (rid, _) in enumerate(order) if rid == "b")
+  a_before_b = sum(1 for rid, _ in order[:first_b] if rid == "a")
+  assert a_before_b >= 2, f"no interleaving: order head {order[:8]}"

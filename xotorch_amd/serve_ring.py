@@ -52,11 +52,11 @@ from xotorch_amd.models.weights import fast_random_init_gpu, load_shard_weights,
 from xotorch_amd.parallel.comm import RingComm, init_distributed
 from xotorch_amd.parallel.ring import equal_ring_shards
 
-OP_IDLE, OP_DECODE, OP_ADMIT, OP_RELEASE, OP_SHUTDOWN = 0, 1, 2, 3, 4
+OP_IDLE, OP_DECODE, OP_ADMIT, OP_RELEASE, OP_SHUTDOWN, OP_PREFILL = 0, 1, 2, 3, 4, 5
 
 MAX_SEQ = int(os.getenv("XOT_SERVE_MAX_SEQ", "4096"))
 SLOTS = int(os.getenv("XOT_SERVE_SLOTS", "8"))
-PREFILL_CHUNK = int(os.getenv("XOT_SERVE_PREFILL_CHUNK", "512"))
+PREFILL_CHUNK = int(os.getenv("XOT_SERVE_PREFILL_CHUNK", "256"))
 
 
 @dataclass
@@ -141,6 +141,8 @@ class RingSlotWorker:
     self.active = [False] * slots
     self.slot_state: List[SlotState] = [SlotState() for _ in range(slots)]
     self._graph: Optional[torch.cuda.CUDAGraph] = None
+    self._pf: Dict[int, dict] = {}  # slot -> in-progress chunked prefill (every rank)
+    self._pf_turn = False
     self.eos_token_id = self.cfg.eos_token_id
 
   # ---------------- decode tick ----------------
@@ -179,12 +181,16 @@ class RingSlotWorker:
     self._reset_inactive()
 
   def _reset_inactive(self):
-    """Pin inactive slots at position 0 so graph-side position advance can
-    never run a dead slot's KV writes past capacity."""
+    """Park inactive slots on their own LAST cache row (max_seq-1): the
+    decode graph appends garbage KV for every slot each tick, and the
+    admission clamp (max_new <= max_seq - plen - 1) guarantees an active
+    slot's attention never reads past row max_seq-2 — so the scratch row
+    can take garbage writes without ever clobbering real prefill data
+    (prerequisite for interleaving chunked prefill with decode ticks)."""
     idle = [i for i in range(self.slots) if not self.active[i]]
     if idle:
       idx = torch.tensor(idle, dtype=torch.int64, device=self.positions.device)
-      self.positions[idx] = 0
+      self.positions[idx] = self.max_seq - 1
       self.seq_lens[idx] = 1
 
   def decode_tick(self):
@@ -227,39 +233,45 @@ class RingSlotWorker:
       out.append(type(layer)(*sl) if hasattr(layer, "_fields") else sl)
     return out
 
-  def prefill_slot(self, slot: int, tokens: Optional[torch.Tensor], plen: int) -> Optional[int]:
-    """Run the prompt through the ring in chunks; returns first token on the
-    last stage (None elsewhere)."""
+  def _prefill_chunk(self, slot: int, caches, tokens, c0: int, c1: int,
+                     plen: int) -> Optional[int]:
+    """One prompt chunk through this rank's layers; returns the first
+    sampled token on the last stage of the last chunk, else None."""
     dev = self.device
+    with torch.inference_mode():
+      pos = torch.arange(c0, c1, dtype=torch.int32, device=dev)
+      if self.is_first:
+        x = tokens[:, c0:c1].to(dev)
+        h = self.model(x, caches=caches, positions=pos, start_pos=c0)
+      else:
+        hbuf = torch.empty(1, c1 - c0, self.cfg.dim, dtype=self.dtype, device=dev)
+        self.comm.recv(hbuf, self.prev_rank)
+        h = self.model(hbuf, caches=caches, positions=pos, start_pos=c0)
+      if not self.is_last:
+        if self.world > 1:
+          self.comm.send(h.contiguous(), self.next_rank)
+        return None
+      if c1 != plen:
+        return None
+      st = self.slot_state[slot]
+      logits = h[:, -1] if h.dim() == 3 else h
+      if st.temp > 0:
+        from xotorch_amd import ops
+        return int(ops.softmax_sample(logits, temperature=st.temp, top_k=35)[0])
+      return int(logits.argmax(dim=-1)[0])
+
+  def prefill_slot(self, slot: int, tokens: Optional[torch.Tensor], plen: int) -> Optional[int]:
+    """Run the whole prompt through the ring in chunks (non-interleaved
+    path, used when the ring has nothing decoding)."""
     caches = self._slot_caches(slot)
     first_tok = None
-    with torch.inference_mode():
-      for c0 in range(0, plen, PREFILL_CHUNK):
-        c1 = min(c0 + PREFILL_CHUNK, plen)
-        pos = torch.arange(c0, c1, dtype=torch.int32, device=dev)
-        last_chunk = c1 == plen
-        if self.is_first:
-          x = tokens[:, c0:c1].to(dev)
-          h = self.model(x, caches=caches, positions=pos, start_pos=c0)
-        else:
-          hbuf = torch.empty(1, c1 - c0, self.cfg.dim, dtype=self.dtype, device=dev)
-          self.comm.recv(hbuf, self.prev_rank)
-          h = self.model(hbuf, caches=caches, positions=pos, start_pos=c0)
-        if not self.is_last:
-          if self.world > 1:
-            self.comm.send(h.contiguous(), self.next_rank)
-        elif last_chunk:
-          st = self.slot_state[slot]
-          if st.temp > 0:
-            from xotorch_amd import ops
-            first_tok = int(ops.softmax_sample(h[:, -1] if h.dim() == 3 else h,
-                                               temperature=st.temp, top_k=35)[0])
-          else:
-            logits = h[:, -1] if h.dim() == 3 else h
-            first_tok = int(logits.argmax(dim=-1)[0])
-      # position state after prefill
-      self.positions[slot] = plen
-      self.seq_lens[slot] = plen + 1
+    for c0 in range(0, plen, PREFILL_CHUNK):
+      c1 = min(c0 + PREFILL_CHUNK, plen)
+      tok = self._prefill_chunk(slot, caches, tokens, c0, c1, plen)
+      if tok is not None:
+        first_tok = tok
+    self.positions[slot] = plen
+    self.seq_lens[slot] = plen + 1
     return first_tok
 
   def _log_done(self, st: SlotState):
@@ -289,6 +301,8 @@ class RingSlotWorker:
     """Run the ring loop. On rank 0, `admit_q` provides AdmitMsg and `emit`
     receives (request_id, token, is_finished, meta). Followers pass None."""
     self._cancelled = set()
+    self._pf.clear()
+    self._pf_turn = False
     idle_ticks = 0
     pending_release: List[int] = []
     inflight: Dict[int, AdmitMsg] = {}
@@ -300,17 +314,21 @@ class RingSlotWorker:
         if self._cancelled:
           for i in range(self.slots):
             st = self.slot_state[i]
-            if self.active[i] and st.request_id in self._cancelled and not st.done:
+            if (self.active[i] or i in self._pf) and st.request_id in self._cancelled and not st.done:
               st.done = True
               if emit:
                 emit(st.request_id, 0, True, {"cancelled": True})
               pending_release.append(i)
           self._cancelled.clear()
         if pending_release:
-          op, slot = OP_RELEASE, pending_release.pop(0)
+          # one RELEASE tick drains the whole backlog (slot = bitmask), so a
+          # finished slot never sees another decode tick's KV advance
+          op, slot = OP_RELEASE, sum(1 << i for i in set(pending_release))
+          pending_release.clear()
         else:
           msg = None
-          free = next((i for i in range(self.slots) if not self.active[i]), None)
+          free = next((i for i in range(self.slots)
+                       if not self.active[i] and i not in self._pf), None)
           if admit_q is not None and free is not None:
             try:
               msg = admit_q.get_nowait()
@@ -322,8 +340,17 @@ class RingSlotWorker:
             op, slot, plen = OP_ADMIT, free, msg.tokens.shape[1]
             extra = int(msg.temp * 1000)
             inflight[free] = msg
+          elif self._pf and (self._pf_turn or not any(self.active)):
+            # alternate one prompt chunk with one decode tick so slots that
+            # are already generating keep emitting during long prefills
+            slot = max(self._pf, key=lambda i: self._pf[i]["c0"])  # finish-first
+            pf = self._pf[slot]
+            op, plen = OP_PREFILL, pf["c0"]
+            extra = min(pf["c0"] + PREFILL_CHUNK, pf["plen"])
+            self._pf_turn = False
           elif any(self.active):
             op = OP_DECODE
+            self._pf_turn = True
           else:
             op = OP_IDLE
       op, slot, plen, extra = self._bcast_hdr(op, slot, plen, extra)
@@ -340,8 +367,11 @@ class RingSlotWorker:
           time.sleep(min(idle_sleep * (1 + idle_ticks // 50), 0.05))
         continue
       if op == OP_RELEASE:
-        self.active[slot] = False
-        self.slot_state[slot] = SlotState()
+        for i in range(self.slots):
+          if slot & (1 << i):
+            self.active[i] = False
+            self.slot_state[i] = SlotState()
+            self._pf.pop(i, None)
         self._reset_inactive()
         continue
       if op == OP_ADMIT:
@@ -354,8 +384,25 @@ class RingSlotWorker:
           self.comm.broadcast(toks, 0)
         st = SlotState(prompt_len=plen, temp=extra / 1000.0)
         self.slot_state[slot] = st
-        first = self.prefill_slot(slot, toks, plen)
-        # ship the first token to rank 0 (it finishes the admit bookkeeping)
+        self._pf[slot] = {"toks": toks, "c0": 0, "plen": plen,
+                          "caches": self._slot_caches(slot)}
+        if self.rank == 0:
+          msg = inflight.pop(slot)
+          st.request_id = msg.request_id
+          st.max_new = min(msg.max_new, self.max_seq - plen - 1)
+          st.t_admit = msg.t_submit
+        continue
+      if op == OP_PREFILL:
+        c0, c1 = plen, extra
+        pf = self._pf[slot]
+        first = self._prefill_chunk(slot, pf["caches"], pf["toks"], c0, c1, pf["plen"])
+        pf["c0"] = c1
+        if c1 < pf["plen"]:
+          continue
+        # last chunk: activate the slot and ship the first token to rank 0
+        del self._pf[slot]
+        self.positions[slot] = pf["plen"]
+        self.seq_lens[slot] = pf["plen"] + 1
         self.tok_bcast.zero_()
         if self.is_last and first is not None:
           self.tok_bcast[slot] = first
@@ -365,16 +412,13 @@ class RingSlotWorker:
           self.tok_buf[slot, 0] = self.tok_bcast[slot]
         self.active[slot] = True
         if self.rank == 0:
-          msg = inflight.pop(slot)
-          st.request_id = msg.request_id
-          st.max_new = min(msg.max_new, self.max_seq - plen - 1)
+          st = self.slot_state[slot]
           st.generated = 1
-          st.ttft_s = time.perf_counter() - msg.t_submit
-          st.t_admit = msg.t_submit
+          st.ttft_s = time.perf_counter() - st.t_admit
           tok = int(self.tok_bcast[slot])
           fin = (self.eos_token_id is not None and tok == self.eos_token_id) or st.generated >= st.max_new
           if emit:
-            emit(msg.request_id, tok, fin, {"ttft_s": st.ttft_s})
+            emit(st.request_id, tok, fin, {"ttft_s": st.ttft_s})
           if fin:
             st.done = True
             self._log_done(st)
