@@ -37,7 +37,14 @@ def main():
   remaining = set()
   done = threading.Event()
 
+  gaps = []
+  last_emit = {}
+
   def emit(rid, tok, fin, meta):
+    now = time.perf_counter()
+    if rid in last_emit:
+      gaps.append(now - last_emit[rid])
+    last_emit[rid] = now
     got.setdefault(rid, []).append(tok)
     if "ttft_s" in meta:
       ttfts[rid] = meta["ttft_s"]
@@ -65,6 +72,10 @@ def main():
         f"{args.slots} slots, graphs={w._graph is not None}: "
         f"{total} tokens in {dt_all:.2f}s = {total/dt_all:.0f} tok/s aggregate; "
         f"TTFT p50 {statistics.median(tl)*1000:.0f} ms / p95 {tl[int(len(tl)*0.95)-1]*1000:.0f} ms")
+  if gaps:
+    gs = sorted(gaps)
+    print(f"inter-token gap: p50 {statistics.median(gs)*1000:.1f} ms / "
+          f"p95 {gs[int(len(gs)*0.95)-1]*1000:.1f} ms / max {gs[-1]*1000:.1f} ms")
 
 
 if __name__ == "__main__":
