@@ -19,9 +19,12 @@ def main():
   p.add_argument("--draft", default="llama-3.2-1b")
   p.add_argument("--gamma", type=int, default=4)
   p.add_argument("--max-new", type=int, default=64)
+  p.add_argument("--dtype", default=None, choices=["bf16", "fp32"])
   args = p.parse_args()
   dev = "cuda" if torch.cuda.is_available() else "cpu"
   dt = torch.bfloat16 if dev == "cuda" else torch.float32
+  if args.dtype:
+    dt = torch.bfloat16 if args.dtype == "bf16" else torch.float32
   sd = SpeculativeDecoder.from_model_ids(args.target, args.draft, device=dev, dtype=dt,
                                          gamma=args.gamma)
   g = torch.Generator().manual_seed(5)
