@@ -1005,3 +1005,20 @@ def test_mla_fp8_decode(hip, monkeypatch):
   err = (out_8 - out_b).abs().max().item()
   ref_scale = out_b.abs().max().item()
   assert err < 0.12 * max(ref_scale, 1.0) + 0.05, (err, ref_scale)
+
+
+@pytest.mark.gpu
+def test_spec_self_draft_full_acceptance():
+  """Draft == target (same seed -> identical weights): acceptance must stay
+  ~1.0 across MANY rounds. Catches the draft-cache hole regression (a fully
+  accepted round used to leave one unwritten KV row in the draft cache,
+  collapsing acceptance to ~0.06 from round 2 on)."""
+  import torch
+  from xotorch_amd.engine.spec import SpeculativeDecoder
+  sd = SpeculativeDecoder.from_model_ids("llama-3-8b", "llama-3-8b",
+                                         device="cuda", dtype=torch.bfloat16, gamma=4)
+  g = torch.Generator().manual_seed(5)
+  prompt = torch.randint(0, 32000, (1, 64), generator=g)
+  toks, stats = sd.generate(prompt, max_new=64)
+  assert stats.rounds >= 10
+  assert stats.accept_rate > 0.9, f"accept {stats.accept_rate:.2f} ({stats.accepted}/{stats.proposed})"

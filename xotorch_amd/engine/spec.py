@@ -116,6 +116,7 @@ class SpeculativeDecoder:
     t_cur = int(t_logits[0, -1].argmax())
     out: List[int] = [t_cur]
     P = S          # both caches truthful through position P-1; t_cur sits at P (unfed)
+    d_next = S     # first cache row the DRAFT has not yet written
     while len(out) < max_new and not (eos_id is not None and t_cur == eos_id):
       g = min(self.gamma, self.max_seq - P - 2)
       if g <= 0:
@@ -124,9 +125,19 @@ class SpeculativeDecoder:
       d_toks: List[int] = []
       cur = t_cur
       dp = P
-      for _ in range(g):
-        dl = self._forward(self.draft, self.d_cache,
-                           torch.tensor([[cur]], dtype=torch.int64, device=dev), dp)
+      for i in range(g):
+        if i == 0 and dp > d_next:
+          # after a fully-accepted round the draft never saw its own last
+          # proposal: feed the accepted-but-unfed positions in one chunk so
+          # the draft cache has no zero-KV hole (hole -> proposals diverge
+          # from the target forever and acceptance collapses)
+          toks = [out[p - S] for p in range(d_next, dp)] + [cur]
+          dl = self._forward(self.draft, self.d_cache,
+                             torch.tensor([toks], dtype=torch.int64, device=dev), d_next)
+        else:
+          dl = self._forward(self.draft, self.d_cache,
+                             torch.tensor([[cur]], dtype=torch.int64, device=dev), dp)
+        d_next = dp + 1
         cur = int(dl[0, -1].argmax())
         d_toks.append(cur)
         dp += 1
