@@ -27,8 +27,9 @@ def main():
   args = p.parse_args()
   dev = "cuda" if torch.cuda.is_available() else "cpu"
   dt = torch.bfloat16 if dev == "cuda" else torch.float32
+  max_seq = max(2048, args.prompt_len + args.max_new + 2)
   w = RingSlotWorker(args.model, 0, 1, device=dev, dtype=dt,
-                     slots=args.slots, max_seq=2048, use_graphs=not args.no_graphs)
+                     slots=args.slots, max_seq=max_seq, use_graphs=not args.no_graphs)
   if dev == "cuda" and not args.no_graphs:
     w._build_graph()
   rng = np.random.default_rng(3)
