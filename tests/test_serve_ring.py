@@ -214,3 +214,35 @@ def test_prefill_interleaves_with_decode(monkeypatch):
   first_b = next(i for i, (rid, _) in enumerate(order) if rid == "b")
   a_before_b = sum(1 for rid, _ in order[:first_b] if rid == "a")
   assert a_before_b >= 2, f"no interleaving: order head {order[:8]}"
+
+
+def test_cancel_during_prefill(monkeypatch):
+  """Cancelling a request whose prompt is still prefilling drops the
+  in-progress chunked prefill and frees the slot."""
+  import time as _t
+  monkeypatch.setattr("xotorch_amd.serve_ring.PREFILL_CHUNK", 1)
+  w = RingSlotWorker(TINY_ID, 0, 1, device="cpu", dtype=torch.float32,
+                     slots=1, max_seq=128, use_graphs=False)
+  q = queue.Queue()
+  got = {}
+  done = threading.Event()
+
+  def emit(rid, tok, fin, meta):
+    got.setdefault(rid, []).append((tok, fin, meta))
+    if fin and rid == "short":
+      done.set()
+
+  rng = np.random.default_rng(31)
+  long_ids = list(rng.integers(0, 200, 64))   # 64 one-token chunks
+  short_ids = list(rng.integers(0, 200, 4))
+  q.put(AdmitMsg("long", torch.tensor([long_ids], dtype=torch.int64), 8, 0.0))
+  t = threading.Thread(target=w.serve_forever, args=(q, emit), daemon=True)
+  t.start()
+  _t.sleep(0.1)          # mid-prefill (64 chunks take a while on CPU)
+  w.cancel("long")
+  q.put(AdmitMsg("short", torch.tensor([short_ids], dtype=torch.int64), 3, 0.0))
+  assert done.wait(60), "slot not freed by mid-prefill cancellation"
+  q.put(AdmitMsg("stop", None, 0, 0.0))
+  t.join(timeout=30)
+  assert [tok for tok, _, _ in got["short"]] == oracle_tokens(short_ids, 3)
+  assert got["long"][-1][2].get("cancelled")
