@@ -62,14 +62,15 @@ def _load_ours_from_oracle(ref, cfg, shard):
   return ours
 
 
-def _build(seed=0):
+def _build(seed=0, extra=None):
   from transformers import DeepseekV3Config, DeepseekV3ForCausalLM
   from xotorch_amd.models.config import config_from_hf
   from xotorch_amd.shard import Shard
   torch.manual_seed(seed)
-  hf_cfg = DeepseekV3Config(**ORACLE_KW)
+  kw = {**ORACLE_KW, **(extra or {})}
+  hf_cfg = DeepseekV3Config(**kw)
   ref = DeepseekV3ForCausalLM(hf_cfg).eval().float()
-  raw = {**ORACLE_KW, "model_type": "deepseek_v3",
+  raw = {**kw, "model_type": "deepseek_v3",
          "rope_theta": hf_cfg.rope_parameters["rope_theta"] if hasattr(hf_cfg, "rope_parameters") else getattr(hf_cfg, "rope_theta", 10000.0),
          "routed_scaling_factor": hf_cfg.routed_scaling_factor,
          "norm_topk_prob": hf_cfg.norm_topk_prob,
@@ -250,3 +251,26 @@ def test_mla_per_row_positions():
     for b in range(B):
       assert torch.allclose(out[b:b + 1], refs[b], atol=1e-5), \
         (b, float((out[b:b + 1] - refs[b]).abs().max()))
+
+
+def test_yarn_rope_matches_oracle():
+  """YaRN long-context scaling (NTK-by-parts frequencies + mscale on cos/sin
+  AND on the softmax scale) against transformers' yarn implementation,
+  prefill and cached decode past original_max_position_embeddings."""
+  yarn = {"rope_scaling": {"rope_type": "yarn", "factor": 4.0,
+                           "beta_fast": 32, "beta_slow": 1,
+                           "mscale": 0.707, "mscale_all_dim": 0.707,
+                           "original_max_position_embeddings": 16}}
+  ref, ours, cfg, shard = _build(seed=3, extra=yarn)
+  assert cfg.rope_scaling is not None and cfg.rope_scaling.rope_type == "yarn"
+  B, S = 2, 24  # past the pretraining window of 16
+  x = torch.randint(0, cfg.vocab_size, (B, S), generator=torch.Generator().manual_seed(9))
+  cache = _latent_caches(cfg, cfg.n_layers, B, S + 4)
+  with torch.no_grad():
+    lref = ref(x).logits
+    lours = ours(x, caches=cache.caches, positions=torch.arange(S), start_pos=0, last_only=False)
+    assert torch.allclose(lours, lref, atol=3e-4, rtol=3e-4), (lours - lref).abs().max()
+    nxt = lref[:, -1].argmax(-1, keepdim=True)
+    lref2 = ref(torch.cat([x, nxt], dim=1)).logits[:, -1]
+    lours2 = ours(nxt, caches=cache.caches, positions=torch.tensor([S]), start_pos=S, is_decode=True)
+    assert torch.allclose(lours2, lref2, atol=3e-4, rtol=3e-4), (lours2 - lref2).abs().max()

@@ -26,12 +26,18 @@ _DTYPE_MAP = {
 
 @dataclass
 class RopeScaling:
-  # llama3-style frequency scaling; None fields → plain RoPE
+  # llama3- or yarn-style frequency scaling; None fields → plain RoPE
   factor: float = 1.0
   low_freq_factor: float = 1.0
   high_freq_factor: float = 4.0
   original_max_position_embeddings: int = 8192
   rope_type: str = "default"
+  # yarn (deepseek long-context) fields — HF _compute_yarn_parameters semantics
+  beta_fast: float = 32.0
+  beta_slow: float = 1.0
+  mscale: float = 1.0
+  mscale_all_dim: float = 0.0
+  truncate: bool = True
 
 
 @dataclass
@@ -115,14 +121,27 @@ def config_from_hf(config_path: Path | str | dict, model_id: str = "unknown") ->
   dim = raw.get("hidden_size", 4096)
   head_dim = raw.get("head_dim") or dim // n_heads
   rope_scaling = None
-  rs = raw.get("rope_scaling")
-  if rs and rs.get("rope_type", rs.get("type", "default")) == "llama3":
+  rs = raw.get("rope_scaling") or raw.get("rope_parameters")
+  rs_type = rs.get("rope_type", rs.get("type", "default")) if rs else "default"
+  if rs and rs_type == "llama3":
     rope_scaling = RopeScaling(
       factor=rs.get("factor", 8.0),
       low_freq_factor=rs.get("low_freq_factor", 1.0),
       high_freq_factor=rs.get("high_freq_factor", 4.0),
       original_max_position_embeddings=rs.get("original_max_position_embeddings", 8192),
       rope_type="llama3",
+    )
+  elif rs and rs_type == "yarn":
+    orig = rs.get("original_max_position_embeddings", 4096)
+    rope_scaling = RopeScaling(
+      factor=rs.get("factor") or raw.get("max_position_embeddings", orig) / orig,
+      original_max_position_embeddings=orig,
+      rope_type="yarn",
+      beta_fast=rs.get("beta_fast") or 32.0,
+      beta_slow=rs.get("beta_slow") or 1.0,
+      mscale=rs.get("mscale") or 1.0,
+      mscale_all_dim=rs.get("mscale_all_dim") or 0.0,
+      truncate=rs.get("truncate", True),
     )
 
   max_seq_len = raw.get("max_position_embeddings", 8192)

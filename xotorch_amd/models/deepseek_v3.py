@@ -10,11 +10,13 @@ with e-score correction bias, group-limited top-k (n_group/topk_group),
 routed scaling and always-on shared experts; the first k layers are dense.
 
 The reference lists deepseek-v3 / r1 cards its torchtune GQA assembly cannot
-run; here the family runs correctly on a plain-torch
-compute path validated against transformers' DeepseekV3ForCausalLM (tiny
-random configs — the full 671B needs the 8-GPU ring). The CDNA4 kernels are
-not used (MLA attention kernel: docs/ROADMAP.md). YaRN rope scaling for the
-full-length context is not implemented (default rope only).
+run; here the family runs the absorbed-latent
+MLA MFMA decode path on GPU (hip_ops.hip: mla_prep_append / mla_q_prep /
+attn_decode_mla) and a plain-torch path on CPU, both validated against
+transformers' DeepseekV3ForCausalLM on tiny random configs (the full 671B
+needs the 8-GPU ring). YaRN long-context rope scaling is implemented
+(NTK-by-parts frequency blend + mscale corrections, validated against
+transformers).
 
 Cache contract: caches[idx] is a LayerKV whose k tensor stores the kv_nope
 latent [B, 1, T, kv_lora_rank] and v tensor stores the roped shared key
@@ -77,7 +79,13 @@ class MLAttention(nn.Module):
     self.kv_a_layernorm = nn.Parameter(torch.ones(cfg.kv_lora_rank))
     self.kv_b_proj = nn.Linear(cfg.kv_lora_rank, H * (cfg.qk_nope_head_dim + cfg.v_head_dim), bias=False)
     self.o_proj = XotLinear(H * cfg.v_head_dim, D, bias=cfg.attn_bias)
-    self.scale = self.qk_head_dim ** -0.5  # default rope (no yarn mscale)
+    self.scale = self.qk_head_dim ** -0.5
+    rs = cfg.rope_scaling
+    if rs is not None and rs.rope_type == "yarn" and rs.mscale_all_dim and rs.factor > 1:
+      # yarn softmax-scale correction (HF yarn_apply_mscale)
+      import math
+      m = 0.1 * rs.mscale_all_dim * math.log(rs.factor) + 1.0
+      self.scale *= m * m
 
   def forward(self, x, cos, sin, positions, kv, start_pos: int,
               is_decode: bool = False, seq_lens=None):
@@ -405,13 +413,13 @@ class DeepseekV3Model(nn.Module):
       self.norm = nn.Parameter(torch.ones(cfg.dim))
       self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
     # per-pair rope tables at the ROPE head dim
-    cos, sin = rope_cos_sin(cfg.qk_rope_head_dim, cfg.max_seq_len, cfg.rope_theta, None)
+    cos, sin = rope_cos_sin(cfg.qk_rope_head_dim, cfg.max_seq_len, cfg.rope_theta, cfg.rope_scaling)
     self.register_buffer("rope_cos", cos, persistent=False)
     self.register_buffer("rope_sin", sin, persistent=False)
 
   def reset_rope(self):
     cos, sin = rope_cos_sin(self.cfg.qk_rope_head_dim, self.cfg.max_seq_len, self.cfg.rope_theta,
-                            None, device=self.rope_cos.device)
+                            self.cfg.rope_scaling, device=self.rope_cos.device)
     self.rope_cos, self.rope_sin = cos, sin
 
   @property

@@ -60,9 +60,39 @@ def rope_cos_sin(
     smoothed = (1 - smooth) * inv_freq / scaling.factor + smooth * inv_freq
     is_medium = (wavelen >= high_freq_wavelen) & (wavelen <= low_freq_wavelen)
     inv_freq = torch.where(is_medium, smoothed, scaled)
+  attn_factor = 1.0
+  if scaling is not None and getattr(scaling, "rope_type", "default") == "yarn":
+    # YaRN (deepseek long-context): NTK-by-parts blend of interpolated and
+    # extrapolated frequencies + a log-factor magnitude correction baked
+    # into cos/sin (HF _compute_yarn_parameters semantics, validated
+    # against transformers' DeepseekV3 in tests/test_deepseek_cpu.py)
+    dim, base, factor = head_dim, theta, scaling.factor
+    orig = scaling.original_max_position_embeddings
+
+    def corr_dim(n_rot):
+      return (dim * math.log(orig / (n_rot * 2 * math.pi))) / (2 * math.log(base))
+
+    low, high = corr_dim(scaling.beta_fast), corr_dim(scaling.beta_slow)
+    if scaling.truncate:
+      low, high = math.floor(low), math.ceil(high)
+    low, high = max(low, 0), min(high, dim - 1)
+    if low == high:
+      high += 0.001
+    ramp = torch.clamp(
+      (torch.arange(head_dim // 2, dtype=torch.float32, device=device) - low) / (high - low), 0, 1)
+    extrap_w = 1.0 - ramp
+    inv_freq = (inv_freq / factor) * (1 - extrap_w) + inv_freq * extrap_w
+
+    def get_mscale(scale, m=1.0):
+      return 1.0 if scale <= 1 else 0.1 * m * math.log(scale) + 1.0
+
+    if scaling.mscale and scaling.mscale_all_dim:
+      attn_factor = get_mscale(factor, scaling.mscale) / get_mscale(factor, scaling.mscale_all_dim)
+    else:
+      attn_factor = get_mscale(factor)
   t = torch.arange(max_seq_len, dtype=torch.float32, device=device)
   freqs = torch.outer(t, inv_freq)
-  return freqs.cos().to(dtype), freqs.sin().to(dtype)
+  return (freqs.cos() * attn_factor).to(dtype), (freqs.sin() * attn_factor).to(dtype)
 
 
 def _rotate_half(x: torch.Tensor) -> torch.Tensor:
