@@ -1022,3 +1022,56 @@ def test_spec_self_draft_full_acceptance():
   toks, stats = sd.generate(prompt, max_new=64)
   assert stats.rounds >= 10
   assert stats.accept_rate > 0.9, f"accept {stats.accept_rate:.2f} ({stats.accepted}/{stats.proposed})"
+
+
+@pytest.mark.gpu
+def test_deepseek_yarn_gpu_matches_eager():
+  """YaRN rope scaling flows through the HIP MLA path (host cos/sin tables +
+  softmax-scale param reach mla_prep_append / mla_q_prep / attn_decode_mla):
+  GPU bf16 decode vs the CPU fp32 eager oracle with the same yarn config."""
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  from xotorch_amd.models.config import config_from_hf
+  from xotorch_amd.models.deepseek_v3 import DeepseekV3Model
+  from xotorch_amd.models.weights import random_init
+  from xotorch_amd.shard import Shard
+  raw = dict(model_type="deepseek_v3", vocab_size=512, hidden_size=256, intermediate_size=512,
+             num_hidden_layers=2, num_attention_heads=8, num_key_value_heads=8,
+             kv_lora_rank=512, qk_rope_head_dim=64, qk_nope_head_dim=128, v_head_dim=128,
+             q_lora_rank=0, first_k_dense_replace=2, n_routed_experts=0,
+             rms_norm_eps=1e-6, rope_theta=10000.0, max_position_embeddings=256,
+             rope_scaling={"rope_type": "yarn", "factor": 8.0, "beta_fast": 32,
+                           "beta_slow": 1, "mscale": 0.707, "mscale_all_dim": 0.707,
+                           "original_max_position_embeddings": 32})
+  cfg = config_from_hf(raw, "ds-yarn")
+  assert cfg.rope_scaling is not None and cfg.rope_scaling.rope_type == "yarn"
+  shard = Shard("ds-yarn", 0, 1, 2)
+  torch.manual_seed(5)
+  m = DeepseekV3Model(cfg, shard).to("cuda").to(torch.bfloat16)
+  random_init(m)
+  m.reset_rope()
+  m.eval()
+  base_scale = (cfg.qk_nope_head_dim + cfg.qk_rope_head_dim) ** -0.5
+  assert m.layers["0"].self_attn.scale > base_scale  # mscale^2 correction applied
+  mc = DeepseekV3Model(cfg, shard).float()
+  mc.load_state_dict({k: v.float().cpu() for k, v in m.state_dict().items()}, strict=False)
+  mc.reset_rope()
+  mc.eval()
+  B, S = 2, 48  # prefill past the 32-token pretraining window
+  toks = torch.randint(0, 512, (B, S), device="cuda")
+  heads, kd, vd = cfg.kv_cache_dims()
+  cg = ShardKVCache(2, B, heads, S + 8, kd, torch.bfloat16, "cuda", v_dim=vd)
+  cc = ShardKVCache(2, B, heads, S + 8, kd, torch.float32, "cpu", v_dim=vd)
+  with torch.inference_mode():
+    pos = torch.arange(S, dtype=torch.int32, device="cuda")
+    lg = m(toks, caches=cg.caches, positions=pos, start_pos=0)
+    lr = mc(toks.cpu(), caches=cc.caches, positions=torch.arange(S), start_pos=0)
+    assert (lg.float().cpu().argmax(-1) == lr.argmax(-1)).all()
+    nxt = lg.argmax(-1, keepdim=True)
+    for step in range(3):
+      p = S + step
+      lg = m(nxt, caches=cg.caches, positions=torch.tensor([p], dtype=torch.int32, device="cuda"),
+             start_pos=p, is_decode=True)
+      lr = mc(nxt.cpu(), caches=cc.caches, positions=torch.tensor([p]), start_pos=p, is_decode=True)
+      assert torch.allclose(lg.float().cpu(), lr.float(), atol=0.5, rtol=0.1), \
+        (step, (lg.float().cpu() - lr.float()).abs().max())
+      nxt = lg.argmax(-1, keepdim=True)
