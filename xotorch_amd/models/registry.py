@@ -61,8 +61,9 @@ def _qwen3_cfg(dim, n_layers, n_heads, n_kv, inter, vocab=151936, tie=False, max
 
 def _deepseek_v3_cfg():
   # DeepSeek-V3 / R1 (671B): MLA + 256-expert MoE; 8x MI355X (2.3 TB HBM)
-  # holds it at bf16 via the ring. YaRN long-context scaling not implemented
-  # (default rope window).
+  # holds it at bf16 via the ring. YaRN params as published (factor 40 over
+  # a 4096 pretraining window); max_position kept at 4096 here so
+  # random-init cards stay cache-affordable — raise with XOT_MAX_SEQ_LEN.
   return {
     "model_type": "deepseek_v3", "hidden_size": 7168, "num_hidden_layers": 61,
     "num_attention_heads": 128, "num_key_value_heads": 128,
@@ -74,6 +75,9 @@ def _deepseek_v3_cfg():
     "qk_nope_head_dim": 128, "v_head_dim": 128, "vocab_size": 129280,
     "rope_theta": 10000.0, "rms_norm_eps": 1e-6,
     "max_position_embeddings": 4096, "torch_dtype": "bfloat16",
+    "rope_scaling": {"rope_type": "yarn", "factor": 40.0, "beta_fast": 32,
+                     "beta_slow": 1, "mscale": 1.0, "mscale_all_dim": 1.0,
+                     "original_max_position_embeddings": 4096},
     "bos_token_id": 0, "eos_token_id": 1,
   }
 
@@ -180,6 +184,9 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
     "qk_nope_head_dim": 128, "v_head_dim": 128, "vocab_size": 102400,
     "rope_theta": 10000.0, "rms_norm_eps": 1e-6,
     "max_position_embeddings": 4096, "torch_dtype": "bfloat16",
+    "rope_scaling": {"rope_type": "yarn", "factor": 40.0, "beta_fast": 32,
+                     "beta_slow": 1, "mscale": 0.707, "mscale_all_dim": 0.707,
+                     "original_max_position_embeddings": 4096},
     "bos_token_id": 100000, "eos_token_id": 100001,
   },
   # phi-4-mini (llama-like enough for the generic decoder)
