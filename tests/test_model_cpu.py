@@ -317,3 +317,39 @@ def test_pack_decode_weight_is_a_permutation():
     got = flat[(n32 * (K // 16) + kt) * 512 + l * 8 + j]
     want = w[n32 * 32 + (l & 31)][kt * 16 + (l >> 5) * 8 + j]
     assert torch.equal(got, want), (n32, kt, l, j)
+
+
+def test_qwen2_yarn_matches_transformers():
+  """YaRN rope on the GQA family (qwen2 long-context configs): our generic
+  yarn branch in rope_cos_sin vs transformers' yarn implementation, prefill
+  past the pretraining window + one cached decode step."""
+  from transformers import Qwen2Config, Qwen2ForCausalLM
+  from xotorch_amd.models.weights import hf_key_map, remap_hf_state
+  kw = dict(vocab_size=151, hidden_size=64, intermediate_size=128, num_hidden_layers=3,
+            num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64,
+            rope_theta=10000.0, rms_norm_eps=1e-6, tie_word_embeddings=False,
+            rope_scaling={"rope_type": "yarn", "factor": 4.0, "beta_fast": 32,
+                          "beta_slow": 1, "original_max_position_embeddings": 16})
+  torch.manual_seed(11)
+  ref = Qwen2ForCausalLM(Qwen2Config(**kw)).eval().float()
+  cfg = config_from_hf({**kw, "model_type": "qwen2"}, "qwen-tiny")
+  assert cfg.rope_scaling is not None and cfg.rope_scaling.rope_type == "yarn"
+  cfg.torch_dtype = torch.float32
+  shard = Shard("qwen-tiny", 0, cfg.n_layers - 1, cfg.n_layers)
+  ours = ShardedModel(cfg, shard).float()
+  sd = ref.state_dict()
+  missing = ours.load_state_dict(remap_hf_state(sd, hf_key_map(shard, cfg)), strict=False).missing_keys
+  assert not [m for m in missing if "rope" not in m], missing
+  ours.eval()
+  B, S = 2, 24  # past the 16-token pretraining window
+  x = torch.randint(0, 151, (B, S), generator=torch.Generator().manual_seed(4))
+  cache = ShardKVCache(cfg.n_layers, B, cfg.n_kv_heads, S + 4, cfg.head_dim,
+                       dtype=torch.float32, device="cpu")
+  with torch.no_grad():
+    lref = ref(x).logits
+    lours = ours(x, caches=cache.caches, positions=torch.arange(S), start_pos=0, last_only=False)
+    torch.testing.assert_close(lours, lref, atol=3e-4, rtol=3e-4)
+    nxt = lref[:, -1].argmax(-1, keepdim=True)
+    lref2 = ref(torch.cat([x, nxt], 1)).logits[:, -1]
+    lours2 = ours(nxt, caches=cache.caches, positions=torch.tensor([S]), start_pos=S, is_decode=True)
+    torch.testing.assert_close(lours2, lref2, atol=3e-4, rtol=3e-4)
