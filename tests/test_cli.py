@@ -1,0 +1,37 @@
+"""CLI surface: parser semantics + the one-shot `xot run` path end to end
+with the dummy engine (no network, no GPU)."""
+import asyncio
+
+import pytest
+
+from xotorch_amd.cli import build_parser, pick_engine, run_model_cli
+
+
+def test_parser_verbs_and_defaults():
+  p = build_parser()
+  a = p.parse_args([])
+  assert a.command is None and a.default_temp == 0.0 and a.chatgpt_api_port == 52415
+  a = p.parse_args(["run", "llama-3.2-1b", "--prompt", "hi", "--max-generate-tokens", "7"])
+  assert (a.command, a.model_name, a.prompt, a.max_generate_tokens) == ("run", "llama-3.2-1b", "hi", 7)
+  a = p.parse_args(["serve", "llama-3-70b", "--gpus", "8", "--slots", "16", "--no-graphs"])
+  assert (a.command, a.gpus, a.slots, a.no_graphs) == ("serve", 8, 16, True)
+  a = p.parse_args(["train", "dummy", "--data", "d.jsonl", "--epochs", "2"])
+  assert (a.command, a.data, a.epochs) == ("train", "d.jsonl", 2)
+
+
+def test_pick_engine_explicit_and_default():
+  p = build_parser()
+  assert pick_engine(p.parse_args(["run", "dummy", "--inference-engine", "dummy"])) == "dummy"
+  # default resolution never raises and returns a known engine name
+  assert pick_engine(p.parse_args(["run", "dummy"])) in ("torch", "hip", "dummy")
+
+
+def test_run_one_shot_dummy(capsys):
+  """`xot run dummy --prompt ...` produces tokens and exits cleanly."""
+  p = build_parser()
+  args = p.parse_args(["run", "dummy", "--prompt", "hello world",
+                       "--inference-engine", "dummy", "--discovery-module", "none",
+                       "--max-generate-tokens", "5", "--disable-tui"])
+  asyncio.new_event_loop().run_until_complete(run_model_cli(args))
+  out = capsys.readouterr().out
+  assert out.strip(), "no generated text printed"
