@@ -89,6 +89,8 @@ class RingSlotWorker:
     self.model_id = model_id
     self.rank, self.world = rank, world
     self.device, self.dtype = device, dtype
+    if not 1 <= slots <= 63:
+      raise ValueError(f"slots must be in [1, 63] (release op is an int64 bitmask), got {slots}")
     self.slots, self.max_seq = slots, max_seq
     self.use_graphs = use_graphs and device == "cuda"
     self.cfg = config_from_hf(builtin_config(model_id) or {}, model_id)
