@@ -191,9 +191,12 @@ BUILTIN_CONFIGS: Dict[str, dict] = {
   },
   # phi-4-mini (llama-like enough for the generic decoder)
   "phi-4-mini": _llama_cfg(3072, 32, 24, 8, 8192, vocab=200064, theta=10000.0, max_pos=131072, tie=True),
+  "phi-4-mini-instruct": _llama_cfg(3072, 32, 24, 8, 8192, vocab=200064, theta=10000.0, max_pos=131072, tie=True),
   # tiny test model
   "dummy": _llama_cfg(64, 4, 4, 2, 128, vocab=256, theta=10000.0, max_pos=256, tie=True),
 }
+# DeepSeek-Coder-V2-Lite shares the V2-Lite architecture exactly
+BUILTIN_CONFIGS["deepseek-coder-v2-lite"] = dict(BUILTIN_CONFIGS["deepseek-v2-lite"])
 
 
 # model cards: layers + HF repo per engine (engine names of THIS framework)
@@ -244,6 +247,12 @@ model_cards: Dict[str, dict] = {
   "deepseek-r1": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-R1", "HIPEngine": "deepseek-ai/DeepSeek-R1"}},
   "deepseek-v3": {"layers": 61, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-V3", "HIPEngine": "deepseek-ai/DeepSeek-V3"}},
   "deepseek-v2-lite": {"layers": 27, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-V2-Lite-Chat", "HIPEngine": "deepseek-ai/DeepSeek-V2-Lite-Chat"}},
+  "deepseek-coder-v2-lite": {"layers": 27, "repo": {"TorchEngine": "deepseek-ai/DeepSeek-Coder-V2-Lite-Instruct", "HIPEngine": "deepseek-ai/DeepSeek-Coder-V2-Lite-Instruct"}},
+  "phi-4-mini-instruct": {"layers": 32, "repo": {"TorchEngine": "microsoft/Phi-4-mini-instruct", "HIPEngine": "microsoft/Phi-4-mini-instruct"}},
+  # reference card id for a bitsandbytes-4bit 405B checkpoint: bnb weight
+  # formats are not supported (use llama-3.1-405b bf16 + the opt-in fp8
+  # modes instead); listed for id parity, filtered like llava
+  "llama-3.1-405b-8bit": {"layers": 126, "repo": {}},
   # llava needs PIL for the image path (not present in this environment);
   # the reference's vision handling is also vestigial
   "llava-1.5-7b-hf": {"layers": 32, "repo": {}},
@@ -291,6 +300,9 @@ pretty_names = {
   "gemma2-27b": "Gemma2 27B",
   "deepseek-r1": "DeepSeek R1",
   "deepseek-v2-lite": "DeepSeek V2 Lite (16B MLA)",
+  "deepseek-coder-v2-lite": "DeepSeek Coder V2 Lite (16B MLA)",
+  "phi-4-mini-instruct": "Phi-4 Mini Instruct",
+  "llama-3.1-405b-8bit": "Llama 3.1 405B bnb-4bit (unsupported weight format)",
   "deepseek-v3": "DeepSeek V3",
   "llava-1.5-7b-hf": "LLaVa 1.5 7B (unsupported arch)",
   "phi-4-mini": "Phi-4 Mini",
