@@ -246,3 +246,22 @@ def test_cancel_during_prefill(monkeypatch):
   t.join(timeout=30)
   assert [tok for tok, _, _ in got["short"]] == oracle_tokens(short_ids, 3)
   assert got["long"][-1][2].get("cancelled")
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_chunked_prefill_interleave(tmp_path, monkeypatch):
+  """World-2 ring with multi-chunk prompts (chunk=2): OP_PREFILL hops each
+  chunk through both stages while an active slot keeps decoding; tokens
+  still match the single-request oracle."""
+  import json
+  import torch.multiprocessing as mp
+  from xotorch_amd.helpers import find_available_port
+  monkeypatch.setenv("XOT_SERVE_PREFILL_CHUNK", "2")  # inherited by spawned ranks
+  rng = np.random.default_rng(41)
+  reqs = [("a", [int(v) for v in rng.integers(0, 200, 3)], 8),   # activates first
+          ("b", [int(v) for v in rng.integers(0, 200, 11)], 4)]  # 6 chunks
+  port = find_available_port("127.0.0.1")
+  mp.spawn(_ring_worker, args=(2, port, str(tmp_path), reqs), nprocs=2, join=True)
+  got = json.loads((tmp_path / "tokens.json").read_text())
+  for rid, ids, max_new in reqs:
+    assert got[rid] == oracle_tokens(ids, max_new), rid
