@@ -35,3 +35,15 @@ def test_run_one_shot_dummy(capsys):
   asyncio.new_event_loop().run_until_complete(run_model_cli(args))
   out = capsys.readouterr().out
   assert out.strip(), "no generated text printed"
+
+
+def test_tools_and_entry_scripts_parse():
+  """Every tools/ script, bench.py and __graft_entry__.py must at least
+  parse — they run remotely on GPU boxes where a syntax error is costly."""
+  import ast
+  from pathlib import Path
+  root = Path(__file__).resolve().parent.parent
+  files = sorted((root / "tools").glob("*.py")) + [root / "bench.py", root / "__graft_entry__.py"]
+  assert len(files) > 8
+  for f in files:
+    ast.parse(f.read_text(), filename=str(f))
