@@ -168,3 +168,65 @@ def test_post_download_route():
     await node.stop()
     return True
   assert run(go())
+
+
+def test_stop_sequences():
+  """OpenAI `stop` param: generation is truncated before the stop string in
+  both non-streaming and streaming modes, and the request is cancelled."""
+  async def go():
+    node, client = await make_client()
+    # discover the untruncated output first
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "messages": [{"role": "user", "content": "hello"}]})
+    full = (await r.json())["choices"][0]["message"]["content"]
+    assert len(full) >= 3, full
+    stop = full[1:3]  # substring from the middle
+    want = full[: full.find(stop)]
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "stop": stop,
+      "messages": [{"role": "user", "content": "hello"}]})
+    data = await r.json()
+    assert data["choices"][0]["message"]["content"] == want
+    assert data["choices"][0]["finish_reason"] == "stop"
+    # streaming: concatenated deltas match the same truncation
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "stream": True, "stop": [stop],
+      "messages": [{"role": "user", "content": "hello"}]})
+    text = ""
+    async for line in r.content:
+      line = line.decode().strip()
+      if not line.startswith("data: ") or line == "data: [DONE]":
+        continue
+      for ch in json.loads(line[6:])["choices"]:
+        text += ch["delta"].get("content") or ""
+    assert text == want, (text, want)
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())
+
+
+def test_streaming_carries_final_batch_text():
+  """Streaming deltas concatenated must equal the non-streaming content —
+  the final chunk carries both text and finish_reason (a chunk with
+  finish_reason used to drop its text, losing the last token batch)."""
+  async def go():
+    node, client = await make_client()
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "messages": [{"role": "user", "content": "same seed"}]})
+    full = (await r.json())["choices"][0]["message"]["content"]
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "stream": True,
+      "messages": [{"role": "user", "content": "same seed"}]})
+    text = ""
+    async for line in r.content:
+      line = line.decode().strip()
+      if not line.startswith("data: ") or line == "data: [DONE]":
+        continue
+      for ch in json.loads(line[6:])["choices"]:
+        text += ch.get("delta", {}).get("content") or ""
+    assert text == full, (text, full)
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())

@@ -67,7 +67,9 @@ def generate_completion(request_id: str, tokens: List[int], decoded: str, model:
     }],
   }
   if stream:
-    completion["choices"][0]["delta"] = {"role": "assistant", "content": decoded} if not finish_reason else {}
+    # the final chunk may carry both text and finish_reason: dropping the
+    # text would lose the last token batch on streaming clients
+    completion["choices"][0]["delta"] = {"role": "assistant", "content": decoded} if decoded else {}
   else:
     completion["choices"][0]["message"] = {"role": "assistant", "content": decoded}
     completion["usage"] = {"completion_tokens": len(tokens)}
@@ -283,32 +285,65 @@ class ChatGPTAPI:
         state["temperature"] = float(chat_request.temperature)
       if data.get("top_p") is not None:
         state["top_p"] = float(data["top_p"])
+      stops = data.get("stop")
+      if isinstance(stops, str):
+        stops = [stops]
+      elif isinstance(stops, list):
+        stops = [x for x in stops if isinstance(x, str) and x][:4]  # OpenAI caps at 4
+      else:
+        stops = None
+      stops = stops or None
       try:
         await self.node.process_prompt(shard, prompt, request_id, state or None)
       except ValueError as e:
         # e.g. prompt longer than the serving context — a client error
         return web.json_response({"detail": str(e)}, status=400)
       if stream:
-        return await self._stream_response(request, request_id, model_id, tokenizer)
-      return await self._full_response(request_id, model_id, tokenizer)
+        return await self._stream_response(request, request_id, model_id, tokenizer, stops)
+      return await self._full_response(request_id, model_id, tokenizer, stops)
     finally:
       self.token_queues.pop(request_id, None)
       self.prev_token_lens.pop(request_id, None)
 
-  async def _stream_response(self, request, request_id, model_id, tokenizer):
+  @staticmethod
+  def _find_stop(text: str, stops) -> int:
+    """Earliest index of any stop string in text, or -1."""
+    cut = -1
+    for st in stops or ():
+      i = text.find(st)
+      if i != -1 and (cut == -1 or i < cut):
+        cut = i
+    return cut
+
+  async def _stream_response(self, request, request_id, model_id, tokenizer, stops=None):
     response = web.StreamResponse(status=200, headers={
       "Content-Type": "text/event-stream", "Cache-Control": "no-cache",
     })
     await response.prepare(request)
     all_tokens: List[int] = []
     finished = False
+    # stop-sequence handling: hold back max(len)-1 chars so a stop string
+    # split across token boundaries is still caught before being emitted
+    text_all, sent = "", 0
+    hold = max((len(st) for st in stops), default=1) - 1 if stops else 0
     try:
       while not finished:
         tokens, finished = await asyncio.wait_for(self.token_queues[request_id].get(), self.response_timeout)
         all_tokens.extend(tokens)
-        decoded = tokenizer.decode(tokens) if tokens else ""
-        chunk = generate_completion(request_id, tokens, decoded, model_id,
-                                    "stop" if finished else None, True, "chat.completion.chunk")
+        text_all += tokenizer.decode(tokens) if tokens else ""
+        cut = self._find_stop(text_all, stops)
+        if cut != -1:
+          delta, fin = text_all[sent:cut], "stop"
+          finished = True
+          cancel = getattr(self.node, "cancel_request", None)
+          if cancel is not None:
+            cancel(request_id)
+        else:
+          safe = len(text_all) if finished else max(sent, len(text_all) - hold)
+          delta, fin = text_all[sent:safe], ("stop" if finished else None)
+        sent += len(delta)
+        chunk = generate_completion(request_id, tokens, delta, model_id,
+                                    fin, True, "chat.completion.chunk")
         await response.write(f"data: {json.dumps(chunk)}\n\n".encode())
     except (asyncio.CancelledError, ConnectionResetError):
       # client went away mid-stream: free the slot/session instead of
@@ -321,12 +356,23 @@ class ChatGPTAPI:
     await response.write_eof()
     return response
 
-  async def _full_response(self, request_id, model_id, tokenizer):
+  async def _full_response(self, request_id, model_id, tokenizer, stops=None):
     all_tokens: List[int] = []
     finished = False
+    text_all = ""
     while not finished:
       tokens, finished = await asyncio.wait_for(self.token_queues[request_id].get(), self.response_timeout)
       all_tokens.extend(tokens)
+      if stops:
+        text_all += tokenizer.decode(tokens) if tokens else ""
+        cut = self._find_stop(text_all, stops)
+        if cut != -1:
+          cancel = getattr(self.node, "cancel_request", None)
+          if cancel is not None:
+            cancel(request_id)
+          return web.json_response(
+            generate_completion(request_id, all_tokens, text_all[:cut], model_id,
+                                "stop", False, "chat.completion"))
     decoded = tokenizer.decode(all_tokens) if all_tokens else ""
     return web.json_response(
       generate_completion(request_id, all_tokens, decoded, model_id, "stop", False, "chat.completion"))
