@@ -230,3 +230,24 @@ def test_streaming_carries_final_batch_text():
     await node.stop()
     return True
   assert run(go())
+
+
+def test_stream_options_include_usage():
+  """OpenAI stream_options.include_usage: a final usage chunk with empty
+  choices precedes [DONE]."""
+  async def go():
+    node, client = await make_client()
+    r = await client.post("/v1/chat/completions", json={
+      "model": "dummy", "stream": True, "stream_options": {"include_usage": True},
+      "messages": [{"role": "user", "content": "hi"}]})
+    chunks = []
+    async for line in r.content:
+      line = line.decode().strip()
+      if line.startswith("data: ") and line != "data: [DONE]":
+        chunks.append(json.loads(line[6:]))
+    assert chunks[-1]["choices"] == []
+    assert chunks[-1]["usage"]["completion_tokens"] >= 1
+    await client.close()
+    await node.stop()
+    return True
+  assert run(go())

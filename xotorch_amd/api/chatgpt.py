@@ -299,7 +299,9 @@ class ChatGPTAPI:
         # e.g. prompt longer than the serving context — a client error
         return web.json_response({"detail": str(e)}, status=400)
       if stream:
-        return await self._stream_response(request, request_id, model_id, tokenizer, stops)
+        include_usage = bool((data.get("stream_options") or {}).get("include_usage"))
+        return await self._stream_response(request, request_id, model_id, tokenizer, stops,
+                                           include_usage)
       return await self._full_response(request_id, model_id, tokenizer, stops)
     finally:
       self.token_queues.pop(request_id, None)
@@ -315,7 +317,8 @@ class ChatGPTAPI:
         cut = i
     return cut
 
-  async def _stream_response(self, request, request_id, model_id, tokenizer, stops=None):
+  async def _stream_response(self, request, request_id, model_id, tokenizer, stops=None,
+                             include_usage=False):
     response = web.StreamResponse(status=200, headers={
       "Content-Type": "text/event-stream", "Cache-Control": "no-cache",
     })
@@ -352,6 +355,12 @@ class ChatGPTAPI:
       if cancel is not None:
         cancel(request_id)
       raise
+    if include_usage:
+      usage_chunk = generate_completion(request_id, [], "", model_id, None, True,
+                                        "chat.completion.chunk")
+      usage_chunk["choices"] = []
+      usage_chunk["usage"] = {"completion_tokens": len(all_tokens)}
+      await response.write(f"data: {json.dumps(usage_chunk)}\n\n".encode())
     await response.write(b"data: [DONE]\n\n")
     await response.write_eof()
     return response
