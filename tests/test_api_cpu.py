@@ -246,7 +246,10 @@ def test_stream_options_include_usage():
       if line.startswith("data: ") and line != "data: [DONE]":
         chunks.append(json.loads(line[6:]))
     assert chunks[-1]["choices"] == []
-    assert chunks[-1]["usage"]["completion_tokens"] >= 1
+    u = chunks[-1]["usage"]
+    assert u["completion_tokens"] >= 1
+    assert u.get("prompt_tokens", 0) >= 1
+    assert u.get("total_tokens") == u["prompt_tokens"] + u["completion_tokens"]
     await client.close()
     await node.stop()
     return True

@@ -53,7 +53,8 @@ def build_prompt(tokenizer, messages: List[Message], tools=None) -> str:
 
 
 def generate_completion(request_id: str, tokens: List[int], decoded: str, model: str,
-                        finish_reason: Optional[str], stream: bool, object_type: str) -> dict:
+                        finish_reason: Optional[str], stream: bool, object_type: str,
+                        prompt_tokens: Optional[int] = None) -> dict:
   completion = {
     "id": f"chatcmpl-{request_id}",
     "object": object_type,
@@ -73,6 +74,9 @@ def generate_completion(request_id: str, tokens: List[int], decoded: str, model:
   else:
     completion["choices"][0]["message"] = {"role": "assistant", "content": decoded}
     completion["usage"] = {"completion_tokens": len(tokens)}
+    if prompt_tokens is not None:
+      completion["usage"]["prompt_tokens"] = prompt_tokens
+      completion["usage"]["total_tokens"] = prompt_tokens + len(tokens)
   return completion
 
 
@@ -285,6 +289,10 @@ class ChatGPTAPI:
         state["temperature"] = float(chat_request.temperature)
       if data.get("top_p") is not None:
         state["top_p"] = float(data["top_p"])
+      try:
+        n_prompt = len(tokenizer.encode(prompt))
+      except Exception:
+        n_prompt = None
       stops = data.get("stop")
       if isinstance(stops, str):
         stops = [stops]
@@ -301,8 +309,8 @@ class ChatGPTAPI:
       if stream:
         include_usage = bool((data.get("stream_options") or {}).get("include_usage"))
         return await self._stream_response(request, request_id, model_id, tokenizer, stops,
-                                           include_usage)
-      return await self._full_response(request_id, model_id, tokenizer, stops)
+                                           include_usage, n_prompt)
+      return await self._full_response(request_id, model_id, tokenizer, stops, n_prompt)
     finally:
       self.token_queues.pop(request_id, None)
       self.prev_token_lens.pop(request_id, None)
@@ -318,7 +326,7 @@ class ChatGPTAPI:
     return cut
 
   async def _stream_response(self, request, request_id, model_id, tokenizer, stops=None,
-                             include_usage=False):
+                             include_usage=False, n_prompt=None):
     response = web.StreamResponse(status=200, headers={
       "Content-Type": "text/event-stream", "Cache-Control": "no-cache",
     })
@@ -360,12 +368,15 @@ class ChatGPTAPI:
                                         "chat.completion.chunk")
       usage_chunk["choices"] = []
       usage_chunk["usage"] = {"completion_tokens": len(all_tokens)}
+      if n_prompt is not None:
+        usage_chunk["usage"]["prompt_tokens"] = n_prompt
+        usage_chunk["usage"]["total_tokens"] = n_prompt + len(all_tokens)
       await response.write(f"data: {json.dumps(usage_chunk)}\n\n".encode())
     await response.write(b"data: [DONE]\n\n")
     await response.write_eof()
     return response
 
-  async def _full_response(self, request_id, model_id, tokenizer, stops=None):
+  async def _full_response(self, request_id, model_id, tokenizer, stops=None, n_prompt=None):
     all_tokens: List[int] = []
     finished = False
     text_all = ""
@@ -381,10 +392,11 @@ class ChatGPTAPI:
             cancel(request_id)
           return web.json_response(
             generate_completion(request_id, all_tokens, text_all[:cut], model_id,
-                                "stop", False, "chat.completion"))
+                                "stop", False, "chat.completion", n_prompt))
     decoded = tokenizer.decode(all_tokens) if all_tokens else ""
     return web.json_response(
-      generate_completion(request_id, all_tokens, decoded, model_id, "stop", False, "chat.completion"))
+      generate_completion(request_id, all_tokens, decoded, model_id, "stop", False,
+                          "chat.completion", n_prompt))
 
   async def run(self, host: str = "0.0.0.0", port: int = 52415):
     self._runner = web.AppRunner(self.app)
