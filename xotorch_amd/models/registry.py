@@ -334,9 +334,13 @@ def build_full_shard(model_id: str, engine_classname: str = "HIPEngine") -> Opti
 
 
 def get_supported_models(supported_engine_lists: Optional[List[List[str]]] = None) -> List[str]:
-  """Model ids runnable by at least one engine in EVERY peer's engine list."""
+  """Model ids runnable by at least one engine in EVERY peer's engine list.
+
+  With no argument: every card that has at least one engine repo — cards
+  listed purely for id parity with the reference (llava vision,
+  bitsandbytes-format checkpoints) are excluded from "supported"."""
   if not supported_engine_lists:
-    return list(model_cards.keys())
+    return [m for m, c in model_cards.items() if c.get("repo")]
   out = []
   for model_id, card in model_cards.items():
     repos = card.get("repo", {})
