@@ -47,3 +47,30 @@ def test_tools_and_entry_scripts_parse():
   assert len(files) > 8
   for f in files:
     ast.parse(f.read_text(), filename=str(f))
+
+
+def test_train_and_eval_cli_end_to_end(tmp_path, capsys):
+  """`xot train`/`xot eval` through the real torch engine (tiny builtin
+  "dummy" llama on CPU) over a tiny jsonl set: losses print, a checkpoint
+  lands in --save-checkpoint-dir."""
+  import json
+  for split in ("train", "valid", "test"):
+    with open(tmp_path / f"{split}.jsonl", "w") as f:
+      for i in range(2):
+        f.write(json.dumps({"text": f"{split} example {i} lorem ipsum"}) + "\n")
+  ckpt = tmp_path / "ckpts"
+  p = build_parser()
+  args = p.parse_args(["train", "dummy", "--inference-engine", "torch",
+                       "--discovery-module", "none", "--data", str(tmp_path),
+                       "--epochs", "1", "--save-every", "1",
+                       "--save-checkpoint-dir", str(ckpt), "--disable-tui"])
+  from xotorch_amd.cli import train_model_cli
+  asyncio.new_event_loop().run_until_complete(train_model_cli(args, train=True))
+  out = capsys.readouterr().out
+  assert "mean loss" in out
+  assert any(ckpt.rglob("*")), "no checkpoint written"
+  args = p.parse_args(["eval", "dummy", "--inference-engine", "torch",
+                       "--discovery-module", "none", "--data", str(tmp_path),
+                       "--disable-tui"])
+  asyncio.new_event_loop().run_until_complete(train_model_cli(args, train=False))
+  assert "eval loss" in capsys.readouterr().out
