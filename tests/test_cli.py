@@ -74,3 +74,11 @@ def test_train_and_eval_cli_end_to_end(tmp_path, capsys):
                        "--disable-tui"])
   asyncio.new_event_loop().run_until_complete(train_model_cli(args, train=False))
   assert "eval loss" in capsys.readouterr().out
+  # resume from the saved checkpoint and keep training
+  saved = next(ckpt.rglob("*.safetensors"))
+  args = p.parse_args(["train", "dummy", "--inference-engine", "torch",
+                       "--discovery-module", "none", "--data", str(tmp_path),
+                       "--epochs", "1", "--save-every", "5",
+                       "--resume-checkpoint", str(saved), "--disable-tui"])
+  asyncio.new_event_loop().run_until_complete(train_model_cli(args, train=True))
+  assert "mean loss" in capsys.readouterr().out
