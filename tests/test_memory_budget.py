@@ -119,3 +119,14 @@ def test_405b_ring_fits_hbm():
   need = ring_rank_bytes("llama-3.1-405b", 8, mb_batch=64, prompt_len=512,
                          max_gen=256, include_packed=False)
   assert need < HBM_BYTES * 0.85, f"{need/2**30:.1f} GiB"
+
+
+@pytest.mark.parametrize("model_id", ["qwen-2.5-72b", "mistral-large", "nemotron-70b"])
+def test_big_dense_cards_fit_one_gpu(model_id):
+  """Every large dense card a user might `xot run` on one MI355X fits 288 GB
+  at bf16 with the bench batch (B=64, 512+64 tokens) WITHOUT packed weights
+  (pack_decode_weights backs off under memory pressure by design)."""
+  if builtin_config(model_id) is None:
+    pytest.skip(f"no builtin config for {model_id}")
+  need = ring_rank_bytes(model_id, 1, 64, 512, 64, include_packed=False)
+  assert need < HBM_BYTES * 0.92, f"{model_id}: {need/2**30:.0f} GiB"
