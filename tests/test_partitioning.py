@@ -85,3 +85,34 @@ def test_empty_partition_skipped():
   shards = map_partitions_to_shards(parts, 4, "m")
   assert sum(s.get_layer_count() for s in shards) == 4
   assert shards[0].start_layer == 0 and shards[-1].end_layer == 3
+
+
+def test_full_coverage_property_fuzz():
+  """Property fuzz over arbitrary positive memory weights and layer counts:
+  shards must always tile [0, n_layers) exactly (the reference's rounding
+  regressions lived exactly here)."""
+  from hypothesis import given, settings, strategies as st
+
+  @settings(max_examples=300, deadline=None)
+  @given(
+    n_layers=st.integers(min_value=1, max_value=200),
+    weights=st.lists(st.floats(min_value=0.01, max_value=1000.0,
+                               allow_nan=False, allow_infinity=False),
+                     min_size=1, max_size=12),
+  )
+  def check(n_layers, weights):
+    total = sum(weights)
+    parts, start = [], 0.0
+    for i, w in enumerate(weights):
+      end = round(start + w / total, 5)
+      parts.append(Partition(f"n{i}", start, end))
+      start = end
+    parts[-1] = Partition(parts[-1].node_id, parts[-1].start, 1.0)
+    shards = map_partitions_to_shards(parts, n_layers, "m")
+    assert shards[0].start_layer == 0
+    assert shards[-1].end_layer == n_layers - 1
+    assert sum(s.get_layer_count() for s in shards) == n_layers
+    for prev, nxt in zip(shards, shards[1:]):
+      assert nxt.start_layer == prev.end_layer + 1
+
+  check()
