@@ -69,3 +69,42 @@ def test_server_unknown_type():
     reply = await srv._dispatch({"type": "nonsense"})
     assert reply["ok"] is False and "unknown" in reply["error"]
   asyncio.run(go())
+
+
+def test_malformed_frames_fail_cleanly():
+  """Garbage length prefixes, truncated payloads and non-msgpack bytes must
+  raise (never hang or return garbage) — a crashing peer cannot wedge the
+  control plane."""
+  async def go():
+    async def serve_bytes(payload):
+      got = {}
+
+      async def handle(reader, writer):
+        try:
+          got["msg"] = await asyncio.wait_for(wire.read_frame(reader), 2)
+        except Exception as e:
+          got["err"] = type(e).__name__
+        writer.close()
+
+      srv = await asyncio.start_server(handle, "127.0.0.1", 0)
+      port = srv.sockets[0].getsockname()[1]
+      r, w = await asyncio.open_connection("127.0.0.1", port)
+      w.write(payload)
+      await w.drain()
+      w.close()
+      await asyncio.sleep(0.1)
+      srv.close()
+      await srv.wait_closed()
+      return got
+
+    # oversize length prefix (wire framing is 4-byte !I)
+    got = await serve_bytes((0xFFFFFFFF).to_bytes(4, "big"))
+    assert got.get("err") == "ValueError", got
+    # truncated payload (claims 100 bytes, sends 3)
+    got = await serve_bytes((100).to_bytes(4, "big") + b"abc")
+    assert "err" in got, got
+    # valid length, non-msgpack garbage
+    body = b"\xc1\xff\x00garbage"
+    got = await serve_bytes(len(body).to_bytes(4, "big") + body)
+    assert "err" in got, got
+  asyncio.run(go())
