@@ -108,3 +108,26 @@ def test_malformed_frames_fail_cleanly():
     got = await serve_bytes(len(body).to_bytes(4, "big") + body)
     assert "err" in got, got
   asyncio.run(go())
+
+
+def test_tensor_roundtrip_property_fuzz():
+  """Hypothesis fuzz: pack/unpack preserves dtype, shape and bytes for all
+  wire-legal dtypes incl. 0-d and empty tensors."""
+  from hypothesis import given, settings, strategies as st
+
+  dtypes = [np.float32, np.float16, np.int64, np.int32, np.uint8, np.bool_]
+
+  @settings(max_examples=200, deadline=None)
+  @given(
+    dt=st.sampled_from(dtypes),
+    shape=st.lists(st.integers(min_value=0, max_value=5), min_size=0, max_size=4),
+    seed=st.integers(min_value=0, max_value=2**31 - 1),
+  )
+  def check(dt, shape, seed):
+    rng = np.random.default_rng(seed)
+    arr = (rng.random(shape) * 100).astype(dt)
+    back = wire.unpack_tensor(wire.pack_tensor(arr))
+    assert back.dtype == arr.dtype and back.shape == arr.shape
+    assert np.array_equal(back, arr)
+
+  check()

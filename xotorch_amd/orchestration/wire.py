@@ -23,8 +23,10 @@ MAX_FRAME = 256 * 1024 * 1024  # parity with reference's 256 MB message cap
 def pack_tensor(arr: Optional[np.ndarray]) -> Optional[dict]:
   if arr is None:
     return None
+  arr = np.asarray(arr)
+  shape = list(arr.shape)  # before ascontiguousarray: it promotes 0-d to 1-d
   arr = np.ascontiguousarray(arr)
-  return {"shape": list(arr.shape), "dtype": str(arr.dtype), "data": arr.tobytes()}
+  return {"shape": shape, "dtype": str(arr.dtype), "data": arr.tobytes()}
 
 
 def unpack_tensor(d: Optional[dict]) -> Optional[np.ndarray]:
