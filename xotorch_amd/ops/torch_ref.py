@@ -298,7 +298,9 @@ def softmax_sample(
 
   logits: [B, V] → token ids [B].
   """
-  if temperature <= 0.0:
+  if temperature <= 1e-4:
+    # sub-epsilon temperature: logits/T overflows to inf -> NaN softmax;
+    # the distribution is argmax to numerical precision anyway
     return logits.argmax(dim=-1)
   logits = logits.float() / temperature
   if top_k and top_k > 0 and top_k < logits.shape[-1]:
@@ -307,8 +309,9 @@ def softmax_sample(
   if top_p and 0.0 < top_p < 1.0:
     sorted_logits, sorted_idx = torch.sort(logits, descending=True, dim=-1)
     cum = torch.softmax(sorted_logits, dim=-1).cumsum(dim=-1)
-    # keep tokens until cumulative prob exceeds top_p (always keep the first)
-    cut = cum - torch.softmax(sorted_logits, dim=-1) >= top_p
+    # keep tokens until cumulative prob exceeds top_p; STRICT inequality so
+    # the argmax (prior cumulative mass 0) survives any top_p > 0
+    cut = cum - torch.softmax(sorted_logits, dim=-1) > top_p
     sorted_logits = sorted_logits.masked_fill(cut, float("-inf"))
     logits = torch.full_like(logits, float("-inf")).scatter(-1, sorted_idx, sorted_logits)
   probs = torch.softmax(logits, dim=-1)
