@@ -205,3 +205,38 @@ def test_softmax_sample_property_fuzz():
       assert (picked >= kth).all(), "sampled outside the top-k set"
 
   check()
+
+
+def test_rope_rotation_invariants_fuzz():
+  """RoPE is a rotation: each (i, i+hd/2) pair keeps its 2-norm, position 0
+  is identity, and q.k dot products depend only on relative position."""
+  import torch
+  from hypothesis import given, settings, strategies as st
+  from xotorch_amd.ops.torch_ref import rope_apply, rope_cos_sin
+
+  @settings(max_examples=60, deadline=None)
+  @given(seed=st.integers(min_value=0, max_value=2**31 - 1),
+         hd=st.sampled_from([8, 32, 64, 128]),
+         pos=st.integers(min_value=0, max_value=100))
+  def check(seed, hd, pos):
+    g = torch.Generator().manual_seed(seed)
+    cos, sin = rope_cos_sin(hd, 256, 10000.0)
+    q = torch.randn(1, 1, 2, hd, generator=g)
+    k = torch.randn(1, 1, 1, hd, generator=g)
+    qp, kp = rope_apply(q, k, cos, sin, torch.tensor([pos]))
+    half = hd // 2
+    pn = lambda x: (x[..., :half] ** 2 + x[..., half:] ** 2)
+    torch.testing.assert_close(pn(qp), pn(q), rtol=1e-4, atol=1e-5)
+    if pos == 0:
+      torch.testing.assert_close(qp, q)
+    # relative-position property: <rope(q,p+d), rope(k,p)> independent of p
+    d = 7
+    qa, ka = rope_apply(q, k, cos, sin, torch.tensor([pos + d]))
+    _, kb = rope_apply(q, k, cos, sin, torch.tensor([pos]))
+    dot1 = (qa[0, 0, 0] * kb[0, 0, 0]).sum()
+    qc, _ = rope_apply(q, k, cos, sin, torch.tensor([d]))
+    _, kd = rope_apply(q, k, cos, sin, torch.tensor([0]))
+    dot2 = (qc[0, 0, 0] * kd[0, 0, 0]).sum()
+    torch.testing.assert_close(dot1, dot2, rtol=1e-3, atol=1e-3)
+
+  check()
