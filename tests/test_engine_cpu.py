@@ -240,3 +240,42 @@ def test_rope_rotation_invariants_fuzz():
     torch.testing.assert_close(dot1, dot2, rtol=1e-3, atol=1e-3)
 
   check()
+
+
+def test_attn_decode_convexity_and_ragged_mask_fuzz():
+  """Decode attention output is a convex combination of the first seq_len
+  value rows — per-slot ragged lengths must never leak later cache rows
+  (the property continuous batching rides on)."""
+  import torch
+  from hypothesis import given, settings, strategies as st
+  from xotorch_amd.ops.torch_ref import attn_decode
+
+  @settings(max_examples=60, deadline=None)
+  @given(seed=st.integers(min_value=0, max_value=2**31 - 1),
+         B=st.integers(min_value=1, max_value=4),
+         T=st.integers(min_value=2, max_value=24))
+  def check(seed, B, T):
+    g = torch.Generator().manual_seed(seed)
+    H, KVH, hd = 4, 2, 16
+    q = torch.randn(B, 1, H, hd, generator=g)
+    k = torch.randn(B, KVH, T, hd, generator=g)
+    v = torch.randn(B, KVH, T, hd, generator=g)
+    seq_lens = torch.randint(1, T + 1, (B,), generator=g)
+    # poison the rows past each slot's seq_len: they must not matter
+    v2 = v.clone()
+    k2 = k.clone()
+    for b in range(B):
+      v2[b, :, seq_lens[b]:] = 1e6
+      k2[b, :, seq_lens[b]:] = 1e6
+    out = attn_decode(q, k, v, seq_lens)
+    out2 = attn_decode(q, k2, v2, seq_lens)
+    torch.testing.assert_close(out, out2, rtol=1e-5, atol=1e-6)
+    # convexity: within [min, max] of the visible value rows per slot/head
+    for b in range(B):
+      vis = v[b].repeat_interleave(H // KVH, dim=0)[:, :seq_lens[b]]  # [H,L,hd]
+      lo = vis.min(dim=1).values - 1e-4
+      hi = vis.max(dim=1).values + 1e-4
+      o = out[b, 0]
+      assert (o >= lo).all() and (o <= hi).all()
+
+  check()
