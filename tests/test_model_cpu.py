@@ -353,3 +353,37 @@ def test_qwen2_yarn_matches_transformers():
     lref2 = ref(torch.cat([x, nxt], 1)).logits[:, -1]
     lours2 = ours(nxt, caches=cache.caches, positions=torch.tensor([S]), start_pos=S, is_decode=True)
     torch.testing.assert_close(lours2, lref2, atol=3e-4, rtol=3e-4)
+
+
+def test_config_parse_key_subsets_fuzz():
+  """config_from_hf must parse any subset of a real config's keys without
+  crashing (HF configs in the wild omit fields freely) and keep derived
+  invariants (head_dim, kv heads <= heads fallback)."""
+  from hypothesis import given, settings, strategies as st
+
+  FULL = {
+    "model_type": "llama", "hidden_size": 128, "num_hidden_layers": 4,
+    "num_attention_heads": 8, "num_key_value_heads": 4, "intermediate_size": 256,
+    "vocab_size": 500, "rope_theta": 10000.0, "rms_norm_eps": 1e-5,
+    "max_position_embeddings": 512, "tie_word_embeddings": True,
+    "head_dim": 16, "attention_bias": False, "torch_dtype": "bfloat16",
+    "eos_token_id": [7, 8], "bos_token_id": 1,
+    "rope_scaling": {"rope_type": "llama3", "factor": 8.0,
+                     "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                     "original_max_position_embeddings": 256},
+  }
+  keys = sorted(FULL)
+
+  @settings(max_examples=120, deadline=None)
+  @given(mask=st.lists(st.booleans(), min_size=len(keys), max_size=len(keys)))
+  def check(mask):
+    raw = {k: FULL[k] for k, keep in zip(keys, mask) if keep}
+    cfg = config_from_hf(raw, "fuzz")
+    assert cfg.n_heads >= 1 and cfg.n_kv_heads >= 1
+    assert cfg.n_kv_heads <= cfg.n_heads
+    assert cfg.head_dim * 1 > 0
+    assert cfg.max_seq_len >= 1
+    if "eos_token_id" in raw:
+      assert cfg.eos_token_id == 7  # list collapses to first
+
+  check()
