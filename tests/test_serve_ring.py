@@ -265,3 +265,50 @@ def test_two_rank_chunked_prefill_interleave(tmp_path, monkeypatch):
   got = json.loads((tmp_path / "tokens.json").read_text())
   for rid, ids, max_new in reqs:
     assert got[rid] == oracle_tokens(ids, max_new), rid
+
+
+@pytest.mark.timeout(300)
+def test_serve_chaos_random_admits_and_cancels(monkeypatch):
+  """Randomized integration: 12 requests with random prompt lengths, budgets
+  and admission times, a few mid-flight cancels, small prefill chunks, 3
+  slots. Every request that ran to completion must match its oracle."""
+  import time as _t
+  monkeypatch.setattr("xotorch_amd.serve_ring.PREFILL_CHUNK", 3)
+  rng = np.random.default_rng(97)
+  w = RingSlotWorker(TINY_ID, 0, 1, device="cpu", dtype=torch.float32,
+                     slots=3, max_seq=128, use_graphs=False)
+  reqs = {}
+  for i in range(12):
+    rid = f"r{i}"
+    plen = int(rng.integers(2, 20))
+    reqs[rid] = (list(rng.integers(0, 200, plen)), int(rng.integers(1, 10)))
+  cancel_ids = {"r3", "r7"}
+  q = queue.Queue()
+  got, done_flags = {}, {}
+  all_done = threading.Event()
+
+  def emit(rid, tok, fin, meta):
+    if not meta.get("cancelled"):
+      got.setdefault(rid, []).append(tok)
+    if fin:
+      done_flags[rid] = meta.get("cancelled", False)
+      if len(done_flags) == len(reqs):
+        all_done.set()
+
+  t = threading.Thread(target=w.serve_forever, args=(q, emit), daemon=True)
+  t.start()
+  for i, (rid, (ids, max_new)) in enumerate(reqs.items()):
+    q.put(AdmitMsg(rid, torch.tensor([ids], dtype=torch.int64), max_new, 0.0))
+    if rid in cancel_ids:
+      _t.sleep(0.02)
+      w.cancel(rid)
+    if i % 3 == 2:
+      _t.sleep(0.05)  # stagger admissions across decode progress
+  assert all_done.wait(180), f"unfinished: {set(reqs) - set(done_flags)}"
+  q.put(AdmitMsg("stop", None, 0, 0.0))
+  t.join(timeout=30)
+  completed = [rid for rid, cancelled in done_flags.items() if not cancelled]
+  assert len(completed) >= len(reqs) - len(cancel_ids)
+  for rid in completed:
+    ids, max_new = reqs[rid]
+    assert got[rid] == oracle_tokens(ids, max_new), rid
