@@ -116,3 +116,27 @@ def test_full_coverage_property_fuzz():
       assert nxt.start_layer == prev.end_layer + 1
 
   check()
+
+
+def test_equal_ring_shards_property_fuzz():
+  """equal_ring_shards tiles [0, n_layers) contiguously with sizes differing
+  by at most 1 (balanced pipeline stages) for every world size."""
+  from hypothesis import given, settings, strategies as st
+  from xotorch_amd.parallel.ring import equal_ring_shards
+
+  @settings(max_examples=200, deadline=None)
+  @given(n_layers=st.integers(min_value=1, max_value=200),
+         world=st.integers(min_value=1, max_value=16))
+  def check(n_layers, world):
+    if world > n_layers:
+      return  # not enough layers for every stage
+    shards = equal_ring_shards("m", n_layers, world)
+    assert len(shards) == world
+    assert shards[0].start_layer == 0
+    assert shards[-1].end_layer == n_layers - 1
+    for a, b in zip(shards, shards[1:]):
+      assert b.start_layer == a.end_layer + 1
+    sizes = [s.get_layer_count() for s in shards]
+    assert max(sizes) - min(sizes) <= 1
+
+  check()
