@@ -387,3 +387,20 @@ def test_config_parse_key_subsets_fuzz():
       assert cfg.eos_token_id == 7  # list collapses to first
 
   check()
+
+
+def test_kv_cache_dims_all_builtin_configs():
+  """Every builtin config yields a constructible KV cache whose per-token
+  byte cost is positive and (for MLA) much smaller than GQA equivalents."""
+  from xotorch_amd.engine.kvcache import ShardKVCache
+  for mid, raw in BUILTIN_CONFIGS.items():
+    cfg = config_from_hf(raw, mid)
+    heads, kd, vd = cfg.kv_cache_dims()
+    assert heads >= 1 and kd >= 1
+    c = ShardKVCache(1, 1, heads, 32, kd, torch.float32, "cpu", v_dim=vd)
+    lk = c.caches[0]
+    assert lk.k.shape[2] >= 32 - 31  # T dim exists (may be padded)
+    if cfg.model_type in ("deepseek_v3", "deepseek_v2"):
+      gqa_bytes = cfg.n_kv_heads * cfg.head_dim * 2 * 2
+      mla_bytes = heads * (kd + (vd or kd)) * 2
+      assert mla_bytes < gqa_bytes / 2, mid  # the latent cache is the point
