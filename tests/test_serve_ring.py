@@ -312,3 +312,51 @@ def test_serve_chaos_random_admits_and_cancels(monkeypatch):
   for rid in completed:
     ids, max_new = reqs[rid]
     assert got[rid] == oracle_tokens(ids, max_new), rid
+
+
+@pytest.mark.timeout(300)
+def test_serve_ring_main_http_end_to_end():
+  """The `xot serve` entry point (serve_ring.py __main__): world-1 CPU
+  worker + ChatGPT API in a subprocess; a completion request round-trips."""
+  import json as _json
+  import subprocess
+  import sys
+  import time as _t
+  import urllib.request
+  from pathlib import Path
+  from xotorch_amd.helpers import find_available_port
+
+  port = find_available_port("127.0.0.1")
+  script = Path(__file__).resolve().parent.parent / "xotorch_amd" / "serve_ring.py"
+  env = dict(os.environ, XOT_OFFLINE="1")
+  proc = subprocess.Popen([sys.executable, str(script), "--model", "dummy",
+                           "--port", str(port), "--no-graphs"],
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT, env=env)
+  try:
+    deadline = _t.time() + 60
+    up = False
+    while _t.time() < deadline:
+      try:
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/healthcheck", timeout=2)
+        up = True
+        break
+      except Exception:
+        if proc.poll() is not None:
+          out = proc.stdout.read().decode(errors="replace")
+          raise AssertionError(f"server died early:\n{out[-2000:]}")
+        _t.sleep(0.3)
+    assert up, "server never came up"
+    body = _json.dumps({"model": "dummy", "max_tokens": 4,
+                        "messages": [{"role": "user", "content": "hi"}]}).encode()
+    req = urllib.request.Request(f"http://127.0.0.1:{port}/v1/chat/completions",
+                                 data=body, headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=60) as r:
+      data = _json.loads(r.read())
+    assert data["object"] == "chat.completion"
+    assert data["choices"][0]["message"]["content"]
+  finally:
+    proc.terminate()
+    try:
+      proc.wait(timeout=15)
+    except subprocess.TimeoutExpired:
+      proc.kill()
