@@ -82,3 +82,46 @@ def test_train_and_eval_cli_end_to_end(tmp_path, capsys):
                        "--resume-checkpoint", str(saved), "--disable-tui"])
   asyncio.new_event_loop().run_until_complete(train_model_cli(args, train=True))
   assert "mean loss" in capsys.readouterr().out
+
+
+def test_daemon_entry_http_smoke():
+  """The bare `xot` daemon entry point in a subprocess: API comes up,
+  healthcheck + models respond, clean shutdown."""
+  import json as _json
+  import os
+  import subprocess
+  import sys
+  import time as _t
+  import urllib.request
+  from xotorch_amd.helpers import find_available_port
+
+  port = find_available_port("127.0.0.1")
+  code = ("import sys; sys.argv = ['xot', '--disable-tui', '--discovery-module', 'none', "
+          f"'--inference-engine', 'dummy', '--chatgpt-api-port', '{port}']; "
+          "from xotorch_amd.cli import run; run()")
+  env = dict(os.environ, XOT_OFFLINE="1")
+  proc = subprocess.Popen([sys.executable, "-c", code],
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT, env=env)
+  try:
+    deadline = _t.time() + 60
+    up = False
+    while _t.time() < deadline:
+      try:
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/healthcheck", timeout=2)
+        up = True
+        break
+      except Exception:
+        if proc.poll() is not None:
+          out = proc.stdout.read().decode(errors="replace")
+          raise AssertionError(f"daemon died early:\n{out[-2000:]}")
+        _t.sleep(0.3)
+    assert up, "daemon API never came up"
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/v1/models", timeout=10) as r:
+      data = _json.loads(r.read())
+    assert any(m["id"] == "llama-3-70b" for m in data["data"])
+  finally:
+    proc.terminate()
+    try:
+      proc.wait(timeout=15)
+    except subprocess.TimeoutExpired:
+      proc.kill()
