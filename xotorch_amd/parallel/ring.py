@@ -35,11 +35,18 @@ from xotorch_amd.shard import Shard
 
 
 def equal_ring_shards(model_id: str, n_layers: int, world: int) -> List[Shard]:
-  """Equal-memory ring partitions (8x identical MI355X) → equal layer shards."""
-  parts = []
+  """Equal-memory ring partitions (8x identical MI355X) → balanced layer
+  shards. Integer split (sizes differ by at most 1): the float-boundary
+  partition map can hand one stage 2 extra layers on worlds that are
+  binary-inexact (e.g. 99 layers / 11 ranks → a 10-layer and an 8-layer
+  stage), and pipeline throughput is set by the slowest stage."""
+  base, rem = divmod(n_layers, world)
+  shards, start = [], 0
   for i in range(world):
-    parts.append(Partition(f"rank{i}", i / world, (i + 1) / world))
-  return map_partitions_to_shards(parts, n_layers, model_id)
+    cnt = base + (1 if i < rem else 0)
+    shards.append(Shard(model_id, start, start + cnt - 1, n_layers))
+    start += cnt
+  return shards
 
 
 @dataclass
