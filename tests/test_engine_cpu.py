@@ -318,3 +318,38 @@ def test_attn_prefill_vs_bruteforce_fuzz():
           torch.testing.assert_close(out[b, i, h], ref, rtol=2e-4, atol=2e-5)
 
   check()
+
+
+def test_attn_decode_vs_bruteforce_fuzz():
+  """Same independent cross-check for the decode attention oracle: ragged
+  per-slot lengths, sliding window, softcap."""
+  import torch
+  from hypothesis import given, settings, strategies as st
+  from xotorch_amd.ops.torch_ref import attn_decode
+
+  @settings(max_examples=40, deadline=None)
+  @given(seed=st.integers(min_value=0, max_value=2**31 - 1),
+         T=st.integers(min_value=1, max_value=12),
+         window=st.sampled_from([0, 3]),
+         softcap=st.sampled_from([0.0, 4.0]))
+  def check(seed, T, window, softcap):
+    g = torch.Generator().manual_seed(seed)
+    B, H, KVH, hd = 2, 4, 2, 8
+    q = torch.randn(B, 1, H, hd, generator=g)
+    kc = torch.randn(B, KVH, T + 3, hd, generator=g)
+    vc = torch.randn(B, KVH, T + 3, hd, generator=g)
+    seq_lens = torch.randint(1, T + 1, (B,), generator=g)
+    out = attn_decode(q, kc, vc, seq_lens, window=window, softcap=softcap)
+    scale = hd ** -0.5
+    for b in range(B):
+      L = int(seq_lens[b])
+      lo = max(0, L - window) if window else 0
+      for h in range(H):
+        kv = h // (H // KVH)
+        sc = (q[b, 0, h] @ kc[b, kv, lo:L].T) * scale
+        if softcap:
+          sc = torch.tanh(sc / softcap) * softcap
+        ref = torch.softmax(sc, dim=-1) @ vc[b, kv, lo:L]
+        torch.testing.assert_close(out[b, 0, h], ref, rtol=2e-4, atol=2e-5)
+
+  check()
